@@ -1,0 +1,172 @@
+"""Byte-budgeted LRU of on-disk models.
+
+Reimplements the contract of the reference's LRUCache
+(/root/reference/pkg/cachemanager/lrucache.go:11-105): Get/Put with
+most-recently-used ordering, `ensure_free_bytes` evicting from the LRU
+tail and deleting the evicted files, `list_models` in MRU order.
+
+Reference bugs intentionally fixed here:
+  * eviction deletes the model directory recursively at its ABSOLUTE path
+    (lrucache.go:73-78 removed a relative path with non-recursive
+    os.Remove and killed the process on failure);
+  * `ensure_free_bytes` is not double-invoked per miss — `put` assumes the
+    caller made room (cachemanager.py does) but still enforces the budget;
+  * optionally rebuilds the index from a cache-dir scan at startup
+    (the reference leaked stale files after restart).
+"""
+from __future__ import annotations
+
+import logging
+import os
+import shutil
+import threading
+from collections import OrderedDict
+from dataclasses import dataclass, field
+from typing import Callable, List, Optional, Tuple
+
+log = logging.getLogger("tfsc.lru")
+
+ModelId = Tuple[str, int]  # (model_name, version)
+
+
+@dataclass
+class Model:
+    name: str
+    version: int
+    path: str            # relative to base_dir, e.g. "mymodel/3"
+    size_on_disk: int
+
+    @property
+    def id(self) -> ModelId:
+        return (self.name, self.version)
+
+
+def dir_size(path: str) -> int:
+    """Recursive on-disk byte size (the reference's DiskModelProvider
+    stat'ed only the directory inode — diskmodelprovider.go:71-83)."""
+    total = 0
+    for root, _dirs, files in os.walk(path):
+        for f in files:
+            try:
+                total += os.path.getsize(os.path.join(root, f))
+            except OSError:
+                pass
+    return total
+
+
+class LRUCache:
+    """Thread-safe byte-budgeted LRU index over a cache directory."""
+
+    def __init__(self, base_dir: str, max_size_bytes: int,
+                 rebuild_from_disk: bool = False,
+                 on_evict: Optional[Callable[[Model], None]] = None):
+        self.base_dir = base_dir
+        self.max_size = int(max_size_bytes)
+        self.current_size = 0
+        self._lock = threading.Lock()
+        self._entries: "OrderedDict[ModelId, Model]" = OrderedDict()
+        self._on_evict = on_evict
+        os.makedirs(base_dir, exist_ok=True)
+        if rebuild_from_disk:
+            self._rebuild()
+
+    # -- queries -----------------------------------------------------------
+    def get(self, name: str, version: int) -> Optional[Model]:
+        with self._lock:
+            entry = self._entries.get((name, version))
+            if entry is None:
+                return None
+            self._entries.move_to_end(entry.id, last=False)  # front = MRU
+            return entry
+
+    def contains(self, name: str, version: int) -> bool:
+        with self._lock:
+            return (name, version) in self._entries
+
+    def list_models(self) -> List[Model]:
+        """MRU-first order (lrucache.go:89-101)."""
+        with self._lock:
+            return list(self._entries.values())
+
+    def model_path(self, model: Model) -> str:
+        return os.path.join(self.base_dir, model.path)
+
+    # -- mutation ------------------------------------------------------------
+    def put(self, model: Model) -> None:
+        with self._lock:
+            old = self._entries.pop(model.id, None)
+            if old is not None:
+                self.current_size -= old.size_on_disk
+            self._ensure_free_locked(model.size_on_disk)
+            self._entries[model.id] = model
+            self._entries.move_to_end(model.id, last=False)
+            self.current_size += model.size_on_disk
+
+    def ensure_free_bytes(self, n_bytes: int) -> None:
+        with self._lock:
+            self._ensure_free_locked(n_bytes)
+
+    def remove(self, name: str, version: int, delete_files: bool = True) -> bool:
+        with self._lock:
+            entry = self._entries.pop((name, version), None)
+            if entry is None:
+                return False
+            self.current_size -= entry.size_on_disk
+            if delete_files:
+                self._delete_files(entry)
+            if self._on_evict:
+                self._on_evict(entry)
+            return True
+
+    # -- internals -----------------------------------------------------------
+    def _ensure_free_locked(self, n_bytes: int) -> None:
+        while self._entries and self.current_size + n_bytes > self.max_size:
+            lru_id, lru_model = self._entries.popitem(last=True)
+            log.info("evicting model %s:%d (%d bytes)", lru_model.name,
+                     lru_model.version, lru_model.size_on_disk)
+            self.current_size -= lru_model.size_on_disk
+            self._delete_files(lru_model)
+            if self._on_evict:
+                self._on_evict(lru_model)
+
+    def _delete_files(self, model: Model) -> None:
+        abs_path = os.path.join(self.base_dir, model.path)
+        try:
+            if os.path.isdir(abs_path):
+                shutil.rmtree(abs_path)
+            elif os.path.exists(abs_path):
+                os.remove(abs_path)
+            # clean the now-possibly-empty model dir
+            parent = os.path.dirname(abs_path)
+            if parent != self.base_dir and os.path.isdir(parent) and \
+                    not os.listdir(parent):
+                os.rmdir(parent)
+        except OSError as e:
+            log.warning("could not delete evicted model files %s: %s",
+                        abs_path, e)
+
+    def _rebuild(self) -> None:
+        """Re-index <base_dir>/<model>/<version> dirs left by a previous run."""
+        try:
+            names = sorted(os.listdir(self.base_dir))
+        except OSError:
+            return
+        for name in names:
+            mdir = os.path.join(self.base_dir, name)
+            if not os.path.isdir(mdir):
+                continue
+            for ver in sorted(os.listdir(mdir)):
+                vdir = os.path.join(mdir, ver)
+                if not os.path.isdir(vdir):
+                    continue
+                try:
+                    version = int(ver)
+                except ValueError:
+                    continue
+                model = Model(name=name, version=version,
+                              path=os.path.join(name, ver),
+                              size_on_disk=dir_size(vdir))
+                with self._lock:
+                    self._ensure_free_locked(model.size_on_disk)
+                    self._entries[model.id] = model
+                    self.current_size += model.size_on_disk

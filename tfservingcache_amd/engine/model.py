@@ -1,0 +1,134 @@
+"""LoadedModel: a compiled SavedModel ready to serve predictions.
+
+CPU path: plan interpreted by executor_cpu (numpy fp32).
+GPU path: plan compiled by the CDNA4 HIP engine (engine/gpu.py) into a
+per-GPU resident model with bf16 MFMA kernels; LoadedModel is the common
+facade the cache tier talks to.
+"""
+from __future__ import annotations
+
+import threading
+from typing import Dict, Optional
+
+import numpy as np
+
+from .executor_cpu import CpuExecutor
+from .planner import Plan, compile_graph, is_sym, resolve_dim
+from .savedmodel import read_saved_model
+from ..wire import graph as g
+from ..wire import messages as m
+
+
+class ModelExecError(Exception):
+    pass
+
+
+class LoadedModel:
+    def __init__(self, name: str, version: int, plan: Plan,
+                 device: str = "cpu", gpu_model=None):
+        self.name = name
+        self.version = version
+        self.plan = plan
+        self.device = device
+        self._cpu = CpuExecutor(plan)
+        self._gpu = gpu_model       # engine/gpu.py GpuModel, if on GPU
+        self._lock = threading.Lock()
+
+    # -- introspection ------------------------------------------------------
+    @property
+    def signature_def(self) -> Optional[m.SignatureDef]:
+        return self.plan.signature_def
+
+    def weight_bytes(self) -> int:
+        return self.plan.weight_bytes()
+
+    # -- execution ----------------------------------------------------------
+    def _infer_batch(self, feeds: Dict[int, np.ndarray]) -> int:
+        for idx, arr in feeds.items():
+            shape = self.plan.tensors[idx].shape
+            for i, d in enumerate(shape):
+                if is_sym(d):
+                    if i >= arr.ndim:
+                        raise ModelExecError(
+                            f"input rank {arr.ndim} < expected {len(shape)}")
+                    k = -d
+                    if arr.shape[i] % k:
+                        raise ModelExecError(
+                            f"input dim {arr.shape[i]} not divisible by {k}")
+                    return arr.shape[i] // k
+        return 1
+
+    def _check_feeds(self, feeds: Dict[int, np.ndarray], batch: int) -> None:
+        for idx, arr in feeds.items():
+            want = self.plan.resolve_shape(self.plan.tensors[idx].shape, batch)
+            if tuple(arr.shape) != want:
+                raise ModelExecError(
+                    f"input {self.plan.tensors[idx].name}: shape "
+                    f"{tuple(arr.shape)} != expected {want}")
+
+    def predict(self, inputs: Dict[str, np.ndarray],
+                output_filter=None) -> Dict[str, np.ndarray]:
+        """Predict with signature-alias-keyed inputs/outputs."""
+        plan = self.plan
+        feeds: Dict[int, np.ndarray] = {}
+        for alias, arr in inputs.items():
+            idx = plan.sig_inputs.get(alias)
+            if idx is None:
+                # also accept raw graph tensor names
+                idx = plan.by_name.get(alias if ":" in alias else alias + ":0")
+            if idx is None:
+                raise ModelExecError(f"unknown input {alias!r}")
+            feeds[idx] = np.asarray(arr)
+        missing = [a for a, i in plan.sig_inputs.items() if i not in feeds]
+        if missing:
+            raise ModelExecError(f"missing inputs: {missing}")
+        batch = self._infer_batch(feeds)
+        self._check_feeds(feeds, batch)
+
+        out_aliases = list(plan.sig_outputs)
+        if output_filter:
+            out_aliases = [a for a in out_aliases if a in set(output_filter)]
+        fetch = [plan.sig_outputs[a] for a in out_aliases]
+
+        if self._gpu is not None:
+            vals = self._gpu.run(feeds, batch, fetch)
+        else:
+            vals = self._cpu.run(feeds, batch, fetch)
+        return {a: vals[plan.sig_outputs[a]] for a in out_aliases}
+
+    def session_run(self, feeds_by_name: Dict[str, np.ndarray],
+                    fetch_names) -> Dict[str, np.ndarray]:
+        plan = self.plan
+        feeds: Dict[int, np.ndarray] = {}
+        for name, arr in feeds_by_name.items():
+            tname = name if ":" in name else name + ":0"
+            idx = plan.by_name.get(tname)
+            if idx is None:
+                raise ModelExecError(f"unknown feed {name!r}")
+            feeds[idx] = np.asarray(arr)
+        fetch = []
+        for name in fetch_names:
+            tname = name if ":" in name else name + ":0"
+            idx = plan.by_name.get(tname)
+            if idx is None:
+                raise ModelExecError(f"unknown fetch {name!r}")
+            fetch.append(idx)
+        batch = self._infer_batch(feeds) if feeds else 1
+        if self._gpu is not None:
+            vals = self._gpu.run(feeds, batch, fetch)
+        else:
+            vals = self._cpu.run(feeds, batch, fetch)
+        return {name: vals[idx] for name, idx in zip(fetch_names, fetch)}
+
+
+def load_model_from_dir(version_dir: str, name: str, version: int,
+                        signature_name: str = g.DEFAULT_SERVING_SIGNATURE
+                        ) -> LoadedModel:
+    graph_def, signatures = read_saved_model(version_dir)
+    sig = signatures.get(signature_name)
+    if sig is None and signatures:
+        sig = next(iter(signatures.values()))
+    if sig is None:
+        raise ModelExecError(f"no signatures in {version_dir}")
+    plan = compile_graph(graph_def, sig)
+    return LoadedModel(name, version, plan)

@@ -1,0 +1,288 @@
+"""SavedModel builders for the benchmark model families.
+
+There is no TensorFlow (and no network) in this environment, so the
+fixtures the reference's docs exercise with public SavedModels
+(half_plus_two: deploy/docker-compose/readme.md:24-60) are generated
+here with random-init weights — the graphs use standard TF op names and
+the standard frozen-graph decompositions (GELU via Erf, LayerNorm via
+Mean/SquaredDifference/Rsqrt) so the planner's pattern matchers see what
+real frozen SavedModels contain.
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional, Sequence
+
+import numpy as np
+
+from ..engine.savedmodel import GraphBuilder, write_saved_model
+
+
+# ---------------------------------------------------------------------------
+# half_plus_two: y = 0.5*x + 2  (functional reference point — BASELINE.md)
+# ---------------------------------------------------------------------------
+def build_half_plus_two():
+    gb = GraphBuilder()
+    x = gb.placeholder("x", np.float32, [-1], signature_name="x")
+    a = gb.const("a", np.float32(0.5))
+    b = gb.const("b", np.float32(2.0))
+    mul = gb.node("Mul", "mul", [x, a], T=gb.a_type(1))
+    y = gb.node("AddV2", "y", [mul, b], T=gb.a_type(1))
+    gb.mark_output("y", y)
+    return gb.build()
+
+
+# ---------------------------------------------------------------------------
+# simple MLP (tests): relu(x@W1+b1)@W2+b2 -> softmax
+# ---------------------------------------------------------------------------
+def build_mlp(d_in=16, d_hidden=32, d_out=8, seed=0):
+    rng = np.random.default_rng(seed)
+    gb = GraphBuilder()
+    x = gb.placeholder("x", np.float32, [-1, d_in], signature_name="x")
+    w1 = gb.const("w1", rng.standard_normal((d_in, d_hidden),
+                                            dtype=np.float32) * 0.3)
+    b1 = gb.const("b1", rng.standard_normal(d_hidden, dtype=np.float32) * 0.1)
+    w2 = gb.const("w2", rng.standard_normal((d_hidden, d_out),
+                                            dtype=np.float32) * 0.3)
+    b2 = gb.const("b2", rng.standard_normal(d_out, dtype=np.float32) * 0.1)
+    mm1 = gb.node("MatMul", "mm1", [x, w1], T=gb.a_type(1))
+    ba1 = gb.node("BiasAdd", "ba1", [mm1, b1], T=gb.a_type(1))
+    r1 = gb.node("Relu", "relu1", [ba1], T=gb.a_type(1))
+    mm2 = gb.node("MatMul", "mm2", [r1, w2], T=gb.a_type(1))
+    ba2 = gb.node("BiasAdd", "ba2", [mm2, b2], T=gb.a_type(1))
+    sm = gb.node("Softmax", "probs", [ba2], T=gb.a_type(1))
+    gb.mark_output("probs", sm)
+    return gb.build()
+
+
+# ---------------------------------------------------------------------------
+# ResNet-50 v1.5 (NHWC, inference): conv stem -> 4 stages of bottlenecks ->
+# global mean -> fc -> softmax. BN emitted as FusedBatchNormV3 (inference),
+# which the planner folds into the conv weights.
+# ---------------------------------------------------------------------------
+def build_resnet50(image_size=224, num_classes=1000, seed=0,
+                   stage_blocks=(3, 4, 6, 3)):
+    rng = np.random.default_rng(seed)
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+
+    def conv(name, x, cin, cout, k, stride, use_relu=True, add_to=None):
+        w = gb.const(f"{name}/w", (rng.standard_normal((k, k, cin, cout))
+                                   * np.sqrt(2.0 / (k * k * cin))
+                                   ).astype(np.float32))
+        c = gb.node("Conv2D", f"{name}/conv", [x, w], T=f32,
+                    strides=gb.a_ints([1, stride, stride, 1]),
+                    padding=gb.a_str("SAME"),
+                    data_format=gb.a_str("NHWC"))
+        scale = gb.const(f"{name}/gamma",
+                         np.abs(rng.standard_normal(cout)).astype(np.float32)
+                         * 0.5 + 0.5)
+        offset = gb.const(f"{name}/beta",
+                          (rng.standard_normal(cout) * 0.1).astype(np.float32))
+        mean = gb.const(f"{name}/mean",
+                        (rng.standard_normal(cout) * 0.1).astype(np.float32))
+        var = gb.const(f"{name}/var",
+                       np.abs(rng.standard_normal(cout)).astype(np.float32)
+                       * 0.5 + 0.5)
+        bn = gb.node("FusedBatchNormV3", f"{name}/bn",
+                     [c, scale, offset, mean, var], T=f32, U=f32,
+                     epsilon=gb.a_float(1.001e-5),
+                     is_training=gb.a_bool(False),
+                     data_format=gb.a_str("NHWC"))
+        cur = bn
+        if add_to is not None:
+            cur = gb.node("AddV2", f"{name}/add", [cur, add_to], T=f32)
+        if use_relu:
+            cur = gb.node("Relu", f"{name}/relu", [cur], T=f32)
+        return cur
+
+    x = gb.placeholder("input", np.float32, [-1, image_size, image_size, 3],
+                       signature_name="input")
+    # stem: 7x7/2 conv + 3x3/2 maxpool
+    h = conv("stem", x, 3, 64, 7, 2)
+    h = gb.node("MaxPool", "stem/pool", [h], T=f32,
+                ksize=gb.a_ints([1, 3, 3, 1]),
+                strides=gb.a_ints([1, 2, 2, 1]),
+                padding=gb.a_str("SAME"),
+                data_format=gb.a_str("NHWC"))
+
+    cin = 64
+    widths = (64, 128, 256, 512)
+    for stage, (blocks, wmid) in enumerate(zip(stage_blocks, widths)):
+        cout = wmid * 4
+        for blk in range(blocks):
+            stride = 2 if (stage > 0 and blk == 0) else 1
+            name = f"s{stage}b{blk}"
+            if blk == 0:
+                shortcut = conv(f"{name}/down", h, cin, cout, 1, stride,
+                                use_relu=False)
+            else:
+                shortcut = h
+            t = conv(f"{name}/c1", h, cin, wmid, 1, 1)
+            t = conv(f"{name}/c2", t, wmid, wmid, 3, stride)
+            h = conv(f"{name}/c3", t, wmid, cout, 1, 1,
+                     use_relu=True, add_to=shortcut)
+            cin = cout
+
+    gap_axes = gb.const("gap/axes", np.array([1, 2], dtype=np.int32))
+    pooled = gb.node("Mean", "gap", [h, gap_axes], T=f32,
+                     keep_dims=gb.a_bool(False))
+    wfc = gb.const("fc/w", (rng.standard_normal((cin, num_classes))
+                            * np.sqrt(1.0 / cin)).astype(np.float32))
+    bfc = gb.const("fc/b", np.zeros(num_classes, dtype=np.float32))
+    logits = gb.node("MatMul", "fc/mm", [pooled, wfc], T=f32)
+    logits = gb.node("BiasAdd", "logits", [logits, bfc], T=f32)
+    probs = gb.node("Softmax", "probs", [logits], T=f32)
+    gb.mark_output("probs", probs)
+    gb.mark_output("logits", logits)
+    return gb.build()
+
+
+# ---------------------------------------------------------------------------
+# BERT-base encoder (inference): embeddings -> 12 transformer layers ->
+# pooler. Frozen-graph decompositions for LayerNorm and GELU.
+# ---------------------------------------------------------------------------
+def build_bert(seq_len=128, hidden=768, layers=12, heads=12,
+               intermediate=3072, vocab=30522, seed=0):
+    rng = np.random.default_rng(seed)
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    i32 = gb.a_type(3)
+    dh = hidden // heads
+
+    def W(name, shape, scale=0.02):
+        return gb.const(name, (rng.standard_normal(shape) * scale
+                               ).astype(np.float32))
+
+    def layernorm(name, x):
+        """Emit the canonical frozen LayerNorm primitive pattern."""
+        axes = gb.const(f"{name}/axes", np.array([-1], dtype=np.int32))
+        axes2 = gb.const(f"{name}/axes2", np.array([-1], dtype=np.int32))
+        mean = gb.node("Mean", f"{name}/mean", [x, axes], T=f32,
+                       keep_dims=gb.a_bool(True))
+        sqd = gb.node("SquaredDifference", f"{name}/sqd", [x, mean], T=f32)
+        var = gb.node("Mean", f"{name}/var", [sqd, axes2], T=f32,
+                      keep_dims=gb.a_bool(True))
+        eps = gb.const(f"{name}/eps", np.float32(1e-12))
+        addeps = gb.node("AddV2", f"{name}/addeps", [var, eps], T=f32)
+        rstd = gb.node("Rsqrt", f"{name}/rsqrt", [addeps], T=f32)
+        sub = gb.node("Sub", f"{name}/sub", [x, mean], T=f32)
+        mul = gb.node("Mul", f"{name}/mul", [sub, rstd], T=f32)
+        gamma = gb.const(f"{name}/gamma", np.ones(hidden, dtype=np.float32))
+        beta = gb.const(f"{name}/beta", np.zeros(hidden, dtype=np.float32))
+        mulg = gb.node("Mul", f"{name}/mulg", [mul, gamma], T=f32)
+        return gb.node("AddV2", f"{name}/out", [mulg, beta], T=f32)
+
+    def gelu(name, x):
+        c1 = gb.const(f"{name}/rsqrt2", np.float32(0.7071067811865476))
+        c2 = gb.const(f"{name}/half", np.float32(0.5))
+        c3 = gb.const(f"{name}/one", np.float32(1.0))
+        m1 = gb.node("Mul", f"{name}/m1", [x, c1], T=f32)
+        erf = gb.node("Erf", f"{name}/erf", [m1], T=f32)
+        a1 = gb.node("AddV2", f"{name}/a1", [erf, c3], T=f32)
+        m2 = gb.node("Mul", f"{name}/m2", [x, c2], T=f32)
+        return gb.node("Mul", f"{name}/out", [m2, a1], T=f32)
+
+    def dense(name, x, din, dout, act_nodes=None):
+        w = W(f"{name}/w", (din, dout))
+        b = gb.const(f"{name}/b", np.zeros(dout, dtype=np.float32))
+        mm = gb.node("MatMul", f"{name}/mm", [x, w], T=f32)
+        return gb.node("BiasAdd", f"{name}/out", [mm, b], T=f32)
+
+    ids = gb.placeholder("input_ids", np.int32, [-1, seq_len],
+                         signature_name="input_ids")
+    tok_emb = W("embeddings/word", (vocab, hidden))
+    pos_emb = W("embeddings/pos", (seq_len, hidden))
+    flat_shape = gb.const("flat_ids_shape", np.array([-1], dtype=np.int32))
+    ids_flat = gb.node("Reshape", "ids_flat", [ids, flat_shape])
+    gax = gb.const("gather_axis", np.array(0, dtype=np.int32))
+    emb = gb.node("GatherV2", "embed", [tok_emb, ids_flat, gax],
+                  Tparams=f32, Tindices=i32)
+    # [B*S, H] + pos broadcast: tile pos via reshape trick — add pos per row
+    to_bsh = gb.const("to_bsh", np.array([-1, seq_len, hidden],
+                                         dtype=np.int32))
+    emb3 = gb.node("Reshape", "embed3", [emb, to_bsh])
+    pos3 = gb.node("Reshape", "pos3",
+                   [pos_emb, gb.const("pos_shape",
+                                      np.array([1, seq_len, hidden],
+                                               dtype=np.int32))])
+    h3 = gb.node("AddV2", "embed_sum", [emb3, pos3], T=f32)
+    to_2d = gb.const("to_2d", np.array([-1, hidden], dtype=np.int32))
+    h = gb.node("Reshape", "embed2d", [h3, to_2d])
+    h = layernorm("embeddings/ln", h)
+
+    to_heads = gb.const("to_heads", np.array([-1, seq_len, heads, dh],
+                                             dtype=np.int32))
+    perm = gb.const("perm0213", np.array([0, 2, 1, 3], dtype=np.int32))
+    to_ctx = gb.const("to_ctx", np.array([-1, hidden], dtype=np.int32))
+
+    for li in range(layers):
+        pre = f"layer{li}"
+        q = dense(f"{pre}/q", h, hidden, hidden)
+        k = dense(f"{pre}/k", h, hidden, hidden)
+        v = dense(f"{pre}/v", h, hidden, hidden)
+
+        def split_heads(name, t):
+            r = gb.node("Reshape", f"{name}/r", [t, to_heads])
+            return gb.node("Transpose", f"{name}/t", [r, perm], T=f32)
+
+        qh = split_heads(f"{pre}/qh", q)   # [B, heads, S, dh]
+        kh = split_heads(f"{pre}/kh", k)
+        vh = split_heads(f"{pre}/vh", v)
+        scale = gb.const(f"{pre}/scale", np.float32(1.0 / np.sqrt(dh)))
+        scores = gb.node("BatchMatMulV2", f"{pre}/scores", [qh, kh],
+                         T=f32, adj_x=gb.a_bool(False), adj_y=gb.a_bool(True))
+        scaled = gb.node("Mul", f"{pre}/scaled", [scores, scale], T=f32)
+        probs = gb.node("Softmax", f"{pre}/probs", [scaled], T=f32)
+        ctx = gb.node("BatchMatMulV2", f"{pre}/ctx", [probs, vh], T=f32)
+        ctx_t = gb.node("Transpose", f"{pre}/ctx_t", [ctx, perm], T=f32)
+        ctx2 = gb.node("Reshape", f"{pre}/ctx2", [ctx_t, to_ctx])
+        att = dense(f"{pre}/att_out", ctx2, hidden, hidden)
+        res1 = gb.node("AddV2", f"{pre}/res1", [att, h], T=f32)
+        h1 = layernorm(f"{pre}/ln1", res1)
+        ffn1 = dense(f"{pre}/ffn1", h1, hidden, intermediate)
+        act = gelu(f"{pre}/gelu", ffn1)
+        ffn2 = dense(f"{pre}/ffn2", act, intermediate, hidden)
+        res2 = gb.node("AddV2", f"{pre}/res2", [ffn2, h1], T=f32)
+        h = layernorm(f"{pre}/ln2", res2)
+
+    # pooler: first token -> dense tanh
+    to_3d = gb.const("to_3d_out", np.array([-1, seq_len, hidden],
+                                           dtype=np.int32))
+    h3out = gb.node("Reshape", "seq_out3", [h, to_3d])
+    pooler_w = W("pooler/w", (hidden, hidden))
+    pooler_b = gb.const("pooler/b", np.zeros(hidden, dtype=np.float32))
+    # take token 0 of each sequence using strided gather via reshape+gather?
+    # simplest frozen form: Slice is unsupported; use GatherV2 over axis 1
+    # after transpose — emit Mean over sequence instead for the benchmark
+    # signature (sequence embedding), plus full sequence output.
+    axes_sp = gb.const("pool_axes", np.array([1], dtype=np.int32))
+    pooled_in = gb.node("Mean", "pool_mean", [h3out, axes_sp], T=f32,
+                        keep_dims=gb.a_bool(False))
+    pmm = gb.node("MatMul", "pooler/mm", [pooled_in, pooler_w], T=f32)
+    pba = gb.node("BiasAdd", "pooler/ba", [pmm, pooler_b], T=f32)
+    pooled = gb.node("Tanh", "pooled", [pba], T=f32)
+
+    gb.mark_output("sequence_output", h3out)
+    gb.mark_output("pooled_output", pooled)
+    return gb.build()
+
+
+# ---------------------------------------------------------------------------
+# repo writer
+# ---------------------------------------------------------------------------
+_BUILDERS = {
+    "half_plus_two": build_half_plus_two,
+    "mlp": build_mlp,
+    "resnet50": build_resnet50,
+    "bert_base": build_bert,
+}
+
+
+def write_model_repo(base_dir: str, models: Sequence[tuple],
+                     builder_kwargs: Optional[dict] = None) -> None:
+    """models: list of (model_name, version, builder_name)."""
+    for name, version, builder in models:
+        kw = (builder_kwargs or {}).get(name, {})
+        sm = _BUILDERS[builder](**kw)
+        write_saved_model(sm, os.path.join(base_dir, name, str(version)))
