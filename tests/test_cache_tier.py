@@ -1,0 +1,277 @@
+"""End-to-end cache-tier tests: CacheManager + LRU + pool + REST + gRPC.
+
+Mirrors the reference's component-test style (tfservingproxy_test.go uses
+real localhost sockets against a fake backend; here the backend is the
+real in-process engine over a temp model repo).
+"""
+import json
+import threading
+import time
+
+import grpc
+import numpy as np
+import pytest
+from aiohttp.test_utils import TestClient, TestServer
+
+from tfservingcache_amd.cachemanager import (CacheManager, LRUCache,
+                                             ModelPool, make_cpu_loader)
+from tfservingcache_amd.cachemanager.providers import DiskModelProvider
+from tfservingcache_amd.models import write_model_repo
+from tfservingcache_amd.tfservingproxy import (LocalServingHandler,
+                                               make_cache_grpc_server,
+                                               make_cache_rest_app)
+from tfservingcache_amd.utils import metrics as mt
+from tfservingcache_amd.wire import messages as m
+from tfservingcache_amd.wire.tensor import (numpy_to_tensorproto,
+                                            tensorproto_to_numpy)
+
+
+@pytest.fixture()
+def stack(tmp_path):
+    repo = tmp_path / "repo"
+    cache_dir = tmp_path / "cache"
+    write_model_repo(str(repo), [
+        ("half_plus_two", 123, "half_plus_two"),
+        ("half_plus_two", 124, "half_plus_two"),
+        ("mlp", 1, "mlp"),
+    ])
+    provider = DiskModelProvider(str(repo))
+    cache = LRUCache(str(cache_dir), max_size_bytes=10 * 1024 * 1024)
+    pool = ModelPool(make_cpu_loader(cache), max_concurrent_models=2)
+    cm = CacheManager(provider, cache, pool, model_fetch_timeout=10.0)
+    handler = LocalServingHandler(cm)
+    return cm, handler
+
+
+def test_predict_via_handler(stack):
+    cm, handler = stack
+    req = m.PredictRequest(
+        model_spec=m.ModelSpec(name="half_plus_two",
+                               version=m.Int64Value(value=123)),
+        inputs={"x": numpy_to_tensorproto(
+            np.array([1.0, 2.0, 5.0], dtype=np.float32))})
+    resp = handler.predict(req)
+    out = tensorproto_to_numpy(resp.outputs["y"])
+    np.testing.assert_allclose(out, [2.5, 3.0, 4.5])
+    assert resp.model_spec.version.value == 123
+
+
+def test_version_latest_resolution(stack):
+    cm, handler = stack
+    req = m.PredictRequest(
+        model_spec=m.ModelSpec(name="half_plus_two"),
+        inputs={"x": numpy_to_tensorproto(
+            np.array([4.0], dtype=np.float32))})
+    resp = handler.predict(req)
+    assert resp.model_spec.version.value == 124  # latest
+
+
+def test_model_status_and_reload(stack):
+    cm, handler = stack
+    handler.predict(m.PredictRequest(
+        model_spec=m.ModelSpec(name="mlp", version=m.Int64Value(value=1)),
+        inputs={"x": numpy_to_tensorproto(
+            np.zeros((1, 16), dtype=np.float32))}))
+    st = handler.get_model_status(m.GetModelStatusRequest(
+        model_spec=m.ModelSpec(name="mlp")))
+    assert st.model_version_status[0].state == m.STATE_AVAILABLE
+
+    with pytest.raises(Exception):
+        handler.get_model_status(m.GetModelStatusRequest(
+            model_spec=m.ModelSpec(name="missing_model")))
+
+
+def test_lru_eviction_on_pool(stack):
+    """maxConcurrentModels=2: a third model evicts the LRU one from the
+    pool (the reference pushed only the MRU prefix — cachemanager.go:168)."""
+    cm, handler = stack
+
+    def predict(name, version, x):
+        return handler.predict(m.PredictRequest(
+            model_spec=m.ModelSpec(name=name,
+                                   version=m.Int64Value(value=version)),
+            inputs={"x": numpy_to_tensorproto(x)}))
+
+    predict("half_plus_two", 123, np.array([1.0], dtype=np.float32))
+    predict("mlp", 1, np.zeros((1, 16), dtype=np.float32))
+    predict("half_plus_two", 124, np.array([1.0], dtype=np.float32))
+    # pool only holds 2: (hpt,123) must be gone
+    time.sleep(0.1)
+    states = {(n, v): s for (n, v), s in cm.pool.model_states().items()}
+    assert (("half_plus_two", 123) not in states or
+            states[("half_plus_two", 123)] != m.STATE_AVAILABLE)
+    assert states[("half_plus_two", 124)] == m.STATE_AVAILABLE
+    assert states[("mlp", 1)] == m.STATE_AVAILABLE
+
+
+def test_concurrent_misses_single_flight(stack):
+    cm, handler = stack
+    results = []
+    errs = []
+
+    def worker():
+        try:
+            resp = handler.predict(m.PredictRequest(
+                model_spec=m.ModelSpec(name="mlp",
+                                       version=m.Int64Value(value=1)),
+                inputs={"x": numpy_to_tensorproto(
+                    np.zeros((2, 16), dtype=np.float32))}))
+            results.append(resp)
+        except Exception as e:      # noqa: BLE001
+            errs.append(e)
+
+    threads = [threading.Thread(target=worker) for _ in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errs
+    assert len(results) == 8
+
+
+# ---------------------------------------------------------------------------
+# REST round trip over a real socket
+# ---------------------------------------------------------------------------
+async def test_rest_predict(stack, aiohttp_client=None):
+    cm, handler = stack
+    app = make_cache_rest_app(handler, metrics_render=mt.render)
+    server = TestServer(app)
+    client = TestClient(server)
+    await client.start_server()
+    try:
+        resp = await client.post(
+            "/v1/models/half_plus_two/versions/123:predict",
+            json={"instances": [1.0, 2.0, 5.0]})
+        assert resp.status == 200
+        body = await resp.json()
+        assert body == {"predictions": [2.5, 3.0, 4.5]}
+
+        # columnar + named
+        resp = await client.post(
+            "/v1/models/mlp:predict",
+            json={"inputs": {"x": [[0.0] * 16]}})
+        assert resp.status == 200
+        body = await resp.json()
+        assert len(body["outputs"][0]) == 8
+
+        # status endpoint
+        resp = await client.get("/v1/models/half_plus_two/versions/123")
+        body = await resp.json()
+        assert body["model_version_status"][0]["state"] == "AVAILABLE"
+
+        # metadata
+        resp = await client.get("/v1/models/half_plus_two/metadata")
+        body = await resp.json()
+        assert "signature_def" in body["metadata"]
+
+        # malformed URL -> 404 (tfservingproxy_test.go:142-170 analog)
+        resp = await client.post("/v1/foo/bar:predict", json={})
+        assert resp.status == 404
+
+        # unknown model -> 404
+        resp = await client.post("/v1/models/nope:predict",
+                                 json={"instances": [1.0]})
+        assert resp.status == 404
+
+        # metrics scrape
+        resp = await client.get("/monitoring/prometheus/metrics")
+        text = await resp.text()
+        assert "tfservingcache_cache_total" in text
+        assert "tfservingcache_proxy_requests_total" in text
+    finally:
+        await client.close()
+
+
+# ---------------------------------------------------------------------------
+# gRPC round trip over a real socket
+# ---------------------------------------------------------------------------
+def test_grpc_predict_and_status(stack):
+    cm, handler = stack
+    server, health = make_cache_grpc_server(handler)
+    port = server.add_insecure_port("127.0.0.1:0")
+    server.start()
+    try:
+        ch = grpc.insecure_channel(f"127.0.0.1:{port}")
+        predict = ch.unary_unary(
+            "/tensorflow.serving.PredictionService/Predict",
+            request_serializer=lambda r: r.encode(),
+            response_deserializer=m.PredictResponse.decode)
+        req = m.PredictRequest(
+            model_spec=m.ModelSpec(name="half_plus_two",
+                                   version=m.Int64Value(value=123)),
+            inputs={"x": numpy_to_tensorproto(
+                np.array([1.0, 2.0, 5.0], dtype=np.float32))})
+        resp = predict(req, timeout=10)
+        np.testing.assert_allclose(
+            tensorproto_to_numpy(resp.outputs["y"]), [2.5, 3.0, 4.5])
+
+        status = ch.unary_unary(
+            "/tensorflow.serving.ModelService/GetModelStatus",
+            request_serializer=lambda r: r.encode(),
+            response_deserializer=m.GetModelStatusResponse.decode)
+        st = status(m.GetModelStatusRequest(
+            model_spec=m.ModelSpec(name="half_plus_two")), timeout=10)
+        assert any(s.state == m.STATE_AVAILABLE
+                   for s in st.model_version_status)
+
+        # health
+        from tfservingcache_amd.tfservingproxy.grpc_server import (
+            HealthCheckRequest, HealthCheckResponse)
+        check = ch.unary_unary(
+            "/grpc.health.v1.Health/Check",
+            request_serializer=lambda r: r.encode(),
+            response_deserializer=HealthCheckResponse.decode)
+        hr = check(HealthCheckRequest(), timeout=10)
+        assert hr.status == 1  # SERVING
+
+        # MultiInference -> UNIMPLEMENTED (tfservingproxy.go:215-217)
+        multi = ch.unary_unary(
+            "/tensorflow.serving.PredictionService/MultiInference",
+            request_serializer=lambda b: b,
+            response_deserializer=lambda b: b)
+        with pytest.raises(grpc.RpcError) as ei:
+            multi(b"", timeout=10)
+        assert ei.value.code() == grpc.StatusCode.UNIMPLEMENTED
+
+        # SessionRun
+        run = ch.unary_unary(
+            "/tensorflow.serving.SessionService/SessionRun",
+            request_serializer=lambda r: r.encode(),
+            response_deserializer=m.SessionRunResponse.decode)
+        rr = run(m.SessionRunRequest(
+            model_spec=m.ModelSpec(name="half_plus_two",
+                                   version=m.Int64Value(value=123)),
+            feed=[m.NamedTensorProto(
+                name="x", tensor=numpy_to_tensorproto(
+                    np.array([10.0], dtype=np.float32)))],
+            fetch=["y:0"]), timeout=10)
+        np.testing.assert_allclose(
+            tensorproto_to_numpy(rr.tensor[0].tensor), [7.0])
+
+        ch.close()
+    finally:
+        server.stop(None)
+
+
+def test_classify_regress(stack):
+    cm, handler = stack
+    ex = m.Example(features=m.Features())
+    ex.features.feature["x"] = m.Feature(
+        float_list=m.FloatList(value=[3.0]))
+    req = m.RegressionRequest(
+        model_spec=m.ModelSpec(name="half_plus_two",
+                               version=m.Int64Value(value=123)),
+        input=m.Input(example_list=m.ExampleList(examples=[ex])))
+    resp = handler.regress(req)
+    assert abs(resp.result.regressions[0].value - 3.5) < 1e-6
+
+    creq = m.ClassificationRequest(
+        model_spec=m.ModelSpec(name="mlp", version=m.Int64Value(value=1)),
+        input=m.Input(example_list=m.ExampleList(examples=[
+            m.Example(features=m.Features(feature={
+                "x": m.Feature(float_list=m.FloatList(
+                    value=[0.1] * 16))}))])))
+    cresp = handler.classify(creq)
+    classes = cresp.result.classifications[0].classes
+    assert len(classes) == 8
+    assert abs(sum(c.score for c in classes) - 1.0) < 1e-4

@@ -53,3 +53,16 @@ class DiskModelProvider(ModelProvider):
 
     def check(self) -> bool:
         return os.path.isdir(self.base_dir)
+
+    def latest_version(self, model_name: str):
+        model_dir = os.path.join(self.base_dir, model_name)
+        if not os.path.isdir(model_dir):
+            return None
+        versions = []
+        for entry in os.listdir(model_dir):
+            if os.path.isdir(os.path.join(model_dir, entry)):
+                try:
+                    versions.append(int(entry))
+                except ValueError:
+                    pass
+        return max(versions) if versions else None
