@@ -1,0 +1,41 @@
+"""Build the CDNA4 HIP engine extension in-tree.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+Produces tfservingcache_amd/engine/_tfsc_engine*.so (gitignored; ships to
+the GPU box with the gpurun snapshot).
+"""
+import os
+import sys
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils import cpp_extension  # noqa: E402
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(ROOT, "tfservingcache_amd", "engine", "csrc")
+
+sources = [
+    os.path.join(CSRC, "executor.cpp"),
+    os.path.join(CSRC, "ops", "ops_memory.hip"),
+    os.path.join(CSRC, "ops", "gemm.hip"),
+    os.path.join(CSRC, "ops", "conv.hip"),
+]
+
+setup(
+    name="tfsc_engine",
+    ext_modules=[
+        cpp_extension.CUDAExtension(
+            name="tfservingcache_amd.engine._tfsc_engine",
+            sources=sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": cpp_extension.BuildExtension.with_options(
+        no_python_abi_suffix=False)},
+)

@@ -132,3 +132,20 @@ def make_cpu_loader(cache: LRUCache) -> Callable[[str, int], LoadedModel]:
         vdir = os.path.join(cache.base_dir, name, str(version))
         return load_model_from_dir(vdir, name, version)
     return loader
+
+
+def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
+                    max_batch: int = 64,
+                    use_graphs: bool = True) -> Callable[[str, int], LoadedModel]:
+    """Loader that compiles the SavedModel onto one MI355X: weights land
+    in the GPU's HBM pool (bf16, GEMM layouts pre-transformed) and the
+    predict path runs the CDNA4 HIP kernels."""
+    def loader(name: str, version: int) -> LoadedModel:
+        from ..engine.gpu import GpuModel
+        vdir = os.path.join(cache.base_dir, name, str(version))
+        lm = load_model_from_dir(vdir, name, version)
+        lm._gpu = GpuModel(lm.plan, device=device, max_batch=max_batch,
+                           use_graphs=use_graphs)
+        lm.device = device
+        return lm
+    return loader

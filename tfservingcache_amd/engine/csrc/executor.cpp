@@ -1,0 +1,254 @@
+// ExecPlan: a compiled sequence of kernel launches for one (model, batch
+// bucket). The Python planner (engine/gpu.py) resolves every shape,
+// buffer offset and fusion decision ahead of time; this layer only
+// launches CDNA4 kernels in order on the current HIP stream, so the
+// per-request overhead is one Python->C++ call. hipGraph capture/replay
+// of the whole sequence removes the per-kernel launch overhead for
+// launch-bound small models (MI355X_MICROARCH.md §price list: boundary).
+//
+// This (plus the .hip kernels) replaces the graph execution the
+// reference delegated to tensorflow_model_server.
+#include <torch/extension.h>
+
+#include <hip/hip_runtime.h>
+
+#include <c10/hip/HIPStream.h>
+
+#include <cstdint>
+#include <stdexcept>
+#include <vector>
+
+#include "kernels.h"
+
+namespace tfsc {
+
+enum CallKind : int {
+  K_ELT_UNARY = 0,
+  K_ELT_BINARY,
+  K_BN_ACT,
+  K_SOFTMAX,
+  K_LAYERNORM,
+  K_MEAN_LAST,
+  K_MEAN_MID,
+  K_POOL,
+  K_TRANSPOSE,
+  K_GATHER,
+  K_PAD_NHWC,
+  K_GEMM,
+  K_BGEMM,
+  K_IM2COL,
+};
+
+struct Call {
+  int kind;
+  std::vector<intptr_t> ptrs;
+  std::vector<int64_t> ints;
+  std::vector<float> floats;
+};
+
+static void launch_call(const Call& c, hipStream_t s) {
+  auto p = [&](int i) { return reinterpret_cast<ushort*>(c.ptrs[i]); };
+  auto cp = [&](int i) { return reinterpret_cast<const ushort*>(c.ptrs[i]); };
+  const auto& I = c.ints;
+  switch (c.kind) {
+    case K_ELT_UNARY:
+      launch_eltwise_unary(s, cp(0), p(1), I[0], int(I[1]));
+      break;
+    case K_ELT_BINARY: {
+      BcastArgs bc;
+      bc.ndim = int(I[2]);
+      for (int d = 0; d < MAX_DIMS; ++d) {
+        bc.dims[d] = d < bc.ndim ? I[3 + d] : 1;
+        bc.sa[d] = d < bc.ndim ? I[3 + MAX_DIMS + d] : 0;
+        bc.sb[d] = d < bc.ndim ? I[3 + 2 * MAX_DIMS + d] : 0;
+      }
+      launch_eltwise_binary(s, cp(0), cp(1), p(2), I[0], bc, int(I[1]));
+      break;
+    }
+    case K_BN_ACT:
+      launch_bn_act(s, cp(0), cp(1), cp(2), p(3), I[0], I[1], int(I[2]));
+      break;
+    case K_SOFTMAX:
+      launch_softmax(s, cp(0), p(1), I[0], I[1]);
+      break;
+    case K_LAYERNORM:
+      launch_layernorm(s, cp(0), cp(1), cp(2), p(3), I[0], I[1],
+                       c.floats[0]);
+      break;
+    case K_MEAN_LAST:
+      launch_reduce_mean_last(s, cp(0), p(1), I[0], I[1]);
+      break;
+    case K_MEAN_MID:
+      launch_reduce_mean_mid(s, cp(0), p(1), I[0], I[1], I[2]);
+      break;
+    case K_POOL:
+      launch_pool(s, cp(0), p(1), I[0] != 0, int(I[1]), int(I[2]),
+                  int(I[3]), int(I[4]), int(I[5]), int(I[6]), int(I[7]),
+                  int(I[8]), int(I[9]), int(I[10]), int(I[11]), int(I[12]));
+      break;
+    case K_TRANSPOSE: {
+      int ndim = int(I[0]);
+      launch_transpose(s, cp(0), p(1), ndim, I.data() + 1,
+                       I.data() + 1 + ndim, I[1 + 2 * ndim]);
+      break;
+    }
+    case K_GATHER:
+      launch_gather_rows(s, cp(0), reinterpret_cast<const int*>(c.ptrs[1]),
+                         p(2), I[0], I[1]);
+      break;
+    case K_PAD_NHWC:
+      launch_pad_nhwc(s, cp(0), p(1), int(I[0]), int(I[1]), int(I[2]),
+                      int(I[3]), int(I[4]), int(I[5]), int(I[6]),
+                      int(I[7]));
+      break;
+    case K_GEMM: {
+      const ushort* bias = c.ptrs[2] ? cp(2) : nullptr;
+      const ushort* res = c.ptrs[3] ? cp(3) : nullptr;
+      launch_gemm(s, cp(0), cp(1), bias, res, p(4), I[0], I[1], I[2],
+                  int(I[3]), c.floats[0]);
+      break;
+    }
+    case K_BGEMM:
+      launch_batched_gemm(s, cp(0), cp(1), p(2), I[0], I[1], I[2], I[3],
+                          I[4], I[5], I[6], I[7] != 0, c.floats[0]);
+      break;
+    case K_IM2COL:
+      launch_im2col(s, cp(0), p(1), int(I[0]), int(I[1]), int(I[2]),
+                    int(I[3]), int(I[4]), int(I[5]), int(I[6]), int(I[7]),
+                    int(I[8]), int(I[9]), int(I[10]), int(I[11]),
+                    int(I[12]));
+      break;
+    default:
+      throw std::runtime_error("unknown call kind " +
+                               std::to_string(c.kind));
+  }
+}
+
+class ExecPlan {
+ public:
+  explicit ExecPlan(const std::vector<Call>& calls) : calls_(calls) {}
+
+  void run() {
+    hipStream_t s = c10::hip::getCurrentHIPStream().stream();
+    for (const auto& c : calls_) launch_call(c, s);
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess)
+      throw std::runtime_error(std::string("kernel launch failed: ") +
+                               hipGetErrorString(e));
+  }
+
+  // capture the whole sequence into a hipGraph bound to fixed buffers;
+  // subsequent run_graph() replays it (no per-kernel launch cost)
+  void capture() {
+    hipStream_t s = c10::hip::getCurrentHIPStream().stream();
+    if (graph_exec_) return;
+    hipGraph_t graph = nullptr;
+    if (hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal) !=
+        hipSuccess)
+      throw std::runtime_error("hipStreamBeginCapture failed");
+    for (const auto& c : calls_) launch_call(c, s);
+    if (hipStreamEndCapture(s, &graph) != hipSuccess)
+      throw std::runtime_error("hipStreamEndCapture failed");
+    hipGraphExec_t ge = nullptr;
+    if (hipGraphInstantiate(&ge, graph, nullptr, nullptr, 0) != hipSuccess) {
+      hipGraphDestroy(graph);
+      throw std::runtime_error("hipGraphInstantiate failed");
+    }
+    hipGraphDestroy(graph);
+    graph_exec_ = ge;
+  }
+
+  bool has_graph() const { return graph_exec_ != nullptr; }
+
+  void run_graph() {
+    if (!graph_exec_) {
+      run();
+      return;
+    }
+    hipStream_t s = c10::hip::getCurrentHIPStream().stream();
+    if (hipGraphLaunch(graph_exec_, s) != hipSuccess)
+      throw std::runtime_error("hipGraphLaunch failed");
+  }
+
+  ~ExecPlan() {
+    if (graph_exec_) hipGraphExecDestroy(graph_exec_);
+  }
+
+  size_t n_calls() const { return calls_.size(); }
+
+ private:
+  std::vector<Call> calls_;
+  hipGraphExec_t graph_exec_ = nullptr;
+};
+
+}  // namespace tfsc
+
+namespace py = pybind11;
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  using tfsc::Call;
+  using tfsc::ExecPlan;
+
+  py::class_<ExecPlan>(mod, "ExecPlan")
+      .def(py::init([](py::list calls) {
+        std::vector<Call> cs;
+        cs.reserve(calls.size());
+        for (auto item : calls) {
+          py::tuple t = item.cast<py::tuple>();
+          Call c;
+          c.kind = t[0].cast<int>();
+          c.ptrs = t[1].cast<std::vector<intptr_t>>();
+          c.ints = t[2].cast<std::vector<int64_t>>();
+          c.floats = t[3].cast<std::vector<float>>();
+          cs.push_back(std::move(c));
+        }
+        return new ExecPlan(cs);
+      }))
+      .def("run", &ExecPlan::run, py::call_guard<py::gil_scoped_release>())
+      .def("capture", &ExecPlan::capture,
+           py::call_guard<py::gil_scoped_release>())
+      .def("run_graph", &ExecPlan::run_graph,
+           py::call_guard<py::gil_scoped_release>())
+      .def("has_graph", &ExecPlan::has_graph)
+      .def("n_calls", &ExecPlan::n_calls);
+
+  mod.attr("K_ELT_UNARY") = int(tfsc::K_ELT_UNARY);
+  mod.attr("K_ELT_BINARY") = int(tfsc::K_ELT_BINARY);
+  mod.attr("K_BN_ACT") = int(tfsc::K_BN_ACT);
+  mod.attr("K_SOFTMAX") = int(tfsc::K_SOFTMAX);
+  mod.attr("K_LAYERNORM") = int(tfsc::K_LAYERNORM);
+  mod.attr("K_MEAN_LAST") = int(tfsc::K_MEAN_LAST);
+  mod.attr("K_MEAN_MID") = int(tfsc::K_MEAN_MID);
+  mod.attr("K_POOL") = int(tfsc::K_POOL);
+  mod.attr("K_TRANSPOSE") = int(tfsc::K_TRANSPOSE);
+  mod.attr("K_GATHER") = int(tfsc::K_GATHER);
+  mod.attr("K_PAD_NHWC") = int(tfsc::K_PAD_NHWC);
+  mod.attr("K_GEMM") = int(tfsc::K_GEMM);
+  mod.attr("K_BGEMM") = int(tfsc::K_BGEMM);
+  mod.attr("K_IM2COL") = int(tfsc::K_IM2COL);
+
+  // elementwise fn codes
+  mod.attr("ELT_ADD") = int(tfsc::ELT_ADD);
+  mod.attr("ELT_SUB") = int(tfsc::ELT_SUB);
+  mod.attr("ELT_MUL") = int(tfsc::ELT_MUL);
+  mod.attr("ELT_DIV") = int(tfsc::ELT_DIV);
+  mod.attr("ELT_MAX") = int(tfsc::ELT_MAX);
+  mod.attr("ELT_MIN") = int(tfsc::ELT_MIN);
+  mod.attr("ELT_SQDIFF") = int(tfsc::ELT_SQDIFF);
+  mod.attr("ELT_RELU") = int(tfsc::ELT_RELU);
+  mod.attr("ELT_TANH") = int(tfsc::ELT_TANH);
+  mod.attr("ELT_SIGMOID") = int(tfsc::ELT_SIGMOID);
+  mod.attr("ELT_ERF") = int(tfsc::ELT_ERF);
+  mod.attr("ELT_SQRT") = int(tfsc::ELT_SQRT);
+  mod.attr("ELT_RSQRT") = int(tfsc::ELT_RSQRT);
+  mod.attr("ELT_EXP") = int(tfsc::ELT_EXP);
+  mod.attr("ELT_NEG") = int(tfsc::ELT_NEG);
+  mod.attr("ELT_SQUARE") = int(tfsc::ELT_SQUARE);
+  mod.attr("ELT_GELU") = int(tfsc::ELT_GELU);
+
+  mod.attr("ACT_NONE") = int(tfsc::ACT_NONE);
+  mod.attr("ACT_RELU") = int(tfsc::ACT_RELU);
+  mod.attr("ACT_TANH") = int(tfsc::ACT_TANH);
+  mod.attr("ACT_SIGMOID") = int(tfsc::ACT_SIGMOID);
+  mod.attr("ACT_GELU") = int(tfsc::ACT_GELU);
+}

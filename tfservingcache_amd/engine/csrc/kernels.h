@@ -1,0 +1,87 @@
+// Host-side kernel launcher declarations. All activation buffers are
+// bf16 (ushort) unless noted; integer buffers are int32. Launches are
+// asynchronous on the given stream.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+namespace tfsc {
+
+// elementwise fn codes (planner 'eltwise' op)
+enum EltFn : int {
+  ELT_ADD = 0, ELT_SUB, ELT_MUL, ELT_DIV, ELT_MAX, ELT_MIN, ELT_SQDIFF,
+  ELT_RELU, ELT_TANH, ELT_SIGMOID, ELT_ERF, ELT_SQRT, ELT_RSQRT, ELT_EXP,
+  ELT_NEG, ELT_SQUARE, ELT_GELU,
+};
+
+// activation codes for fused epilogues
+enum Act : int { ACT_NONE = 0, ACT_RELU, ACT_TANH, ACT_SIGMOID, ACT_GELU };
+
+constexpr int MAX_DIMS = 6;
+
+struct BcastArgs {          // broadcast strides (elements), 0 = broadcast
+  int ndim;
+  int64_t dims[MAX_DIMS];
+  int64_t sa[MAX_DIMS];
+  int64_t sb[MAX_DIMS];
+};
+
+void launch_eltwise_unary(hipStream_t s, const ushort* x, ushort* y,
+                          int64_t n, int fn);
+void launch_eltwise_binary(hipStream_t s, const ushort* a, const ushort* b,
+                           ushort* y, int64_t n_out, const BcastArgs& bc,
+                           int fn);
+void launch_bn_act(hipStream_t s, const ushort* x, const ushort* scale,
+                   const ushort* shift, ushort* y, int64_t rows, int64_t c,
+                   int act);
+void launch_softmax(hipStream_t s, const ushort* x, ushort* y,
+                    int64_t rows, int64_t cols);
+void launch_layernorm(hipStream_t s, const ushort* x, const ushort* gamma,
+                      const ushort* beta, ushort* y, int64_t rows,
+                      int64_t cols, float eps);
+void launch_reduce_mean_last(hipStream_t s, const ushort* x, ushort* y,
+                             int64_t rows, int64_t cols);
+// mean over axis 1 of [d0, d1, d2]
+void launch_reduce_mean_mid(hipStream_t s, const ushort* x, ushort* y,
+                            int64_t d0, int64_t d1, int64_t d2);
+// global mean over H,W of NHWC
+void launch_global_mean(hipStream_t s, const ushort* x, ushort* y,
+                        int64_t n, int64_t hw, int64_t c);
+void launch_pool(hipStream_t s, const ushort* x, ushort* y, bool is_max,
+                 int N, int H, int W, int C, int Ho, int Wo, int kh, int kw,
+                 int sh, int sw, int pt, int pl);
+void launch_transpose(hipStream_t s, const ushort* x, ushort* y, int ndim,
+                      const int64_t* out_dims, const int64_t* in_strides,
+                      int64_t n_out);
+void launch_gather_rows(hipStream_t s, const ushort* table, const int* idx,
+                        ushort* y, int64_t n_idx, int64_t row_elems);
+void launch_pad_nhwc(hipStream_t s, const ushort* x, ushort* y,
+                     int N, int H, int W, int C, int pt, int pb, int pl,
+                     int pr);
+
+// GEMM: C[M,N] = act(A[M,K] @ B[K,N] + bias [+ residual]).
+// A row-major bf16, B row-major bf16 (pre-transposed at load if the graph
+// wanted B^T), C bf16. f32 accumulate via MFMA.
+void launch_gemm(hipStream_t s, const ushort* A, const ushort* B,
+                 const ushort* bias, const ushort* residual, ushort* C,
+                 int64_t M, int64_t N, int64_t K, int act, float alpha);
+
+// Batched GEMM over leading batch dim: [bat, M, K] @ [bat, K, N] (operands
+// contiguous; strides in elements; stride 0 broadcasts an operand).
+void launch_batched_gemm(hipStream_t s, const ushort* A, const ushort* B,
+                         ushort* C, int64_t bat, int64_t M, int64_t N,
+                         int64_t K, int64_t strideA, int64_t strideB,
+                         int64_t strideC, bool trans_b, float alpha);
+
+// im2col for NHWC conv: out[N*Ho*Wo, K_pad] where the first R*S*C columns
+// hold the patch elements (r,s,c order, c fastest) and columns >= R*S*C
+// are zero (K_pad is the GEMM's 64-multiple). Out-of-image patch elements
+// are zero. The conv then runs as launch_gemm with pre-transposed weights
+// [Kc][K_pad].
+void launch_im2col(hipStream_t s, const ushort* x, ushort* y,
+                   int N, int H, int W, int C, int R, int S,
+                   int sh, int sw, int pt, int pl, int Ho, int Wo,
+                   int k_pad);
+
+}  // namespace tfsc

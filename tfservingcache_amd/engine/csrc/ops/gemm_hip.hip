@@ -1,0 +1,275 @@
+#include "hip/hip_runtime.h"
+// bf16 MFMA GEMM for gfx950 — the serving hot path.
+//
+// C[M,N] = act(alpha * A[M,K] @ B^T[N,K] + bias [+ residual])
+//
+// Structure follows the CDNA4 canonical GEMM anatomy
+// (/opt/skills/guides/cdna_hip_programming.md §5): 128x128 block tile,
+// BK=64, 4 waves (one 64x64 wave tile each), double-buffered LDS filled
+// with 16-byte `global_load_lds` (lane-linear dest, XOR-swizzled SOURCE
+// address + matching XOR on the ds_read — §5.4 rule 21 / T2), MFMA
+// v_mfma_f32_16x16x32_bf16 accumulating f32, fused
+// bias/activation/residual epilogue.
+//
+// B is expected TRANSPOSED ([N][K] row-major): the planner pre-transposes
+// weights at model-load time so both A and B fragments read contiguously
+// along K. A variant with B in [K][N] (register-staged transpose on the
+// fly) serves the batched attention P@V case.
+//
+// Row indices are clamped at the staging loads (no out-of-bounds reads on
+// partial tiles); stores are masked by (row < M && col < N). K must be a
+// multiple of 64 — the planner zero-pads weights and im2col buffers.
+#include "../common.h"
+#include "../kernels.h"
+
+#include <stdexcept>
+
+namespace tfsc {
+
+using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4_t = __attribute__((ext_vector_type(4))) float;
+
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int N_WAVES = 4;                // 2x2 wave grid of 64x64 tiles
+constexpr int THREADS = N_WAVES * WAVE;   // 256
+
+// LDS tile: [128 rows][64 cols] bf16, 128-byte rows = 8 chunks of 16B.
+// XOR swizzle: chunk' = chunk ^ (row & 7)  (T2: <=2-way conflicts).
+TFSC_DEV int lds_off(int row, int chunk) {   // byte offset into a tile
+  return row * 128 + ((chunk ^ (row & 7)) << 4);
+}
+
+TFSC_DEV void stage_tile_glds(const ushort* __restrict__ src, int64_t ld,
+                              int row_limit, int row0, int k0,
+                              char* lds_tile, int wave, int lane) {
+  // one wave stages 32 rows (4 glds x 8 rows); lane l covers
+  // (row = 8*i + l/8, chunk = l%8) of its wave's 32-row slice. The glds
+  // LDS destination is wave-uniform-base + lane*16 (lane-linear), which
+  // matches the [row][chunk] image exactly.
+  int r_in = (lane >> 3);           // 0..7
+  int chunk = lane & 7;             // 0..7
+  #pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row = wave * 32 + i * 8 + r_in;
+    int grow = row0 + row;
+    grow = grow < row_limit ? grow : row_limit;
+    // source chunk is pre-swizzled so the LINEAR lds image holds the
+    // swizzled layout (rule 21: swizzle source + read, never the dest)
+    int chunk_src = chunk ^ (row & 7);
+    const ushort* gptr = src + (int64_t)grow * ld + k0 + chunk_src * 8;
+    // wave-uniform base for this glds: start of the wave's 8-row slice
+    char* lds_base = lds_tile + (wave * 32 + i * 8) * 128;
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const uint32_t*>(gptr),
+        reinterpret_cast<uint32_t*>(lds_base), 16, 0, 0);
+  }
+}
+
+// register-staged transpose path for B given as [K][N] (attention P@V):
+// each lane loads 8 bf16 along N (coalesced) and scatters them into the
+// [N][K]-image LDS tile with the same XOR swizzle.
+TFSC_DEV void stage_tile_transposed(const ushort* __restrict__ src,
+                                    int64_t ld, int k_limit, int n0, int k0,
+                                    ushort* lds_tile, int wave, int lane,
+                                    int n_limit) {
+  // tile wanted: rows = n (128), cols = k (64). Source element (k, n).
+  // 256 threads x 8 elems = 2048 elems/pass; tile = 8192 elems -> 4 passes.
+  int tid = wave * WAVE + lane;
+  #pragma unroll
+  for (int pass = 0; pass < 4; ++pass) {
+    int idx = pass * THREADS + tid;       // covers k-major: 64 k x 16 ngroups
+    int k = idx >> 4;                     // 0..63
+    int ng = idx & 15;                    // n-group of 8
+    int gk = k0 + k;
+    gk = gk < k_limit ? gk : k_limit;
+    const ushort* gptr = src + (int64_t)gk * ld + n0 + ng * 8;
+    ushort vals[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int gn = n0 + ng * 8 + j;
+      vals[j] = (gn < n_limit) ? gptr[j] : ushort(0);
+    }
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int n = ng * 8 + j;
+      int byte_off = lds_off(n, k >> 3) + (k & 7) * 2;
+      *reinterpret_cast<ushort*>(
+          reinterpret_cast<char*>(lds_tile) + byte_off) = vals[j];
+    }
+  }
+}
+
+TFSC_DEV float act_apply(float v, int act) {
+  switch (act) {
+    case ACT_RELU: return v > 0.f ? v : 0.f;
+    case ACT_TANH: return tanhf(v);
+    case ACT_SIGMOID: return 1.f / (1.f + __expf(-v));
+    case ACT_GELU: return 0.5f * v * (1.f + erff(v * 0.70710678f));
+    default: return v;
+  }
+}
+
+template <bool TRANS_B, bool HAS_BIAS, bool HAS_RES>
+__global__ __launch_bounds__(THREADS)
+void gemm_bf16_kernel(const ushort* __restrict__ A,
+                      const ushort* __restrict__ B,
+                      const ushort* __restrict__ bias,
+                      const ushort* __restrict__ residual,
+                      ushort* __restrict__ C,
+                      int M, int N, int K, int act, float alpha,
+                      int64_t strideA, int64_t strideB, int64_t strideC,
+                      int n_tiles_m) {
+  __shared__ __attribute__((aligned(16))) char smem[4 * BM * BK * 2];
+  // [buf][A|B] tiles, 16KB each: A0 | B0 | A1 | B1
+  auto lds_a = [&](int buf) -> char* { return smem + buf * 32768; };
+  auto lds_b = [&](int buf) -> char* { return smem + 16384 + buf * 32768; };
+
+  const int bat = blockIdx.y;
+  const ushort* Ab = A + bat * strideA;
+  const ushort* Bb = B + bat * strideB;
+  ushort* Cb = C + bat * strideC;
+
+  const int tile_m = blockIdx.x % n_tiles_m;
+  const int tile_n = blockIdx.x / n_tiles_m;
+  const int m0 = tile_m * BM;
+  const int n0 = tile_n * BN;
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int wm = wave >> 1, wn = wave & 1;      // 2x2 wave grid
+
+  f32x4_t acc[4][4] = {};
+
+  const int n_ktiles = K / BK;
+
+  // prologue: stage tile 0 into buf 0
+  {
+    stage_tile_glds(Ab, K, M - 1, m0, 0, lds_a(0), wave, lane);
+    if (TRANS_B) {
+      stage_tile_glds(Bb, K, N - 1, n0, 0, lds_b(0), wave, lane);
+    } else {
+      stage_tile_transposed(Bb, N, K - 1, n0, 0,
+                            reinterpret_cast<ushort*>(lds_b(0)), wave, lane,
+                            N);
+    }
+  }
+
+  // 2-phase schedule (T3 minimum form, cdna_hip_programming.md §5.5):
+  // one barrier per K-tile; the next tile's staging loads are issued
+  // BEFORE this tile's ds_read+MFMA so the glds overlaps the compute.
+  // __syncthreads() emits the vmcnt(0) drain for the in-flight glds.
+  int cur = 0;
+  for (int kt = 0; kt < n_ktiles; ++kt) {
+    __syncthreads();
+    if (kt + 1 < n_ktiles) {
+      int k0 = (kt + 1) * BK;
+      stage_tile_glds(Ab, K, M - 1, m0, k0, lds_a(cur ^ 1), wave, lane);
+      if (TRANS_B) {
+        stage_tile_glds(Bb, K, N - 1, n0, k0, lds_b(cur ^ 1), wave, lane);
+      } else {
+        stage_tile_transposed(Bb, N, K - 1, n0, k0,
+                              reinterpret_cast<ushort*>(lds_b(cur ^ 1)),
+                              wave, lane, N);
+      }
+    }
+
+    const char* at = lds_a(cur);
+    const char* bt = lds_b(cur);
+    const int frow = lane & 15;           // fragment row/col within 16
+    const int kgrp = lane >> 4;           // 0..3
+
+    #pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {      // two K=32 steps per 64-tile
+      bf16x8_t a_frag[4], b_frag[4];
+      const int chunk = ks * 4 + kgrp;
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        int row = wm * 64 + mi * 16 + frow;
+        a_frag[mi] = *reinterpret_cast<const bf16x8_t*>(
+            at + lds_off(row, chunk));
+      }
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = wn * 64 + ni * 16 + frow;
+        b_frag[ni] = *reinterpret_cast<const bf16x8_t*>(
+            bt + lds_off(row, chunk));
+      }
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    cur ^= 1;
+  }
+
+  // epilogue: C/D layout for 16x16x32: col = lane&15,
+  // row = (lane>>4)*4 + reg (cdna_hip_programming.md §3)
+  const int col_in = lane & 15;
+  const int row_base = (lane >> 4) * 4;
+  #pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int gcol = n0 + wn * 64 + ni * 16 + col_in;
+      if (gcol >= N) continue;
+      float bv = HAS_BIAS ? bf2f(bias[gcol]) : 0.f;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int grow = m0 + wm * 64 + mi * 16 + row_base + r;
+        if (grow >= M) continue;
+        float v = acc[mi][ni][r] * alpha + bv;
+        if (HAS_RES)
+          v += bf2f(residual[(int64_t)grow * N + gcol]);
+        Cb[(int64_t)grow * N + gcol] = f2bf(act_apply(v, act));
+      }
+    }
+  }
+}
+
+template <bool TRANS_B>
+static void gemm_dispatch(hipStream_t s, const ushort* A, const ushort* B,
+                          const ushort* bias, const ushort* residual,
+                          ushort* C, int64_t bat, int64_t M, int64_t N,
+                          int64_t K, int act, float alpha, int64_t sA,
+                          int64_t sB, int64_t sC) {
+  if (K % BK != 0)
+    throw std::runtime_error("gemm: K must be a multiple of 64 (got " +
+                             std::to_string(K) + ")");
+  int ntm = int(ceil_div(M, BM)), ntn = int(ceil_div(N, BN));
+  dim3 grid(ntm * ntn, (unsigned)bat);
+  dim3 block(THREADS);
+  bool hb = bias != nullptr, hr = residual != nullptr;
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, block, 0, s, A, B, bias, residual, C,
+                       (int)M, (int)N, (int)K, act, alpha, sA, sB, sC, ntm);
+  };
+  if (hb && hr)       launch(gemm_bf16_kernel<TRANS_B, true, true>);
+  else if (hb)        launch(gemm_bf16_kernel<TRANS_B, true, false>);
+  else if (hr)        launch(gemm_bf16_kernel<TRANS_B, false, true>);
+  else                launch(gemm_bf16_kernel<TRANS_B, false, false>);
+}
+
+void launch_gemm(hipStream_t s, const ushort* A, const ushort* B,
+                 const ushort* bias, const ushort* residual, ushort* C,
+                 int64_t M, int64_t N, int64_t K, int act, float alpha) {
+  gemm_dispatch<true>(s, A, B, bias, residual, C, 1, M, N, K, act, alpha,
+                      0, 0, 0);
+}
+
+void launch_batched_gemm(hipStream_t s, const ushort* A, const ushort* B,
+                         ushort* C, int64_t bat, int64_t M, int64_t N,
+                         int64_t K, int64_t strideA, int64_t strideB,
+                         int64_t strideC, bool trans_b, float alpha) {
+  if (trans_b)
+    gemm_dispatch<true>(s, A, B, nullptr, nullptr, C, bat, M, N, K,
+                        ACT_NONE, alpha, strideA, strideB, strideC);
+  else
+    gemm_dispatch<false>(s, A, B, nullptr, nullptr, C, bat, M, N, K,
+                         ACT_NONE, alpha, strideA, strideB, strideC);
+}
+
+}  // namespace tfsc

@@ -1,0 +1,441 @@
+#include "hip/hip_runtime.h"
+// Memory-bound serving kernels for gfx950: elementwise, fused BN+act,
+// softmax, layernorm, reductions, pooling, transpose, gather, pad.
+//
+// All are HBM-bandwidth-bound: bf16 I/O is vectorized as ushort4/8 where
+// the layout permits (cdna_hip_programming.md G13), grids are capped and
+// grid-stride (G11), accumulation is f32.
+#include "../common.h"
+#include "../kernels.h"
+
+#include <stdexcept>
+#include <string>
+
+namespace tfsc {
+
+constexpr int TPB = 256;
+constexpr int MAX_BLOCKS = 2048;   // 256 CU * 8 blocks/CU
+
+static inline int grid_for(int64_t n, int per_thread = 1) {
+  int64_t blocks = ceil_div(n, int64_t(TPB) * per_thread);
+  return int(blocks < MAX_BLOCKS ? (blocks > 0 ? blocks : 1) : MAX_BLOCKS);
+}
+
+// ---------------------------------------------------------------------------
+// elementwise
+// ---------------------------------------------------------------------------
+TFSC_DEV float apply_unary(float v, int fn) {
+  switch (fn) {
+    case ELT_RELU: return v > 0.f ? v : 0.f;
+    case ELT_TANH: return tanhf(v);
+    case ELT_SIGMOID: return 1.f / (1.f + __expf(-v));
+    case ELT_ERF: return erff(v);
+    case ELT_SQRT: return sqrtf(v);
+    case ELT_RSQRT: return rsqrtf(v);
+    case ELT_EXP: return __expf(v);
+    case ELT_NEG: return -v;
+    case ELT_SQUARE: return v * v;
+    case ELT_GELU: return 0.5f * v * (1.f + erff(v * 0.70710678f));
+    default: return v;
+  }
+}
+
+TFSC_DEV float apply_binary(float a, float b, int fn) {
+  switch (fn) {
+    case ELT_ADD: return a + b;
+    case ELT_SUB: return a - b;
+    case ELT_MUL: return a * b;
+    case ELT_DIV: return a / b;
+    case ELT_MAX: return fmaxf(a, b);
+    case ELT_MIN: return fminf(a, b);
+    case ELT_SQDIFF: { float d = a - b; return d * d; }
+    default: return a;
+  }
+}
+
+__global__ void k_unary(const ushort* __restrict__ x, ushort* __restrict__ y,
+                        int64_t n, int fn) {
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  // vectorized over 4 bf16 per thread
+  int64_t n4 = n / 4;
+  for (int64_t i = i0; i < n4; i += stride) {
+    short4_t v = reinterpret_cast<const short4_t*>(x)[i];
+    short4_t r;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      r[j] = (short)f2bf(apply_unary(bf2f((ushort)v[j]), fn));
+    reinterpret_cast<short4_t*>(y)[i] = r;
+  }
+  for (int64_t i = n4 * 4 + i0; i < n; i += stride)
+    y[i] = f2bf(apply_unary(bf2f(x[i]), fn));
+}
+
+__global__ void k_binary_same(const ushort* __restrict__ a,
+                              const ushort* __restrict__ b,
+                              ushort* __restrict__ y, int64_t n, int fn) {
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  int64_t n4 = n / 4;
+  for (int64_t i = i0; i < n4; i += stride) {
+    short4_t va = reinterpret_cast<const short4_t*>(a)[i];
+    short4_t vb = reinterpret_cast<const short4_t*>(b)[i];
+    short4_t r;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      r[j] = (short)f2bf(apply_binary(bf2f((ushort)va[j]),
+                                      bf2f((ushort)vb[j]), fn));
+    reinterpret_cast<short4_t*>(y)[i] = r;
+  }
+  for (int64_t i = n4 * 4 + i0; i < n; i += stride)
+    y[i] = f2bf(apply_binary(bf2f(a[i]), bf2f(b[i]), fn));
+}
+
+__global__ void k_binary_bcast(const ushort* __restrict__ a,
+                               const ushort* __restrict__ b,
+                               ushort* __restrict__ y, int64_t n,
+                               BcastArgs bc, int fn) {
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n; i += stride) {
+    int64_t rem = i, ia = 0, ib = 0;
+    #pragma unroll
+    for (int d = MAX_DIMS - 1; d >= 0; --d) {
+      if (d >= bc.ndim) continue;
+      int64_t c = rem % bc.dims[d];
+      rem /= bc.dims[d];
+      ia += c * bc.sa[d];
+      ib += c * bc.sb[d];
+    }
+    y[i] = f2bf(apply_binary(bf2f(a[ia]), bf2f(b[ib]), fn));
+  }
+}
+
+void launch_eltwise_unary(hipStream_t s, const ushort* x, ushort* y,
+                          int64_t n, int fn) {
+  hipLaunchKernelGGL(k_unary, dim3(grid_for(n, 4)), dim3(TPB), 0, s,
+                     x, y, n, fn);
+}
+
+void launch_eltwise_binary(hipStream_t s, const ushort* a, const ushort* b,
+                           ushort* y, int64_t n_out, const BcastArgs& bc,
+                           int fn) {
+  bool same = true;
+  for (int d = 0; d < bc.ndim; ++d) {
+    // contiguous same-shape iff both stride patterns are the canonical
+    // contiguous ones (marked by sa/sb equal to out strides; python sets
+    // sa=sb=contig in that case and ndim=1)
+  }
+  if (bc.ndim == 1 && bc.sa[0] == 1 && bc.sb[0] == 1) {
+    hipLaunchKernelGGL(k_binary_same, dim3(grid_for(n_out, 4)), dim3(TPB),
+                       0, s, a, b, y, n_out, fn);
+  } else {
+    hipLaunchKernelGGL(k_binary_bcast, dim3(grid_for(n_out)), dim3(TPB),
+                       0, s, a, b, y, n_out, bc, fn);
+  }
+  (void)same;
+}
+
+// ---------------------------------------------------------------------------
+// fused scale/shift (+act): y = act(x * scale[c] + shift[c]) over [rows, C]
+// ---------------------------------------------------------------------------
+__global__ void k_bn_act(const ushort* __restrict__ x,
+                         const ushort* __restrict__ scale,
+                         const ushort* __restrict__ shift,
+                         ushort* __restrict__ y, int64_t n, int64_t c,
+                         int act) {
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n; i += stride) {
+    int64_t ch = i % c;
+    float v = bf2f(x[i]) * bf2f(scale[ch]) + bf2f(shift[ch]);
+    switch (act) {
+      case ACT_RELU: v = v > 0.f ? v : 0.f; break;
+      case ACT_TANH: v = tanhf(v); break;
+      case ACT_SIGMOID: v = 1.f / (1.f + __expf(-v)); break;
+      case ACT_GELU: v = 0.5f * v * (1.f + erff(v * 0.70710678f)); break;
+    }
+    y[i] = f2bf(v);
+  }
+}
+
+void launch_bn_act(hipStream_t s, const ushort* x, const ushort* scale,
+                   const ushort* shift, ushort* y, int64_t rows, int64_t c,
+                   int act) {
+  int64_t n = rows * c;
+  hipLaunchKernelGGL(k_bn_act, dim3(grid_for(n)), dim3(TPB), 0, s,
+                     x, scale, shift, y, n, c, act);
+}
+
+// ---------------------------------------------------------------------------
+// row softmax: one block per row (grid-stride over rows), online max+sum
+// ---------------------------------------------------------------------------
+template <typename Reduce>
+TFSC_DEV float block_reduce(float v, Reduce red, float init) {
+  __shared__ float smem[TPB / WAVE];
+  #pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    v = red(v, __shfl_down(v, off, WAVE));
+  int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) smem[wid] = v;
+  __syncthreads();
+  v = (threadIdx.x < TPB / WAVE) ? smem[threadIdx.x] : init;
+  #pragma unroll
+  for (int off = TPB / WAVE / 2; off > 0; off >>= 1)
+    v = red(v, __shfl_down(v, off, WAVE));
+  v = __shfl(v, 0, WAVE);
+  if (threadIdx.x == 0) smem[0] = v;
+  __syncthreads();
+  v = smem[0];
+  __syncthreads();
+  return v;
+}
+
+struct MaxOp { TFSC_DEV float operator()(float a, float b) const { return fmaxf(a, b); } };
+struct SumOp { TFSC_DEV float operator()(float a, float b) const { return a + b; } };
+
+__global__ void k_softmax(const ushort* __restrict__ x,
+                          ushort* __restrict__ y, int64_t rows,
+                          int64_t cols) {
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const ushort* xr = x + row * cols;
+    ushort* yr = y + row * cols;
+    float mx = -3.0e38f;
+    for (int64_t i = threadIdx.x; i < cols; i += TPB)
+      mx = fmaxf(mx, bf2f(xr[i]));
+    mx = block_reduce(mx, MaxOp(), -3.0e38f);
+    float sum = 0.f;
+    for (int64_t i = threadIdx.x; i < cols; i += TPB)
+      sum += __expf(bf2f(xr[i]) - mx);
+    sum = block_reduce(sum, SumOp(), 0.f);
+    float inv = 1.f / sum;
+    for (int64_t i = threadIdx.x; i < cols; i += TPB)
+      yr[i] = f2bf(__expf(bf2f(xr[i]) - mx) * inv);
+  }
+}
+
+void launch_softmax(hipStream_t s, const ushort* x, ushort* y,
+                    int64_t rows, int64_t cols) {
+  int blocks = int(rows < MAX_BLOCKS ? rows : MAX_BLOCKS);
+  hipLaunchKernelGGL(k_softmax, dim3(blocks), dim3(TPB), 0, s,
+                     x, y, rows, cols);
+}
+
+// ---------------------------------------------------------------------------
+// layernorm (last dim): one block per row
+// ---------------------------------------------------------------------------
+__global__ void k_layernorm(const ushort* __restrict__ x,
+                            const ushort* __restrict__ gamma,
+                            const ushort* __restrict__ beta,
+                            ushort* __restrict__ y, int64_t rows,
+                            int64_t cols, float eps) {
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const ushort* xr = x + row * cols;
+    ushort* yr = y + row * cols;
+    float sum = 0.f, sq = 0.f;
+    for (int64_t i = threadIdx.x; i < cols; i += TPB) {
+      float v = bf2f(xr[i]);
+      sum += v;
+      sq += v * v;
+    }
+    sum = block_reduce(sum, SumOp(), 0.f);
+    sq = block_reduce(sq, SumOp(), 0.f);
+    float mean = sum / cols;
+    float var = sq / cols - mean * mean;
+    float rstd = rsqrtf(var > 0.f ? var + eps : eps);
+    for (int64_t i = threadIdx.x; i < cols; i += TPB) {
+      float v = (bf2f(xr[i]) - mean) * rstd;
+      yr[i] = f2bf(v * bf2f(gamma[i]) + bf2f(beta[i]));
+    }
+  }
+}
+
+void launch_layernorm(hipStream_t s, const ushort* x, const ushort* gamma,
+                      const ushort* beta, ushort* y, int64_t rows,
+                      int64_t cols, float eps) {
+  int blocks = int(rows < MAX_BLOCKS ? rows : MAX_BLOCKS);
+  hipLaunchKernelGGL(k_layernorm, dim3(blocks), dim3(TPB), 0, s,
+                     x, gamma, beta, y, rows, cols, eps);
+}
+
+// ---------------------------------------------------------------------------
+// reductions
+// ---------------------------------------------------------------------------
+__global__ void k_mean_last(const ushort* __restrict__ x,
+                            ushort* __restrict__ y, int64_t rows,
+                            int64_t cols) {
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const ushort* xr = x + row * cols;
+    float sum = 0.f;
+    for (int64_t i = threadIdx.x; i < cols; i += TPB)
+      sum += bf2f(xr[i]);
+    sum = block_reduce(sum, SumOp(), 0.f);
+    if (threadIdx.x == 0) y[row] = f2bf(sum / cols);
+  }
+}
+
+void launch_reduce_mean_last(hipStream_t s, const ushort* x, ushort* y,
+                             int64_t rows, int64_t cols) {
+  int blocks = int(rows < MAX_BLOCKS ? rows : MAX_BLOCKS);
+  hipLaunchKernelGGL(k_mean_last, dim3(blocks), dim3(TPB), 0, s,
+                     x, y, rows, cols);
+}
+
+// mean over axis 1 of [d0, d1, d2]: out[d0, d2]. Threads over (d0, d2) —
+// coalesced on d2.
+__global__ void k_mean_mid(const ushort* __restrict__ x,
+                           ushort* __restrict__ y, int64_t d0, int64_t d1,
+                           int64_t d2) {
+  int64_t n = d0 * d2;
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n; i += stride) {
+    int64_t o = i / d2, c = i % d2;
+    const ushort* xp = x + o * d1 * d2 + c;
+    float sum = 0.f;
+    for (int64_t j = 0; j < d1; ++j) sum += bf2f(xp[j * d2]);
+    y[i] = f2bf(sum / d1);
+  }
+}
+
+void launch_reduce_mean_mid(hipStream_t s, const ushort* x, ushort* y,
+                            int64_t d0, int64_t d1, int64_t d2) {
+  hipLaunchKernelGGL(k_mean_mid, dim3(grid_for(d0 * d2)), dim3(TPB), 0, s,
+                     x, y, d0, d1, d2);
+}
+
+void launch_global_mean(hipStream_t s, const ushort* x, ushort* y,
+                        int64_t n, int64_t hw, int64_t c) {
+  launch_reduce_mean_mid(s, x, y, n, hw, c);
+}
+
+// ---------------------------------------------------------------------------
+// pooling (NHWC; one thread per output element, coalesced over C)
+// ---------------------------------------------------------------------------
+__global__ void k_pool(const ushort* __restrict__ x, ushort* __restrict__ y,
+                       bool is_max, int N, int H, int W, int C, int Ho,
+                       int Wo, int kh, int kw, int sh, int sw, int pt,
+                       int pl) {
+  int64_t n_out = (int64_t)N * Ho * Wo * C;
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n_out; i += stride) {
+    int c = int(i % C);
+    int64_t t = i / C;
+    int wo = int(t % Wo); t /= Wo;
+    int ho = int(t % Ho); int n = int(t / Ho);
+    float acc = is_max ? -3.0e38f : 0.f;
+    int count = 0;
+    for (int r = 0; r < kh; ++r) {
+      int hi = ho * sh + r - pt;
+      if (hi < 0 || hi >= H) continue;
+      for (int q = 0; q < kw; ++q) {
+        int wi = wo * sw + q - pl;
+        if (wi < 0 || wi >= W) continue;
+        float v = bf2f(x[(((int64_t)n * H + hi) * W + wi) * C + c]);
+        if (is_max) acc = fmaxf(acc, v);
+        else acc += v;
+        ++count;
+      }
+    }
+    y[i] = f2bf(is_max ? acc : (count ? acc / count : 0.f));
+  }
+}
+
+void launch_pool(hipStream_t s, const ushort* x, ushort* y, bool is_max,
+                 int N, int H, int W, int C, int Ho, int Wo, int kh, int kw,
+                 int sh, int sw, int pt, int pl) {
+  int64_t n_out = (int64_t)N * Ho * Wo * C;
+  hipLaunchKernelGGL(k_pool, dim3(grid_for(n_out)), dim3(TPB), 0, s,
+                     x, y, is_max, N, H, W, C, Ho, Wo, kh, kw, sh, sw,
+                     pt, pl);
+}
+
+// ---------------------------------------------------------------------------
+// transpose / gather / pad
+// ---------------------------------------------------------------------------
+struct TransArgs {
+  int ndim;
+  int64_t out_dims[MAX_DIMS];
+  int64_t in_strides[MAX_DIMS];   // input element stride per OUTPUT dim
+};
+
+__global__ void k_transpose(const ushort* __restrict__ x,
+                            ushort* __restrict__ y, TransArgs ta,
+                            int64_t n) {
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n; i += stride) {
+    int64_t rem = i, src = 0;
+    #pragma unroll
+    for (int d = MAX_DIMS - 1; d >= 0; --d) {
+      if (d >= ta.ndim) continue;
+      int64_t c = rem % ta.out_dims[d];
+      rem /= ta.out_dims[d];
+      src += c * ta.in_strides[d];
+    }
+    y[i] = x[src];
+  }
+}
+
+void launch_transpose(hipStream_t s, const ushort* x, ushort* y, int ndim,
+                      const int64_t* out_dims, const int64_t* in_strides,
+                      int64_t n_out) {
+  TransArgs ta;
+  ta.ndim = ndim;
+  for (int d = 0; d < MAX_DIMS; ++d) {
+    ta.out_dims[d] = d < ndim ? out_dims[d] : 1;
+    ta.in_strides[d] = d < ndim ? in_strides[d] : 0;
+  }
+  hipLaunchKernelGGL(k_transpose, dim3(grid_for(n_out)), dim3(TPB), 0, s,
+                     x, y, ta, n_out);
+}
+
+__global__ void k_gather_rows(const ushort* __restrict__ table,
+                              const int* __restrict__ idx,
+                              ushort* __restrict__ y, int64_t n_idx,
+                              int64_t row) {
+  int64_t n = n_idx * row;
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n; i += stride) {
+    int64_t r = i / row, c = i % row;
+    y[i] = table[(int64_t)idx[r] * row + c];
+  }
+}
+
+void launch_gather_rows(hipStream_t s, const ushort* table, const int* idx,
+                        ushort* y, int64_t n_idx, int64_t row_elems) {
+  hipLaunchKernelGGL(k_gather_rows, dim3(grid_for(n_idx * row_elems)),
+                     dim3(TPB), 0, s, table, idx, y, n_idx, row_elems);
+}
+
+__global__ void k_pad_nhwc(const ushort* __restrict__ x,
+                           ushort* __restrict__ y, int N, int H, int W,
+                           int C, int pt, int pl, int Ho, int Wo) {
+  int64_t n_out = (int64_t)N * Ho * Wo * C;
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n_out; i += stride) {
+    int c = int(i % C);
+    int64_t t = i / C;
+    int wo = int(t % Wo); t /= Wo;
+    int ho = int(t % Ho); int n = int(t / Ho);
+    int hi = ho - pt, wi = wo - pl;
+    ushort v = 0;
+    if (hi >= 0 && hi < H && wi >= 0 && wi < W)
+      v = x[(((int64_t)n * H + hi) * W + wi) * C + c];
+    y[i] = v;
+  }
+}
+
+void launch_pad_nhwc(hipStream_t s, const ushort* x, ushort* y,
+                     int N, int H, int W, int C, int pt, int pb, int pl,
+                     int pr) {
+  int Ho = H + pt + pb, Wo = W + pl + pr;
+  int64_t n_out = (int64_t)N * Ho * Wo * C;
+  hipLaunchKernelGGL(k_pad_nhwc, dim3(grid_for(n_out)), dim3(TPB), 0, s,
+                     x, y, N, H, W, C, pt, pl, Ho, Wo);
+}
+
+}  // namespace tfsc

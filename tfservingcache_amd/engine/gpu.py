@@ -1,0 +1,645 @@
+"""GPU lowering: Plan -> CDNA4 kernel-call sequences.
+
+Responsibilities (all ahead of the hot path):
+  * weight layout transforms at model-load time — GEMM/conv weights are
+    pre-transposed to [N][K] bf16 with K zero-padded to a multiple of 64
+    so both MFMA fragments read contiguously along K (see csrc/ops/gemm.hip);
+  * per-batch-bucket ExecContexts: shapes resolved, every activation
+    assigned an offset in one workspace arena (liveness-based reuse),
+    kernel calls emitted with raw pointers; optionally hipGraph-captured;
+  * the per-request path is: H2D feed copies -> ExecPlan.run ->
+    D2H fetches. Weights stay resident (the HBM3E model pool).
+
+Torch is used for memory management and H2D/D2H only — all math runs in
+the hand-written HIP kernels (the extension fails loudly if missing on a
+GPU box; no silent eager fallback).
+"""
+from __future__ import annotations
+
+import logging
+import math
+import threading
+from dataclasses import dataclass
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+from .planner import Plan, PlanOp, is_sym, resolve_dim
+
+log = logging.getLogger("tfsc.gpu")
+
+_torch = None
+_ext = None
+_import_lock = threading.Lock()
+
+
+class GpuUnavailable(RuntimeError):
+    pass
+
+
+def _load_backend():
+    """Import torch + the HIP extension. Loud failure on GPU boxes."""
+    global _torch, _ext
+    with _import_lock:
+        if _ext is not None:
+            return _torch, _ext
+        import torch
+        if not torch.cuda.is_available():
+            raise GpuUnavailable("no ROCm GPU visible")
+        try:
+            from . import _tfsc_engine as ext
+        except ImportError as e:
+            raise GpuUnavailable(
+                "GPU present but the tfsc HIP engine extension is not "
+                "built — run `PYTORCH_ROCM_ARCH=gfx950 python setup.py "
+                f"build_ext --inplace` ({e})") from e
+        _torch, _ext = torch, ext
+        return torch, ext
+
+
+def gpu_available() -> bool:
+    try:
+        _load_backend()
+        return True
+    except Exception:       # noqa: BLE001
+        return False
+
+
+ALIGN = 256  # byte alignment of workspace slices
+
+
+def _pad64(k: int) -> int:
+    return (k + 63) // 64 * 64
+
+
+_ACT_CODE = {"none": 0, "relu": 1, "tanh": 2, "sigmoid": 3, "gelu": 4}
+
+_ELT_CODE = {
+    "add": 0, "sub": 1, "mul": 2, "div": 3, "max": 4, "min": 5,
+    "sqdiff": 6, "relu": 7, "tanh": 8, "sigmoid": 9, "erf": 10,
+    "sqrt": 11, "rsqrt": 12, "exp": 13, "neg": 14, "square": 15,
+    "gelu": 16,
+}
+
+
+@dataclass
+class _BufInfo:
+    offset: int = -1          # byte offset in workspace
+    nbytes: int = 0
+    shape: Tuple[int, ...] = ()
+    is_int: bool = False
+
+
+class ExecContext:
+    """Shapes + workspace + calls for one batch bucket."""
+
+    def __init__(self, gm: "GpuModel", batch: int):
+        self.gm = gm
+        self.batch = batch
+        torch, ext = _load_backend()
+        plan = gm.plan
+        dev = gm.device
+
+        self.shapes: List[Tuple[int, ...]] = [
+            plan.resolve_shape(t.shape, batch) for t in plan.tensors]
+        self.root = [t.alias_of if t.alias_of is not None else t.idx
+                     for t in plan.tensors]
+        # resolve alias chains
+        for i, r in enumerate(self.root):
+            seen = 0
+            while plan.tensors[r].alias_of is not None and seen < 16:
+                r = plan.tensors[r].alias_of
+                seen += 1
+            self.root[i] = r
+
+        self._plan_buffers()
+        self.workspace = torch.empty(self.total_bytes, dtype=torch.uint8,
+                                     device=dev)
+        calls = self._emit_calls()
+        self.exec_plan = ext.ExecPlan(calls)
+        self.captured = False
+        self._views: Dict[int, object] = {}
+        # dedicated non-default stream (the default stream cannot be
+        # hipGraph-captured; copies + kernels + D2H all run here)
+        self.stream = torch.cuda.Stream(device=dev)
+
+    # -- buffer planning ---------------------------------------------------
+    def _buf_bytes(self, idx: int) -> int:
+        t = self.gm.plan.tensors[idx]
+        n = int(np.prod(self.shapes[idx])) if self.shapes[idx] else 1
+        esize = 4 if t.dtype == "i32" else 2
+        return max(n * esize, esize)
+
+    def _plan_buffers(self) -> None:
+        plan = self.gm.plan
+        live_end: Dict[int, int] = {}
+        scratch_sizes: Dict[int, int] = {}   # op_idx -> scratch bytes
+
+        def root(i):
+            return self.root[i]
+
+        n_ops = len(plan.ops)
+        for oi, op in enumerate(plan.ops):
+            for i in op.inputs + op.outputs:
+                live_end[root(i)] = oi
+        for i in plan.sig_outputs.values():
+            live_end[root(i)] = n_ops + 1
+        for alias, i in plan.sig_inputs.items():
+            live_end.setdefault(root(i), -1)
+            live_end[root(i)] = max(live_end[root(i)], 0)
+
+        # im2col scratch sizes
+        for oi, op in enumerate(plan.ops):
+            if op.kind == "conv2d":
+                R, S, Cin, Kc = op.params["rsck"]
+                if not (R == 1 and S == 1 and op.params["stride"] == (1, 1)
+                        and op.params["pads"] == (0, 0, 0, 0)):
+                    Ho, Wo = op.params["out_hw"]
+                    Mrows = self.batch * Ho * Wo
+                    scratch_sizes[oi] = Mrows * _pad64(R * S * Cin) * 2
+
+        self.bufs: Dict[int, _BufInfo] = {}
+        self.scratch_off: Dict[int, int] = {}
+        free: List[Tuple[int, int]] = []     # (nbytes, offset)
+        cursor = 0
+
+        def alloc(nbytes: int) -> int:
+            nonlocal cursor
+            nbytes = (nbytes + ALIGN - 1) // ALIGN * ALIGN
+            best = None
+            for j, (sz, off) in enumerate(free):
+                if sz >= nbytes and (best is None or sz < free[best][0]):
+                    best = j
+            if best is not None:
+                sz, off = free.pop(best)
+                if sz > nbytes:
+                    free.append((sz - nbytes, off + nbytes))
+                return off
+            off = cursor
+            cursor += nbytes
+            return off
+
+        def release(off: int, nbytes: int) -> None:
+            nbytes = (nbytes + ALIGN - 1) // ALIGN * ALIGN
+            free.append((nbytes, off))
+
+        def ensure(idx: int) -> None:
+            r = root(idx)
+            if r in self.bufs:
+                return
+            nb = self._buf_bytes(r)
+            self.bufs[r] = _BufInfo(alloc(nb), nb, self.shapes[r],
+                                    plan.tensors[r].dtype == "i32")
+
+        # inputs first (persistent through the run)
+        for i in plan.sig_inputs.values():
+            ensure(i)
+        for oi, op in enumerate(plan.ops):
+            for i in op.outputs:
+                ensure(i)
+            if oi in scratch_sizes:
+                self.scratch_off[oi] = alloc(scratch_sizes[oi])
+                release(self.scratch_off[oi], scratch_sizes[oi])
+                # NOTE: scratch freed immediately after alloc so the NEXT
+                # op's outputs can reuse it; but it must survive through
+                # this op — handled because outputs were allocated first.
+            for i in op.inputs:
+                r = root(i)
+                if r in self.bufs and live_end.get(r, -1) == oi and \
+                        plan.tensors[r].kind != "input" and \
+                        live_end[r] <= n_ops:
+                    release(self.bufs[r].offset, self.bufs[r].nbytes)
+        self.total_bytes = max(cursor, ALIGN)
+
+    # -- call emission -----------------------------------------------------
+    def _ptr(self, idx: int) -> int:
+        r = self.root[idx]
+        t = self.gm.plan.tensors[r]
+        if t.kind == "weight":
+            return self.gm.weight_ptr(r)
+        return self.workspace.data_ptr() + self.bufs[r].offset
+
+    def _emit_calls(self) -> list:
+        torch, ext = _load_backend()
+        plan = self.gm.plan
+        calls = []
+        for oi, op in enumerate(plan.ops):
+            k = op.kind
+            p = op.params
+            if k == "eltwise":
+                calls.extend(self._c_eltwise(op, ext))
+            elif k == "gemm":
+                calls.extend(self._c_gemm(op, ext))
+            elif k == "conv2d":
+                calls.extend(self._c_conv(op, oi, ext))
+            elif k == "batched_gemm":
+                calls.extend(self._c_bgemm(op, ext))
+            elif k == "softmax":
+                shape = self.shapes[op.inputs[0]]
+                rows = int(np.prod(shape[:-1])) if len(shape) > 1 else 1
+                calls.append((ext.K_SOFTMAX,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.outputs[0])],
+                              [rows, shape[-1]], []))
+            elif k == "layernorm":
+                shape = self.shapes[op.inputs[0]]
+                rows = int(np.prod(shape[:-1])) if len(shape) > 1 else 1
+                calls.append((ext.K_LAYERNORM,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.inputs[1]),
+                               self._ptr(op.inputs[2]),
+                               self._ptr(op.outputs[0])],
+                              [rows, shape[-1]], [float(p["eps"])]))
+            elif k == "bn_act":
+                shape = self.shapes[op.inputs[0]]
+                c = shape[-1]
+                rows = int(np.prod(shape)) // c
+                calls.append((ext.K_BN_ACT,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.inputs[1]),
+                               self._ptr(op.inputs[2]),
+                               self._ptr(op.outputs[0])],
+                              [rows, c, _ACT_CODE[p.get("act", "none")]],
+                              []))
+            elif k == "global_mean":
+                n, h, w, c = self.shapes[op.inputs[0]]
+                calls.append((ext.K_MEAN_MID,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.outputs[0])],
+                              [n, h * w, c], []))
+            elif k == "reduce_mean_mid":
+                d0, d1, d2 = self.shapes[op.inputs[0]]
+                calls.append((ext.K_MEAN_MID,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.outputs[0])],
+                              [d0, d1, d2], []))
+            elif k == "reduce_mean_last":
+                shape = self.shapes[op.inputs[0]]
+                rows = int(np.prod(shape[:-1])) if len(shape) > 1 else 1
+                calls.append((ext.K_MEAN_LAST,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.outputs[0])],
+                              [rows, shape[-1]], []))
+            elif k == "pool":
+                n, h, w, c = self.shapes[op.inputs[0]]
+                ho, wo = p["out_hw"]
+                kh, kw = p["ksize"]
+                sh, sw = p["stride"]
+                pt, pb, pl, pr = p["pads"]
+                calls.append((ext.K_POOL,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.outputs[0])],
+                              [1 if p["mode"] == "max" else 0,
+                               n, h, w, c, ho, wo, kh, kw, sh, sw, pt, pl],
+                              []))
+            elif k == "transpose":
+                in_shape = self.shapes[op.inputs[0]]
+                perm = p["perm"]
+                out_shape = self.shapes[op.outputs[0]]
+                in_strides = [1] * len(in_shape)
+                for d in range(len(in_shape) - 2, -1, -1):
+                    in_strides[d] = in_strides[d + 1] * in_shape[d + 1]
+                strides_out = [in_strides[q] for q in perm]
+                n_out = int(np.prod(out_shape))
+                calls.append((ext.K_TRANSPOSE,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.outputs[0])],
+                              [len(out_shape)] + list(out_shape) +
+                              strides_out + [n_out], []))
+            elif k == "gather":
+                tshape = self.shapes[op.inputs[0]]
+                ishape = self.shapes[op.inputs[1]]
+                row = int(np.prod(tshape[1:]))
+                n_idx = int(np.prod(ishape)) if ishape else 1
+                calls.append((ext.K_GATHER,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.inputs[1]),
+                               self._ptr(op.outputs[0])],
+                              [n_idx, row], []))
+            elif k == "pad":
+                shape = self.shapes[op.inputs[0]]
+                pads = p["pads"]
+                if len(shape) != 4 or any(p_[0] or p_[1]
+                                          for p_ in (pads[0], pads[3])):
+                    raise RuntimeError("GPU pad: only NHWC H/W pads")
+                calls.append((ext.K_PAD_NHWC,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.outputs[0])],
+                              [shape[0], shape[1], shape[2], shape[3],
+                               pads[1][0], pads[1][1], pads[2][0],
+                               pads[2][1]], []))
+            else:
+                raise RuntimeError(f"GPU lowering: unsupported op {k}")
+        return calls
+
+    def _bcast_ints(self, out_shape, a_shape, b_shape):
+        nd = len(out_shape)
+        if nd > 6:
+            raise RuntimeError("eltwise rank > 6")
+
+        def strides_for(shape):
+            shape = list(shape)
+            shape = [1] * (nd - len(shape)) + shape
+            st = [0] * nd
+            acc = 1
+            for d in range(nd - 1, -1, -1):
+                if shape[d] == out_shape[d] and shape[d] != 1:
+                    st[d] = acc
+                elif shape[d] == 1 and out_shape[d] != 1:
+                    st[d] = 0
+                elif shape[d] == out_shape[d]:
+                    st[d] = acc   # both 1
+                else:
+                    raise RuntimeError(
+                        f"cannot broadcast {shape} to {out_shape}")
+                acc *= shape[d]
+            return st
+
+        return strides_for(a_shape), strides_for(b_shape)
+
+    def _c_eltwise(self, op: PlanOp, ext):
+        fn = _ELT_CODE[op.params["fn"]]
+        out = op.outputs[0]
+        out_shape = self.shapes[out]
+        n_out = int(np.prod(out_shape)) if out_shape else 1
+        if len(op.inputs) == 1:
+            return [(ext.K_ELT_UNARY,
+                     [self._ptr(op.inputs[0]), self._ptr(out)],
+                     [n_out, fn], [])]
+        a, b = op.inputs
+        sa_shape, sb_shape = self.shapes[a], self.shapes[b]
+        if tuple(sa_shape) == tuple(out_shape) == tuple(sb_shape):
+            return [(ext.K_ELT_BINARY,
+                     [self._ptr(a), self._ptr(b), self._ptr(out)],
+                     [n_out, fn, 1, n_out, 1, 1], [])]
+        sa, sb = self._bcast_ints(out_shape, sa_shape, sb_shape)
+        nd = len(out_shape)
+        ints = [n_out, fn, nd] + list(out_shape) + sa + sb
+        return [(ext.K_ELT_BINARY,
+                 [self._ptr(a), self._ptr(b), self._ptr(out)], ints, [])]
+
+    def _c_gemm(self, op: PlanOp, ext):
+        p = op.params
+        a = op.inputs[0]
+        w_plan_idx = op.inputs[1]
+        a_shape = self.shapes[a]
+        M = int(np.prod(a_shape[:-1]))
+        K = a_shape[-1]
+        if p.get("trans_a"):
+            raise RuntimeError("gemm trans_a unsupported on GPU")
+        wt = self.gm.gemm_weight(w_plan_idx, bool(p.get("trans_b")))
+        N = wt.shape[0]
+        Kp = wt.shape[1]
+        ni = 2
+        bias_ptr = 0
+        res_ptr = 0
+        if p.get("has_bias"):
+            bias_ptr = self.gm.weight_ptr(op.inputs[ni])
+            ni += 1
+        if p.get("residual"):
+            res_ptr = self._ptr(op.inputs[ni])
+        if Kp != K:
+            # activations K not 64-aligned: pad path unsupported (planner
+            # pads weights only) — requires K % 64 == 0
+            raise RuntimeError(f"gemm K={K} must be 64-aligned "
+                               f"(weight padded to {Kp})")
+        return [(ext.K_GEMM,
+                 [self._ptr(a), wt.data_ptr(), bias_ptr, res_ptr,
+                  self._ptr(op.outputs[0])],
+                 [M, N, K, _ACT_CODE[p.get("act", "none")]],
+                 [1.0])]
+
+    def _c_conv(self, op: PlanOp, oi: int, ext):
+        p = op.params
+        R, S, Cin, Kc = p["rsck"]
+        sh, sw = p["stride"]
+        pt, pb, pl, pr = p["pads"]
+        Ho, Wo = p["out_hw"]
+        x = op.inputs[0]
+        n, H, W, C = self.shapes[x]
+        M = n * Ho * Wo
+        bias_ptr = self.gm.weight_ptr(op.inputs[2])
+        res_ptr = self._ptr(op.inputs[3]) if p.get("residual") else 0
+        wt = self.gm.conv_weight(op.inputs[1])
+        Kp = wt.shape[1]
+        act = _ACT_CODE[p.get("act", "none")]
+        if R == 1 and S == 1 and (sh, sw) == (1, 1) and \
+                (pt, pb, pl, pr) == (0, 0, 0, 0) and C % 64 == 0:
+            return [(ext.K_GEMM,
+                     [self._ptr(x), wt.data_ptr(), bias_ptr, res_ptr,
+                      self._ptr(op.outputs[0])],
+                     [M, Kc, C, act], [1.0])]
+        scratch = self.workspace.data_ptr() + self.scratch_off[oi]
+        return [
+            (ext.K_IM2COL, [self._ptr(x), scratch],
+             [n, H, W, C, R, S, sh, sw, pt, pl, Ho, Wo, Kp], []),
+            (ext.K_GEMM,
+             [scratch, wt.data_ptr(), bias_ptr, res_ptr,
+              self._ptr(op.outputs[0])],
+             [M, Kc, Kp, act], [1.0]),
+        ]
+
+    def _c_bgemm(self, op: PlanOp, ext):
+        p = op.params
+        a, b = op.inputs
+        a_shape = self.shapes[a]
+        b_shape = self.shapes[b]
+        if p.get("trans_a"):
+            raise RuntimeError("batched_gemm trans_a unsupported")
+        trans_b = bool(p.get("trans_b"))
+        M, K = a_shape[-2], a_shape[-1]
+        if trans_b:
+            N = b_shape[-2]
+        else:
+            N = b_shape[-1]
+        bat_a = int(np.prod(a_shape[:-2])) if len(a_shape) > 2 else 1
+        bat_b = int(np.prod(b_shape[:-2])) if len(b_shape) > 2 else 1
+        bat = max(bat_a, bat_b)
+        sA = M * K if bat_a > 1 else 0
+        sB = (b_shape[-1] * b_shape[-2]) if bat_b > 1 else 0
+        sC = M * N
+        return [(ext.K_BGEMM,
+                 [self._ptr(a), self._ptr(b), self._ptr(op.outputs[0])],
+                 [bat, M, N, K, sA, sB, sC, 1 if trans_b else 0],
+                 [1.0])]
+
+    # -- runtime -----------------------------------------------------------
+    def view(self, idx: int):
+        torch, _ = _load_backend()
+        r = self.root[idx]
+        v = self._views.get(idx)
+        if v is not None:
+            return v
+        info = self.bufs[r]
+        t = self.gm.plan.tensors[r]
+        base = self.workspace[info.offset:info.offset + info.nbytes]
+        if t.dtype == "i32":
+            v = base.view(torch.int32)[:int(np.prod(self.shapes[idx]))]
+        else:
+            v = base.view(torch.bfloat16)[:int(np.prod(self.shapes[idx]))]
+        v = v.view(self.shapes[idx])
+        self._views[idx] = v
+        return v
+
+    def run(self, feeds: Dict[int, np.ndarray],
+            fetch: List[int]) -> Dict[int, np.ndarray]:
+        torch, _ = _load_backend()
+        with torch.cuda.device(self.gm.device), \
+                torch.cuda.stream(self.stream):
+            for idx, arr in feeds.items():
+                v = self.view(idx)
+                t = torch.from_numpy(np.ascontiguousarray(arr))
+                if v.dtype == torch.int32:
+                    t = t.to(torch.int32)
+                else:
+                    t = t.to(torch.float32)
+                pad_rows = v.shape[0] - t.shape[0] if v.ndim else 0
+                dt = t.to(self.gm.device, non_blocking=True)
+                if v.dtype == torch.bfloat16:
+                    dt = dt.to(torch.bfloat16)
+                if pad_rows > 0:
+                    v[:t.shape[0]].copy_(dt)
+                    v[t.shape[0]:].zero_()
+                else:
+                    v.copy_(dt)
+            if self.gm.use_graphs and not self.captured:
+                # warm-up eager run, then capture on this stream
+                self.exec_plan.run()
+                self.stream.synchronize()
+                try:
+                    self.exec_plan.capture()
+                except Exception:       # noqa: BLE001
+                    log.exception("hipGraph capture failed; staying eager")
+                self.captured = True
+            if self.captured and self.exec_plan.has_graph():
+                self.exec_plan.run_graph()
+            else:
+                self.exec_plan.run()
+            out = {}
+            for idx in fetch:
+                v = self.view(idx)
+                out[idx] = v.float().cpu().numpy()
+            self.stream.synchronize()
+            return out
+
+
+class GpuModel:
+    """A plan resident on one GPU (weights in HBM + per-bucket contexts)."""
+
+    BUCKETS = (1, 2, 4, 8, 16, 32, 64)
+
+    def __init__(self, plan: Plan, device: str = "cuda:0",
+                 max_batch: int = 64, use_graphs: bool = True):
+        torch, _ = _load_backend()
+        self.plan = plan
+        self.device = device
+        self.max_batch = max_batch
+        self.use_graphs = use_graphs
+        self._weights: Dict[int, object] = {}
+        self._gemm_weights: Dict[Tuple[int, bool], object] = {}
+        self._conv_weights: Dict[int, object] = {}
+        self._contexts: Dict[int, ExecContext] = {}
+        self._lock = threading.Lock()
+        with torch.cuda.device(device):
+            self._upload_weights()
+
+    # -- weights -----------------------------------------------------------
+    def _upload_weights(self) -> None:
+        torch, _ = _load_backend()
+        for t in self.plan.tensors:
+            if t.kind != "weight" or t.weight is None:
+                continue
+            w = t.weight
+            if w.dtype in (np.int32, np.int64):
+                wt = torch.from_numpy(np.ascontiguousarray(
+                    w.astype(np.int32))).to(self.device)
+            else:
+                wt = torch.from_numpy(np.ascontiguousarray(
+                    w.astype(np.float32))).to(self.device).to(torch.bfloat16)
+            self._weights[t.idx] = wt
+
+    def weight_ptr(self, idx: int) -> int:
+        return self._weights[idx].data_ptr()
+
+    def gemm_weight(self, idx: int, graph_trans_b: bool):
+        """[N][Kpad] bf16 pre-transposed weight for the GEMM kernel."""
+        torch, _ = _load_backend()
+        key = (idx, graph_trans_b)
+        wt = self._gemm_weights.get(key)
+        if wt is not None:
+            return wt
+        if idx not in self._weights:
+            raise RuntimeError(
+                "GPU gemm requires a constant weight operand; "
+                "activation x activation MatMul must be BatchMatMul")
+        w = self._weights[idx].float()
+        if not graph_trans_b:
+            w = w.t().contiguous()      # [K,N] -> [N,K]
+        N, K = w.shape
+        Kp = _pad64(K)
+        if Kp != K:
+            w = torch.nn.functional.pad(w, (0, Kp - K))
+        wt = w.to(torch.bfloat16).contiguous()
+        self._gemm_weights[key] = wt
+        return wt
+
+    def conv_weight(self, idx: int):
+        """[Kc][pad64(R*S*C)] bf16 from the [R,S,C,K] master."""
+        torch, _ = _load_backend()
+        wt = self._conv_weights.get(idx)
+        if wt is not None:
+            return wt
+        w = self._weights[idx].float()          # [R,S,C,K]
+        R, S, C, Kc = w.shape
+        w = w.reshape(R * S * C, Kc).t().contiguous()   # [K][RSC]
+        Kp = _pad64(R * S * C)
+        if Kp != R * S * C:
+            w = torch.nn.functional.pad(w, (0, Kp - R * S * C))
+        wt = w.to(torch.bfloat16).contiguous()
+        self._conv_weights[idx] = wt
+        return wt
+
+    def weight_bytes(self) -> int:
+        total = 0
+        for d in (self._weights, self._gemm_weights, self._conv_weights):
+            for wt in d.values():
+                total += wt.numel() * wt.element_size()
+        return total
+
+    # -- execution ---------------------------------------------------------
+    def _bucket(self, batch: int) -> int:
+        for b in self.BUCKETS:
+            if b >= batch and b <= self.max_batch:
+                return b
+        return self.max_batch
+
+    def context(self, batch: int) -> ExecContext:
+        b = self._bucket(batch)
+        with self._lock:
+            ctx = self._contexts.get(b)
+            if ctx is None:
+                ctx = ExecContext(self, b)
+                self._contexts[b] = ctx
+            return ctx
+
+    def run(self, feeds: Dict[int, np.ndarray], batch: int,
+            fetch: List[int]) -> Dict[int, np.ndarray]:
+        ctx = self.context(batch)
+        with self._lock:
+            out = ctx.run(feeds, fetch)
+        if ctx.batch != batch:
+            # un-pad the batch dimension of fetched outputs
+            plan = self.plan
+            for idx in list(out):
+                shape = plan.tensors[idx].shape
+                if shape and is_sym(shape[0]):
+                    rows = resolve_dim(shape[0], batch)
+                    out[idx] = out[idx][:rows]
+        return out
+
+    def release(self) -> None:
+        with self._lock:
+            self._contexts.clear()
+            self._weights.clear()
+            self._gemm_weights.clear()
+            self._conv_weights.clear()

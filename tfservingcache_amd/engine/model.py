@@ -40,7 +40,14 @@ class LoadedModel:
         return self.plan.signature_def
 
     def weight_bytes(self) -> int:
+        if self._gpu is not None:
+            return self._gpu.weight_bytes()
         return self.plan.weight_bytes()
+
+    def release(self) -> None:
+        if self._gpu is not None:
+            self._gpu.release()
+            self._gpu = None
 
     # -- execution ----------------------------------------------------------
     def _infer_batch(self, feeds: Dict[int, np.ndarray]) -> int:
