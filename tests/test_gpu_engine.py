@@ -135,6 +135,7 @@ def test_graph_replay_consistency(tmp_path):
 
 def test_native_extension_is_loaded():
     """The serving math must run in our HIP kernels, not a fallback."""
+    import torch  # noqa: F401  (provides libc10 for the extension)
     from tfservingcache_amd.engine import _tfsc_engine as ext
     assert hasattr(ext, "ExecPlan")
     assert ext.__file__.endswith(".so")
