@@ -55,12 +55,13 @@ static void launch_call(const Call& c, hipStream_t s) {
       launch_eltwise_unary(s, cp(0), p(1), I[0], int(I[1]));
       break;
     case K_ELT_BINARY: {
+      // ints layout: [n_out, fn, ndim, dims[ndim], sa[ndim], sb[ndim]]
       BcastArgs bc;
       bc.ndim = int(I[2]);
       for (int d = 0; d < MAX_DIMS; ++d) {
         bc.dims[d] = d < bc.ndim ? I[3 + d] : 1;
-        bc.sa[d] = d < bc.ndim ? I[3 + MAX_DIMS + d] : 0;
-        bc.sb[d] = d < bc.ndim ? I[3 + 2 * MAX_DIMS + d] : 0;
+        bc.sa[d] = d < bc.ndim ? I[3 + bc.ndim + d] : 0;
+        bc.sb[d] = d < bc.ndim ? I[3 + 2 * bc.ndim + d] : 0;
       }
       launch_eltwise_binary(s, cp(0), cp(1), p(2), I[0], bc, int(I[1]));
       break;
