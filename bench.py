@@ -1,0 +1,250 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark — the driver contract.
+
+    python bench.py --gpus N --steps K --warmup W
+
+Measures the BASELINE.json headline: predict req/sec (whole node) plus
+p50 cold-load latency. One rank per GPU (the driver launches ranks via
+torch.distributed.run for N>1); each rank runs the full node-local
+serving stack — disk model repo -> DiskModelProvider -> byte-budget LRU
+-> GPU model pool (CDNA4 HIP engine, bf16) — and drives it with
+concurrent Predict requests through the gRPC message path (protobuf
+decode -> CacheManager -> engine -> protobuf encode; TCP loopback
+excluded). A "step" is REQS_PER_STEP completed requests per rank.
+
+Modes (--mode):
+  warm  ResNet-50 warm-cache predict loop (BASELINE configs[1]; default)
+  lru   --models N ResNet-50s, pool capped at --pool-size, Zipf access:
+        the evict/reload path (BASELINE configs[2]); cold-load latencies
+        recorded from the pool.
+
+Data: synthetic (random normal images), random-init weights — no
+network/datasets in this environment; stated in the "data" field.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import tempfile
+import threading
+import time
+from concurrent.futures import ThreadPoolExecutor
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from tfservingcache_amd.cachemanager import (CacheManager, LRUCache,  # noqa: E402
+                                             ModelPool, make_cpu_loader,
+                                             make_gpu_loader)
+from tfservingcache_amd.cachemanager.providers import DiskModelProvider  # noqa: E402
+from tfservingcache_amd.models.builders import (build_resnet50,  # noqa: E402
+                                                build_bert)
+from tfservingcache_amd.engine.savedmodel import write_saved_model  # noqa: E402
+from tfservingcache_amd.tfservingproxy import LocalServingHandler  # noqa: E402
+from tfservingcache_amd.wire import messages as m  # noqa: E402
+from tfservingcache_amd.wire.tensor import numpy_to_tensorproto  # noqa: E402
+
+REQS_PER_STEP = 50
+
+
+def _link_tree(src: str, dst: str) -> None:
+    for root, _dirs, files in os.walk(src):
+        rel = os.path.relpath(root, src)
+        os.makedirs(os.path.join(dst, rel), exist_ok=True)
+        for f in files:
+            os.link(os.path.join(root, f), os.path.join(dst, rel, f))
+
+
+def build_repo(base: str, n_models: int, image_size: int,
+               model_kind: str = "resnet50") -> list:
+    """Write one SavedModel, hard-link it into n_models model dirs."""
+    proto_dir = os.path.join(base, "_proto", "1")
+    if model_kind == "resnet50":
+        sm = build_resnet50(image_size=image_size, num_classes=1000)
+    else:
+        sm = build_bert()
+    write_saved_model(sm, proto_dir)
+    names = []
+    for i in range(n_models):
+        name = f"{model_kind}_{i:04d}"
+        _link_tree(proto_dir, os.path.join(base, name, "1"))
+        names.append(name)
+    return names
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--mode", choices=["warm", "lru"], default="warm")
+    ap.add_argument("--batch", type=int, default=8,
+                    help="images per predict request")
+    ap.add_argument("--models", type=int, default=100,
+                    help="models in the repo (lru mode)")
+    ap.add_argument("--pool-size", type=int, default=10,
+                    help="models resident per GPU (lru mode)")
+    ap.add_argument("--image-size", type=int, default=224)
+    ap.add_argument("--threads", type=int, default=4,
+                    help="concurrent client threads per rank")
+    ap.add_argument("--zipf", type=float, default=1.1)
+    ap.add_argument("--cpu", action="store_true",
+                    help="CPU engine (CI smoke only; not a benchmark)")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    dist = None
+    torch = None
+    if not args.cpu:
+        import torch  # noqa: F811
+    if world > 1:
+        import torch  # noqa: F811
+        import torch.distributed as dist  # noqa: F811
+        backend = "gloo" if (args.cpu or not torch.cuda.is_available()) \
+            else "nccl"
+        if backend == "nccl":
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
+
+    device = f"cuda:{local_rank}" if not args.cpu else "cpu"
+
+    tmp = tempfile.mkdtemp(prefix=f"tfsc_bench_r{rank}_")
+    repo = os.path.join(tmp, "repo")
+    cache_dir = os.path.join(tmp, "cache")
+    n_models = args.models if args.mode == "lru" else 1
+    names = build_repo(repo, n_models, args.image_size)
+
+    provider = DiskModelProvider(repo)
+    cache = LRUCache(cache_dir, max_size_bytes=200 * 10 ** 9)
+    pool_cap = args.pool_size if args.mode == "lru" else 4
+    if args.cpu:
+        loader = make_cpu_loader(cache)
+    else:
+        loader = make_gpu_loader(cache, device=device,
+                                 max_batch=max(args.batch, 1))
+    pool = ModelPool(loader, max_concurrent_models=pool_cap, device=device)
+    cm = CacheManager(provider, cache, pool, model_fetch_timeout=300.0)
+    handler = LocalServingHandler(cm)
+
+    # pre-encoded request bytes (shared image payload; model name swapped)
+    img = (np.random.default_rng(rank).standard_normal(
+        (args.batch, args.image_size, args.image_size, 3)) * 0.5
+    ).astype(np.float32)
+    tp = numpy_to_tensorproto(img)
+
+    def request_bytes(name: str) -> bytes:
+        return m.PredictRequest(
+            model_spec=m.ModelSpec(name=name, version=m.Int64Value(value=1)),
+            inputs={"input": tp}).encode()
+
+    req_cache = {n: request_bytes(n) for n in names}
+
+    rng = np.random.default_rng(1234 + rank)
+    if args.mode == "lru":
+        # Zipf over models, fixed request schedule
+        probs = 1.0 / np.arange(1, n_models + 1) ** args.zipf
+        probs /= probs.sum()
+        order = rng.permutation(n_models)
+
+        def pick(i):
+            return names[order[rng.choice(n_models, p=probs)]]
+    else:
+        def pick(i):
+            return names[0]
+
+    def one_request(i: int) -> None:
+        data = req_cache[pick(i)]
+        req = m.PredictRequest.decode(data)
+        resp = handler.predict(req)
+        resp.encode()
+
+    # initial load (timed -> cold-load sample even in warm mode)
+    t0 = time.monotonic()
+    one_request(0)
+    first_load_s = time.monotonic() - t0
+
+    pool_executor = ThreadPoolExecutor(max_workers=args.threads)
+
+    def step(base: int) -> None:
+        futs = [pool_executor.submit(one_request, base + j)
+                for j in range(REQS_PER_STEP)]
+        for f in futs:
+            f.result()
+
+    def barrier_sync():
+        if not args.cpu and torch is not None and torch.cuda.is_available():
+            torch.cuda.synchronize(device)
+        if dist is not None:
+            dist.barrier()
+
+    for w in range(args.warmup):
+        step(w * REQS_PER_STEP)
+
+    barrier_sync()
+    t_start = time.monotonic()
+    for k in range(args.steps):
+        step((args.warmup + k) * REQS_PER_STEP)
+    barrier_sync()
+    elapsed = time.monotonic() - t_start
+
+    # MAX over ranks
+    if dist is not None:
+        import torch  # noqa: F811
+        te = torch.tensor([elapsed], dtype=torch.float64,
+                          device="cuda" if dist.get_backend() == "nccl"
+                          else "cpu")
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        elapsed = float(te.item())
+
+    total_requests = args.steps * REQS_PER_STEP * world
+    req_per_sec = total_requests / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    loads = sorted(pool.load_durations) or [first_load_s]
+    cold_p50 = loads[len(loads) // 2] * 1000.0
+
+    if rank == 0:
+        result = {
+            "metric": "predict req/sec (whole node)",
+            "value": round(req_per_sec, 2),
+            "unit": "req/s",
+            "n_gpus": world if not args.cpu else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if not args.cpu else "f32",
+            "data": "synthetic",
+            "config": {
+                "model": "resnet50_v1.5",
+                "global_batch": args.batch * world,
+                "seq_len": args.image_size,
+                "parallelism": f"ring-sharded serving, 1 replica/GPU x{world}",
+                "mode": args.mode,
+                "requests_per_step": REQS_PER_STEP,
+                "batch_per_request": args.batch,
+                "images_per_sec": round(req_per_sec * args.batch, 1),
+                "n_models": n_models,
+                "pool_size": pool_cap,
+                "cold_load_p50_ms": round(cold_p50, 1),
+                "n_cold_loads": len(loads),
+                "transport": "in-process gRPC message path "
+                             "(protobuf decode/encode included)",
+            },
+        }
+        print(json.dumps(result))
+    if dist is not None:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -57,6 +57,8 @@ class ModelPool:
         self._lock = threading.Lock()
         self._cond = threading.Condition(self._lock)
         self._entries: Dict[ModelId, PoolEntry] = {}
+        # seconds per completed load — benchmark/observability hook
+        self.load_durations: List[float] = []
 
     # -- introspection (GetModelStatus semantics) --------------------------
     def get_status(self, name: str, version: Optional[int] = None
@@ -154,6 +156,7 @@ class ModelPool:
                 e.model = model
                 e.state = AVAILABLE
                 e.load_finished = time.monotonic()
+                self.load_durations.append(e.load_finished - e.load_started)
             mt.engine_pool_models.labels(self.device).set(
                 sum(1 for x in self._entries.values()
                     if x.state == AVAILABLE))
