@@ -58,7 +58,8 @@ class CacheManager:
         mt.cache_total.labels(*labels).inc()
         try:
             model = self.pool.get_model(name, version)
-            if model is not None and self.cache.contains(name, version):
+            # cache.get (not contains) so hits refresh the MRU order
+            if model is not None and self.cache.get(name, version) is not None:
                 mt.cache_hits.labels(*labels).inc()
                 return model
             mt.cache_misses.labels(*labels).inc()
@@ -74,20 +75,37 @@ class CacheManager:
         with self._flight_lock:
             lock = self._in_flight.setdefault(mid, threading.Lock())
         with lock:
-            # re-check after winning the lock
-            model = self.pool.get_model(name, version)
-            if model is not None and self.cache.contains(name, version):
-                return model
-            if not self.cache.contains(name, version):
-                size = self.provider.model_size(name, version)
-                self.cache.ensure_free_bytes(size)
-                entry = self.provider.load_model(name, version,
-                                                 self.cache.base_dir)
-                self.cache.put(entry)
-            self._reload_pool()
             try:
-                return self.pool.wait_available(name, version,
-                                                self.fetch_timeout)
+                # re-check after winning the lock
+                model = self.pool.get_model(name, version)
+                if model is not None and self.cache.contains(name, version):
+                    return model
+                if not self.cache.contains(name, version):
+                    size = self.provider.model_size(name, version)
+                    self.cache.ensure_free_bytes(size)
+                    entry = self.provider.load_model(name, version,
+                                                     self.cache.base_dir)
+                    self.cache.put(entry)
+                # a concurrent reload may evict this model between our
+                # reload and the wait — retry a bounded number of times
+                deadline = time.monotonic() + self.fetch_timeout
+                attempt = 0
+                while True:
+                    self.cache.get(name, version)     # keep it MRU
+                    self._reload_pool()
+                    remaining = deadline - time.monotonic()
+                    if remaining <= 0:
+                        raise TimeoutError(
+                            f"model {name}:{version} not available after "
+                            f"{self.fetch_timeout}s")
+                    try:
+                        return self.pool.wait_available(
+                            name, version, min(remaining,
+                                               self.fetch_timeout / 4 + 1))
+                    except TimeoutError:
+                        attempt += 1
+                        if attempt >= 4:
+                            raise
             finally:
                 with self._flight_lock:
                     self._in_flight.pop(mid, None)

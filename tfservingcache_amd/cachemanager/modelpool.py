@@ -99,8 +99,11 @@ class ModelPool:
         to_load: List[Tuple[str, int]] = []
         with self._lock:
             for mid in list(self._entries):
-                if mid not in want and \
-                        self._entries[mid].state in (AVAILABLE, START, LOADING):
+                # never cancel an in-flight load: a concurrent request is
+                # blocked in wait_available on it; it will be reconciled by
+                # the next reload once AVAILABLE (the pool may transiently
+                # exceed max_concurrent_models by in-flight loads)
+                if mid not in want and self._entries[mid].state == AVAILABLE:
                     self._unload_locked(mid)
             for mid in desired:
                 e = self._entries.get(mid)

@@ -37,6 +37,10 @@ class GpuUnavailable(RuntimeError):
     pass
 
 
+class ModelReleasedError(RuntimeError):
+    """Raised when a request races a pool eviction; callers re-fetch."""
+
+
 def _load_backend():
     """Import torch + the HIP extension. Loud failure on GPU boxes."""
     global _torch, _ext
@@ -122,6 +126,11 @@ class ExecContext:
         # dedicated non-default stream (the default stream cannot be
         # hipGraph-captured; copies + kernels + D2H all run here)
         self.stream = torch.cuda.Stream(device=dev)
+        # pinned host staging per feed/fetch tensor: numpy -> pinned
+        # (with dtype conversion on CPU) -> one async DMA, instead of a
+        # pageable f32 copy + on-device convert per request
+        self._pinned_in: Dict[int, object] = {}
+        self._pinned_out: Dict[int, object] = {}
 
     # -- buffer planning ---------------------------------------------------
     def _buf_bytes(self, idx: int) -> int:
@@ -488,20 +497,20 @@ class ExecContext:
                 torch.cuda.stream(self.stream):
             for idx, arr in feeds.items():
                 v = self.view(idx)
-                t = torch.from_numpy(np.ascontiguousarray(arr))
-                if v.dtype == torch.int32:
-                    t = t.to(torch.int32)
+                arr = np.ascontiguousarray(arr)
+                t = torch.from_numpy(arr)
+                pin = self._pinned_in.get(idx)
+                if pin is None or pin.shape != v.shape:
+                    pin = torch.empty(v.shape, dtype=v.dtype,
+                                      pin_memory=True)
+                    self._pinned_in[idx] = pin
+                rows = t.shape[0] if t.ndim else 0
+                if t.ndim and rows < v.shape[0]:
+                    pin[:rows].copy_(t)      # CPU-side dtype convert
+                    pin[rows:].zero_()
                 else:
-                    t = t.to(torch.float32)
-                pad_rows = v.shape[0] - t.shape[0] if v.ndim else 0
-                dt = t.to(self.gm.device, non_blocking=True)
-                if v.dtype == torch.bfloat16:
-                    dt = dt.to(torch.bfloat16)
-                if pad_rows > 0:
-                    v[:t.shape[0]].copy_(dt)
-                    v[t.shape[0]:].zero_()
-                else:
-                    v.copy_(dt)
+                    pin.copy_(t.reshape(v.shape))
+                v.copy_(pin, non_blocking=True)
             if self.gm.use_graphs and not self.captured:
                 # warm-up eager run, then capture on this stream
                 self.exec_plan.run()
@@ -515,12 +524,18 @@ class ExecContext:
                 self.exec_plan.run_graph()
             else:
                 self.exec_plan.run()
-            out = {}
             for idx in fetch:
                 v = self.view(idx)
-                out[idx] = v.float().cpu().numpy()
+                po = self._pinned_out.get(idx)
+                if po is None or po.shape != v.shape:
+                    po = torch.empty(v.shape, dtype=torch.float32
+                                     if v.dtype == torch.bfloat16
+                                     else v.dtype, pin_memory=True)
+                    self._pinned_out[idx] = po
+                po.copy_(v, non_blocking=True)
             self.stream.synchronize()
-            return out
+            return {idx: self._pinned_out[idx].numpy().copy()
+                    for idx in fetch}
 
 
 class GpuModel:
@@ -540,6 +555,7 @@ class GpuModel:
         self._conv_weights: Dict[int, object] = {}
         self._contexts: Dict[int, ExecContext] = {}
         self._lock = threading.Lock()
+        self._released = False
         with torch.cuda.device(device):
             self._upload_weights()
 
@@ -624,8 +640,13 @@ class GpuModel:
 
     def run(self, feeds: Dict[int, np.ndarray], batch: int,
             fetch: List[int]) -> Dict[int, np.ndarray]:
+        if self._released:
+            raise ModelReleasedError("model was evicted from the GPU pool")
         ctx = self.context(batch)
         with self._lock:
+            if self._released:
+                raise ModelReleasedError(
+                    "model was evicted from the GPU pool")
             out = ctx.run(feeds, fetch)
         if ctx.batch != batch:
             # un-pad the batch dimension of fetched outputs
@@ -638,7 +659,11 @@ class GpuModel:
         return out
 
     def release(self) -> None:
+        # taking _lock first means any in-flight run() completes before
+        # its weights are freed; subsequent runs fail fast and the cache
+        # manager re-fetches
         with self._lock:
+            self._released = True
             self._contexts.clear()
             self._weights.clear()
             self._gemm_weights.clear()

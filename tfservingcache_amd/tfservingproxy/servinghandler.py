@@ -89,6 +89,16 @@ class LocalServingHandler:
             out = model.predict(inputs, output_filter)
         except ModelExecError as e:
             raise ServingError(str(e))
+        except Exception as e:          # noqa: BLE001
+            # request raced a pool eviction -> one re-fetch
+            from ..engine.gpu import ModelReleasedError
+            if not isinstance(e, ModelReleasedError):
+                raise
+            model, version = self.get_model(name, version)
+            try:
+                out = model.predict(inputs, output_filter)
+            except ModelExecError as e2:
+                raise ServingError(str(e2))
         return out, version
 
     def predict(self, req: m.PredictRequest) -> m.PredictResponse:
