@@ -165,10 +165,18 @@ def _decode_scalar(kind: str, data: bytes, pos: int, wire_type: int):
         return _u64.unpack_from(data, pos)[0], pos + 8
     if wire_type == WIRE_LEN:
         n, pos = read_varint(data, pos)
-        raw = data[pos:pos + n]
-        pos += n
         if kind == "string":
+            raw = data[pos:pos + n]
+            pos += n
             return raw.decode("utf-8", errors="surrogateescape"), pos
+        # bytes fields are returned zero-copy for large payloads
+        # (tensor_content on the predict hot path): a memoryview keeps
+        # the source buffer alive and feeds np.frombuffer directly
+        if n > 4096:
+            raw = memoryview(data)[pos:pos + n]
+        else:
+            raw = data[pos:pos + n]
+        pos += n
         return raw, pos
     raise ValueError(f"wire type {wire_type} for kind {kind}")
 
