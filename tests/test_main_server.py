@@ -1,0 +1,156 @@
+"""End-to-end Server (composition root) tests on real ports."""
+import os
+import socket
+import time
+
+import grpc
+import numpy as np
+import pytest
+import requests
+
+from tfservingcache_amd.config import Config
+from tfservingcache_amd.main import Server
+from tfservingcache_amd.models import write_model_repo
+from tfservingcache_amd.wire import messages as m
+from tfservingcache_amd.wire.tensor import (numpy_to_tensorproto,
+                                            tensorproto_to_numpy)
+
+
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def make_cfg(tmp_path, with_discovery=False, name="n1"):
+    repo = tmp_path / "repo"
+    if not repo.exists():
+        write_model_repo(str(repo), [("half_plus_two", 1, "half_plus_two")])
+    values = {
+        "cacheRestPort": free_port(),
+        "cacheGrpcPort": free_port(),
+        "proxyRestPort": free_port(),
+        "proxyGrpcPort": free_port(),
+        "modelProvider": {"type": "diskProvider",
+                          "diskProvider": {"baseDir": str(repo)}},
+        "modelCache": {"hostModelPath": str(tmp_path / f"cache_{name}"),
+                       "size": 10 ** 8},
+        "serving": {"maxConcurrentModels": 2},
+        "proxy": {"replicasPerModel": 1, "advertiseHost": "127.0.0.1"},
+    }
+    if with_discovery:
+        values["serviceDiscovery"] = {
+            "type": "file", "heartbeatTTL": 0.5,
+            "file": {"directory": str(tmp_path / "cluster")},
+        }
+    return Config(values)
+
+
+def wait_http(port, path="/healthz", timeout=10):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        try:
+            r = requests.get(f"http://127.0.0.1:{port}{path}", timeout=1)
+            if r.status_code == 200:
+                return
+        except requests.RequestException:
+            time.sleep(0.05)
+    raise TimeoutError(f"port {port} never became ready")
+
+
+def test_server_cache_tier_only(tmp_path):
+    cfg = make_cfg(tmp_path)
+    server = Server(cfg)
+    server.start()
+    try:
+        assert server.cluster is None     # proxy disabled (main.go:103-105)
+        wait_http(server.cache_rest_port)
+        r = requests.post(
+            f"http://127.0.0.1:{server.cache_rest_port}"
+            "/v1/models/half_plus_two:predict",
+            json={"instances": [1.0, 2.0, 5.0]}, timeout=10)
+        assert r.status_code == 200
+        assert r.json() == {"predictions": [2.5, 3.0, 4.5]}
+
+        # metrics endpoint
+        r = requests.get(
+            f"http://127.0.0.1:{server.cache_rest_port}"
+            "/monitoring/prometheus/metrics", timeout=5)
+        assert "tfservingcache_cache_total" in r.text
+
+        # gRPC health
+        ch = grpc.insecure_channel(
+            f"127.0.0.1:{server.cache_grpc_port}")
+        from tfservingcache_amd.tfservingproxy.grpc_server import (
+            HealthCheckRequest, HealthCheckResponse)
+        check = ch.unary_unary(
+            "/grpc.health.v1.Health/Check",
+            request_serializer=lambda x: x.encode(),
+            response_deserializer=HealthCheckResponse.decode)
+        assert check(HealthCheckRequest(), timeout=5).status == 1
+        ch.close()
+    finally:
+        server.stop()
+
+
+def test_server_proxy_tier_two_nodes(tmp_path):
+    cfg1 = make_cfg(tmp_path, with_discovery=True, name="n1")
+    cfg2 = make_cfg(tmp_path, with_discovery=True, name="n2")
+    s1 = Server(cfg1)
+    s2 = Server(cfg2)
+    s1.start()
+    s2.start()
+    try:
+        wait_http(s1.cache_rest_port)
+        wait_http(s2.cache_rest_port)
+        deadline = time.time() + 10
+        while time.time() < deadline and (
+                s1.cluster.n_members() < 2 or s2.cluster.n_members() < 2):
+            time.sleep(0.1)
+        assert s1.cluster.n_members() == 2
+        assert s2.cluster.n_members() == 2
+
+        # REST predict through EITHER proxy reaches the owning node
+        for srv in (s1, s2):
+            r = requests.post(
+                f"http://127.0.0.1:{srv.proxy_rest_port}"
+                "/v1/models/half_plus_two/versions/1:predict",
+                json={"instances": [4.0]}, timeout=15)
+            assert r.status_code == 200, r.text
+            assert r.json() == {"predictions": [4.0]}
+
+        # gRPC predict through the proxy tier
+        ch = grpc.insecure_channel(f"127.0.0.1:{s1.proxy_grpc_port}")
+        predict = ch.unary_unary(
+            "/tensorflow.serving.PredictionService/Predict",
+            request_serializer=lambda x: x.encode(),
+            response_deserializer=m.PredictResponse.decode)
+        resp = predict(m.PredictRequest(
+            model_spec=m.ModelSpec(name="half_plus_two",
+                                   version=m.Int64Value(value=1)),
+            inputs={"x": numpy_to_tensorproto(
+                np.array([10.0], dtype=np.float32))}), timeout=15)
+        np.testing.assert_allclose(
+            tensorproto_to_numpy(resp.outputs["y"]), [7.0])
+        ch.close()
+
+        # both proxies route the same model to the SAME owner (ring
+        # determinism across nodes)
+        owner1 = s1.cluster.node_for_key("some_model", 1)
+        owner2 = s2.cluster.node_for_key("some_model", 1)
+        assert owner1.serialize() == owner2.serialize()
+    finally:
+        s1.stop()
+        s2.stop()
+
+
+def test_config_env_override(tmp_path, monkeypatch):
+    monkeypatch.setenv("TFSC_SERVING_GRPCHOST", "otherhost:9999")
+    monkeypatch.setenv("TFSC_MODELCACHE_SIZE", "12345")
+    cfg = Config({"serving": {"grpcHost": "localhost:8500"}})
+    assert cfg.get_string("serving.grpcHost") == "otherhost:9999"
+    assert cfg.get_int("modelCache.size") == 12345
+    assert cfg.get_string("healthProbe.modelName") == \
+        "__TFSERVINGCACHE_PROBE_CHECK__"

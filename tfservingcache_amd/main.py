@@ -1,0 +1,290 @@
+"""Composition root — `python -m tfservingcache_amd.main`.
+
+Mirrors the reference's cmd/taskhandler/main.go:
+  * config.yaml from CWD + TFSC_ env overrides (cfg.go:10-26);
+  * cache tier on cacheRestPort/cacheGrpcPort, proxy tier on
+    proxyRestPort/proxyGrpcPort (config.yaml:1-4, main.go:45-113);
+  * proxy tier disabled when serviceDiscovery.type is unset
+    ("Proxy is disabled", main.go:103-105);
+  * 30s health loop feeding the gRPC health servers from the probe-model
+    check + provider check (main.go:35-42, cachemanager.go:76-89);
+  * factories for model providers and discovery backends
+    (main.go:115-192).
+
+MI355X specifics: the serving backend is the in-process CDNA4 engine
+(one model pool per visible GPU, models sharded over GPUs by the ring's
+(node, gpu) slots; engine.gpus=-1 uses all).
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import socket
+import threading
+import time
+from typing import List, Optional
+
+from aiohttp import web
+
+from .cachemanager import (CacheManager, LRUCache, ModelPool,
+                           make_cpu_loader, make_gpu_loader)
+from .cachemanager.providers import DiskModelProvider
+from .config import Config
+from .taskhandler import ClusterConnection, MetricsMerger
+from .taskhandler.discovery.base import (DiscoveryService, FileDiscovery,
+                                         ServingService, StaticDiscovery)
+from .tfservingproxy import (LocalServingHandler, make_cache_grpc_server,
+                             make_cache_rest_app, make_proxy_grpc_server,
+                             make_proxy_rest_app)
+from .tfservingproxy.grpc_server import HealthState
+from .utils import metrics as mt
+
+log = logging.getLogger("tfsc.main")
+
+
+def create_model_provider(cfg: Config):
+    ptype = cfg.get_string("modelProvider.type")
+    if ptype in ("diskProvider", "disk", ""):
+        base = cfg.get_string("modelProvider.diskProvider.baseDir") or \
+            cfg.get_string("modelProvider.diskProvider.basePath") or \
+            "./model_repo"
+        return DiskModelProvider(base)
+    if ptype in ("s3Provider", "s3"):
+        from .cachemanager.providers.s3 import S3ModelProvider
+        return S3ModelProvider(
+            bucket=cfg.get_string("modelProvider.s3.bucket"),
+            base_path=cfg.get_string("modelProvider.s3.basePath"),
+            endpoint_url=cfg.get_string("modelProvider.s3.endpoint") or None,
+            region=cfg.get_string("modelProvider.s3.region") or "us-east-1")
+    if ptype in ("azBlobProvider", "azblob"):
+        from .cachemanager.providers.azblob import AZBlobModelProvider
+        return AZBlobModelProvider(
+            account=cfg.get_string("modelProvider.azBlob.account"),
+            container=cfg.get_string("modelProvider.azBlob.container"),
+            base_path=cfg.get_string("modelProvider.azBlob.basePath"),
+            account_key=cfg.get_string("modelProvider.azBlob.accountKey"),
+            endpoint=cfg.get_string("modelProvider.azBlob.endpoint") or None)
+    raise ValueError(f"unknown modelProvider.type {ptype!r}")
+
+
+def create_discovery_service(cfg: Config, health_check) -> Optional[DiscoveryService]:
+    dtype = cfg.get_string("serviceDiscovery.type")
+    ttl = cfg.get_float("serviceDiscovery.heartbeatTTL") or 5.0
+    if not dtype:
+        return None
+    if dtype == "static":
+        return StaticDiscovery(
+            [str(m) for m in cfg.get_list("serviceDiscovery.static.members")])
+    if dtype == "file":
+        return FileDiscovery(
+            cfg.get_string("serviceDiscovery.file.directory") or
+            "/tmp/tfsc-cluster", heartbeat_ttl=ttl)
+    if dtype == "consul":
+        from .taskhandler.discovery.consul import ConsulDiscovery
+        return ConsulDiscovery(
+            service_name=cfg.get_string(
+                "serviceDiscovery.consul.serviceName") or "tfservingcache",
+            service_id=cfg.get_string("serviceDiscovery.consul.serviceId"),
+            address=cfg.get_string("serviceDiscovery.consul.address") or
+            "http://127.0.0.1:8500",
+            heartbeat_ttl=ttl, health_check=health_check)
+    if dtype == "etcd":
+        from .taskhandler.discovery.etcd import EtcdDiscovery
+        auth = cfg.get_dict("serviceDiscovery.etcd.authorization")
+        return EtcdDiscovery(
+            service_name=cfg.get_string(
+                "serviceDiscovery.etcd.serviceName") or "tfservingcache",
+            endpoints=[str(e) for e in
+                       cfg.get_list("serviceDiscovery.etcd.endpoints")],
+            heartbeat_ttl=ttl,
+            username=str(auth.get("username", "")),
+            password=str(auth.get("password", "")))
+    if dtype == "k8s":
+        from .taskhandler.discovery.kubernetes import KubernetesDiscovery
+        return KubernetesDiscovery(
+            field_selector=cfg.get_dict(
+                "serviceDiscovery.k8s.fieldSelector"),
+            port_names=cfg.get_dict("serviceDiscovery.k8s.portNames"))
+    raise ValueError(f"unknown serviceDiscovery.type {dtype!r}")
+
+
+def _gpu_devices(cfg: Config) -> List[str]:
+    n = cfg.get_int("engine.gpus")
+    try:
+        import torch
+        if not torch.cuda.is_available():
+            return []
+        avail = torch.cuda.device_count()
+    except Exception:       # noqa: BLE001
+        return []
+    if n < 0 or n > avail:
+        n = avail
+    return [f"cuda:{i}" for i in range(n)]
+
+
+def create_cache_manager(cfg: Config) -> CacheManager:
+    provider = create_model_provider(cfg)
+    cache = LRUCache(
+        cfg.get_string("modelCache.hostModelPath") or "./models",
+        max_size_bytes=cfg.get_int("modelCache.size") or 10 ** 9,
+        rebuild_from_disk=cfg.get_bool("modelCache.rebuildFromDisk"))
+    max_models = cfg.get_int("serving.maxConcurrentModels") or 2
+    devices = _gpu_devices(cfg)
+    if devices:
+        # one pool spanning the node's GPUs is modeled as one pool per
+        # process for now; device cuda:0 default, multi-GPU sharding via
+        # per-GPU ring slots + one process per GPU (see bench.py)
+        loader = make_gpu_loader(cache, device=devices[0],
+                                 max_batch=cfg.get_int("engine.maxbatch"))
+        device = devices[0]
+    else:
+        loader = make_cpu_loader(cache)
+        device = "cpu"
+    pool = ModelPool(loader, max_concurrent_models=max_models,
+                     device=device)
+    fetch_timeout = cfg.get_float("serving.grpcConfigTimeout") or 10.0
+    return CacheManager(provider, cache, pool,
+                        model_fetch_timeout=fetch_timeout,
+                        model_labels=cfg.get_bool("metrics.modelLabels"))
+
+
+class Server:
+    def __init__(self, cfg: Config):
+        self.cfg = cfg
+        self.cm = create_cache_manager(cfg)
+        self.handler = LocalServingHandler(self.cm)
+        self.health = HealthState()
+        self.cluster: Optional[ClusterConnection] = None
+        self.discovery: Optional[DiscoveryService] = None
+        self._stop = threading.Event()
+        self._grpc_servers = []
+        self._runners = []
+        self._loop = None
+
+        self.cache_rest_port = cfg.get_int("cacheRestPort") or 8094
+        self.cache_grpc_port = cfg.get_int("cacheGrpcPort") or 8095
+        self.proxy_rest_port = cfg.get_int("proxyRestPort") or 8093
+        self.proxy_grpc_port = cfg.get_int("proxyGrpcPort") or 8100
+
+    def is_healthy(self) -> bool:
+        probe = self.cfg.get_string("healthProbe.modelName")
+        return self.cm.is_healthy(probe)
+
+    # -- startup -----------------------------------------------------------
+    def start(self) -> None:
+        metrics_path = self.cfg.get_string("metrics.path") or \
+            "/monitoring/prometheus/metrics"
+        merger = MetricsMerger(
+            self.cfg.get_string("serving.metricsScrapeUrl") or None,
+            timeout=self.cfg.get_float("metrics.timeout") or 3.0)
+
+        # cache tier gRPC
+        cache_grpc, _ = make_cache_grpc_server(
+            self.handler, health=self.health,
+            max_msg=self.cfg.get_int("serving.grpcMaxMsgSize") or
+            16 * 1024 * 1024)
+        cache_grpc.add_insecure_port(f"[::]:{self.cache_grpc_port}")
+        cache_grpc.start()
+        self._grpc_servers.append(cache_grpc)
+
+        # proxy tier (only with service discovery — main.go:103-105)
+        self.discovery = create_discovery_service(self.cfg, self.is_healthy)
+        proxy_apps = []
+        if self.discovery is not None:
+            replicas = self.cfg.get_int("proxy.replicasPerModel") or 1
+            self.cluster = ClusterConnection(self.discovery, replicas)
+            host = self.cfg.get_string("proxy.advertiseHost") or \
+                socket.gethostbyname(socket.gethostname())
+            self.cluster.connect(ServingService(
+                host, self.cache_rest_port, self.cache_grpc_port))
+
+            def pick_rest(model, version):
+                return self.cluster.node_for_key(model, version).rest_addr
+
+            def pick_grpc(model, version):
+                return self.cluster.node_for_key(model, version).grpc_addr
+
+            proxy_grpc, _, self._fwd = make_proxy_grpc_server(
+                pick_grpc, health=self.health,
+                max_msg=self.cfg.get_int("serving.grpcMaxMsgSize") or
+                16 * 1024 * 1024)
+            proxy_grpc.add_insecure_port(f"[::]:{self.proxy_grpc_port}")
+            proxy_grpc.start()
+            self._grpc_servers.append(proxy_grpc)
+            proxy_apps.append((make_proxy_rest_app(
+                pick_rest, metrics_path=metrics_path,
+                metrics_render=merger.render,
+                timeout_s=self.cfg.get_float("proxy.grpcTimeout") or 60.0),
+                self.proxy_rest_port))
+        else:
+            log.info("Proxy is disabled (no serviceDiscovery.type)")
+
+        cache_app = make_cache_rest_app(self.handler,
+                                        metrics_path=metrics_path,
+                                        metrics_render=merger.render)
+
+        # REST servers on a dedicated asyncio loop thread
+        apps = [(cache_app, self.cache_rest_port)] + proxy_apps
+        t = threading.Thread(target=self._run_rest, args=(apps,),
+                             daemon=True)
+        t.start()
+
+        # health loop (30s — main.go:35-42)
+        threading.Thread(target=self._health_loop, daemon=True).start()
+
+    def _run_rest(self, apps) -> None:
+        self._loop = asyncio.new_event_loop()
+        asyncio.set_event_loop(self._loop)
+
+        async def boot():
+            for app, port in apps:
+                runner = web.AppRunner(app)
+                await runner.setup()
+                site = web.TCPSite(runner, "0.0.0.0", port)
+                await site.start()
+                self._runners.append(runner)
+        self._loop.run_until_complete(boot())
+        self._loop.run_forever()
+
+    def _health_loop(self) -> None:
+        while not self._stop.wait(30.0):
+            try:
+                self.health.set_serving(self.is_healthy())
+            except Exception:       # noqa: BLE001
+                self.health.set_serving(False)
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self.cluster is not None:
+            self.cluster.disconnect()
+        for s in self._grpc_servers:
+            s.stop(grace=1.0)
+        if self._loop is not None:
+            async def shutdown():
+                for r in self._runners:
+                    await r.cleanup()
+            fut = asyncio.run_coroutine_threadsafe(shutdown(), self._loop)
+            try:
+                fut.result(timeout=5)
+            except Exception:       # noqa: BLE001
+                pass
+            self._loop.call_soon_threadsafe(self._loop.stop)
+        self.cm.close()
+
+
+def main() -> int:
+    cfg = Config.load()
+    server = Server(cfg)
+    server.start()
+    log.info("tfservingcache-amd serving: cacheRest=%d cacheGrpc=%d",
+             server.cache_rest_port, server.cache_grpc_port)
+    try:
+        while True:
+            time.sleep(3600)
+    except KeyboardInterrupt:
+        server.stop()
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())
