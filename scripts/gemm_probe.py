@@ -137,6 +137,33 @@ def main():
         padding=1).permute(0, 2, 3, 1)
     ok &= check("conv3x3 (im2col+gemm)", out, want, rtol=0.1, atol=0.1)
 
+    # --- fused implicit-GEMM conv ----------------------------------------
+    for (N_, H, W, C, R, S, Kc, st) in [(2, 16, 16, 64, 3, 3, 64, 1),
+                                        (2, 15, 15, 64, 3, 3, 128, 2),
+                                        (1, 14, 14, 256, 3, 3, 256, 1)]:
+        x = bfbuf(torch.randn(N_, H, W, C, device=dev) * 0.5)
+        w = torch.randn(R, S, C, Kc, device=dev) * 0.2
+        Kp = ((R * S * C + 63) // 64) * 64
+        wt = bfbuf(torch.nn.functional.pad(
+            w.reshape(R * S * C, Kc).t(), (0, Kp - R * S * C)))
+        bias = bfbuf(torch.randn(Kc, device=dev) * 0.1)
+        zeros = torch.zeros(64, device=dev, dtype=torch.uint8)
+        pad = (R - 1) // 2
+        Ho = (H + 2 * pad - R) // st + 1
+        Wo = (W + 2 * pad - S) // st + 1
+        out = torch.empty(N_, Ho, Wo, Kc, device=dev, dtype=torch.bfloat16)
+        run_plan([(ext.K_CONV,
+                   [x.data_ptr(), wt.data_ptr(), bias.data_ptr(), 0,
+                    zeros.data_ptr(), out.data_ptr()],
+                   [N_, H, W, C, Kc, R, S, st, st, pad, pad, Ho, Wo, Kp,
+                    ext.ACT_RELU], [])])
+        want = torch.relu(torch.nn.functional.conv2d(
+            x.float().permute(0, 3, 1, 2), w.cuda().permute(3, 2, 0, 1),
+            bias=bias.float(), padding=pad, stride=st
+        ).permute(0, 2, 3, 1))
+        ok &= check(f"conv_igemm {H}x{W}x{C}->{Kc} s{st}", out, want,
+                    rtol=0.1, atol=0.1)
+
     # --- pool -------------------------------------------------------------
     x = bfbuf(torch.randn(2, 16, 16, 32, device=dev))
     out = torch.empty(2, 8, 8, 32, device=dev, dtype=torch.bfloat16)

@@ -37,6 +37,7 @@ enum CallKind : int {
   K_GEMM,
   K_BGEMM,
   K_IM2COL,
+  K_CONV,
 };
 
 struct Call {
@@ -119,6 +120,17 @@ static void launch_call(const Call& c, hipStream_t s) {
                     int(I[8]), int(I[9]), int(I[10]), int(I[11]),
                     int(I[12]));
       break;
+    case K_CONV: {
+      // ptrs: [x, w, bias, residual|0, zeros, y]
+      // ints: [N,H,W,C,Kc,R,S,sh,sw,pt,pl,Ho,Wo,k_pad,act]
+      const ushort* res = c.ptrs[3] ? cp(3) : nullptr;
+      launch_conv_igemm(s, cp(0), cp(1), cp(2), res, cp(4), p(5),
+                        int(I[0]), int(I[1]), int(I[2]), int(I[3]),
+                        int(I[4]), int(I[5]), int(I[6]), int(I[7]),
+                        int(I[8]), int(I[9]), int(I[10]), int(I[11]),
+                        int(I[12]), int(I[13]), int(I[14]));
+      break;
+    }
     default:
       throw std::runtime_error("unknown call kind " +
                                std::to_string(c.kind));
@@ -227,6 +239,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("K_GEMM") = int(tfsc::K_GEMM);
   mod.attr("K_BGEMM") = int(tfsc::K_BGEMM);
   mod.attr("K_IM2COL") = int(tfsc::K_IM2COL);
+  mod.attr("K_CONV") = int(tfsc::K_CONV);
 
   // elementwise fn codes
   mod.attr("ELT_ADD") = int(tfsc::ELT_ADD);

@@ -74,6 +74,19 @@ void launch_batched_gemm(hipStream_t s, const ushort* A, const ushort* B,
                          int64_t K, int64_t strideA, int64_t strideB,
                          int64_t strideC, bool trans_b, float alpha);
 
+// Fused implicit-GEMM NHWC Conv2D (C % 8 == 0):
+//   y[N*Ho*Wo, Kc] = act(x (*) w + bias [+ residual])
+// with w pre-transposed [Kc][pad64(R*S*C)] like launch_gemm's B. The
+// A-tile gather happens in the GEMM's LDS staging (per-lane source
+// address from the conv geometry; out-of-image patches read from
+// `zeros`, a >=16-byte zeroed device buffer).
+void launch_conv_igemm(hipStream_t s, const ushort* x, const ushort* w,
+                       const ushort* bias, const ushort* residual,
+                       const ushort* zeros, ushort* y,
+                       int N, int H, int W, int C, int Kc, int R, int S,
+                       int sh, int sw, int pt, int pl, int Ho, int Wo,
+                       int k_pad, int act);
+
 // im2col for NHWC conv: out[N*Ho*Wo, K_pad] where the first R*S*C columns
 // hold the patch elements (r,s,c order, c fastest) and columns >= R*S*C
 // are zero (K_pad is the GEMM's 64-multiple). Out-of-image patch elements

@@ -161,8 +161,12 @@ class ExecContext:
         for oi, op in enumerate(plan.ops):
             if op.kind == "conv2d":
                 R, S, Cin, Kc = op.params["rsck"]
-                if not (R == 1 and S == 1 and op.params["stride"] == (1, 1)
-                        and op.params["pads"] == (0, 0, 0, 0)):
+                is_1x1 = (R == 1 and S == 1 and
+                          op.params["stride"] == (1, 1) and
+                          op.params["pads"] == (0, 0, 0, 0) and
+                          Cin % 64 == 0)
+                # im2col scratch only for the C%8!=0 fallback (conv stem)
+                if not is_1x1 and Cin % 8 != 0:
                     Ho, Wo = op.params["out_hw"]
                     Mrows = self.batch * Ho * Wo
                     scratch_sizes[oi] = Mrows * _pad64(R * S * Cin) * 2
@@ -438,6 +442,13 @@ class ExecContext:
                      [self._ptr(x), wt.data_ptr(), bias_ptr, res_ptr,
                       self._ptr(op.outputs[0])],
                      [M, Kc, C, act], [1.0])]
+        if C % 8 == 0:
+            # fused implicit-GEMM (no im2col materialization)
+            return [(ext.K_CONV,
+                     [self._ptr(x), wt.data_ptr(), bias_ptr, res_ptr,
+                      self.gm.zeros_ptr(), self._ptr(op.outputs[0])],
+                     [n, H, W, C, Kc, R, S, sh, sw, pt, pl, Ho, Wo, Kp,
+                      act], [])]
         scratch = self.workspace.data_ptr() + self.scratch_off[oi]
         return [
             (ext.K_IM2COL, [self._ptr(x), scratch],
@@ -576,6 +587,15 @@ class GpuModel:
 
     def weight_ptr(self, idx: int) -> int:
         return self._weights[idx].data_ptr()
+
+    def zeros_ptr(self) -> int:
+        """A small zeroed device buffer used as the gather source for
+        out-of-image conv patches (csrc/ops/gemm.hip stage_tile_conv_a)."""
+        torch, _ = _load_backend()
+        if getattr(self, "_zeros", None) is None:
+            self._zeros = torch.zeros(64, dtype=torch.uint8,
+                                      device=self.device)
+        return self._zeros.data_ptr()
 
     def gemm_weight(self, idx: int, graph_trans_b: bool):
         """[N][Kpad] bf16 pre-transposed weight for the GEMM kernel."""
