@@ -75,20 +75,32 @@ def build_repo(base: str, n_models: int, image_size: int,
     return names
 
 
+def _graph_status(pool):
+    try:
+        for e in pool._entries.values():      # noqa: SLF001
+            gm = getattr(e.model, "_gpu", None)
+            if gm is not None and gm._contexts:  # noqa: SLF001
+                return all(c.captured and c.exec_plan.has_graph()
+                           for c in gm._contexts.values())
+    except Exception:       # noqa: BLE001
+        pass
+    return None
+
+
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--mode", choices=["warm", "lru"], default="warm")
-    ap.add_argument("--batch", type=int, default=8,
+    ap.add_argument("--batch", type=int, default=16,
                     help="images per predict request")
     ap.add_argument("--models", type=int, default=100,
                     help="models in the repo (lru mode)")
     ap.add_argument("--pool-size", type=int, default=10,
                     help="models resident per GPU (lru mode)")
     ap.add_argument("--image-size", type=int, default=224)
-    ap.add_argument("--threads", type=int, default=4,
+    ap.add_argument("--threads", type=int, default=6,
                     help="concurrent client threads per rank")
     ap.add_argument("--zipf", type=float, default=1.1)
     ap.add_argument("--cpu", action="store_true",
@@ -238,6 +250,7 @@ def main() -> int:
                 "n_cold_loads": len(loads),
                 "transport": "in-process gRPC message path "
                              "(protobuf decode/encode included)",
+                "hipgraph": _graph_status(pool),
             },
         }
         print(json.dumps(result))

@@ -38,6 +38,7 @@ enum CallKind : int {
   K_BGEMM,
   K_IM2COL,
   K_CONV,
+  K_PAD_LAST,
 };
 
 struct Call {
@@ -131,6 +132,9 @@ static void launch_call(const Call& c, hipStream_t s) {
                         int(I[12]), int(I[13]), int(I[14]));
       break;
     }
+    case K_PAD_LAST:
+      launch_pad_last(s, cp(0), p(1), I[0], int(I[1]), int(I[2]));
+      break;
     default:
       throw std::runtime_error("unknown call kind " +
                                std::to_string(c.kind));
@@ -240,6 +244,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("K_BGEMM") = int(tfsc::K_BGEMM);
   mod.attr("K_IM2COL") = int(tfsc::K_IM2COL);
   mod.attr("K_CONV") = int(tfsc::K_CONV);
+  mod.attr("K_PAD_LAST") = int(tfsc::K_PAD_LAST);
 
   // elementwise fn codes
   mod.attr("ELT_ADD") = int(tfsc::ELT_ADD);

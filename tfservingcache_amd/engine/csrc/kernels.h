@@ -59,6 +59,8 @@ void launch_gather_rows(hipStream_t s, const ushort* table, const int* idx,
 void launch_pad_nhwc(hipStream_t s, const ushort* x, ushort* y,
                      int N, int H, int W, int C, int pt, int pb, int pl,
                      int pr);
+void launch_pad_last(hipStream_t s, const ushort* x, ushort* y,
+                     int64_t rows, int c_in, int c_out);
 
 // GEMM: C[M,N] = act(A[M,K] @ B[K,N] + bias [+ residual]).
 // A row-major bf16, B row-major bf16 (pre-transposed at load if the graph

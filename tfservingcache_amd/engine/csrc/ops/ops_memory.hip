@@ -409,6 +409,28 @@ void launch_gather_rows(hipStream_t s, const ushort* table, const int* idx,
                      dim3(TPB), 0, s, table, idx, y, n_idx, row_elems);
 }
 
+// widen the channel dim with zeros: [rows, c_in] -> [rows, c_out]
+// (used to lift C%8!=0 conv inputs — e.g. RGB stems — onto the fused
+// implicit-GEMM path)
+__global__ void k_pad_last(const ushort* __restrict__ x,
+                           ushort* __restrict__ y, int64_t rows, int c_in,
+                           int c_out) {
+  int64_t n = rows * c_out;
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n; i += stride) {
+    int c = int(i % c_out);
+    int64_t r = i / c_out;
+    y[i] = c < c_in ? x[r * c_in + c] : ushort(0);
+  }
+}
+
+void launch_pad_last(hipStream_t s, const ushort* x, ushort* y,
+                     int64_t rows, int c_in, int c_out) {
+  hipLaunchKernelGGL(k_pad_last, dim3(grid_for(rows * c_out)), dim3(TPB),
+                     0, s, x, y, rows, c_in, c_out);
+}
+
 __global__ void k_pad_nhwc(const ushort* __restrict__ x,
                            ushort* __restrict__ y, int N, int H, int W,
                            int C, int pt, int pl, int Ho, int Wo) {

@@ -165,11 +165,12 @@ class ExecContext:
                           op.params["stride"] == (1, 1) and
                           op.params["pads"] == (0, 0, 0, 0) and
                           Cin % 64 == 0)
-                # im2col scratch only for the C%8!=0 fallback (conv stem)
+                # channel-pad scratch for the C%8!=0 case (RGB stem):
+                # input widened to C8 channels, then fused conv
                 if not is_1x1 and Cin % 8 != 0:
-                    Ho, Wo = op.params["out_hw"]
-                    Mrows = self.batch * Ho * Wo
-                    scratch_sizes[oi] = Mrows * _pad64(R * S * Cin) * 2
+                    H, W = op.params["hw"]
+                    c8 = (Cin + 7) // 8 * 8
+                    scratch_sizes[oi] = self.batch * H * W * c8 * 2
 
         self.bufs: Dict[int, _BufInfo] = {}
         self.scratch_off: Dict[int, int] = {}
@@ -449,14 +450,20 @@ class ExecContext:
                       self.gm.zeros_ptr(), self._ptr(op.outputs[0])],
                      [n, H, W, C, Kc, R, S, sh, sw, pt, pl, Ho, Wo, Kp,
                       act], [])]
+        # C % 8 != 0 (RGB stem): widen channels to C8 with zeros, then the
+        # fused conv with channel-padded weights
+        c8 = (C + 7) // 8 * 8
+        wt8 = self.gm.conv_weight_cpad(op.inputs[1], c8)
+        Kp8 = wt8.shape[1]
         scratch = self.workspace.data_ptr() + self.scratch_off[oi]
         return [
-            (ext.K_IM2COL, [self._ptr(x), scratch],
-             [n, H, W, C, R, S, sh, sw, pt, pl, Ho, Wo, Kp], []),
-            (ext.K_GEMM,
-             [scratch, wt.data_ptr(), bias_ptr, res_ptr,
-              self._ptr(op.outputs[0])],
-             [M, Kc, Kp, act], [1.0]),
+            (ext.K_PAD_LAST, [self._ptr(x), scratch],
+             [n * H * W, C, c8], []),
+            (ext.K_CONV,
+             [scratch, wt8.data_ptr(), bias_ptr, res_ptr,
+              self.gm.zeros_ptr(), self._ptr(op.outputs[0])],
+             [n, H, W, c8, Kc, R, S, sh, sw, pt, pl, Ho, Wo, Kp8, act],
+             []),
         ]
 
     def _c_bgemm(self, op: PlanOp, ext):
@@ -633,6 +640,25 @@ class GpuModel:
             w = torch.nn.functional.pad(w, (0, Kp - R * S * C))
         wt = w.to(torch.bfloat16).contiguous()
         self._conv_weights[idx] = wt
+        return wt
+
+    def conv_weight_cpad(self, idx: int, c8: int):
+        """[Kc][pad64(R*S*c8)] bf16 with the input-channel dim widened to
+        c8 (zeros) — pairs with the K_PAD_LAST channel widening."""
+        torch, _ = _load_backend()
+        key = (idx, "cpad", c8)
+        wt = self._conv_weights.get(key)
+        if wt is not None:
+            return wt
+        w = self._weights[idx].float()          # [R,S,C,K]
+        R, S, C, Kc = w.shape
+        w = torch.nn.functional.pad(w, (0, 0, 0, c8 - C))  # pad C dim
+        w = w.reshape(R * S * c8, Kc).t().contiguous()
+        Kp = _pad64(R * S * c8)
+        if Kp != R * S * c8:
+            w = torch.nn.functional.pad(w, (0, Kp - R * S * c8))
+        wt = w.to(torch.bfloat16).contiguous()
+        self._conv_weights[key] = wt
         return wt
 
     def weight_bytes(self) -> int:
