@@ -105,6 +105,8 @@ def main() -> int:
     ap.add_argument("--zipf", type=float, default=1.1)
     ap.add_argument("--cpu", action="store_true",
                     help="CPU engine (CI smoke only; not a benchmark)")
+    ap.add_argument("--dyn-batch", action="store_true",
+                    help="enable server-side dynamic batching")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -139,7 +141,9 @@ def main() -> int:
         loader = make_cpu_loader(cache)
     else:
         loader = make_gpu_loader(cache, device=device,
-                                 max_batch=max(args.batch, 1))
+                                 max_batch=max(args.batch, 64)
+                                 if args.dyn_batch else max(args.batch, 1),
+                                 batching=args.dyn_batch)
     pool = ModelPool(loader, max_concurrent_models=pool_cap, device=device)
     cm = CacheManager(provider, cache, pool, model_fetch_timeout=300.0)
     handler = LocalServingHandler(cm)
@@ -251,6 +255,7 @@ def main() -> int:
                 "transport": "in-process gRPC message path "
                              "(protobuf decode/encode included)",
                 "hipgraph": _graph_status(pool),
+                "dynamic_batching": bool(args.dyn_batch),
             },
         }
         print(json.dumps(result))

@@ -153,11 +153,15 @@ def make_cpu_loader(cache: LRUCache) -> Callable[[str, int], LoadedModel]:
 
 
 def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
-                    max_batch: int = 64,
-                    use_graphs: bool = True) -> Callable[[str, int], LoadedModel]:
+                    max_batch: int = 64, use_graphs: bool = True,
+                    batching: bool = False,
+                    batch_timeout_s: float = 0.002
+                    ) -> Callable[[str, int], LoadedModel]:
     """Loader that compiles the SavedModel onto one MI355X: weights land
     in the GPU's HBM pool (bf16, GEMM layouts pre-transformed) and the
-    predict path runs the CDNA4 HIP kernels."""
+    predict path runs the CDNA4 HIP kernels. With batching=True,
+    concurrent Predicts merge server-side (TF Serving --enable_batching
+    analog)."""
     def loader(name: str, version: int) -> LoadedModel:
         from ..engine.gpu import GpuModel
         vdir = os.path.join(cache.base_dir, name, str(version))
@@ -165,5 +169,8 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
         lm._gpu = GpuModel(lm.plan, device=device, max_batch=max_batch,
                            use_graphs=use_graphs)
         lm.device = device
+        if batching:
+            lm.enable_batching(max_batch=max_batch,
+                               timeout_s=batch_timeout_s)
         return lm
     return loader

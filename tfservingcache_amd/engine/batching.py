@@ -1,0 +1,188 @@
+"""Dynamic request batching (leader-follower).
+
+TF Serving provides server-side batching behind the reference
+(--enable_batching); here it is implemented at the engine boundary:
+concurrent Predict calls for the same model whose inputs differ only in
+the batch dimension are merged into one kernel-plan execution and the
+outputs are split back per caller.
+
+Leader-follower, no extra threads: the first request into an empty
+queue becomes the leader, waits up to `timeout_s` (or until the merged
+batch would exceed `max_batch`) for followers, then executes the merged
+batch once. Configured via serving.batching.{enabled,maxBatchSize,
+batchTimeoutMicros} (TF Serving's knob names).
+"""
+from __future__ import annotations
+
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+
+@dataclass(eq=False)
+class _Item:
+    inputs: Dict[str, np.ndarray]
+    rows: int
+    event: threading.Event = field(default_factory=threading.Event)
+    out: Optional[Dict[str, np.ndarray]] = None
+    error: Optional[Exception] = None
+
+
+class _Group:
+    __slots__ = ("items", "leader_active")
+
+    def __init__(self):
+        self.items: List[_Item] = []
+        self.leader_active = False
+
+
+class DynamicBatcher:
+    def __init__(self, run_fn, batch_dims: Dict[str, int],
+                 max_batch: int = 64, timeout_s: float = 0.002):
+        """run_fn(inputs, output_filter) -> outputs (the unbatched call);
+        batch_dims: input alias -> index of its batch dimension (must be
+        0 for all aliases for batching to engage)."""
+        self._run = run_fn
+        self._batchable = bool(batch_dims) and all(
+            d == 0 for d in batch_dims.values())
+        self.max_batch = max_batch
+        self.timeout_s = timeout_s
+        self._lock = threading.Lock()
+        self._cond = threading.Condition(self._lock)
+        self._groups: Dict[tuple, _Group] = {}
+
+    @staticmethod
+    def _key(inputs: Dict[str, np.ndarray], output_filter) -> tuple:
+        shape_sig = tuple(sorted(
+            (k, v.shape[1:], str(v.dtype)) for k, v in inputs.items()))
+        filt = tuple(sorted(output_filter)) if output_filter else None
+        return (shape_sig, filt)
+
+    def predict(self, inputs: Dict[str, np.ndarray],
+                output_filter=None) -> Dict[str, np.ndarray]:
+        if not self._batchable:
+            return self._run(inputs, output_filter)
+        rows_set = {v.shape[0] for v in inputs.values() if v.ndim}
+        if len(rows_set) != 1:
+            return self._run(inputs, output_filter)
+        rows = rows_set.pop()
+        if rows >= self.max_batch:
+            return self._run(inputs, output_filter)
+
+        key = self._key(inputs, output_filter)
+        item = _Item(inputs, rows)
+        with self._lock:
+            group = self._groups.setdefault(key, _Group())
+            group.items.append(item)
+            if group.leader_active:
+                leader = False
+            else:
+                group.leader_active = True
+                leader = True
+            self._cond.notify_all()
+
+        if not leader:
+            item.event.wait()
+            if item.error is not None:
+                raise item.error
+            return item.out
+
+        # leader: wait for followers until timeout or the batch is full
+        deadline = time.monotonic() + self.timeout_s
+        with self._lock:
+            while True:
+                total = sum(i.rows for i in group.items)
+                remaining = deadline - time.monotonic()
+                if total >= self.max_batch or remaining <= 0:
+                    break
+                self._cond.wait(remaining)
+            batch: List[_Item] = []
+            total = 0
+            for i in list(group.items):
+                if total + i.rows > self.max_batch and batch:
+                    break
+                batch.append(i)
+                total += i.rows
+            group.items = [i for i in group.items if i not in batch]
+            if group.items:
+                # promote a new leader for the remainder
+                group.leader_active = True
+                promoted = group.items[0]
+            else:
+                group.leader_active = False
+                promoted = None
+                self._groups.pop(key, None)
+
+        if promoted is not None:
+            # the promoted item's caller is blocked in event.wait; run its
+            # batch on a helper thread so this leader can proceed
+            threading.Thread(target=self._lead_remainder,
+                             args=(key,), daemon=True).start()
+
+        try:
+            if len(batch) == 1:
+                out = self._run(batch[0].inputs, output_filter)
+                batch[0].out = out
+            else:
+                merged = {
+                    k: np.concatenate([i.inputs[k] for i in batch], axis=0)
+                    for k in batch[0].inputs
+                }
+                out = self._run(merged, output_filter)
+                off = 0
+                for i in batch:
+                    i.out = {k: v[off:off + i.rows] for k, v in out.items()}
+                    off += i.rows
+        except Exception as e:          # noqa: BLE001
+            for i in batch:
+                i.error = e
+            for i in batch[1:]:
+                i.event.set()
+            raise
+        for i in batch[1:]:
+            i.event.set()
+        return batch[0].out
+
+    def _lead_remainder(self, key: tuple) -> None:
+        """Execute the left-over items of an over-full batch window."""
+        with self._lock:
+            group = self._groups.get(key)
+            if group is None or not group.items:
+                if group is not None:
+                    group.leader_active = False
+                return
+            batch = list(group.items[: ])
+            total = 0
+            take = []
+            for i in batch:
+                if total + i.rows > self.max_batch and take:
+                    break
+                take.append(i)
+                total += i.rows
+            group.items = [i for i in group.items if i not in take]
+            more = bool(group.items)
+            if not more:
+                group.leader_active = False
+                self._groups.pop(key, None)
+        filt = list(key[1]) if key[1] else None
+        try:
+            if len(take) == 1:
+                take[0].out = self._run(take[0].inputs, filt)
+            else:
+                merged = {k: np.concatenate([i.inputs[k] for i in take],
+                                            axis=0) for k in take[0].inputs}
+                out = self._run(merged, filt)
+                off = 0
+                for i in take:
+                    i.out = {k: v[off:off + i.rows] for k, v in out.items()}
+                    off += i.rows
+        except Exception as e:          # noqa: BLE001
+            for i in take:
+                i.error = e
+        for i in take:
+            i.event.set()
+        if more:
+            self._lead_remainder(key)

@@ -33,6 +33,23 @@ class LoadedModel:
         self._cpu = CpuExecutor(plan)
         self._gpu = gpu_model       # engine/gpu.py GpuModel, if on GPU
         self._lock = threading.Lock()
+        self._batcher = None
+
+    def enable_batching(self, max_batch: int = 64,
+                        timeout_s: float = 0.002) -> None:
+        """Server-side dynamic batching (TF Serving --enable_batching
+        analog): concurrent Predicts merge into one plan execution."""
+        from .batching import DynamicBatcher
+        batch_dims = {}
+        for alias, idx in self.plan.sig_inputs.items():
+            shape = self.plan.tensors[idx].shape
+            if not shape or not is_sym(shape[0]):
+                batch_dims = {}
+                break
+            batch_dims[alias] = 0
+        self._batcher = DynamicBatcher(self._predict_impl, batch_dims,
+                                       max_batch=max_batch,
+                                       timeout_s=timeout_s)
 
     # -- introspection ------------------------------------------------------
     @property
@@ -75,7 +92,16 @@ class LoadedModel:
 
     def predict(self, inputs: Dict[str, np.ndarray],
                 output_filter=None) -> Dict[str, np.ndarray]:
-        """Predict with signature-alias-keyed inputs/outputs."""
+        """Predict with signature-alias-keyed inputs/outputs (batched
+        server-side when enable_batching was called)."""
+        if self._batcher is not None:
+            return self._batcher.predict(
+                {k: np.asarray(v) for k, v in inputs.items()},
+                output_filter)
+        return self._predict_impl(inputs, output_filter)
+
+    def _predict_impl(self, inputs: Dict[str, np.ndarray],
+                      output_filter=None) -> Dict[str, np.ndarray]:
         plan = self.plan
         feeds: Dict[int, np.ndarray] = {}
         for alias, arr in inputs.items():

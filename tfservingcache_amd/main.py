@@ -134,8 +134,12 @@ def create_cache_manager(cfg: Config) -> CacheManager:
         # one pool spanning the node's GPUs is modeled as one pool per
         # process for now; device cuda:0 default, multi-GPU sharding via
         # per-GPU ring slots + one process per GPU (see bench.py)
-        loader = make_gpu_loader(cache, device=devices[0],
-                                 max_batch=cfg.get_int("engine.maxbatch"))
+        loader = make_gpu_loader(
+            cache, device=devices[0],
+            max_batch=cfg.get_int("engine.maxbatch"),
+            batching=cfg.get_bool("serving.batching.enabled"),
+            batch_timeout_s=(cfg.get_float(
+                "serving.batching.batchTimeoutMicros") or 2000.0) / 1e6)
         device = devices[0]
     else:
         loader = make_cpu_loader(cache)
