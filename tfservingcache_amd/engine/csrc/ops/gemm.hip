@@ -108,6 +108,78 @@ TFSC_DEV float act_apply(float v, int act) {
   }
 }
 
+// Coalesced epilogue: the MFMA C/D fragment layout scatters a lane's
+// values over 2-byte strides, so direct stores are scalar ushort
+// (32-byte wave segments — issue- and coalescing-poor, cf. guide T21).
+// Instead each wave restages its 64x64 tile (acc*alpha + bias, bf16)
+// through its own 8 KB LDS region, then streams it out row-major with
+// 16-byte lanes — and reads the optional residual coalesced in the same
+// pass. One __syncthreads() guards the LDS reuse of the staging tiles.
+template <bool HAS_BIAS, bool HAS_RES>
+TFSC_DEV void epilogue_store(f32x4_t (&acc)[4][4], char* smem,
+                             const ushort* __restrict__ bias,
+                             const ushort* __restrict__ residual,
+                             ushort* __restrict__ Cb,
+                             int M, int N, int m0, int n0, int wave,
+                             int lane, int wm, int wn, int act,
+                             float alpha) {
+  __syncthreads();              // tiles are dead; reuse as staging
+  ushort* stage = reinterpret_cast<ushort*>(smem + wave * 8192);
+  const int col_in = lane & 15;
+  const int row_base = (lane >> 4) * 4;
+  #pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int col = ni * 16 + col_in;
+      float bv = HAS_BIAS ? bf2f(bias[n0 + wn * 64 + col]) : 0.f;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = mi * 16 + row_base + r;
+        // pad column stride by 0 (64 cols x 2B = 128 B rows; lanes of a
+        // 16-lane group write 2B at 2B stride -> conflict-free halves)
+        stage[row * 64 + col] = f2bf(acc[mi][ni][r] * alpha + bv);
+      }
+    }
+  }
+  __syncthreads();
+  // stream out: iteration i, lane l -> linear 16B chunk i*64+l of the
+  // wave's 64x64 tile (8 chunks per row) — consecutive lanes write
+  // consecutive chunks (coalesced 1 KB per wave-instruction)
+  #pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int chunk_id = i * 64 + lane;
+    int row = chunk_id >> 3;
+    int chunk = chunk_id & 7;
+    int grow = m0 + wm * 64 + row;
+    int gcol0 = n0 + wn * 64 + chunk * 8;
+    if (grow >= M) continue;
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(stage + row * 64 +
+                                                chunk * 8);
+    int64_t goff = (int64_t)grow * N + gcol0;
+    if (HAS_RES || act != ACT_NONE) {
+      bf16x8 rv = {};
+      if (HAS_RES && gcol0 + 7 < N)
+        rv = *reinterpret_cast<const bf16x8*>(residual + goff);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float x = bf2f((ushort)v[j]);
+        if (HAS_RES) {
+          if (gcol0 + 7 < N) x += bf2f((ushort)rv[j]);
+          else if (gcol0 + j < N) x += bf2f(residual[goff + j]);
+        }
+        v[j] = (short)f2bf(act_apply(x, act));
+      }
+    }
+    if (gcol0 + 7 < N) {
+      *reinterpret_cast<bf16x8*>(Cb + goff) = v;
+    } else {
+      for (int j = 0; j < 8 && gcol0 + j < N; ++j)
+        Cb[goff + j] = (ushort)v[j];
+    }
+  }
+}
+
 template <bool TRANS_B, bool HAS_BIAS, bool HAS_RES>
 __global__ __launch_bounds__(THREADS)
 void gemm_bf16_kernel(const ushort* __restrict__ A,
@@ -205,28 +277,9 @@ void gemm_bf16_kernel(const ushort* __restrict__ A,
     cur ^= 1;
   }
 
-  // epilogue: C/D layout for 16x16x32: col = lane&15,
-  // row = (lane>>4)*4 + reg (cdna_hip_programming.md §3)
-  const int col_in = lane & 15;
-  const int row_base = (lane >> 4) * 4;
-  #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
-    #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-      int gcol = n0 + wn * 64 + ni * 16 + col_in;
-      if (gcol >= N) continue;
-      float bv = HAS_BIAS ? bf2f(bias[gcol]) : 0.f;
-      #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int grow = m0 + wm * 64 + mi * 16 + row_base + r;
-        if (grow >= M) continue;
-        float v = acc[mi][ni][r] * alpha + bv;
-        if (HAS_RES)
-          v += bf2f(residual[(int64_t)grow * N + gcol]);
-        Cb[(int64_t)grow * N + gcol] = f2bf(act_apply(v, act));
-      }
-    }
-  }
+  epilogue_store<HAS_BIAS, HAS_RES>(acc, smem, bias, residual, Cb, M, N,
+                                    m0, n0, wave, lane, wm, wn, act,
+                                    alpha);
 }
 
 template <bool TRANS_B>
@@ -371,26 +424,8 @@ void conv_igemm_kernel(const ushort* __restrict__ x,
     cur ^= 1;
   }
 
-  const int col_in = lane & 15;
-  const int row_base = (lane >> 4) * 4;
-  #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
-    #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-      int gcol = n0 + wn * 64 + ni * 16 + col_in;
-      if (gcol >= N) continue;
-      float bv = bf2f(bias[gcol]);
-      #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int grow = m0 + wm * 64 + mi * 16 + row_base + r;
-        if (grow >= M) continue;
-        float v = acc[mi][ni][r] + bv;
-        if (HAS_RES)
-          v += bf2f(residual[(int64_t)grow * N + gcol]);
-        Cout[(int64_t)grow * N + gcol] = f2bf(act_apply(v, act));
-      }
-    }
-  }
+  epilogue_store<true, HAS_RES>(acc, smem, bias, residual, Cout, M, N,
+                                m0, n0, wave, lane, wm, wn, act, 1.0f);
 }
 
 void launch_conv_igemm(hipStream_t s, const ushort* x, const ushort* w,
