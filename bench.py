@@ -17,6 +17,12 @@ Modes (--mode):
   lru   --models N ResNet-50s, pool capped at --pool-size, Zipf access:
         the evict/reload path (BASELINE configs[2]); cold-load latencies
         recorded from the pool.
+  ring  consistent-hash ring over the N ranks/GPUs with
+        --replicas replicas per model; model files fan out to replica
+        GPUs over the RCCL/xGMI replica plane at preload; each rank
+        serves Zipf-weighted requests for the models it owns
+        (BASELINE configs[3]/[4]). --model mixed serves BERT-base +
+        ResNet-50 together (configs[4]).
 
 Data: synthetic (random normal images), random-init weights — no
 network/datasets in this environment; stated in the "data" field.
@@ -59,19 +65,26 @@ def _link_tree(src: str, dst: str) -> None:
 
 
 def build_repo(base: str, n_models: int, image_size: int,
-               model_kind: str = "resnet50") -> list:
-    """Write one SavedModel, hard-link it into n_models model dirs."""
-    proto_dir = os.path.join(base, "_proto", "1")
-    if model_kind == "resnet50":
-        sm = build_resnet50(image_size=image_size, num_classes=1000)
-    else:
-        sm = build_bert()
-    write_saved_model(sm, proto_dir)
+               model_kind: str = "resnet50", seq_len: int = 128) -> list:
+    """Write one SavedModel per family, hard-link into n_models dirs.
+    Returns [(name, kind)]."""
+    kinds = (["resnet50", "bert_base"] if model_kind == "mixed"
+             else [model_kind])
+    protos = {}
+    for kind in kinds:
+        proto_dir = os.path.join(base, f"_proto_{kind}", "1")
+        if kind == "resnet50":
+            sm = build_resnet50(image_size=image_size, num_classes=1000)
+        else:
+            sm = build_bert(seq_len=seq_len)
+        write_saved_model(sm, proto_dir)
+        protos[kind] = proto_dir
     names = []
     for i in range(n_models):
-        name = f"{model_kind}_{i:04d}"
-        _link_tree(proto_dir, os.path.join(base, name, "1"))
-        names.append(name)
+        kind = kinds[i % len(kinds)]
+        name = f"{kind}_{i:04d}"
+        _link_tree(protos[kind], os.path.join(base, name, "1"))
+        names.append((name, kind))
     return names
 
 
@@ -92,7 +105,14 @@ def main() -> int:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--mode", choices=["warm", "lru"], default="warm")
+    ap.add_argument("--mode", choices=["warm", "lru", "ring"],
+                    default="warm")
+    ap.add_argument("--model", choices=["resnet50", "bert_base", "mixed"],
+                    default="resnet50")
+    ap.add_argument("--replicas", type=int, default=2,
+                    help="replicasPerModel (ring mode)")
+    ap.add_argument("--seq-len", type=int, default=128,
+                    help="BERT sequence length")
     ap.add_argument("--batch", type=int, default=16,
                     help="images per predict request")
     ap.add_argument("--models", type=int, default=100,
@@ -131,12 +151,15 @@ def main() -> int:
     tmp = tempfile.mkdtemp(prefix=f"tfsc_bench_r{rank}_")
     repo = os.path.join(tmp, "repo")
     cache_dir = os.path.join(tmp, "cache")
-    n_models = args.models if args.mode == "lru" else 1
-    names = build_repo(repo, n_models, args.image_size)
+    n_models = args.models if args.mode in ("lru", "ring") else 1
+    named = build_repo(repo, n_models, args.image_size, args.model,
+                       args.seq_len)
+    names = [n for n, _k in named]
+    kind_of = dict(named)
 
     provider = DiskModelProvider(repo)
     cache = LRUCache(cache_dir, max_size_bytes=200 * 10 ** 9)
-    pool_cap = args.pool_size if args.mode == "lru" else 4
+    pool_cap = args.pool_size if args.mode in ("lru", "ring") else 4
     if args.cpu:
         loader = make_cpu_loader(cache)
     else:
@@ -148,28 +171,68 @@ def main() -> int:
     cm = CacheManager(provider, cache, pool, model_fetch_timeout=300.0)
     handler = LocalServingHandler(cm)
 
-    # pre-encoded request bytes (shared image payload; model name swapped)
+    # pre-encoded request bytes (shared payloads; model name swapped)
     img = (np.random.default_rng(rank).standard_normal(
         (args.batch, args.image_size, args.image_size, 3)) * 0.5
     ).astype(np.float32)
-    tp = numpy_to_tensorproto(img)
+    tp_img = numpy_to_tensorproto(img)
+    ids = np.random.default_rng(rank).integers(
+        0, 30000, (args.batch, args.seq_len)).astype(np.int32)
+    tp_ids = numpy_to_tensorproto(ids)
 
     def request_bytes(name: str) -> bytes:
+        if kind_of[name] == "bert_base":
+            inputs = {"input_ids": tp_ids}
+        else:
+            inputs = {"input": tp_img}
         return m.PredictRequest(
             model_spec=m.ModelSpec(name=name, version=m.Int64Value(value=1)),
-            inputs={"input": tp}).encode()
+            inputs=inputs).encode()
 
     req_cache = {n: request_bytes(n) for n in names}
 
     rng = np.random.default_rng(1234 + rank)
-    if args.mode == "lru":
-        # Zipf over models, fixed request schedule
-        probs = 1.0 / np.arange(1, n_models + 1) ** args.zipf
-        probs /= probs.sum()
-        order = rng.permutation(n_models)
+    probs_all = 1.0 / np.arange(1, n_models + 1) ** args.zipf
+    probs_all /= probs_all.sum()
+    order = rng.permutation(n_models)
+
+    if args.mode == "ring":
+        # ring over ranks; this rank serves the models it owns
+        from tfservingcache_amd.taskhandler import (ConsistentHashRing,
+                                                    model_key)
+        from tfservingcache_amd.parallel import ReplicaPlane
+        ring = ConsistentHashRing()
+        ring.set_members([f"rank{r}" for r in range(world)])
+
+        def owners_of(name, version=1):
+            mems = ring.get_n(model_key(name, version),
+                              min(args.replicas, world))
+            return sorted(int(mm[4:]) for mm in mems)
+
+        owned_idx = [i for i, n in enumerate(names)
+                     if rank in owners_of(n)]
+        if world > 1:
+            # stage-once + RCCL fan-out of the model files (the stager
+            # fetches from the provider; replicas receive over xGMI)
+            plane = ReplicaPlane(device=device if not args.cpu else None)
+
+            def fetch_local(name, version):
+                entry = provider.load_model(name, version, cache.base_dir)
+                cache.put(entry)
+
+            plane.preload_replicated([(n, 1) for n in names], owners_of,
+                                     cache.base_dir, fetch_local)
+        if not owned_idx:
+            owned_idx = [rank % n_models]
+        w = probs_all[owned_idx]
+        w = w / w.sum()
+        owned_names = [names[i] for i in owned_idx]
 
         def pick(i):
-            return names[order[rng.choice(n_models, p=probs)]]
+            return owned_names[rng.choice(len(owned_names), p=w)]
+    elif args.mode == "lru":
+        def pick(i):
+            return names[order[rng.choice(n_models, p=probs_all)]]
     else:
         def pick(i):
             return names[0]
@@ -240,10 +303,14 @@ def main() -> int:
             "dtype": "bf16" if not args.cpu else "f32",
             "data": "synthetic",
             "config": {
-                "model": "resnet50_v1.5",
+                "model": {"resnet50": "resnet50_v1.5",
+                          "bert_base": "bert_base",
+                          "mixed": "bert_base+resnet50_v1.5"}[args.model],
                 "global_batch": args.batch * world,
                 "seq_len": args.image_size,
-                "parallelism": f"ring-sharded serving, 1 replica/GPU x{world}",
+                "parallelism": f"ring-sharded serving x{world}" +
+                               (f", replicas={args.replicas}"
+                                if args.mode == "ring" else ""),
                 "mode": args.mode,
                 "requests_per_step": REQS_PER_STEP,
                 "batch_per_request": args.batch,
