@@ -229,10 +229,10 @@ def main() -> int:
         owned_names = [names[i] for i in owned_idx]
 
         def pick(i):
-            return owned_names[rng.choice(len(owned_names), p=w)]
+            return owned_names[int(rng.choice(np.arange(len(owned_names)), p=w))]
     elif args.mode == "lru":
         def pick(i):
-            return names[order[rng.choice(n_models, p=probs_all)]]
+            return names[order[int(rng.choice(np.arange(n_models), p=probs_all))]]
     else:
         def pick(i):
             return names[0]
@@ -307,7 +307,7 @@ def main() -> int:
                           "bert_base": "bert_base",
                           "mixed": "bert_base+resnet50_v1.5"}[args.model],
                 "global_batch": args.batch * world,
-                "seq_len": args.image_size,
+                "seq_len": args.seq_len if args.model != "resnet50" else args.image_size,
                 "parallelism": f"ring-sharded serving x{world}" +
                                (f", replicas={args.replicas}"
                                 if args.mode == "ring" else ""),
