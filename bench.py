@@ -228,11 +228,16 @@ def main() -> int:
         w = w / w.sum()
         owned_names = [names[i] for i in owned_idx]
 
+        # precomputed schedule: Generator.choice is not thread-safe
+        sched = rng.choice(np.arange(len(owned_names)), size=65536, p=w)
+
         def pick(i):
-            return owned_names[int(rng.choice(np.arange(len(owned_names)), p=w))]
+            return owned_names[sched[i % 65536]]
     elif args.mode == "lru":
+        sched = rng.choice(np.arange(n_models), size=65536, p=probs_all)
+
         def pick(i):
-            return names[order[int(rng.choice(np.arange(n_models), p=probs_all))]]
+            return names[order[sched[i % 65536]]]
     else:
         def pick(i):
             return names[0]
