@@ -160,9 +160,14 @@ class ModelPool:
                 e.state = AVAILABLE
                 e.load_finished = time.monotonic()
                 self.load_durations.append(e.load_finished - e.load_started)
-            mt.engine_pool_models.labels(self.device).set(
-                sum(1 for x in self._entries.values()
-                    if x.state == AVAILABLE))
+            avail = [x for x in self._entries.values()
+                     if x.state == AVAILABLE and x.model is not None]
+            mt.engine_pool_models.labels(self.device).set(len(avail))
+            try:
+                mt.engine_pool_bytes.labels(self.device).set(
+                    sum(x.model.weight_bytes() for x in avail))
+            except Exception:       # noqa: BLE001
+                pass
             self._cond.notify_all()
 
     # -- event-driven wait (replaces the 500 ms poll loop) -----------------

@@ -33,6 +33,19 @@ constexpr int BM = 128, BN = 128, BK = 64;
 constexpr int N_WAVES = 4;                // 2x2 wave grid of 64x64 tiles
 constexpr int THREADS = N_WAVES * WAVE;   // 256
 
+// XCD-aware blockIdx remap (guide T1, bijective variant): the dispatcher
+// places block b on XCD b % 8, so consecutive tile ids (which share A/B
+// panels) land on different per-XCD L2s; remapping gives each XCD a
+// contiguous chunk of the grid. Applies when the kernel is HBM-bound
+// (+10% on large GEMMs); bijective for any grid size.
+TFSC_DEV int xcd_swizzle(int bid, int nblocks) {
+  constexpr int NXCD = 8;
+  if (nblocks < 2 * NXCD) return bid;
+  int q = nblocks / NXCD, r = nblocks % NXCD;
+  int xcd = bid % NXCD, idx = bid / NXCD;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
 // LDS tile: [128 rows][64 cols] bf16, 128-byte rows = 8 chunks of 16B.
 // XOR swizzle: chunk' = chunk ^ (row & 7)  (T2: <=2-way conflicts).
 TFSC_DEV int lds_off(int row, int chunk) {   // byte offset into a tile
@@ -201,8 +214,9 @@ void gemm_bf16_kernel(const ushort* __restrict__ A,
   const ushort* Bb = B + bat * strideB;
   ushort* Cb = C + bat * strideC;
 
-  const int tile_m = blockIdx.x % n_tiles_m;
-  const int tile_n = blockIdx.x / n_tiles_m;
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int tile_m = bid % n_tiles_m;
+  const int tile_n = bid / n_tiles_m;
   const int m0 = tile_m * BM;
   const int n0 = tile_n * BN;
 
@@ -375,8 +389,9 @@ void conv_igemm_kernel(const ushort* __restrict__ x,
   auto lds_a = [&](int buf) -> char* { return smem + buf * 32768; };
   auto lds_b = [&](int buf) -> char* { return smem + 16384 + buf * 32768; };
 
-  const int tile_m = blockIdx.x % n_tiles_m;
-  const int tile_n = blockIdx.x / n_tiles_m;
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int tile_m = bid % n_tiles_m;
+  const int tile_n = bid / n_tiles_m;
   const int m0 = tile_m * BM;
   const int n0 = tile_n * BN;
   const int wave = threadIdx.x / WAVE;

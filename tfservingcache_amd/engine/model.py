@@ -102,6 +102,14 @@ class LoadedModel:
 
     def _predict_impl(self, inputs: Dict[str, np.ndarray],
                       output_filter=None) -> Dict[str, np.ndarray]:
+        from ..utils import metrics as mt
+        with mt.engine_predict_duration.labels(self.name,
+                                               str(self.version),
+                                               self.device).time():
+            return self._predict_inner(inputs, output_filter)
+
+    def _predict_inner(self, inputs: Dict[str, np.ndarray],
+                       output_filter=None) -> Dict[str, np.ndarray]:
         plan = self.plan
         feeds: Dict[int, np.ndarray] = {}
         for alias, arr in inputs.items():
