@@ -127,6 +127,10 @@ def main() -> int:
                     help="CPU engine (CI smoke only; not a benchmark)")
     ap.add_argument("--dyn-batch", action="store_true",
                     help="enable server-side dynamic batching")
+    ap.add_argument("--transport", choices=["inproc", "grpc"],
+                    default="inproc",
+                    help="inproc: gRPC message path without sockets; "
+                         "grpc: real gRPC server+client over loopback")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -242,11 +246,32 @@ def main() -> int:
         def pick(i):
             return names[0]
 
-    def one_request(i: int) -> None:
-        data = req_cache[pick(i)]
-        req = m.PredictRequest.decode(data)
-        resp = handler.predict(req)
-        resp.encode()
+    grpc_server = None
+    if args.transport == "grpc":
+        import grpc as grpc_mod
+        from tfservingcache_amd.tfservingproxy import make_cache_grpc_server
+        grpc_server, _health = make_cache_grpc_server(
+            handler, max_workers=max(args.threads * 2, 8),
+            max_msg=256 * 1024 * 1024)
+        gport = grpc_server.add_insecure_port("127.0.0.1:0")
+        grpc_server.start()
+        channel = grpc_mod.insecure_channel(
+            f"127.0.0.1:{gport}",
+            options=[("grpc.max_receive_message_length", 256 * 1024 * 1024),
+                     ("grpc.max_send_message_length", 256 * 1024 * 1024)])
+        predict_rpc = channel.unary_unary(
+            "/tensorflow.serving.PredictionService/Predict",
+            request_serializer=lambda b: b,
+            response_deserializer=m.PredictResponse.decode)
+
+        def one_request(i: int) -> None:
+            predict_rpc(req_cache[pick(i)], timeout=300)
+    else:
+        def one_request(i: int) -> None:
+            data = req_cache[pick(i)]
+            req = m.PredictRequest.decode(data)
+            resp = handler.predict(req)
+            resp.encode()
 
     # initial load (timed -> cold-load sample even in warm mode)
     t0 = time.monotonic()
@@ -324,8 +349,10 @@ def main() -> int:
                 "pool_size": pool_cap,
                 "cold_load_p50_ms": round(cold_p50, 1),
                 "n_cold_loads": len(loads),
-                "transport": "in-process gRPC message path "
-                             "(protobuf decode/encode included)",
+                "transport": ("real gRPC server+client over TCP loopback"
+                              if args.transport == "grpc" else
+                              "in-process gRPC message path "
+                              "(protobuf decode/encode included)"),
                 "hipgraph": _graph_status(pool),
                 "dynamic_batching": bool(args.dyn_batch),
             },

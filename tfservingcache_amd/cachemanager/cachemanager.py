@@ -155,20 +155,32 @@ def make_cpu_loader(cache: LRUCache) -> Callable[[str, int], LoadedModel]:
 def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
                     max_batch: int = 64, use_graphs: bool = True,
                     batching: bool = False,
-                    batch_timeout_s: float = 0.002
+                    batch_timeout_s: float = 0.002,
+                    devices: Optional[List[str]] = None
                     ) -> Callable[[str, int], LoadedModel]:
-    """Loader that compiles the SavedModel onto one MI355X: weights land
+    """Loader that compiles the SavedModel onto an MI355X: weights land
     in the GPU's HBM pool (bf16, GEMM layouts pre-transformed) and the
     predict path runs the CDNA4 HIP kernels. With batching=True,
     concurrent Predicts merge server-side (TF Serving --enable_batching
-    analog)."""
+    analog). With `devices` (several GPUs in one process), each model is
+    assigned a GPU by consistent hash of model##version — the
+    single-process analog of the ring's (node, GPU) slots."""
+    import zlib
+
+    def pick_device(name: str, version: int) -> str:
+        if not devices:
+            return device
+        key = f"{name}##{version}"
+        return devices[zlib.crc32(key.encode()) % len(devices)]
+
     def loader(name: str, version: int) -> LoadedModel:
         from ..engine.gpu import GpuModel
         vdir = os.path.join(cache.base_dir, name, str(version))
         lm = load_model_from_dir(vdir, name, version)
-        lm._gpu = GpuModel(lm.plan, device=device, max_batch=max_batch,
+        dev = pick_device(name, version)
+        lm._gpu = GpuModel(lm.plan, device=dev, max_batch=max_batch,
                            use_graphs=use_graphs)
-        lm.device = device
+        lm.device = dev
         if batching:
             lm.enable_batching(max_batch=max_batch,
                                timeout_s=batch_timeout_s)

@@ -131,16 +131,17 @@ def create_cache_manager(cfg: Config) -> CacheManager:
     max_models = cfg.get_int("serving.maxConcurrentModels") or 2
     devices = _gpu_devices(cfg)
     if devices:
-        # one pool spanning the node's GPUs is modeled as one pool per
-        # process for now; device cuda:0 default, multi-GPU sharding via
-        # per-GPU ring slots + one process per GPU (see bench.py)
+        # models are hash-sharded over the node's visible GPUs (the
+        # single-process analog of per-GPU ring slots; for one process
+        # per GPU, set engine.gpus=1 and launch with CUDA_VISIBLE_DEVICES)
         loader = make_gpu_loader(
-            cache, device=devices[0],
+            cache, device=devices[0], devices=devices,
             max_batch=cfg.get_int("engine.maxbatch"),
             batching=cfg.get_bool("serving.batching.enabled"),
             batch_timeout_s=(cfg.get_float(
                 "serving.batching.batchTimeoutMicros") or 2000.0) / 1e6)
-        device = devices[0]
+        device = devices[0] if len(devices) == 1 else \
+            f"cuda[0-{len(devices) - 1}]"
     else:
         loader = make_cpu_loader(cache)
         device = "cpu"
