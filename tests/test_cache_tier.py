@@ -275,3 +275,20 @@ def test_classify_regress(stack):
     classes = cresp.result.classifications[0].classes
     assert len(classes) == 8
     assert abs(sum(c.score for c in classes) - 1.0) < 1e-4
+
+
+def test_version_labels(stack):
+    cm, handler = stack
+    req = m.ReloadConfigRequest(config=m.ModelServerConfig(
+        model_config_list=m.ModelConfigList(config=[m.ModelConfig(
+            name="half_plus_two", base_path="/x",
+            model_version_policy=m.ServableVersionPolicy(
+                specific=m.ServableVersionPolicySpecific(versions=[123])),
+            version_labels={"stable": 123})])))
+    handler.handle_reload_config(req)
+    presp = handler.predict(m.PredictRequest(
+        model_spec=m.ModelSpec(name="half_plus_two",
+                               version_label="stable"),
+        inputs={"x": numpy_to_tensorproto(
+            np.array([2.0], dtype=np.float32))}))
+    assert presp.model_spec.version.value == 123

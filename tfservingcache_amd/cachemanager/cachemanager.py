@@ -102,7 +102,10 @@ class CacheManager:
                         return self.pool.wait_available(
                             name, version, min(remaining,
                                                self.fetch_timeout / 4 + 1))
-                    except TimeoutError:
+                    except (TimeoutError, RuntimeError):
+                        # RuntimeError: the pool load failed (e.g. a
+                        # ReloadConfig raced the disk fetch) — the END
+                        # entry is re-created by the next reload
                         attempt += 1
                         if attempt >= 4:
                             raise

@@ -33,8 +33,14 @@ class ServingError(Exception):
 class LocalServingHandler:
     def __init__(self, cache_manager: CacheManager):
         self.cm = cache_manager
+        # model name -> {label: version}, populated by ReloadConfig
+        # (ModelConfig.version_labels, model_server_config.proto field 8)
+        self._version_labels: Dict[str, Dict[str, int]] = {}
 
     # -- helpers -----------------------------------------------------------
+    def _resolve_label(self, name: str, label: str) -> Optional[int]:
+        return self._version_labels.get(name, {}).get(label)
+
     def _resolve_version(self, name: str, version: int) -> int:
         if version:
             return version
@@ -52,9 +58,17 @@ class LocalServingHandler:
         raise ServingError(f"no versions of model {name} found",
                            m.ERROR_NOT_FOUND)
 
-    def get_model(self, name: str, version: int) -> Tuple[LoadedModel, int]:
+    def get_model(self, name: str, version: int,
+                  version_label: str = "") -> Tuple[LoadedModel, int]:
         if not name:
             raise ServingError("missing model name")
+        if not version and version_label:
+            labeled = self._resolve_label(name, version_label)
+            if labeled is None:
+                raise ServingError(
+                    f"unknown version label {version_label!r} for model "
+                    f"{name}", m.ERROR_NOT_FOUND)
+            version = labeled
         version = self._resolve_version(name, version)
         try:
             return self.cm.ensure_loaded(name, version), version
@@ -72,8 +86,10 @@ class LocalServingHandler:
     # -- Predict -----------------------------------------------------------
     def predict_arrays(self, name: str, version: int,
                        inputs: Dict[str, np.ndarray],
-                       output_filter=None) -> Tuple[Dict[str, np.ndarray], int]:
-        model, version = self.get_model(name, version)
+                       output_filter=None,
+                       version_label: str = ""
+                       ) -> Tuple[Dict[str, np.ndarray], int]:
+        model, version = self.get_model(name, version, version_label)
         sig = model.signature_def
         # anonymous single input ('' key) -> sole signature input
         if "" in inputs:
@@ -111,7 +127,8 @@ class LocalServingHandler:
                 raise ServingError(f"bad tensor for input {alias!r}: {e}")
         outputs, version = self.predict_arrays(
             spec.name, spec.version_value(), inputs,
-            list(req.output_filter) or None)
+            list(req.output_filter) or None,
+            version_label=spec.version_label)
         resp = m.PredictResponse(model_spec=m.ModelSpec(
             name=spec.name, version=m.Int64Value(value=version),
             signature_name=spec.signature_name or
@@ -232,6 +249,9 @@ class LocalServingHandler:
         desired: List[Tuple[str, int]] = []
         if cfg is not None and cfg.model_config_list is not None:
             for mc in cfg.model_config_list.config:
+                if mc.version_labels:
+                    self._version_labels[mc.name] = {
+                        k: int(v) for k, v in mc.version_labels.items()}
                 pol = mc.model_version_policy
                 versions: List[int] = []
                 if pol is not None and pol.specific is not None:
@@ -241,9 +261,23 @@ class LocalServingHandler:
                     versions = [v]
                 for v in versions:
                     desired.append((mc.name, v))
+        # fetch missing models to the disk cache in the background, then
+        # the pool reload picks them up (declarative desired-state push)
+        import threading
+        for name, v in desired:
+            if not self.cm.cache.contains(name, v):
+                threading.Thread(target=self._prefetch, args=(name, v),
+                                 daemon=True).start()
         self.cm.pool.reload(desired, self.cm._version_dir)
         return m.ReloadConfigResponse(status=m.StatusProto(
             error_code=m.ERROR_OK))
+
+    def _prefetch(self, name: str, version: int) -> None:
+        try:
+            self.cm.ensure_loaded(name, version)
+        except Exception:       # noqa: BLE001
+            log.warning("prefetch of %s:%d failed", name, version,
+                        exc_info=True)
 
     # -- SessionRun --------------------------------------------------------
     def session_run(self, req: m.SessionRunRequest) -> m.SessionRunResponse:
