@@ -214,8 +214,63 @@ __global__ void k_softmax(const ushort* __restrict__ x,
   }
 }
 
+// small-row softmax: one WAVE per row, the whole row in registers
+// (cols <= 8*VEC*... handled for cols <= 512, 8 bf16 per lane), single
+// read of x — the block-per-row kernel re-reads x three times and idles
+// (256-cols) threads (attention S=128 rows ran at ~0.85 TB/s before).
+template <int VPL>   // values per lane (8 -> cols<=512)
+__global__ void k_softmax_wave(const ushort* __restrict__ x,
+                               ushort* __restrict__ y, int64_t rows,
+                               int cols) {
+  int64_t row0 = (int64_t)blockIdx.x * (TPB / WAVE) + threadIdx.x / WAVE;
+  int64_t wstride = (int64_t)gridDim.x * (TPB / WAVE);
+  int lane = threadIdx.x % WAVE;
+  for (int64_t row = row0; row < rows; row += wstride) {
+    const ushort* xr = x + row * cols;
+    ushort* yr = y + row * cols;
+    float v[VPL];
+    int n = (cols + WAVE - 1) / WAVE;   // elements this lane handles
+    float mx = -3.0e38f;
+    #pragma unroll
+    for (int j = 0; j < VPL; ++j) {
+      int i = j * WAVE + lane;
+      v[j] = (j < n && i < cols) ? bf2f(xr[i]) : -3.0e38f;
+      mx = fmaxf(mx, v[j]);
+    }
+    #pragma unroll
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+      mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+    float sum = 0.f;
+    #pragma unroll
+    for (int j = 0; j < VPL; ++j) {
+      v[j] = __expf(v[j] - mx);
+      sum += v[j];
+    }
+    #pragma unroll
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+      sum += __shfl_xor(sum, off, WAVE);
+    float inv = 1.f / sum;
+    #pragma unroll
+    for (int j = 0; j < VPL; ++j) {
+      int i = j * WAVE + lane;
+      if (j < n && i < cols) yr[i] = f2bf(v[j] * inv);
+    }
+  }
+}
+
 void launch_softmax(hipStream_t s, const ushort* x, ushort* y,
                     int64_t rows, int64_t cols) {
+  if (cols <= 512) {
+    int64_t wgs = ceil_div(rows, (int64_t)(TPB / WAVE));
+    int blocks = int(wgs < MAX_BLOCKS ? (wgs > 0 ? wgs : 1) : MAX_BLOCKS);
+    if (cols <= 128)
+      hipLaunchKernelGGL(k_softmax_wave<2>, dim3(blocks), dim3(TPB), 0, s,
+                         x, y, rows, int(cols));
+    else
+      hipLaunchKernelGGL(k_softmax_wave<8>, dim3(blocks), dim3(TPB), 0, s,
+                         x, y, rows, int(cols));
+    return;
+  }
   int blocks = int(rows < MAX_BLOCKS ? rows : MAX_BLOCKS);
   hipLaunchKernelGGL(k_softmax, dim3(blocks), dim3(TPB), 0, s,
                      x, y, rows, cols);
@@ -378,6 +433,31 @@ __global__ void k_transpose(const ushort* __restrict__ x,
   }
 }
 
+// fast path: the permutation keeps the LAST dim innermost (e.g. the
+// attention head split/merge perm [0,2,1,3]) and it is a multiple of 8
+// bf16 — move 16-byte chunks instead of scalars
+__global__ void k_transpose_v8(const ushort* __restrict__ x,
+                               ushort* __restrict__ y, TransArgs ta,
+                               int64_t n_grp, int last) {
+  int grp_per_row = last >> 3;
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n_grp; i += stride) {
+    int g = int(i % grp_per_row);
+    int64_t rem = i / grp_per_row;        // linear index over outer dims
+    int64_t src = 0;
+    #pragma unroll
+    for (int d = MAX_DIMS - 1; d >= 0; --d) {
+      if (d >= ta.ndim - 1) continue;     // skip last dim
+      int64_t c = rem % ta.out_dims[d];
+      rem /= ta.out_dims[d];
+      src += c * ta.in_strides[d];
+    }
+    *reinterpret_cast<uint4*>(y + i * 8) =
+        *reinterpret_cast<const uint4*>(x + src + g * 8);
+  }
+}
+
 void launch_transpose(hipStream_t s, const ushort* x, ushort* y, int ndim,
                       const int64_t* out_dims, const int64_t* in_strides,
                       int64_t n_out) {
@@ -386,6 +466,13 @@ void launch_transpose(hipStream_t s, const ushort* x, ushort* y, int ndim,
   for (int d = 0; d < MAX_DIMS; ++d) {
     ta.out_dims[d] = d < ndim ? out_dims[d] : 1;
     ta.in_strides[d] = d < ndim ? in_strides[d] : 0;
+  }
+  int last = ndim > 0 ? int(out_dims[ndim - 1]) : 0;
+  if (ndim >= 2 && in_strides[ndim - 1] == 1 && last % 8 == 0) {
+    int64_t n_grp = n_out / 8;
+    hipLaunchKernelGGL(k_transpose_v8, dim3(grid_for(n_grp)), dim3(TPB),
+                       0, s, x, y, ta, n_grp, last);
+    return;
   }
   hipLaunchKernelGGL(k_transpose, dim3(grid_for(n_out)), dim3(TPB), 0, s,
                      x, y, ta, n_out);
