@@ -51,18 +51,22 @@ TFSC_DEV int lds_off(int row, int chunk) {   // byte offset into a tile
   return row * 128 + ((chunk ^ (row & 7)) << 4);
 }
 
+template <int ROWS, int NWAVES>
 TFSC_DEV void stage_tile_glds(const ushort* __restrict__ src, int64_t ld,
                               int row_limit, int row0, int k0,
                               char* lds_tile, int wave, int lane) {
-  // one wave stages 32 rows (4 glds x 8 rows); lane l covers
-  // (row = 8*i + l/8, chunk = l%8) of its wave's 32-row slice. The glds
-  // LDS destination is wave-uniform-base + lane*16 (lane-linear), which
-  // matches the [row][chunk] image exactly.
+  // the block's NWAVES waves stage ROWS rows together (each glds covers
+  // 8 rows x 64 cols = 1 KiB); lane l covers (row = 8*i + l/8,
+  // chunk = l%8) of its wave's slice. The glds LDS destination is
+  // wave-uniform-base + lane*16 (lane-linear), which matches the
+  // [row][chunk] image exactly.
+  constexpr int ROWS_PER_WAVE = ROWS / NWAVES;
+  constexpr int N_GLDS = ROWS_PER_WAVE / 8;
   int r_in = (lane >> 3);           // 0..7
   int chunk = lane & 7;             // 0..7
   #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    int row = wave * 32 + i * 8 + r_in;
+  for (int i = 0; i < N_GLDS; ++i) {
+    int row = wave * ROWS_PER_WAVE + i * 8 + r_in;
     int grow = row0 + row;
     grow = grow < row_limit ? grow : row_limit;
     // source chunk is pre-swizzled so the LINEAR lds image holds the
@@ -70,7 +74,7 @@ TFSC_DEV void stage_tile_glds(const ushort* __restrict__ src, int64_t ld,
     int chunk_src = chunk ^ (row & 7);
     const ushort* gptr = src + (int64_t)grow * ld + k0 + chunk_src * 8;
     // wave-uniform base for this glds: start of the wave's 8-row slice
-    char* lds_base = lds_tile + (wave * 32 + i * 8) * 128;
+    char* lds_base = lds_tile + (wave * ROWS_PER_WAVE + i * 8) * 128;
     __builtin_amdgcn_global_load_lds(
         reinterpret_cast<const uint32_t*>(gptr),
         reinterpret_cast<uint32_t*>(lds_base), 16, 0, 0);
@@ -80,18 +84,21 @@ TFSC_DEV void stage_tile_glds(const ushort* __restrict__ src, int64_t ld,
 // register-staged transpose path for B given as [K][N] (attention P@V):
 // each lane loads 8 bf16 along N (coalesced) and scatters them into the
 // [N][K]-image LDS tile with the same XOR swizzle.
+template <int ROWS, int NWAVES>
 TFSC_DEV void stage_tile_transposed(const ushort* __restrict__ src,
                                     int64_t ld, int k_limit, int n0, int k0,
                                     ushort* lds_tile, int wave, int lane,
                                     int n_limit) {
-  // tile wanted: rows = n (128), cols = k (64). Source element (k, n).
-  // 256 threads x 8 elems = 2048 elems/pass; tile = 8192 elems -> 4 passes.
+  // tile wanted: rows = n (ROWS), cols = k (64). Source element (k, n).
+  constexpr int NG = ROWS / 8;            // n-groups of 8
+  constexpr int THREADS_ = NWAVES * WAVE;
+  constexpr int PASSES = (ROWS * 64) / (THREADS_ * 8);
   int tid = wave * WAVE + lane;
   #pragma unroll
-  for (int pass = 0; pass < 4; ++pass) {
-    int idx = pass * THREADS + tid;       // covers k-major: 64 k x 16 ngroups
-    int k = idx >> 4;                     // 0..63
-    int ng = idx & 15;                    // n-group of 8
+  for (int pass = 0; pass < PASSES; ++pass) {
+    int idx = pass * THREADS_ + tid;      // covers k-major: 64 k x NG groups
+    int k = idx / NG;                     // 0..63
+    int ng = idx % NG;                    // n-group of 8
     int gk = k0 + k;
     gk = gk < k_limit ? gk : k_limit;
     const ushort* gptr = src + (int64_t)gk * ld + n0 + ng * 8;
@@ -193,8 +200,12 @@ TFSC_DEV void epilogue_store(f32x4_t (&acc)[4][4], char* smem,
   }
 }
 
-template <bool TRANS_B, bool HAS_BIAS, bool HAS_RES>
-__global__ __launch_bounds__(THREADS)
+// Geometry: WMW x WNW waves, each owning a 64x64 output tile
+// (BM = WMW*64, BN = WNW*64). (2,2) = the full 128x128 tile; (1,2) and
+// (1,1) keep medium shapes (e.g. BERT dense layers, attention batched
+// GEMMs) from under-filling the 256-CU chip.
+template <int WMW, int WNW, bool TRANS_B, bool HAS_BIAS, bool HAS_RES>
+__global__ __launch_bounds__(WMW * WNW * WAVE)
 void gemm_bf16_kernel(const ushort* __restrict__ A,
                       const ushort* __restrict__ B,
                       const ushort* __restrict__ bias,
@@ -203,10 +214,14 @@ void gemm_bf16_kernel(const ushort* __restrict__ A,
                       int M, int N, int K, int act, float alpha,
                       int64_t strideA, int64_t strideB, int64_t strideC,
                       int n_tiles_m) {
-  __shared__ __attribute__((aligned(16))) char smem[4 * BM * BK * 2];
-  // [buf][A|B] tiles, 16KB each: A0 | B0 | A1 | B1
-  auto lds_a = [&](int buf) -> char* { return smem + buf * 32768; };
-  auto lds_b = [&](int buf) -> char* { return smem + 16384 + buf * 32768; };
+  constexpr int BM_ = WMW * 64, BN_ = WNW * 64;
+  constexpr int NW = WMW * WNW;
+  constexpr int A_BYTES = BM_ * BK * 2, B_BYTES = BN_ * BK * 2;
+  __shared__ __attribute__((aligned(16))) char smem[2 * (A_BYTES + B_BYTES)];
+  auto lds_a = [&](int buf) -> char* {
+    return smem + buf * (A_BYTES + B_BYTES); };
+  auto lds_b = [&](int buf) -> char* {
+    return smem + buf * (A_BYTES + B_BYTES) + A_BYTES; };
 
   const int bat = blockIdx.y;
   const ushort* Ab = A + bat * strideA;
@@ -216,12 +231,12 @@ void gemm_bf16_kernel(const ushort* __restrict__ A,
   const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tile_m = bid % n_tiles_m;
   const int tile_n = bid / n_tiles_m;
-  const int m0 = tile_m * BM;
-  const int n0 = tile_n * BN;
+  const int m0 = tile_m * BM_;
+  const int n0 = tile_n * BN_;
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const int wm = wave >> 1, wn = wave & 1;      // 2x2 wave grid
+  const int wm = wave / WNW, wn = wave % WNW;
 
   f32x4_t acc[4][4] = {};
 
@@ -229,13 +244,13 @@ void gemm_bf16_kernel(const ushort* __restrict__ A,
 
   // prologue: stage tile 0 into buf 0
   {
-    stage_tile_glds(Ab, K, M - 1, m0, 0, lds_a(0), wave, lane);
+    stage_tile_glds<BM_, NW>(Ab, K, M - 1, m0, 0, lds_a(0), wave, lane);
     if (TRANS_B) {
-      stage_tile_glds(Bb, K, N - 1, n0, 0, lds_b(0), wave, lane);
+      stage_tile_glds<BN_, NW>(Bb, K, N - 1, n0, 0, lds_b(0), wave, lane);
     } else {
-      stage_tile_transposed(Bb, N, K - 1, n0, 0,
-                            reinterpret_cast<ushort*>(lds_b(0)), wave, lane,
-                            N);
+      stage_tile_transposed<BN_, NW>(
+          Bb, N, K - 1, n0, 0, reinterpret_cast<ushort*>(lds_b(0)), wave,
+          lane, N);
     }
   }
 
@@ -248,13 +263,15 @@ void gemm_bf16_kernel(const ushort* __restrict__ A,
     __syncthreads();
     if (kt + 1 < n_ktiles) {
       int k0 = (kt + 1) * BK;
-      stage_tile_glds(Ab, K, M - 1, m0, k0, lds_a(cur ^ 1), wave, lane);
+      stage_tile_glds<BM_, NW>(Ab, K, M - 1, m0, k0, lds_a(cur ^ 1),
+                               wave, lane);
       if (TRANS_B) {
-        stage_tile_glds(Bb, K, N - 1, n0, k0, lds_b(cur ^ 1), wave, lane);
+        stage_tile_glds<BN_, NW>(Bb, K, N - 1, n0, k0, lds_b(cur ^ 1),
+                                 wave, lane);
       } else {
-        stage_tile_transposed(Bb, N, K - 1, n0, k0,
-                              reinterpret_cast<ushort*>(lds_b(cur ^ 1)),
-                              wave, lane, N);
+        stage_tile_transposed<BN_, NW>(
+            Bb, N, K - 1, n0, k0,
+            reinterpret_cast<ushort*>(lds_b(cur ^ 1)), wave, lane, N);
       }
     }
 
@@ -296,6 +313,27 @@ void gemm_bf16_kernel(const ushort* __restrict__ A,
                                     alpha);
 }
 
+template <int WMW, int WNW, bool TRANS_B>
+static void gemm_launch_geom(hipStream_t s, const ushort* A,
+                             const ushort* B, const ushort* bias,
+                             const ushort* residual, ushort* C,
+                             int64_t bat, int64_t M, int64_t N, int64_t K,
+                             int act, float alpha, int64_t sA, int64_t sB,
+                             int64_t sC) {
+  int ntm = int(ceil_div(M, WMW * 64)), ntn = int(ceil_div(N, WNW * 64));
+  dim3 grid(ntm * ntn, (unsigned)bat);
+  dim3 block(WMW * WNW * WAVE);
+  bool hb = bias != nullptr, hr = residual != nullptr;
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, block, 0, s, A, B, bias, residual, C,
+                       (int)M, (int)N, (int)K, act, alpha, sA, sB, sC, ntm);
+  };
+  if (hb && hr)  launch(gemm_bf16_kernel<WMW, WNW, TRANS_B, true, true>);
+  else if (hb)   launch(gemm_bf16_kernel<WMW, WNW, TRANS_B, true, false>);
+  else if (hr)   launch(gemm_bf16_kernel<WMW, WNW, TRANS_B, false, true>);
+  else           launch(gemm_bf16_kernel<WMW, WNW, TRANS_B, false, false>);
+}
+
 template <bool TRANS_B>
 static void gemm_dispatch(hipStream_t s, const ushort* A, const ushort* B,
                           const ushort* bias, const ushort* residual,
@@ -305,18 +343,22 @@ static void gemm_dispatch(hipStream_t s, const ushort* A, const ushort* B,
   if (K % BK != 0)
     throw std::runtime_error("gemm: K must be a multiple of 64 (got " +
                              std::to_string(K) + ")");
-  int ntm = int(ceil_div(M, BM)), ntn = int(ceil_div(N, BN));
-  dim3 grid(ntm * ntn, (unsigned)bat);
-  dim3 block(THREADS);
-  bool hb = bias != nullptr, hr = residual != nullptr;
-  auto launch = [&](auto kern) {
-    hipLaunchKernelGGL(kern, grid, block, 0, s, A, B, bias, residual, C,
-                       (int)M, (int)N, (int)K, act, alpha, sA, sB, sC, ntm);
+  // geometry by fill: prefer 128x128 tiles; shrink while the grid
+  // under-fills the chip (2 blocks/CU at 64KB LDS -> target >=512,
+  // accept >=232 = ~1 block/CU)
+  auto blocks = [&](int bm, int bn) {
+    return bat * ceil_div(M, bm) * ceil_div(N, bn);
   };
-  if (hb && hr)       launch(gemm_bf16_kernel<TRANS_B, true, true>);
-  else if (hb)        launch(gemm_bf16_kernel<TRANS_B, true, false>);
-  else if (hr)        launch(gemm_bf16_kernel<TRANS_B, false, true>);
-  else                launch(gemm_bf16_kernel<TRANS_B, false, false>);
+  if (blocks(128, 128) >= 232) {
+    gemm_launch_geom<2, 2, TRANS_B>(s, A, B, bias, residual, C, bat, M, N,
+                                    K, act, alpha, sA, sB, sC);
+  } else if (blocks(64, 128) >= 232) {
+    gemm_launch_geom<1, 2, TRANS_B>(s, A, B, bias, residual, C, bat, M, N,
+                                    K, act, alpha, sA, sB, sC);
+  } else {
+    gemm_launch_geom<1, 1, TRANS_B>(s, A, B, bias, residual, C, bat, M, N,
+                                    K, act, alpha, sA, sB, sC);
+  }
 }
 
 void launch_gemm(hipStream_t s, const ushort* A, const ushort* B,
@@ -401,7 +443,7 @@ void conv_igemm_kernel(const ushort* __restrict__ x,
   const int n_ktiles = K / BK;
 
   stage_tile_conv_a(x, zeros, g, M, m0, 0, lds_a(0), wave, lane);
-  stage_tile_glds(B, K, N - 1, n0, 0, lds_b(0), wave, lane);
+  stage_tile_glds<128, 4>(B, K, N - 1, n0, 0, lds_b(0), wave, lane);
 
   int cur = 0;
   for (int kt = 0; kt < n_ktiles; ++kt) {
@@ -409,7 +451,8 @@ void conv_igemm_kernel(const ushort* __restrict__ x,
     if (kt + 1 < n_ktiles) {
       int k0 = (kt + 1) * BK;
       stage_tile_conv_a(x, zeros, g, M, m0, k0, lds_a(cur ^ 1), wave, lane);
-      stage_tile_glds(B, K, N - 1, n0, k0, lds_b(cur ^ 1), wave, lane);
+      stage_tile_glds<128, 4>(B, K, N - 1, n0, k0, lds_b(cur ^ 1), wave,
+                              lane);
     }
     const char* at = lds_a(cur);
     const char* bt = lds_b(cur);
