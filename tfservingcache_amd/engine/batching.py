@@ -22,6 +22,25 @@ from typing import Dict, List, Optional, Tuple
 import numpy as np
 
 
+class SegmentedBatch:
+    """A batch-dim concatenation deferred until the H2D copy: the engine
+    copies each segment into its pinned buffer at its row offset instead
+    of materializing np.concatenate (which costs ~0.5 ms for 8 image
+    requests)."""
+
+    __slots__ = ("segments", "shape", "dtype", "ndim")
+
+    def __init__(self, segments):
+        self.segments = segments
+        rows = sum(s.shape[0] for s in segments)
+        self.shape = (rows,) + tuple(segments[0].shape[1:])
+        self.dtype = segments[0].dtype
+        self.ndim = segments[0].ndim
+
+    def materialize(self) -> np.ndarray:
+        return np.concatenate(self.segments, axis=0)
+
+
 @dataclass(eq=False)
 class _Item:
     inputs: Dict[str, np.ndarray]
@@ -127,8 +146,10 @@ class DynamicBatcher:
                 out = self._run(batch[0].inputs, output_filter)
                 batch[0].out = out
             else:
+                # segmented merge: the engine copies each segment straight
+                # into its pinned staging buffer (no concatenate pass)
                 merged = {
-                    k: np.concatenate([i.inputs[k] for i in batch], axis=0)
+                    k: SegmentedBatch([i.inputs[k] for i in batch])
                     for k in batch[0].inputs
                 }
                 out = self._run(merged, output_filter)
@@ -172,8 +193,8 @@ class DynamicBatcher:
             if len(take) == 1:
                 take[0].out = self._run(take[0].inputs, filt)
             else:
-                merged = {k: np.concatenate([i.inputs[k] for i in take],
-                                            axis=0) for k in take[0].inputs}
+                merged = {k: SegmentedBatch([i.inputs[k] for i in take])
+                          for k in take[0].inputs}
                 out = self._run(merged, filt)
                 off = 0
                 for i in take:

@@ -62,6 +62,8 @@ class CpuExecutor:
             if t.kind == "weight" and t.weight is not None:
                 vals[t.idx] = t.weight
         for idx, arr in feeds.items():
+            if hasattr(arr, "materialize"):
+                arr = arr.materialize()
             want = np.int32 if plan.tensors[idx].dtype == "i32" else np.float32
             vals[idx] = np.asarray(arr, dtype=want)
         # resolve aliases of fed tensors lazily below via _get

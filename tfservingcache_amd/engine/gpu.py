@@ -515,19 +515,29 @@ class ExecContext:
                 torch.cuda.stream(self.stream):
             for idx, arr in feeds.items():
                 v = self.view(idx)
-                arr = np.ascontiguousarray(arr)
-                t = torch.from_numpy(arr)
                 pin = self._pinned_in.get(idx)
                 if pin is None or pin.shape != v.shape:
                     pin = torch.empty(v.shape, dtype=v.dtype,
                                       pin_memory=True)
                     self._pinned_in[idx] = pin
-                rows = t.shape[0] if t.ndim else 0
-                if t.ndim and rows < v.shape[0]:
-                    pin[:rows].copy_(t)      # CPU-side dtype convert
-                    pin[rows:].zero_()
+                if hasattr(arr, "segments"):
+                    # segmented batch: copy each request's rows straight
+                    # into pinned staging (skips np.concatenate)
+                    r0 = 0
+                    for seg in arr.segments:
+                        t = torch.from_numpy(np.ascontiguousarray(seg))
+                        pin[r0:r0 + seg.shape[0]].copy_(t)
+                        r0 += seg.shape[0]
+                    if r0 < v.shape[0]:
+                        pin[r0:].zero_()
                 else:
-                    pin.copy_(t.reshape(v.shape))
+                    t = torch.from_numpy(np.ascontiguousarray(arr))
+                    rows = t.shape[0] if t.ndim else 0
+                    if t.ndim and rows < v.shape[0]:
+                        pin[:rows].copy_(t)      # CPU-side dtype convert
+                        pin[rows:].zero_()
+                    else:
+                        pin.copy_(t.reshape(v.shape))
                 v.copy_(pin, non_blocking=True)
             if self.gm.use_graphs and not self.captured:
                 # warm-up eager run, then capture on this stream

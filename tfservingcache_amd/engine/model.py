@@ -96,7 +96,8 @@ class LoadedModel:
         server-side when enable_batching was called)."""
         if self._batcher is not None:
             return self._batcher.predict(
-                {k: np.asarray(v) for k, v in inputs.items()},
+                {k: np.asarray(v) if not hasattr(v, "segments") else v
+                 for k, v in inputs.items()},
                 output_filter)
         return self._predict_impl(inputs, output_filter)
 
@@ -119,7 +120,7 @@ class LoadedModel:
                 idx = plan.by_name.get(alias if ":" in alias else alias + ":0")
             if idx is None:
                 raise ModelExecError(f"unknown input {alias!r}")
-            feeds[idx] = np.asarray(arr)
+            feeds[idx] = arr if hasattr(arr, "segments") else np.asarray(arr)
         missing = [a for a, i in plan.sig_inputs.items() if i not in feeds]
         if missing:
             raise ModelExecError(f"missing inputs: {missing}")
