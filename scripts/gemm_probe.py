@@ -164,6 +164,25 @@ def main():
         ok &= check(f"conv_igemm {H}x{W}x{C}->{Kc} s{st}", out, want,
                     rtol=0.1, atol=0.1)
 
+    # --- fused attention (flash-style) -----------------------------------
+    for (B, S, H) in [(2, 128, 4), (1, 100, 2), (3, 64, 12)]:
+        D = 64
+        q = bfbuf(torch.randn(B, S, H, D, device=dev) * 0.5)
+        k = bfbuf(torch.randn(B, S, H, D, device=dev) * 0.5)
+        vv = bfbuf(torch.randn(B, S, H, D, device=dev) * 0.5)
+        o = torch.empty(B, S, H, D, device=dev, dtype=torch.bfloat16)
+        scale = 1.0 / (D ** 0.5)
+        run_plan([(ext.K_ATTENTION,
+                   [q.data_ptr(), k.data_ptr(), vv.data_ptr(),
+                    o.data_ptr()], [B, S, H, D], [scale])])
+        qf = q.float().permute(0, 2, 1, 3)     # [B,H,S,D]
+        kf = k.float().permute(0, 2, 1, 3)
+        vf = vv.float().permute(0, 2, 1, 3)
+        want = torch.softmax(qf @ kf.transpose(-1, -2) * scale, dim=-1) @ vf
+        want = want.permute(0, 2, 1, 3)
+        ok &= check(f"attention B{B} S{S} H{H}", o, want, rtol=0.08,
+                    atol=0.02)
+
     # --- pool -------------------------------------------------------------
     x = bfbuf(torch.randn(2, 16, 16, 32, device=dev))
     out = torch.empty(2, 8, 8, 32, device=dev, dtype=torch.bfloat16)

@@ -22,6 +22,7 @@ sources = [
     os.path.join(CSRC, "ops", "ops_memory.hip"),
     os.path.join(CSRC, "ops", "gemm.hip"),
     os.path.join(CSRC, "ops", "conv.hip"),
+    os.path.join(CSRC, "ops", "attention.hip"),
 ]
 
 setup(

@@ -139,3 +139,20 @@ def test_native_extension_is_loaded():
     from tfservingcache_amd.engine import _tfsc_engine as ext
     assert hasattr(ext, "ExecPlan")
     assert ext.__file__.endswith(".so")
+
+
+def test_bert_fused_attention_gpu_vs_cpu(tmp_path):
+    """head_dim=64 BERT uses the fused flash-attention kernel on GPU."""
+    sm = build_bert(seq_len=128, hidden=256, layers=2, heads=4,
+                    intermediate=512, vocab=500, seed=21)
+    gm = _gpu_model(tmp_path, sm, name="bfa")
+    cm = _cpu_model(tmp_path, sm, name="bfa_cpu")
+    assert any(op.kind == "attention" for op in gm.plan.ops)
+    ids = np.random.default_rng(5).integers(0, 500, (2, 128)).astype(
+        np.int32)
+    g = gm.predict({"input_ids": ids})
+    c = cm.predict({"input_ids": ids})
+    _compare({"pooled_output": g["pooled_output"]},
+             {"pooled_output": c["pooled_output"]}, rtol=0.1, atol=0.08)
+    _compare({"sequence_output": g["sequence_output"]},
+             {"sequence_output": c["sequence_output"]}, rtol=0.2, atol=0.15)

@@ -105,3 +105,35 @@ def test_bert_layernorm_numerics(tmp_path):
     var = out.var(-1)
     np.testing.assert_allclose(mean, np.zeros_like(mean), atol=1e-5)
     np.testing.assert_allclose(var, np.ones_like(var), rtol=1e-3)
+
+
+def test_attention_fusion(tmp_path):
+    """head_dim==64 attention collapses to one fused op; numerics match
+    the unfused plan."""
+    from tfservingcache_amd.engine.planner import _Lowerer
+    from tfservingcache_amd.engine.savedmodel import read_saved_model
+    from tfservingcache_amd.engine.executor_cpu import CpuExecutor
+    from tfservingcache_amd.engine.model import LoadedModel
+
+    sm = build_bert(seq_len=16, hidden=128, layers=1, heads=2,
+                    intermediate=64, vocab=60, seed=13)
+    d = tmp_path / "bf" / "1"
+    write_saved_model(sm, str(d))
+    model = load_model_from_dir(str(d), "bf", 1)   # fused (default path)
+    kinds = [op.kind for op in model.plan.ops]
+    assert kinds.count("attention") == 1
+    assert "softmax" not in kinds
+    assert "batched_gemm" not in kinds
+    assert "transpose" not in kinds
+
+    gd, sigs = read_saved_model(str(d))
+    unfused_plan = _Lowerer(gd, next(iter(sigs.values()))).run()
+    unfused = LoadedModel("bf", 1, unfused_plan)
+
+    ids = np.random.default_rng(3).integers(0, 60, (2, 16)).astype(np.int32)
+    a = model.predict({"input_ids": ids})
+    b = unfused.predict({"input_ids": ids})
+    np.testing.assert_allclose(a["sequence_output"], b["sequence_output"],
+                               rtol=1e-5, atol=1e-6)
+    np.testing.assert_allclose(a["pooled_output"], b["pooled_output"],
+                               rtol=1e-5, atol=1e-6)

@@ -39,6 +39,7 @@ enum CallKind : int {
   K_IM2COL,
   K_CONV,
   K_PAD_LAST,
+  K_ATTENTION,
 };
 
 struct Call {
@@ -134,6 +135,11 @@ static void launch_call(const Call& c, hipStream_t s) {
     }
     case K_PAD_LAST:
       launch_pad_last(s, cp(0), p(1), I[0], int(I[1]), int(I[2]));
+      break;
+    case K_ATTENTION:
+      // ptrs: [q, k, v, out]; ints: [B, S, H, D]; floats: [scale]
+      launch_attention(s, cp(0), cp(1), cp(2), p(3), int(I[0]), int(I[1]),
+                       int(I[2]), int(I[3]), c.floats[0]);
       break;
     default:
       throw std::runtime_error("unknown call kind " +
@@ -245,6 +251,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("K_IM2COL") = int(tfsc::K_IM2COL);
   mod.attr("K_CONV") = int(tfsc::K_CONV);
   mod.attr("K_PAD_LAST") = int(tfsc::K_PAD_LAST);
+  mod.attr("K_ATTENTION") = int(tfsc::K_ATTENTION);
 
   // elementwise fn codes
   mod.attr("ELT_ADD") = int(tfsc::ELT_ADD);

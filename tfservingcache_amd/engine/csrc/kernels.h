@@ -99,4 +99,10 @@ void launch_im2col(hipStream_t s, const ushort* x, ushort* y,
                    int sh, int sw, int pt, int pl, int Ho, int Wo,
                    int k_pad);
 
+// Fused multi-head attention over the natural [B*S, H*D] QKV layout
+// (flash-style online softmax; D must be 64). See ops/attention.hip.
+void launch_attention(hipStream_t s, const ushort* Q, const ushort* K,
+                      const ushort* V, ushort* O, int B, int S, int H,
+                      int D_, float scale);
+
 }  // namespace tfsc

@@ -1,0 +1,281 @@
+// Fused multi-head attention (flash-style, online softmax) for gfx950.
+//
+//   out[b,s,h,:] = softmax(scale * Q[b,s,h,:] . K[b,:,h,:]^T) @ V[b,:,h,:]
+//
+// Consumes Q/K/V in their NATURAL layout [B*S, H*D] (the output of the
+// QKV projection GEMMs) and writes out in the same layout — replacing
+// the unfused plan (4 transposes + 2 batched GEMMs + standalone softmax
+// per layer, ~31% of BERT kernel time) with ONE kernel and no S x S
+// score materialization.
+//
+// Geometry: one workgroup (4 waves) per (b, h, 64-row q-block). Q block
+// kept in registers (2 A-fragments per wave), K and V^T tiles
+// double-buffered in LDS with the gemm.hip XOR swizzle; per kv-tile:
+// QK^T on MFMA -> in-register online softmax (16-lane-group shfl
+// reduction per q-row) -> P bounced through LDS into A-fragment layout
+// -> P@V on MFMA into the O accumulator. Requires head_dim D == 64
+// (BERT-base class); the planner keeps the unfused path otherwise.
+#include "../common.h"
+#include "../kernels.h"
+
+#include <stdexcept>
+
+namespace tfsc {
+
+namespace attn {
+
+using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4_t = __attribute__((ext_vector_type(4))) float;
+
+constexpr int D = 64;          // head dim
+constexpr int QBLK = 64;       // q rows per workgroup (16 per wave)
+constexpr int KVBLK = 64;      // kv rows per tile
+constexpr int NW = 4;          // waves
+constexpr int THREADS = NW * WAVE;
+
+// shared 64x64 bf16 tile image with the gemm.hip chunk-XOR swizzle
+TFSC_DEV int t_off(int row, int chunk) {
+  return row * 128 + ((chunk ^ (row & 7)) << 4);
+}
+
+// stage a [64][64] tile from rows of a [rows_total, H*D] matrix
+// (row = s index, stride ld elements, cols = one head's D slice)
+TFSC_DEV void stage_rows_glds(const ushort* __restrict__ src, int64_t ld,
+                              int row0, int row_limit, char* lds_tile,
+                              int wave, int lane) {
+  int r_in = lane >> 3, chunk = lane & 7;
+  #pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int row = wave * 16 + i * 8 + r_in;
+    int grow = row0 + row;
+    grow = grow < row_limit ? grow : row_limit;
+    int chunk_src = chunk ^ (row & 7);
+    const ushort* gptr = src + (int64_t)grow * ld + chunk_src * 8;
+    char* lds_base = lds_tile + (wave * 16 + i * 8) * 128;
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const uint32_t*>(gptr),
+        reinterpret_cast<uint32_t*>(lds_base), 16, 0, 0);
+  }
+}
+
+// stage V^T: [d][kv] image from V rows [kv, H*D] (d-contiguous reads,
+// scattered 2B LDS writes — the bgemm no-trans pattern)
+TFSC_DEV void stage_vt(const ushort* __restrict__ src, int64_t ld,
+                       int row0, int row_limit, ushort* lds_tile,
+                       int wave, int lane) {
+  int tid = wave * WAVE + lane;
+  #pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    int idx = pass * THREADS + tid;       // 512 groups: kv(64) x dgrp(8)
+    int kv = idx >> 3;
+    int dg = idx & 7;
+    int gkv = row0 + kv;
+    gkv = gkv < row_limit ? gkv : row_limit;
+    const ushort* gptr = src + (int64_t)gkv * ld + dg * 8;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int d = dg * 8 + j;
+      int byte_off = t_off(d, kv >> 3) + (kv & 7) * 2;
+      *reinterpret_cast<ushort*>(
+          reinterpret_cast<char*>(lds_tile) + byte_off) = gptr[j];
+    }
+  }
+}
+
+__global__ __launch_bounds__(THREADS)
+void attention_kernel(const ushort* __restrict__ Q,
+                      const ushort* __restrict__ K,
+                      const ushort* __restrict__ V,
+                      ushort* __restrict__ O,
+                      int B, int S, int H, float scale, int n_qblk) {
+  // LDS: Q[64][64] | K dbuf 2x | V^T dbuf 2x | P bounce 4x[16][64]
+  __shared__ __attribute__((aligned(16))) char smem[8192 * 6];
+  char* q_lds = smem;
+  auto k_lds = [&](int buf) -> char* { return smem + 8192 + buf * 8192; };
+  auto vt_lds = [&](int buf) -> char* { return smem + 24576 + buf * 8192; };
+  char* p_lds = smem + 40960;           // 4 waves x 2KB
+
+  const int flat = blockIdx.x;
+  const int qb = flat % n_qblk;
+  const int h = (flat / n_qblk) % H;
+  const int b = flat / (n_qblk * H);
+  const int64_t ld = (int64_t)H * D;
+  const ushort* Qb = Q + (int64_t)b * S * ld + h * D;
+  const ushort* Kb = K + (int64_t)b * S * ld + h * D;
+  const ushort* Vb = V + (int64_t)b * S * ld + h * D;
+  ushort* Ob = O + (int64_t)b * S * ld + h * D;
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int frow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int q0 = qb * QBLK;
+
+  // ---- load Q block to LDS, then this wave's 16 rows into registers
+  stage_rows_glds(Qb, ld, q0, S - 1, q_lds, wave, lane);
+  __syncthreads();
+  bf16x8_t qf[2];      // A-fragments for the 2 K=32 steps over D
+  #pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    int row = wave * 16 + frow;
+    qf[ks] = *reinterpret_cast<const bf16x8_t*>(
+        q_lds + t_off(row, ks * 4 + kgrp));
+  }
+  __syncthreads();   // q_lds no longer needed (could be reused)
+
+  // ---- online softmax state: this lane covers rows wave*16 + kgrp*4+r
+  float m_run[4], l_run[4];
+  f32x4_t o_acc[4] = {};      // O fragments over d (4 x 16 cols)
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -3.0e38f;
+    l_run[r] = 0.f;
+  }
+
+  const int n_kv = (S + KVBLK - 1) / KVBLK;
+  // prologue staging of kv tile 0
+  stage_rows_glds(Kb, ld, 0, S - 1, k_lds(0), wave, lane);
+  stage_vt(Vb, ld, 0, S - 1, reinterpret_cast<ushort*>(vt_lds(0)), wave,
+           lane);
+
+  int cur = 0;
+  for (int t = 0; t < n_kv; ++t) {
+    __syncthreads();          // current tile staged (glds drained here)
+    if (t + 1 < n_kv) {
+      stage_rows_glds(Kb, ld, (t + 1) * KVBLK, S - 1, k_lds(cur ^ 1),
+                      wave, lane);
+      stage_vt(Vb, ld, (t + 1) * KVBLK, S - 1,
+               reinterpret_cast<ushort*>(vt_lds(cur ^ 1)), wave, lane);
+    }
+
+    // ---- QK^T: scores[16 q rows][64 kv] = 1x4 fragments
+    f32x4_t sc[4] = {};
+    {
+      const char* kt = k_lds(cur);
+      #pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+          bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
+              kt + t_off(ni * 16 + frow, ks * 4 + kgrp));
+          sc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qf[ks], kf, sc[ni], 0, 0, 0);
+        }
+      }
+    }
+    // scale + mask the kv tail (rows beyond S were staged as clamps)
+    const int kv_base = t * KVBLK;
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int kv_col = kv_base + ni * 16 + frow;   // this lane's kv column
+      bool valid = kv_col < S;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        sc[ni][r] = valid ? sc[ni][r] * scale : -3.0e38f;
+    }
+
+    // ---- per-row max over the 64 kv columns:
+    // row r lives in reg r across the 16 lanes of this lane's quarter
+    // group and across the 4 fragments -> local max over frags, then
+    // 4-step shfl_xor within the 16-lane group
+    float pmax[4], psum[4];
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float v = fmaxf(fmaxf(sc[0][r], sc[1][r]),
+                      fmaxf(sc[2][r], sc[3][r]));
+      #pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        v = fmaxf(v, __shfl_xor(v, off, 16));
+      pmax[r] = v;
+    }
+    // online update
+    float alpha[4];
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float m_new = fmaxf(m_run[r], pmax[r]);
+      alpha[r] = __expf(m_run[r] - m_new);
+      m_run[r] = m_new;
+    }
+    // exponentiate P and row-sum
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float sum = 0.f;
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        sc[ni][r] = __expf(sc[ni][r] - m_run[r]);
+        sum += sc[ni][r];
+      }
+      #pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        sum += __shfl_xor(sum, off, 16);
+      psum[r] = sum;
+      l_run[r] = l_run[r] * alpha[r] + sum;
+    }
+    // rescale O by alpha (rows of O fragments match reg index)
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        o_acc[ni][r] *= alpha[r];
+
+    // ---- P (C-layout) -> LDS -> A-fragment layout
+    {
+      ushort* pw = reinterpret_cast<ushort*>(p_lds + wave * 2048);
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = kgrp * 4 + r;            // 0..15
+          int col = ni * 16 + frow;          // 0..63
+          // swizzled [16][64] image (row stride 128B)
+          *reinterpret_cast<ushort*>(
+              reinterpret_cast<char*>(pw) + t_off(row, col >> 3) +
+              (col & 7) * 2) = f2bf(sc[ni][r]);
+        }
+      }
+      // wave-local write->read; compiler orders via lgkmcnt
+      const char* pr = reinterpret_cast<const char*>(pw);
+      const char* vt = vt_lds(cur);
+      #pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8_t pf = *reinterpret_cast<const bf16x8_t*>(
+            pr + t_off(frow, ks * 4 + kgrp));
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+          bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+              vt + t_off(ni * 16 + frow, ks * 4 + kgrp));
+          o_acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pf, vf, o_acc[ni], 0, 0, 0);
+        }
+      }
+    }
+    cur ^= 1;
+  }
+
+  // ---- epilogue: O /= l, store rows q0 + wave*16 + kgrp*4 + r
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int row = q0 + wave * 16 + kgrp * 4 + r;
+    if (row >= S) continue;
+    float inv = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+    ushort* orow = Ob + (int64_t)row * ld;
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      orow[ni * 16 + frow] = f2bf(o_acc[ni][r] * inv);
+  }
+}
+
+}  // namespace attn
+
+void launch_attention(hipStream_t s, const ushort* Q, const ushort* K,
+                      const ushort* V, ushort* O, int B, int S, int H,
+                      int D_, float scale) {
+  if (D_ != attn::D)
+    throw std::runtime_error("fused attention requires head_dim == 64");
+  int n_qblk = (S + attn::QBLK - 1) / attn::QBLK;
+  dim3 grid(B * H * n_qblk);
+  hipLaunchKernelGGL(attn::attention_kernel, grid, dim3(attn::THREADS), 0,
+                     s, Q, K, V, O, B, S, H, scale, n_qblk);
+}
+
+}  // namespace tfsc

@@ -169,6 +169,17 @@ class CpuExecutor:
             table = _get(op.inputs[0])
             idx = _get(op.inputs[1]).astype(np.int64)
             vals[op.outputs[0]] = table[idx].astype(np.float32)
+        elif k == "attention":
+            q = _get(op.inputs[0])      # [B, S, H, D]
+            kk_ = _get(op.inputs[1])
+            v = _get(op.inputs[2])
+            scale = p["scale"]
+            scores = np.einsum("bqhd,bkhd->bhqk", q, kk_) * scale
+            mx = scores.max(-1, keepdims=True)
+            e = np.exp(scores - mx)
+            probs = e / e.sum(-1, keepdims=True)
+            ctx = np.einsum("bhqk,bkhd->bqhd", probs, v)
+            vals[op.outputs[0]] = ctx.astype(np.float32)
         elif k == "concat":
             parts = [_get(i) for i in op.inputs]
             vals[op.outputs[0]] = np.concatenate(parts, axis=p["axis"])

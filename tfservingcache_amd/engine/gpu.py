@@ -248,6 +248,15 @@ class ExecContext:
                 calls.extend(self._c_conv(op, oi, ext))
             elif k == "batched_gemm":
                 calls.extend(self._c_bgemm(op, ext))
+            elif k == "attention":
+                b_, s_, h_, d_ = self.shapes[op.inputs[0]]
+                calls.append((ext.K_ATTENTION,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.inputs[1]),
+                               self._ptr(op.inputs[2]),
+                               self._ptr(op.outputs[0])],
+                              [b_, s_, h_, d_],
+                              [float(p["scale"])]))
             elif k == "softmax":
                 shape = self.shapes[op.inputs[0]]
                 rows = int(np.prod(shape[:-1])) if len(shape) > 1 else 1

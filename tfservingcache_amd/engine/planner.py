@@ -871,6 +871,101 @@ def _broadcast(sa: Sequence[Dim], sb: Sequence[Dim]) -> Shape:
     return tuple(reversed(out))
 
 
+def fuse_attention(plan: Plan) -> Plan:
+    """Post-pass: collapse the lowered multi-head-attention op chain
+
+        transpose(q4d) , transpose(k4d) , transpose(v4d)   [perm 0,2,1,3]
+        batched_gemm(qh, kh, trans_b=True) -> eltwise mul(scalar)
+        -> softmax -> batched_gemm(probs, vh) -> transpose [perm 0,2,1,3]
+
+    into one `attention` op operating on the PRE-transpose [B,S,H,D]
+    views (flash-style fused kernel on GPU; no S x S materialization,
+    no transposes). Fused only when head_dim == 64 (the kernel's
+    constraint); otherwise left as-is."""
+    ops = plan.ops
+    produced_by: Dict[int, int] = {}
+    for oi, op in enumerate(ops):
+        for o in op.outputs:
+            produced_by[o] = oi
+    consumers: Dict[int, List[int]] = {}
+    for oi, op in enumerate(ops):
+        for i in op.inputs:
+            consumers.setdefault(i, []).append(oi)
+
+    def sole_consumer(tensor_idx):
+        c = consumers.get(tensor_idx, [])
+        return c[0] if len(c) == 1 else None
+
+    def producer(tensor_idx):
+        oi = produced_by.get(tensor_idx)
+        return ops[oi] if oi is not None else None
+
+    to_remove: set = set()
+    replacements: Dict[int, PlanOp] = {}
+
+    for oi, op in enumerate(ops):
+        if op.kind != "batched_gemm" or not op.params.get("trans_b"):
+            continue
+        tq, tk = producer(op.inputs[0]), producer(op.inputs[1])
+        if tq is None or tk is None or tq.kind != "transpose" or \
+                tk.kind != "transpose":
+            continue
+        if tq.params.get("perm") != [0, 2, 1, 3] or \
+                tk.params.get("perm") != [0, 2, 1, 3]:
+            continue
+        mul_oi = sole_consumer(op.outputs[0])
+        if mul_oi is None or ops[mul_oi].kind != "eltwise" or \
+                ops[mul_oi].params.get("fn") != "mul":
+            continue
+        mul_op = ops[mul_oi]
+        scale_idx = mul_op.inputs[1] if mul_op.inputs[0] == op.outputs[0] \
+            else mul_op.inputs[0]
+        scale_t = plan.tensors[scale_idx]
+        if scale_t.weight is None or scale_t.weight.size != 1:
+            continue
+        scale = float(np.asarray(scale_t.weight).reshape(-1)[0])
+        sm_oi = sole_consumer(mul_op.outputs[0])
+        if sm_oi is None or ops[sm_oi].kind != "softmax":
+            continue
+        pv_oi = sole_consumer(ops[sm_oi].outputs[0])
+        if pv_oi is None or ops[pv_oi].kind != "batched_gemm" or \
+                ops[pv_oi].params.get("trans_b"):
+            continue
+        pv = ops[pv_oi]
+        tv = producer(pv.inputs[1])
+        if tv is None or tv.kind != "transpose" or \
+                tv.params.get("perm") != [0, 2, 1, 3]:
+            continue
+        tc_oi = sole_consumer(pv.outputs[0])
+        if tc_oi is None or ops[tc_oi].kind != "transpose" or \
+                ops[tc_oi].params.get("perm") != [0, 2, 1, 3]:
+            continue
+        tc = ops[tc_oi]
+        q4d, k4d, v4d = tq.inputs[0], tk.inputs[0], tv.inputs[0]
+        shape = plan.tensors[q4d].shape        # [B, S, H, D]
+        if len(shape) != 4 or shape[3] != 64:
+            continue
+        for rm in (produced_by[tq.outputs[0]], produced_by[tk.outputs[0]],
+                   produced_by[tv.outputs[0]], oi, mul_oi, sm_oi, pv_oi,
+                   tc_oi):
+            to_remove.add(rm)
+        replacements[tc_oi] = PlanOp(
+            "attention", [q4d, k4d, v4d], [tc.outputs[0]],
+            {"scale": scale, "seq": shape[1], "heads": shape[2],
+             "head_dim": shape[3]})
+
+    if not to_remove:
+        return plan
+    new_ops: List[PlanOp] = []
+    for oi, op in enumerate(ops):
+        if oi in replacements:
+            new_ops.append(replacements[oi])
+        elif oi not in to_remove:
+            new_ops.append(op)
+    plan.ops = new_ops
+    return plan
+
+
 def compile_graph(graph_def: g.GraphDef,
                   signature: m.SignatureDef) -> Plan:
-    return _Lowerer(graph_def, signature).run()
+    return fuse_attention(_Lowerer(graph_def, signature).run())
