@@ -93,8 +93,9 @@ def _graph_status(pool):
         for e in pool._entries.values():      # noqa: SLF001
             gm = getattr(e.model, "_gpu", None)
             if gm is not None and gm._contexts:  # noqa: SLF001
-                return all(c.captured and c.exec_plan.has_graph()
-                           for c in gm._contexts.values())
+                ctxs = [c for lst in gm._contexts.values() for c in lst]
+                return bool(ctxs) and all(
+                    c.captured and c.exec_plan.has_graph() for c in ctxs)
     except Exception:       # noqa: BLE001
         pass
     return None

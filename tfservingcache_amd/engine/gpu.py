@@ -123,6 +123,7 @@ class ExecContext:
         self.exec_plan = ext.ExecPlan(calls)
         self.captured = False
         self._views: Dict[int, object] = {}
+        self.lock = threading.Lock()
         # dedicated non-default stream (the default stream cannot be
         # hipGraph-captured; copies + kernels + D2H all run here)
         self.stream = torch.cuda.Stream(device=dev)
@@ -581,16 +582,21 @@ class GpuModel:
     BUCKETS = (1, 2, 4, 8, 16, 32, 64)
 
     def __init__(self, plan: Plan, device: str = "cuda:0",
-                 max_batch: int = 64, use_graphs: bool = True):
+                 max_batch: int = 64, use_graphs: bool = True,
+                 n_streams: int = 2):
         torch, _ = _load_backend()
         self.plan = plan
         self.device = device
         self.max_batch = max_batch
         self.use_graphs = use_graphs
+        # up to n_streams ExecContexts per batch bucket, each with its
+        # own HIP stream + workspace: concurrent requests overlap one
+        # context's H2D/D2H with another's kernels
+        self.n_streams = max(1, n_streams)
         self._weights: Dict[int, object] = {}
         self._gemm_weights: Dict[Tuple[int, bool], object] = {}
         self._conv_weights: Dict[int, object] = {}
-        self._contexts: Dict[int, ExecContext] = {}
+        self._contexts: Dict[int, List[ExecContext]] = {}
         self._lock = threading.Lock()
         self._released = False
         with torch.cuda.device(device):
@@ -694,25 +700,40 @@ class GpuModel:
                 return b
         return self.max_batch
 
-    def context(self, batch: int) -> ExecContext:
+    def _acquire_context(self, batch: int) -> ExecContext:
+        """Returns a LOCKED ExecContext for the bucket: an idle one if
+        available, a freshly built one while under n_streams, else
+        blocks on the least-loaded."""
         b = self._bucket(batch)
         with self._lock:
-            ctx = self._contexts.get(b)
-            if ctx is None:
+            if self._released:
+                raise ModelReleasedError(
+                    "model was evicted from the GPU pool")
+            ctxs = self._contexts.setdefault(b, [])
+            for ctx in ctxs:
+                if ctx.lock.acquire(blocking=False):
+                    return ctx
+            if len(ctxs) < self.n_streams:
                 ctx = ExecContext(self, b)
-                self._contexts[b] = ctx
-            return ctx
+                ctxs.append(ctx)
+                ctx.lock.acquire()
+                return ctx
+            ctx = ctxs[0]
+        ctx.lock.acquire()
+        return ctx
 
     def run(self, feeds: Dict[int, np.ndarray], batch: int,
             fetch: List[int]) -> Dict[int, np.ndarray]:
         if self._released:
             raise ModelReleasedError("model was evicted from the GPU pool")
-        ctx = self.context(batch)
-        with self._lock:
+        ctx = self._acquire_context(batch)
+        try:
             if self._released:
                 raise ModelReleasedError(
                     "model was evicted from the GPU pool")
             out = ctx.run(feeds, fetch)
+        finally:
+            ctx.lock.release()
         if ctx.batch != batch:
             # un-pad the batch dimension of fetched outputs
             plan = self.plan
@@ -724,12 +745,18 @@ class GpuModel:
         return out
 
     def release(self) -> None:
-        # taking _lock first means any in-flight run() completes before
-        # its weights are freed; subsequent runs fail fast and the cache
-        # manager re-fetches
+        # mark released (new runs fail fast; the cache manager
+        # re-fetches), then wait for each in-flight context before
+        # freeing the weights under it
         with self._lock:
             self._released = True
+            ctx_lists = list(self._contexts.values())
             self._contexts.clear()
+        for ctxs in ctx_lists:
+            for ctx in ctxs:
+                ctx.lock.acquire()
+                ctx.lock.release()
+        with self._lock:
             self._weights.clear()
             self._gemm_weights.clear()
             self._conv_weights.clear()

@@ -144,6 +144,7 @@ def clone_gpu_model(gpu_model, device: str):
     clone.max_batch = gpu_model.max_batch
     clone.use_graphs = gpu_model.use_graphs
     clone._contexts = {}
+    clone.n_streams = getattr(gpu_model, "n_streams", 2)
     import threading
     clone._lock = threading.Lock()
     clone._released = False
