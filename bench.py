@@ -128,6 +128,8 @@ def main() -> int:
                     help="CPU engine (CI smoke only; not a benchmark)")
     ap.add_argument("--dyn-batch", action="store_true",
                     help="enable server-side dynamic batching")
+    ap.add_argument("--streams", type=int, default=2,
+                    help="execution contexts (HIP streams) per model")
     ap.add_argument("--transport", choices=["inproc", "grpc"],
                     default="inproc",
                     help="inproc: gRPC message path without sockets; "
@@ -171,7 +173,8 @@ def main() -> int:
         loader = make_gpu_loader(cache, device=device,
                                  max_batch=max(args.batch, 64)
                                  if args.dyn_batch else max(args.batch, 1),
-                                 batching=args.dyn_batch)
+                                 batching=args.dyn_batch,
+                                 n_streams=args.streams)
     pool = ModelPool(loader, max_concurrent_models=pool_cap, device=device)
     cm = CacheManager(provider, cache, pool, model_fetch_timeout=300.0)
     handler = LocalServingHandler(cm)

@@ -159,7 +159,8 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
                     max_batch: int = 64, use_graphs: bool = True,
                     batching: bool = False,
                     batch_timeout_s: float = 0.002,
-                    devices: Optional[List[str]] = None
+                    devices: Optional[List[str]] = None,
+                    n_streams: int = 2
                     ) -> Callable[[str, int], LoadedModel]:
     """Loader that compiles the SavedModel onto an MI355X: weights land
     in the GPU's HBM pool (bf16, GEMM layouts pre-transformed) and the
@@ -182,7 +183,7 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
         lm = load_model_from_dir(vdir, name, version)
         dev = pick_device(name, version)
         lm._gpu = GpuModel(lm.plan, device=dev, max_batch=max_batch,
-                           use_graphs=use_graphs)
+                           use_graphs=use_graphs, n_streams=n_streams)
         lm.device = dev
         if batching:
             lm.enable_batching(max_batch=max_batch,
