@@ -128,7 +128,7 @@ def main() -> int:
                     help="CPU engine (CI smoke only; not a benchmark)")
     ap.add_argument("--dyn-batch", action="store_true",
                     help="enable server-side dynamic batching")
-    ap.add_argument("--streams", type=int, default=2,
+    ap.add_argument("--streams", type=int, default=4,
                     help="execution contexts (HIP streams) per model")
     ap.add_argument("--transport", choices=["inproc", "grpc"],
                     default="inproc",

@@ -583,7 +583,7 @@ class GpuModel:
 
     def __init__(self, plan: Plan, device: str = "cuda:0",
                  max_batch: int = 64, use_graphs: bool = True,
-                 n_streams: int = 2):
+                 n_streams: int = 4):
         torch, _ = _load_backend()
         self.plan = plan
         self.device = device
