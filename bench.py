@@ -147,13 +147,19 @@ def main() -> int:
     if world > 1:
         import torch  # noqa: F811
         import torch.distributed as dist  # noqa: F811
-        backend = "gloo" if (args.cpu or not torch.cuda.is_available()) \
-            else "nccl"
+        backend = os.environ.get("TFSC_BENCH_BACKEND") or (
+            "gloo" if (args.cpu or not torch.cuda.is_available())
+            else "nccl")
         if backend == "nccl":
             torch.cuda.set_device(local_rank)
         dist.init_process_group(backend=backend)
 
-    device = f"cuda:{local_rank}" if not args.cpu else "cpu"
+    if args.cpu:
+        device = "cpu"
+    else:
+        import torch as _t
+        n_dev = max(_t.cuda.device_count(), 1)
+        device = f"cuda:{local_rank % n_dev}"
 
     tmp = tempfile.mkdtemp(prefix=f"tfsc_bench_r{rank}_")
     repo = os.path.join(tmp, "repo")
