@@ -23,6 +23,7 @@ sources = [
     os.path.join(CSRC, "ops", "gemm.hip"),
     os.path.join(CSRC, "ops", "conv.hip"),
     os.path.join(CSRC, "ops", "attention.hip"),
+    os.path.join(CSRC, "fastpath.cpp"),
 ]
 
 setup(

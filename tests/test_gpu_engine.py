@@ -156,3 +156,52 @@ def test_bert_fused_attention_gpu_vs_cpu(tmp_path):
              {"pooled_output": c["pooled_output"]}, rtol=0.1, atol=0.08)
     _compare({"sequence_output": g["sequence_output"]},
              {"sequence_output": c["sequence_output"]}, rtol=0.2, atol=0.15)
+
+
+def test_fast_predict_path_matches_python(tmp_path):
+    """C++ fast predict (bytes->bytes) must match the Python path."""
+    from tfservingcache_amd.wire import messages as m
+    from tfservingcache_amd.wire.tensor import (numpy_to_tensorproto,
+                                                tensorproto_to_numpy)
+    from tfservingcache_amd.engine import _tfsc_engine as ext
+
+    sm = build_mlp(d_in=64, d_hidden=128, d_out=64, seed=5)
+    gm = _gpu_model(tmp_path, sm, name="fastm")
+    gm._gpu.model_name = "fastm"
+    gm._gpu.model_version = 1
+    x = np.random.default_rng(0).standard_normal((4, 64)).astype(np.float32)
+    req = m.PredictRequest(
+        model_spec=m.ModelSpec(name="fastm", version=m.Int64Value(value=1)),
+        inputs={"x": numpy_to_tensorproto(x)}).encode()
+
+    # warm up via the python path (registers the fast context)
+    py_out = gm.predict({"x": x})["probs"]
+    assert gm._gpu._fast.has_bucket(4)
+
+    resp_bytes = gm._gpu.fast_predict(req)
+    resp = m.PredictResponse.decode(resp_bytes)
+    fast_out = tensorproto_to_numpy(resp.outputs["probs"])
+    assert fast_out.shape == (4, 64)
+    np.testing.assert_allclose(fast_out, py_out, rtol=1e-3, atol=1e-4)
+    assert resp.model_spec.name == "fastm"
+    assert resp.model_spec.version.value == 1
+
+    # smaller batch into the same bucket (padded rows dropped)
+    x2 = x[:2]
+    req2 = m.PredictRequest(
+        model_spec=m.ModelSpec(name="fastm", version=m.Int64Value(value=1)),
+        inputs={"x": numpy_to_tensorproto(x2)}).encode()
+    resp2 = m.PredictResponse.decode(gm._gpu.fast_predict(req2))
+    out2 = tensorproto_to_numpy(resp2.outputs["probs"])
+    assert out2.shape == (2, 64)
+    np.testing.assert_allclose(out2, py_out[:2], rtol=1e-3, atol=1e-4)
+
+    # typed-val request falls back
+    import pytest as _pytest
+    bad = m.PredictRequest(
+        model_spec=m.ModelSpec(name="fastm", version=m.Int64Value(value=1)),
+        inputs={"x": m.TensorProto(
+            dtype=m.DT_FLOAT, tensor_shape=m.TensorShapeProto.of([1, 64]),
+            float_val=[0.0] * 64)}).encode()
+    with _pytest.raises(ext.FastFallback):
+        gm._gpu.fast_predict(bad)

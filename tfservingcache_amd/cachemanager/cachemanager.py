@@ -183,7 +183,8 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
         lm = load_model_from_dir(vdir, name, version)
         dev = pick_device(name, version)
         lm._gpu = GpuModel(lm.plan, device=dev, max_batch=max_batch,
-                           use_graphs=use_graphs, n_streams=n_streams)
+                           use_graphs=use_graphs, n_streams=n_streams,
+                           model_name=name, model_version=version)
         lm.device = dev
         if batching:
             lm.enable_batching(max_batch=max_batch,

@@ -197,6 +197,20 @@ class ExecPlan {
     if (graph_exec_) hipGraphExecDestroy(graph_exec_);
   }
 
+  // fastpath entry: replay (or launch eagerly) on an explicit stream
+  void run_on(hipStream_t s) {
+    if (graph_exec_) {
+      if (hipGraphLaunch(graph_exec_, s) != hipSuccess)
+        throw std::runtime_error("hipGraphLaunch failed");
+      return;
+    }
+    for (const auto& c : calls_) launch_call(c, s);
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess)
+      throw std::runtime_error(std::string("kernel launch failed: ") +
+                               hipGetErrorString(e));
+  }
+
   size_t n_calls() const { return calls_.size(); }
 
  private:
@@ -204,9 +218,15 @@ class ExecPlan {
   hipGraphExec_t graph_exec_ = nullptr;
 };
 
+void fast_run_plan(void* plan, hipStream_t s) {
+  reinterpret_cast<ExecPlan*>(plan)->run_on(s);
+}
+
 }  // namespace tfsc
 
 namespace py = pybind11;
+
+void register_fastpath(py::module_& mod);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   using tfsc::Call;
@@ -233,7 +253,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
       .def("run_graph", &ExecPlan::run_graph,
            py::call_guard<py::gil_scoped_release>())
       .def("has_graph", &ExecPlan::has_graph)
-      .def("n_calls", &ExecPlan::n_calls);
+      .def("n_calls", &ExecPlan::n_calls)
+      .def("ptr", [](ExecPlan& p) {
+        return reinterpret_cast<uintptr_t>(&p);
+      });
 
   mod.attr("K_ELT_UNARY") = int(tfsc::K_ELT_UNARY);
   mod.attr("K_ELT_BINARY") = int(tfsc::K_ELT_BINARY);
@@ -252,6 +275,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("K_CONV") = int(tfsc::K_CONV);
   mod.attr("K_PAD_LAST") = int(tfsc::K_PAD_LAST);
   mod.attr("K_ATTENTION") = int(tfsc::K_ATTENTION);
+
+  register_fastpath(mod);
 
   // elementwise fn codes
   mod.attr("ELT_ADD") = int(tfsc::ELT_ADD);

@@ -1,0 +1,486 @@
+// C++ fast predict path: PredictRequest bytes -> PredictResponse bytes
+// with the GIL released for the whole call.
+//
+// The Python handler routes by model (ring/cache bookkeeping, metrics)
+// and then hands the RAW request bytes here; this layer does the
+// protobuf wire parse (the subset TF-Serving Predict uses:
+// tensor_content-carrying TensorProtos), copies each input into pinned
+// f32/i32 staging, DMAs to a device staging buffer, casts to the bf16
+// workspace on-GPU, replays the ExecPlan hipGraph, casts/DMAs the
+// outputs back and serializes the response — removing the per-request
+// Python work that capped small-batch throughput.
+//
+// Anything it cannot handle (typed *_val fields, unknown dtypes,
+// missing inputs) raises FastFallback and the Python path serves the
+// request instead.
+#include <torch/extension.h>
+
+#include <hip/hip_runtime.h>
+
+#include <cstring>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "kernels.h"
+
+namespace tfsc {
+
+class ExecPlan;  // from executor.cpp
+void fast_run_plan(void* plan, hipStream_t s);  // defined in executor.cpp
+
+struct FastFallback : std::runtime_error {
+  using std::runtime_error::runtime_error;
+};
+
+#define HIPCHK(x)                                                        \
+  do {                                                                   \
+    hipError_t e_ = (x);                                                 \
+    if (e_ != hipSuccess)                                                \
+      throw std::runtime_error(std::string("fastpath HIP error: ") +     \
+                               hipGetErrorString(e_));                   \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// minimal protobuf wire reader
+// ---------------------------------------------------------------------------
+struct Reader {
+  const uint8_t* p;
+  const uint8_t* end;
+
+  uint64_t varint() {
+    uint64_t v = 0;
+    int shift = 0;
+    while (p < end) {
+      uint8_t b = *p++;
+      v |= uint64_t(b & 0x7f) << shift;
+      if (!(b & 0x80)) return v;
+      shift += 7;
+      if (shift >= 70) break;
+    }
+    throw FastFallback("bad varint");
+  }
+
+  void skip(int wt) {
+    switch (wt) {
+      case 0: varint(); break;
+      case 1: p += 8; break;
+      case 2: { uint64_t n = varint(); p += n; break; }
+      case 5: p += 4; break;
+      default: throw FastFallback("bad wire type");
+    }
+    if (p > end) throw FastFallback("truncated");
+  }
+};
+
+struct ParsedTensor {
+  int dtype = 0;
+  std::vector<int64_t> dims;
+  const uint8_t* content = nullptr;
+  size_t content_len = 0;
+  bool has_typed_vals = false;
+};
+
+static ParsedTensor parse_tensor(const uint8_t* p, const uint8_t* end) {
+  ParsedTensor t;
+  Reader r{p, end};
+  while (r.p < r.end) {
+    uint64_t tag = r.varint();
+    int fno = int(tag >> 3), wt = int(tag & 7);
+    if (fno == 1 && wt == 0) {
+      t.dtype = int(r.varint());
+    } else if (fno == 2 && wt == 2) {            // tensor_shape
+      uint64_t n = r.varint();
+      Reader rs{r.p, r.p + n};
+      r.p += n;
+      while (rs.p < rs.end) {
+        uint64_t stag = rs.varint();
+        if ((stag >> 3) == 2 && (stag & 7) == 2) {   // dim
+          uint64_t dn = rs.varint();
+          Reader rd{rs.p, rs.p + dn};
+          rs.p += dn;
+          int64_t size = 0;
+          while (rd.p < rd.end) {
+            uint64_t dtag = rd.varint();
+            if ((dtag >> 3) == 1 && (dtag & 7) == 0)
+              size = int64_t(rd.varint());
+            else
+              rd.skip(int(dtag & 7));
+          }
+          t.dims.push_back(size);
+        } else {
+          rs.skip(int(stag & 7));
+        }
+      }
+    } else if (fno == 4 && wt == 2) {            // tensor_content
+      uint64_t n = r.varint();
+      t.content = r.p;
+      t.content_len = size_t(n);
+      r.p += n;
+    } else if (fno >= 5 && fno <= 17) {
+      t.has_typed_vals = true;
+      r.skip(wt);
+    } else {
+      r.skip(wt);
+    }
+  }
+  return t;
+}
+
+struct ParsedRequest {
+  std::map<std::string, ParsedTensor> inputs;
+  std::vector<std::string> output_filter;
+};
+
+static ParsedRequest parse_request(const uint8_t* p, size_t len) {
+  ParsedRequest req;
+  Reader r{p, p + len};
+  while (r.p < r.end) {
+    uint64_t tag = r.varint();
+    int fno = int(tag >> 3), wt = int(tag & 7);
+    if (fno == 2 && wt == 2) {        // inputs map entry
+      uint64_t n = r.varint();
+      Reader re{r.p, r.p + n};
+      r.p += n;
+      std::string key;
+      const uint8_t* vptr = nullptr;
+      size_t vlen = 0;
+      while (re.p < re.end) {
+        uint64_t etag = re.varint();
+        if ((etag >> 3) == 1 && (etag & 7) == 2) {
+          uint64_t kn = re.varint();
+          key.assign(reinterpret_cast<const char*>(re.p), kn);
+          re.p += kn;
+        } else if ((etag >> 3) == 2 && (etag & 7) == 2) {
+          uint64_t vn = re.varint();
+          vptr = re.p;
+          vlen = size_t(vn);
+          re.p += vn;
+        } else {
+          re.skip(int(etag & 7));
+        }
+      }
+      if (vptr) req.inputs[key] = parse_tensor(vptr, vptr + vlen);
+    } else if (fno == 3 && wt == 2) {  // output_filter
+      uint64_t n = r.varint();
+      req.output_filter.emplace_back(
+          reinterpret_cast<const char*>(r.p), n);
+      r.p += n;
+    } else {
+      r.skip(wt);
+    }
+  }
+  return req;
+}
+
+// ---------------------------------------------------------------------------
+// minimal protobuf writer (PredictResponse)
+// ---------------------------------------------------------------------------
+static void w_varint(std::string& s, uint64_t v) {
+  while (true) {
+    uint8_t b = v & 0x7f;
+    v >>= 7;
+    if (v) s.push_back(char(b | 0x80));
+    else { s.push_back(char(b)); return; }
+  }
+}
+
+static void w_tag(std::string& s, int fno, int wt) {
+  w_varint(s, uint64_t(fno) << 3 | wt);
+}
+
+static void w_len_prefixed(std::string& s, int fno,
+                           const std::string& payload) {
+  w_tag(s, fno, 2);
+  w_varint(s, payload.size());
+  s += payload;
+}
+
+// ---------------------------------------------------------------------------
+// fast context/model
+// ---------------------------------------------------------------------------
+struct FastIO {
+  std::string alias;
+  bool is_int = false;            // i32 input (ids) vs f32<->bf16
+  uintptr_t pin = 0;              // pinned host buffer (f32/i32)
+  uintptr_t dev_stage = 0;        // device f32/i32 staging
+  uintptr_t dev = 0;              // bf16 (or i32) workspace buffer
+  int64_t row_elems = 0;          // elements per batch row
+  std::vector<int64_t> tail_dims; // dims after the batch dim
+};
+
+struct FastContext {
+  int bucket = 0;
+  void* exec_plan = nullptr;      // tfsc::ExecPlan*
+  hipStream_t stream = nullptr;
+  std::vector<FastIO> ins;
+  std::vector<FastIO> outs;
+  std::mutex mu;
+  bool disabled = false;
+};
+
+class FastModel {
+ public:
+  FastModel(std::string model_name, int64_t version)
+      : name_(std::move(model_name)), version_(version) {}
+
+  int add_context(int bucket, uintptr_t exec_plan, uintptr_t stream,
+                  std::vector<FastIO> ins, std::vector<FastIO> outs) {
+    auto ctx = std::make_unique<FastContext>();
+    ctx->bucket = bucket;
+    ctx->exec_plan = reinterpret_cast<void*>(exec_plan);
+    ctx->stream = reinterpret_cast<hipStream_t>(stream);
+    ctx->ins = std::move(ins);
+    ctx->outs = std::move(outs);
+    std::lock_guard<std::mutex> g(mu_);
+    ctxs_.push_back(std::move(ctx));
+    return int(ctxs_.size()) - 1;
+  }
+
+  // shared-lock interface: the Python execution path locks the same
+  // per-context mutex the fast path uses, so both paths serialize on
+  // the context's stream/buffers
+  void lock_ctx(int id) { ctxs_.at(id)->mu.lock(); }
+  void unlock_ctx(int id) { ctxs_.at(id)->mu.unlock(); }
+
+  // release path: waits out in-flight fast predicts, then blocks the
+  // fast path permanently (buffers are about to be freed)
+  void disable() {
+    std::lock_guard<std::mutex> g(mu_);
+    for (auto& c : ctxs_) {
+      std::lock_guard<std::mutex> cg(c->mu);
+      c->disabled = true;
+    }
+  }
+
+  bool has_bucket(int batch) {
+    std::lock_guard<std::mutex> g(mu_);
+    for (auto& c : ctxs_)
+      if (c->bucket >= batch) return true;
+    return false;
+  }
+
+  std::string predict(const uint8_t* data, size_t len) {
+    ParsedRequest req = parse_request(data, len);
+    if (req.inputs.empty()) throw FastFallback("no inputs");
+    // batch from the first input's leading dim
+    int64_t batch = -1;
+    for (auto& kv : req.inputs) {
+      if (kv.second.has_typed_vals && !kv.second.content)
+        throw FastFallback("typed-val tensor");
+      if (!kv.second.content) throw FastFallback("no tensor_content");
+      if (kv.second.dims.empty()) throw FastFallback("scalar input");
+      if (batch < 0) batch = kv.second.dims[0];
+      else if (batch != kv.second.dims[0])
+        throw FastFallback("inconsistent batch");
+    }
+
+    FastContext* ctx = acquire(int(batch));
+    std::lock_guard<std::mutex> g2(ctx->mu, std::adopt_lock);
+    if (ctx->disabled) throw FastFallback("model released");
+
+    hipStream_t s = ctx->stream;
+    // inputs
+    for (auto& io : ctx->ins) {
+      auto it = req.inputs.find(io.alias);
+      if (it == req.inputs.end()) throw FastFallback("missing input");
+      const ParsedTensor& t = it->second;
+      // validate trailing dims
+      if (int64_t(t.dims.size()) != int64_t(io.tail_dims.size()) + 1)
+        throw FastFallback("rank mismatch");
+      int64_t row = 1;
+      for (size_t d = 0; d < io.tail_dims.size(); ++d) {
+        if (t.dims[d + 1] != io.tail_dims[d])
+          throw FastFallback("dim mismatch");
+        row *= t.dims[d + 1];
+      }
+      size_t esz = 4;            // f32 or i32 on the wire
+      int dt = t.dtype;
+      if (io.is_int) {
+        if (dt != 3) throw FastFallback("want DT_INT32");
+      } else if (dt != 1) {
+        throw FastFallback("want DT_FLOAT");
+      }
+      size_t want = size_t(batch) * row * esz;
+      if (t.content_len != want) throw FastFallback("content size");
+      std::memcpy(reinterpret_cast<void*>(io.pin), t.content, want);
+      if (io.is_int) {
+        HIPCHK(hipMemcpyAsync(reinterpret_cast<void*>(io.dev),
+                              reinterpret_cast<void*>(io.pin), want,
+                              hipMemcpyHostToDevice, s));
+        size_t cap = size_t(ctx->bucket) * row * 4;
+        if (want < cap)
+          HIPCHK(hipMemsetAsync(
+              reinterpret_cast<char*>(io.dev) + want, 0, cap - want, s));
+      } else {
+        HIPCHK(hipMemcpyAsync(reinterpret_cast<void*>(io.dev_stage),
+                              reinterpret_cast<void*>(io.pin), want,
+                              hipMemcpyHostToDevice, s));
+        launch_f32_to_bf16(s,
+                           reinterpret_cast<const float*>(io.dev_stage),
+                           reinterpret_cast<ushort*>(io.dev),
+                           batch * row);
+        size_t cap2 = size_t(ctx->bucket) * row * 2;
+        size_t used2 = size_t(batch) * row * 2;
+        if (used2 < cap2)
+          HIPCHK(hipMemsetAsync(
+              reinterpret_cast<char*>(io.dev) + used2, 0, cap2 - used2,
+              s));
+      }
+    }
+
+    fast_run_plan(ctx->exec_plan, s);
+
+    // outputs: cast + D2H
+    std::vector<const FastIO*> wanted;
+    for (auto& io : ctx->outs) {
+      if (!req.output_filter.empty()) {
+        bool keep = false;
+        for (auto& f : req.output_filter)
+          if (f == io.alias) keep = true;
+        if (!keep) continue;
+      }
+      wanted.push_back(&io);
+    }
+    for (auto* io : wanted) {
+      launch_bf16_to_f32(s, reinterpret_cast<const ushort*>(io->dev),
+                         reinterpret_cast<float*>(io->dev_stage),
+                         batch * io->row_elems);
+      HIPCHK(hipMemcpyAsync(reinterpret_cast<void*>(io->pin),
+                            reinterpret_cast<void*>(io->dev_stage),
+                            size_t(batch) * io->row_elems * 4,
+                            hipMemcpyDeviceToHost, s));
+    }
+    HIPCHK(hipStreamSynchronize(s));
+    hipError_t ke = hipGetLastError();
+    if (ke != hipSuccess)
+      throw std::runtime_error(std::string("fastpath kernel error: ") +
+                               hipGetErrorString(ke));
+
+    // serialize PredictResponse
+    std::string out;
+    out.reserve(wanted.size() * 64 + 1024);
+    for (auto* io : wanted) {
+      std::string tp;
+      w_tag(tp, 1, 0);                 // dtype = DT_FLOAT
+      w_varint(tp, 1);
+      {
+        std::string shape;
+        {
+          std::string dim;
+          w_tag(dim, 1, 0);
+          w_varint(dim, uint64_t(batch));
+          w_len_prefixed(shape, 2, dim);
+        }
+        for (int64_t d : io->tail_dims) {
+          std::string dim;
+          w_tag(dim, 1, 0);
+          w_varint(dim, uint64_t(d));
+          w_len_prefixed(shape, 2, dim);
+        }
+        w_len_prefixed(tp, 2, shape);
+      }
+      size_t nbytes = size_t(batch) * io->row_elems * 4;
+      w_tag(tp, 4, 2);
+      w_varint(tp, nbytes);
+      tp.append(reinterpret_cast<const char*>(io->pin), nbytes);
+
+      std::string entry;
+      w_tag(entry, 1, 2);
+      w_varint(entry, io->alias.size());
+      entry += io->alias;
+      w_len_prefixed(entry, 2, tp);
+      w_len_prefixed(out, 1, entry);   // outputs map
+    }
+    {
+      std::string spec;
+      w_tag(spec, 1, 2);
+      w_varint(spec, name_.size());
+      spec += name_;
+      std::string ver;
+      w_tag(ver, 1, 0);
+      w_varint(ver, uint64_t(version_));
+      w_len_prefixed(spec, 2, ver);
+      w_len_prefixed(out, 2, spec);    // model_spec
+    }
+    return out;
+  }
+
+ private:
+  FastContext* acquire(int batch) {
+    // smallest registered bucket >= batch; prefer an idle context of
+    // that bucket, else block on its first one. Returns LOCKED.
+    int bucket = -1;
+    FastContext* fallback = nullptr;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      for (auto& c : ctxs_)
+        if (c->bucket >= batch && (bucket < 0 || c->bucket < bucket))
+          bucket = c->bucket;
+      if (bucket < 0) throw FastFallback("no context for batch");
+      for (auto& c : ctxs_) {
+        if (c->bucket != bucket) continue;
+        if (fallback == nullptr) fallback = c.get();
+        if (c->mu.try_lock()) return c.get();
+      }
+    }
+    fallback->mu.lock();
+    return fallback;
+  }
+
+  std::string name_;
+  int64_t version_;
+  std::mutex mu_;
+  std::vector<std::unique_ptr<FastContext>> ctxs_;
+};
+
+}  // namespace tfsc
+
+namespace py = pybind11;
+
+void register_fastpath(py::module_& mod) {
+  using tfsc::FastIO;
+  using tfsc::FastModel;
+
+  py::register_exception<tfsc::FastFallback>(mod, "FastFallback");
+
+  py::class_<FastIO>(mod, "FastIO")
+      .def(py::init([](std::string alias, bool is_int, uintptr_t pin,
+                       uintptr_t dev_stage, uintptr_t dev,
+                       int64_t row_elems, std::vector<int64_t> tail) {
+        FastIO io;
+        io.alias = std::move(alias);
+        io.is_int = is_int;
+        io.pin = pin;
+        io.dev_stage = dev_stage;
+        io.dev = dev;
+        io.row_elems = row_elems;
+        io.tail_dims = std::move(tail);
+        return io;
+      }));
+
+  py::class_<FastModel>(mod, "FastModel")
+      .def(py::init<std::string, int64_t>())
+      .def("add_context", &FastModel::add_context)
+      .def("has_bucket", &FastModel::has_bucket)
+      .def("lock_ctx", &FastModel::lock_ctx,
+           py::call_guard<py::gil_scoped_release>())
+      .def("unlock_ctx", &FastModel::unlock_ctx)
+      .def("disable", &FastModel::disable,
+           py::call_guard<py::gil_scoped_release>())
+      .def("predict", [](FastModel& fm, py::bytes data) {
+        char* buf = nullptr;
+        Py_ssize_t len = 0;
+        PyBytes_AsStringAndSize(data.ptr(), &buf, &len);
+        std::string out;
+        {
+          py::gil_scoped_release rel;
+          out = fm.predict(reinterpret_cast<const uint8_t*>(buf),
+                           size_t(len));
+        }
+        return py::bytes(out);
+      });
+}

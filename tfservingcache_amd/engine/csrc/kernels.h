@@ -61,6 +61,10 @@ void launch_pad_nhwc(hipStream_t s, const ushort* x, ushort* y,
                      int pr);
 void launch_pad_last(hipStream_t s, const ushort* x, ushort* y,
                      int64_t rows, int c_in, int c_out);
+void launch_f32_to_bf16(hipStream_t s, const float* x, ushort* y,
+                        int64_t n);
+void launch_bf16_to_f32(hipStream_t s, const ushort* x, float* y,
+                        int64_t n);
 
 // GEMM: C[M,N] = act(A[M,K] @ B[K,N] + bias [+ residual]).
 // A row-major bf16, B row-major bf16 (pre-transposed at load if the graph

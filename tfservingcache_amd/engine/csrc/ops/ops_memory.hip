@@ -496,6 +496,52 @@ void launch_gather_rows(hipStream_t s, const ushort* table, const int* idx,
                      dim3(TPB), 0, s, table, idx, y, n_idx, row_elems);
 }
 
+// dtype casts for the C++ fast predict path (f32 staging <-> bf16
+// workspace; conversion on-GPU instead of host-side)
+__global__ void k_f32_to_bf16(const float* __restrict__ x,
+                              ushort* __restrict__ y, int64_t n) {
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  int64_t n4 = n / 4;
+  for (int64_t i = i0; i < n4; i += stride) {
+    float4 v = reinterpret_cast<const float4*>(x)[i];
+    short4_t r;
+    r[0] = (short)f2bf(v.x); r[1] = (short)f2bf(v.y);
+    r[2] = (short)f2bf(v.z); r[3] = (short)f2bf(v.w);
+    reinterpret_cast<short4_t*>(y)[i] = r;
+  }
+  for (int64_t i = n4 * 4 + i0; i < n; i += stride)
+    y[i] = f2bf(x[i]);
+}
+
+__global__ void k_bf16_to_f32(const ushort* __restrict__ x,
+                              float* __restrict__ y, int64_t n) {
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  int64_t n4 = n / 4;
+  for (int64_t i = i0; i < n4; i += stride) {
+    short4_t v = reinterpret_cast<const short4_t*>(x)[i];
+    float4 r;
+    r.x = bf2f((ushort)v[0]); r.y = bf2f((ushort)v[1]);
+    r.z = bf2f((ushort)v[2]); r.w = bf2f((ushort)v[3]);
+    reinterpret_cast<float4*>(y)[i] = r;
+  }
+  for (int64_t i = n4 * 4 + i0; i < n; i += stride)
+    y[i] = bf2f(x[i]);
+}
+
+void launch_f32_to_bf16(hipStream_t s, const float* x, ushort* y,
+                        int64_t n) {
+  hipLaunchKernelGGL(k_f32_to_bf16, dim3(grid_for(n, 4)), dim3(TPB), 0, s,
+                     x, y, n);
+}
+
+void launch_bf16_to_f32(hipStream_t s, const ushort* x, float* y,
+                        int64_t n) {
+  hipLaunchKernelGGL(k_bf16_to_f32, dim3(grid_for(n, 4)), dim3(TPB), 0, s,
+                     x, y, n);
+}
+
 // widen the channel dim with zeros: [rows, c_in] -> [rows, c_out]
 // (used to lift C%8!=0 conv inputs — e.g. RGB stems — onto the fused
 // implicit-GEMM path)

@@ -124,6 +124,7 @@ class ExecContext:
         self.captured = False
         self._views: Dict[int, object] = {}
         self.lock = threading.Lock()
+        self.fast_id: Optional[int] = None   # set once fast-registered
         # dedicated non-default stream (the default stream cannot be
         # hipGraph-captured; copies + kernels + D2H all run here)
         self.stream = torch.cuda.Stream(device=dev)
@@ -520,6 +521,19 @@ class ExecContext:
 
     def run(self, feeds: Dict[int, np.ndarray],
             fetch: List[int]) -> Dict[int, np.ndarray]:
+        # once fast-registered, the C++ per-context mutex is the lock for
+        # this context's stream/buffers — take it so the Python and C++
+        # paths serialize against each other
+        if self.fast_id is not None and self.fast_id >= 0:
+            self.gm._fast.lock_ctx(self.fast_id)
+            try:
+                return self._run_inner(feeds, fetch)
+            finally:
+                self.gm._fast.unlock_ctx(self.fast_id)
+        return self._run_inner(feeds, fetch)
+
+    def _run_inner(self, feeds: Dict[int, np.ndarray],
+                   fetch: List[int]) -> Dict[int, np.ndarray]:
         torch, _ = _load_backend()
         with torch.cuda.device(self.gm.device), \
                 torch.cuda.stream(self.stream):
@@ -583,10 +597,17 @@ class GpuModel:
 
     def __init__(self, plan: Plan, device: str = "cuda:0",
                  max_batch: int = 64, use_graphs: bool = True,
-                 n_streams: int = 4):
-        torch, _ = _load_backend()
+                 n_streams: int = 4, model_name: str = "",
+                 model_version: int = 0):
+        torch, ext = _load_backend()
         self.plan = plan
         self.device = device
+        self.model_name = model_name
+        self.model_version = model_version
+        # C++ fast predict path (request bytes in, response bytes out);
+        # contexts register after their first Python-path run + capture
+        self._fast = ext.FastModel(model_name or "model",
+                                   int(model_version))
         self.max_batch = max_batch
         self.use_graphs = use_graphs
         # up to n_streams ExecContexts per batch bucket, each with its
@@ -732,6 +753,13 @@ class GpuModel:
                 raise ModelReleasedError(
                     "model was evicted from the GPU pool")
             out = ctx.run(feeds, fetch)
+            # after the first captured run completes (stream idle),
+            # expose the context to the C++ fast path
+            if ctx.fast_id is None and ctx.captured:
+                try:
+                    self._register_fast(ctx)
+                except Exception:       # noqa: BLE001
+                    log.exception("fast-path registration failed")
         finally:
             ctx.lock.release()
         if ctx.batch != batch:
@@ -744,6 +772,59 @@ class GpuModel:
                     out[idx] = out[idx][:rows]
         return out
 
+    def fast_predict(self, request_bytes: bytes) -> bytes:
+        """C++ end-to-end predict (raises ext.FastFallback when the
+        request shape/bucket isn't registered yet — callers then take
+        the Python path, which builds and registers the context)."""
+        if self._released:
+            raise ModelReleasedError("model was evicted from the GPU pool")
+        return self._fast.predict(request_bytes)
+
+    def _register_fast(self, ctx: "ExecContext") -> None:
+        torch, ext = _load_backend()
+        plan = self.plan
+        keep = []
+        ins = []
+        for alias, idx in plan.sig_inputs.items():
+            v = ctx.view(idx)
+            is_int = v.dtype == torch.int32
+            rows = v.shape[0] if v.ndim else 1
+            row_elems = int(np.prod(v.shape[1:])) if v.ndim > 1 else 1
+            pin = torch.empty((rows, row_elems),
+                              dtype=torch.int32 if is_int
+                              else torch.float32, pin_memory=True)
+            if is_int:
+                stage_ptr = 0
+            else:
+                stage = torch.empty((rows, row_elems), dtype=torch.float32,
+                                    device=self.device)
+                keep.append(stage)
+                stage_ptr = stage.data_ptr()
+            keep.append(pin)
+            ins.append(ext.FastIO(alias, bool(is_int), pin.data_ptr(),
+                                  stage_ptr, v.data_ptr(), row_elems,
+                                  [int(d) for d in v.shape[1:]]))
+        outs = []
+        for alias, idx in plan.sig_outputs.items():
+            v = ctx.view(idx)
+            if v.dtype != torch.bfloat16:
+                ctx.fast_id = -1     # non-bf16 output: python path only
+                return
+            rows = v.shape[0] if v.ndim else 1
+            row_elems = int(np.prod(v.shape[1:])) if v.ndim > 1 else 1
+            pin = torch.empty((rows, row_elems), dtype=torch.float32,
+                              pin_memory=True)
+            stage = torch.empty((rows, row_elems), dtype=torch.float32,
+                                device=self.device)
+            keep.extend([pin, stage])
+            outs.append(ext.FastIO(alias, False, pin.data_ptr(),
+                                   stage.data_ptr(), v.data_ptr(),
+                                   row_elems, [int(d) for d in v.shape[1:]]))
+        ctx._fast_keep = keep
+        ctx.fast_id = self._fast.add_context(
+            ctx.batch, ctx.exec_plan.ptr(), ctx.stream.cuda_stream, ins,
+            outs)
+
     def release(self) -> None:
         # mark released (new runs fail fast; the cache manager
         # re-fetches), then wait for each in-flight context before
@@ -752,6 +833,10 @@ class GpuModel:
             self._released = True
             ctx_lists = list(self._contexts.values())
             self._contexts.clear()
+        try:
+            self._fast.disable()     # waits out in-flight fast predicts
+        except Exception:       # noqa: BLE001
+            pass
         for ctxs in ctx_lists:
             for ctx in ctxs:
                 ctx.lock.acquire()

@@ -116,7 +116,10 @@ def make_cache_grpc_server(handler: LocalServingHandler,
 
     prediction = grpc.method_handlers_generic_handler(
         m.PREDICTION_SERVICE, {
-            "Predict": uu(handler.predict, m.PredictRequest),
+            # Predict runs at the bytes level: the C++ fast path consumes
+            # and produces raw wire bytes (servinghandler.predict_bytes)
+            "Predict": grpc.unary_unary_rpc_method_handler(
+                _wrap(handler.predict_bytes), bytes, lambda b: b),
             "Classify": uu(handler.classify, m.ClassificationRequest),
             "Regress": uu(handler.regress, m.RegressionRequest),
             "GetModelMetadata": uu(handler.get_model_metadata,

@@ -297,3 +297,46 @@ class LocalServingHandler:
                     np.ascontiguousarray(arr, dtype=np.float32)
                     if arr.dtype.kind == "f" else arr)))
         return resp
+
+
+def _fast_fallback_cls():
+    try:
+        from ..engine import _tfsc_engine as ext
+        return ext.FastFallback
+    except Exception:       # noqa: BLE001
+        return None
+
+
+def _predict_bytes(self, data: bytes) -> bytes:
+    """Raw-request Predict: C++ end-to-end fast path when the model's
+    engine has a registered fast context for the request's batch bucket;
+    Python decode/execute/encode otherwise (also the warm-up path that
+    builds and registers the context)."""
+    spec = m.peek_model_spec(data)
+    model, _version = self.get_model(spec.name, spec.version_value(),
+                                     spec.version_label)
+    gpu = getattr(model, "_gpu", None)
+    if gpu is not None and model._batcher is None:
+        fb = _fast_fallback_cls()
+        if fb is not None:
+            try:
+                return gpu.fast_predict(data)
+            except fb:
+                pass
+            except Exception as e:      # noqa: BLE001
+                from ..engine.gpu import ModelReleasedError
+                if not isinstance(e, ModelReleasedError):
+                    raise
+                model, _version = self.get_model(spec.name,
+                                                 spec.version_value())
+                gpu = getattr(model, "_gpu", None)
+                if gpu is not None:
+                    try:
+                        return gpu.fast_predict(data)
+                    except fb:
+                        pass
+    req = m.PredictRequest.decode(data)
+    return self.predict(req).encode()
+
+
+LocalServingHandler.predict_bytes = _predict_bytes

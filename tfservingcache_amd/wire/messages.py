@@ -396,6 +396,21 @@ class SessionRunResponse(Message):
     ]
 
 
+def peek_model_spec(data: bytes) -> ModelSpec:
+    """Fast partial parse: extract only the model_spec (field 1) of a
+    PredictRequest without touching the tensor payloads — the routing
+    hot path for the C++ fast predict."""
+    from .pb import read_varint, skip_field
+    pos, end = 0, len(data)
+    while pos < end:
+        tag, pos = read_varint(data, pos)
+        if (tag >> 3) == 1 and (tag & 7) == 2:
+            n, pos = read_varint(data, pos)
+            return ModelSpec.decode(data, pos, pos + n)
+        pos = skip_field(data, pos, tag & 7)
+    return ModelSpec()
+
+
 # --------------------------------------------------------------------------
 # gRPC service/method names (the wire-level routing contract)
 # --------------------------------------------------------------------------
