@@ -20,7 +20,8 @@ def _gpu_model(tmp_path, sm, name="m", version=1, max_batch=64):
     d = tmp_path / name / str(version)
     write_saved_model(sm, str(d))
     lm = load_model_from_dir(str(d), name, version)
-    lm._gpu = GpuModel(lm.plan, device="cuda:0", max_batch=max_batch)
+    lm._gpu = GpuModel(lm.plan, device="cuda:0", max_batch=max_batch,
+                       model_name=name, model_version=version)
     return lm
 
 
@@ -167,8 +168,6 @@ def test_fast_predict_path_matches_python(tmp_path):
 
     sm = build_mlp(d_in=64, d_hidden=128, d_out=64, seed=5)
     gm = _gpu_model(tmp_path, sm, name="fastm")
-    gm._gpu.model_name = "fastm"
-    gm._gpu.model_version = 1
     x = np.random.default_rng(0).standard_normal((4, 64)).astype(np.float32)
     req = m.PredictRequest(
         model_spec=m.ModelSpec(name="fastm", version=m.Int64Value(value=1)),

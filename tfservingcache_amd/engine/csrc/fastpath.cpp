@@ -18,6 +18,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstring>
+#include <atomic>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -224,8 +225,9 @@ struct FastContext {
 
 class FastModel {
  public:
-  FastModel(std::string model_name, int64_t version)
-      : name_(std::move(model_name)), version_(version) {}
+  FastModel(std::string model_name, int64_t version, int target_ctxs)
+      : name_(std::move(model_name)), version_(version),
+        target_(target_ctxs < 1 ? 1 : target_ctxs) {}
 
   int add_context(int bucket, uintptr_t exec_plan, uintptr_t stream,
                   std::vector<FastIO> ins, std::vector<FastIO> outs) {
@@ -411,10 +413,14 @@ class FastModel {
 
  private:
   FastContext* acquire(int batch) {
-    // smallest registered bucket >= batch; prefer an idle context of
-    // that bucket, else block on its first one. Returns LOCKED.
+    // smallest registered bucket >= batch; prefer an idle context.
+    // While fewer than `target_` contexts exist for that bucket and all
+    // are busy, raise FastFallback so the Python path runs instead —
+    // its contention handling builds (and then registers) the
+    // remaining multi-stream contexts. At steady state, block
+    // round-robin.
     int bucket = -1;
-    FastContext* fallback = nullptr;
+    std::vector<FastContext*> cand;
     {
       std::lock_guard<std::mutex> g(mu_);
       for (auto& c : ctxs_)
@@ -423,16 +429,21 @@ class FastModel {
       if (bucket < 0) throw FastFallback("no context for batch");
       for (auto& c : ctxs_) {
         if (c->bucket != bucket) continue;
-        if (fallback == nullptr) fallback = c.get();
         if (c->mu.try_lock()) return c.get();
+        cand.push_back(c.get());
       }
     }
-    fallback->mu.lock();
-    return fallback;
+    if (int(cand.size()) < target_)
+      throw FastFallback("contexts warming");
+    FastContext* c = cand[rr_++ % cand.size()];
+    c->mu.lock();
+    return c;
   }
 
   std::string name_;
   int64_t version_;
+  int target_;
+  std::atomic<unsigned> rr_{0};
   std::mutex mu_;
   std::vector<std::unique_ptr<FastContext>> ctxs_;
 };
@@ -463,7 +474,7 @@ void register_fastpath(py::module_& mod) {
       }));
 
   py::class_<FastModel>(mod, "FastModel")
-      .def(py::init<std::string, int64_t>())
+      .def(py::init<std::string, int64_t, int>())
       .def("add_context", &FastModel::add_context)
       .def("has_bucket", &FastModel::has_bucket)
       .def("lock_ctx", &FastModel::lock_ctx,

@@ -604,16 +604,16 @@ class GpuModel:
         self.device = device
         self.model_name = model_name
         self.model_version = model_version
-        # C++ fast predict path (request bytes in, response bytes out);
-        # contexts register after their first Python-path run + capture
-        self._fast = ext.FastModel(model_name or "model",
-                                   int(model_version))
-        self.max_batch = max_batch
-        self.use_graphs = use_graphs
         # up to n_streams ExecContexts per batch bucket, each with its
         # own HIP stream + workspace: concurrent requests overlap one
         # context's H2D/D2H with another's kernels
         self.n_streams = max(1, n_streams)
+        # C++ fast predict path (request bytes in, response bytes out);
+        # contexts register after their first Python-path run + capture
+        self._fast = ext.FastModel(model_name or "model",
+                                   int(model_version), self.n_streams)
+        self.max_batch = max_batch
+        self.use_graphs = use_graphs
         self._weights: Dict[int, object] = {}
         self._gemm_weights: Dict[Tuple[int, bool], object] = {}
         self._conv_weights: Dict[int, object] = {}
