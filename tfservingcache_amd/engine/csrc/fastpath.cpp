@@ -229,10 +229,15 @@ class FastModel {
       : name_(std::move(model_name)), version_(version),
         target_(target_ctxs < 1 ? 1 : target_ctxs) {}
 
-  int add_context(int bucket, uintptr_t exec_plan, uintptr_t stream,
-                  std::vector<FastIO> ins, std::vector<FastIO> outs) {
+  int add_context(int bucket, int target, uintptr_t exec_plan,
+                  uintptr_t stream, std::vector<FastIO> ins,
+                  std::vector<FastIO> outs) {
     auto ctx = std::make_unique<FastContext>();
     ctx->bucket = bucket;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      if (target > 0) bucket_target_[bucket] = target;
+    }
     ctx->exec_plan = reinterpret_cast<void*>(exec_plan);
     ctx->stream = reinterpret_cast<hipStream_t>(stream);
     ctx->ins = std::move(ins);
@@ -433,7 +438,13 @@ class FastModel {
         cand.push_back(c.get());
       }
     }
-    if (int(cand.size()) < target_)
+    int want = target_;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      auto it = bucket_target_.find(bucket);
+      if (it != bucket_target_.end()) want = it->second;
+    }
+    if (int(cand.size()) < want)
       throw FastFallback("contexts warming");
     FastContext* c = cand[rr_++ % cand.size()];
     c->mu.lock();
@@ -443,6 +454,7 @@ class FastModel {
   std::string name_;
   int64_t version_;
   int target_;
+  std::map<int, int> bucket_target_;
   std::atomic<unsigned> rr_{0};
   std::mutex mu_;
   std::vector<std::unique_ptr<FastContext>> ctxs_;

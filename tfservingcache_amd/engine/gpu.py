@@ -721,10 +721,17 @@ class GpuModel:
                 return b
         return self.max_batch
 
+    def streams_for(self, bucket: int) -> int:
+        # small batches under-fill the chip AND have cheap workspaces:
+        # give them proportionally more concurrent contexts
+        if bucket <= 4:
+            return self.n_streams * 2
+        return self.n_streams
+
     def _acquire_context(self, batch: int) -> ExecContext:
         """Returns a LOCKED ExecContext for the bucket: an idle one if
-        available, a freshly built one while under n_streams, else
-        blocks on the least-loaded."""
+        available, a freshly built one while under the bucket's stream
+        budget, else blocks on the least-loaded."""
         b = self._bucket(batch)
         with self._lock:
             if self._released:
@@ -734,7 +741,7 @@ class GpuModel:
             for ctx in ctxs:
                 if ctx.lock.acquire(blocking=False):
                     return ctx
-            if len(ctxs) < self.n_streams:
+            if len(ctxs) < self.streams_for(b):
                 ctx = ExecContext(self, b)
                 ctxs.append(ctx)
                 ctx.lock.acquire()
@@ -822,8 +829,8 @@ class GpuModel:
                                    row_elems, [int(d) for d in v.shape[1:]]))
         ctx._fast_keep = keep
         ctx.fast_id = self._fast.add_context(
-            ctx.batch, ctx.exec_plan.ptr(), ctx.stream.cuda_stream, ins,
-            outs)
+            ctx.batch, self.streams_for(ctx.batch), ctx.exec_plan.ptr(),
+            ctx.stream.cuda_stream, ins, outs)
 
     def release(self) -> None:
         # mark released (new runs fail fast; the cache manager
