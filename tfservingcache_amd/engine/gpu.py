@@ -635,7 +635,8 @@ class GpuModel:
                     w.astype(np.int32))).to(self.device)
             else:
                 wt = torch.from_numpy(np.ascontiguousarray(
-                    w.astype(np.float32))).to(self.device).to(torch.bfloat16)
+                    np.asarray(w, dtype=np.float32))).to(
+                        self.device).to(torch.bfloat16)
             self._weights[t.idx] = wt
 
     def weight_ptr(self, idx: int) -> int:

@@ -441,12 +441,12 @@ class _Lowerer:
             if nxt.op == "BiasAdd" and bias is None and bn is None and \
                     residual is None and act == ACT_NONE and \
                     self.const_value(nxt.input[1]) is not None:
-                bias = self.weight_of(nxt.input[1]).astype(np.float32)
+                bias = np.asarray(self.weight_of(nxt.input[1]), dtype=np.float32)
             elif nxt.op in ("FusedBatchNorm", "FusedBatchNormV3") and \
                     bn is None and residual is None and act == ACT_NONE and \
                     not _attr_b(nxt, "is_training", False) and \
                     all(self.const_value(r) is not None for r in nxt.input[1:5]):
-                bn = tuple(self.weight_of(r).astype(np.float32)
+                bn = tuple(np.asarray(self.weight_of(r), dtype=np.float32)
                            for r in nxt.input[1:5]) + \
                     (_attr_f(nxt, "epsilon", 1e-3),)
             elif nxt.op in ("Add", "AddV2") and residual is None and \
@@ -478,7 +478,7 @@ class _Lowerer:
         x = self.tid(x_ref)
         if _attr_s(nd, "data_format", "NHWC") != "NHWC":
             raise PlanError("only NHWC Conv2D supported")
-        w = self.weight_of(w_ref).astype(np.float32)   # [R,S,Cin,K]
+        w = np.asarray(self.weight_of(w_ref), dtype=np.float32)  # [R,S,Cin,K]
         strides = _attr_ints(nd, "strides") or [1, 1, 1, 1]
         padding = _attr_s(nd, "padding", "SAME")
         xs = self.tensors[x].shape                      # [B,H,W,C]
@@ -815,7 +815,8 @@ class _Lowerer:
         """FusedBatchNorm whose producer isn't a conv — lower to the fused
         scale/shift (+ optional downstream relu) kernel."""
         x = self.tid(nd.input[0])
-        scale, offset, mean, var = (self.weight_of(r).astype(np.float32)
+        scale, offset, mean, var = (np.asarray(self.weight_of(r),
+                                               dtype=np.float32)
                                     for r in nd.input[1:5])
         eps = _attr_f(nd, "epsilon", 1e-3)
         g_ = scale / np.sqrt(var + eps)
