@@ -159,8 +159,14 @@ class ExecContext:
             live_end.setdefault(root(i), -1)
             live_end[root(i)] = max(live_end[root(i)], 0)
 
-        # im2col scratch sizes
+        # scratch sizes (conv channel-pad, gemm K-pad)
         for oi, op in enumerate(plan.ops):
+            if op.kind == "gemm":
+                a_shape = self.shapes[op.inputs[0]]
+                K = a_shape[-1]
+                if K % 64 != 0:
+                    M = int(np.prod(a_shape[:-1]))
+                    scratch_sizes[oi] = M * _pad64(K) * 2
             if op.kind == "conv2d":
                 R, S, Cin, Kc = op.params["rsck"]
                 is_1x1 = (R == 1 and S == 1 and
@@ -245,7 +251,7 @@ class ExecContext:
             if k == "eltwise":
                 calls.extend(self._c_eltwise(op, ext))
             elif k == "gemm":
-                calls.extend(self._c_gemm(op, ext))
+                calls.extend(self._c_gemm(op, oi, ext))
             elif k == "conv2d":
                 calls.extend(self._c_conv(op, oi, ext))
             elif k == "batched_gemm":
@@ -403,7 +409,7 @@ class ExecContext:
         return [(ext.K_ELT_BINARY,
                  [self._ptr(a), self._ptr(b), self._ptr(out)], ints, [])]
 
-    def _c_gemm(self, op: PlanOp, ext):
+    def _c_gemm(self, op: PlanOp, oi: int, ext):
         p = op.params
         a = op.inputs[0]
         w_plan_idx = op.inputs[1]
@@ -423,16 +429,19 @@ class ExecContext:
             ni += 1
         if p.get("residual"):
             res_ptr = self._ptr(op.inputs[ni])
+        calls = []
+        a_ptr = self._ptr(a)
         if Kp != K:
-            # activations K not 64-aligned: pad path unsupported (planner
-            # pads weights only) — requires K % 64 == 0
-            raise RuntimeError(f"gemm K={K} must be 64-aligned "
-                               f"(weight padded to {Kp})")
-        return [(ext.K_GEMM,
-                 [self._ptr(a), wt.data_ptr(), bias_ptr, res_ptr,
-                  self._ptr(op.outputs[0])],
-                 [M, N, K, _ACT_CODE[p.get("act", "none")]],
-                 [1.0])]
+            # zero-pad the activation's K to the weight's 64-aligned Kp
+            scratch = self.workspace.data_ptr() + self.scratch_off[oi]
+            calls.append((ext.K_PAD_LAST, [a_ptr, scratch], [M, K, Kp], []))
+            a_ptr = scratch
+        calls.append((ext.K_GEMM,
+                      [a_ptr, wt.data_ptr(), bias_ptr, res_ptr,
+                       self._ptr(op.outputs[0])],
+                      [M, N, Kp, _ACT_CODE[p.get("act", "none")]],
+                      [1.0]))
+        return calls
 
     def _c_conv(self, op: PlanOp, oi: int, ext):
         p = op.params
