@@ -121,14 +121,14 @@ def main() -> int:
     ap.add_argument("--pool-size", type=int, default=10,
                     help="models resident per GPU (lru mode)")
     ap.add_argument("--image-size", type=int, default=224)
-    ap.add_argument("--threads", type=int, default=6,
+    ap.add_argument("--threads", type=int, default=10,
                     help="concurrent client threads per rank")
     ap.add_argument("--zipf", type=float, default=1.1)
     ap.add_argument("--cpu", action="store_true",
                     help="CPU engine (CI smoke only; not a benchmark)")
     ap.add_argument("--dyn-batch", action="store_true",
                     help="enable server-side dynamic batching")
-    ap.add_argument("--streams", type=int, default=4,
+    ap.add_argument("--streams", type=int, default=6,
                     help="execution contexts (HIP streams) per model")
     ap.add_argument("--transport", choices=["inproc", "grpc"],
                     default="inproc",

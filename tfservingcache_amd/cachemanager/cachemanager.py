@@ -160,7 +160,7 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
                     batching: bool = False,
                     batch_timeout_s: float = 0.002,
                     devices: Optional[List[str]] = None,
-                    n_streams: int = 4
+                    n_streams: int = 6
                     ) -> Callable[[str, int], LoadedModel]:
     """Loader that compiles the SavedModel onto an MI355X: weights land
     in the GPU's HBM pool (bf16, GEMM layouts pre-transformed) and the

@@ -606,7 +606,7 @@ class GpuModel:
 
     def __init__(self, plan: Plan, device: str = "cuda:0",
                  max_batch: int = 64, use_graphs: bool = True,
-                 n_streams: int = 4, model_name: str = "",
+                 n_streams: int = 6, model_name: str = "",
                  model_version: int = 0):
         torch, ext = _load_backend()
         self.plan = plan
