@@ -227,3 +227,31 @@ def test_metrics_merger_merges_external():
         assert "tfservingcache_proxy_requests" in text
     finally:
         srv.shutdown()
+
+
+def test_file_discovery_member_expiry(tmp_path):
+    """A member that stops heartbeating drops out after 3*TTL (the
+    lease-expiry liveness contract, etcd.go:139-146 analog)."""
+    d1 = FileDiscovery(str(tmp_path / "c"), heartbeat_ttl=0.3,
+                       poll_interval=0.1)
+    d2 = FileDiscovery(str(tmp_path / "c"), heartbeat_ttl=0.3,
+                       poll_interval=0.1)
+    seen = []
+    d1.add_listener(lambda ms: seen.append([s.host for s in ms]))
+    try:
+        d1.register(ServingService("alive", 1, 2))
+        d2.register(ServingService("dying", 3, 4))
+        deadline = time.time() + 5
+        while time.time() < deadline and (
+                not seen or sorted(seen[-1]) != ["alive", "dying"]):
+            time.sleep(0.05)
+        assert sorted(seen[-1]) == ["alive", "dying"]
+        # kill d2's heartbeat without deregistering (simulated crash)
+        d2._stop.set()
+        deadline = time.time() + 6
+        while time.time() < deadline and seen[-1] != ["alive"]:
+            time.sleep(0.1)
+        assert seen[-1] == ["alive"]
+    finally:
+        d1.unregister()
+        d2.unregister()

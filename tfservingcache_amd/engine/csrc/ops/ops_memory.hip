@@ -396,9 +396,74 @@ __global__ void k_pool(const ushort* __restrict__ x, ushort* __restrict__ y,
   }
 }
 
+// vectorized pool (C % 8 == 0): one thread per (n,ho,wo,c8-group),
+// 16-byte loads/stores — 1/8 the index math of the scalar kernel
+__global__ void k_pool_v8(const ushort* __restrict__ x,
+                          ushort* __restrict__ y, bool is_max, int N,
+                          int H, int W, int C, int Ho, int Wo, int kh,
+                          int kw, int sh, int sw, int pt, int pl) {
+  int c8 = C >> 3;
+  int64_t n_grp = (int64_t)N * Ho * Wo * c8;
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n_grp; i += stride) {
+    int g = int(i % c8);
+    int64_t t = i / c8;
+    int wo = int(t % Wo); t /= Wo;
+    int ho = int(t % Ho); int n = int(t / Ho);
+    float acc[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] = is_max ? -3.0e38f : 0.f;
+    int count = 0;
+    for (int r = 0; r < kh; ++r) {
+      int hi = ho * sh + r - pt;
+      if (hi < 0 || hi >= H) continue;
+      for (int q = 0; q < kw; ++q) {
+        int wi = wo * sw + q - pl;
+        if (wi < 0 || wi >= W) continue;
+        short4_t lo = *reinterpret_cast<const short4_t*>(
+            x + (((int64_t)n * H + hi) * W + wi) * C + g * 8);
+        short4_t hi4 = *reinterpret_cast<const short4_t*>(
+            x + (((int64_t)n * H + hi) * W + wi) * C + g * 8 + 4);
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float a = bf2f((ushort)lo[j]);
+          float b = bf2f((ushort)hi4[j]);
+          if (is_max) {
+            acc[j] = fmaxf(acc[j], a);
+            acc[j + 4] = fmaxf(acc[j + 4], b);
+          } else {
+            acc[j] += a;
+            acc[j + 4] += b;
+          }
+        }
+        ++count;
+      }
+    }
+    short4_t out_lo, out_hi;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float a = is_max ? acc[j] : (count ? acc[j] / count : 0.f);
+      float b = is_max ? acc[j + 4] : (count ? acc[j + 4] / count : 0.f);
+      out_lo[j] = (short)f2bf(a);
+      out_hi[j] = (short)f2bf(b);
+    }
+    ushort* dst = y + i * 8;
+    *reinterpret_cast<short4_t*>(dst) = out_lo;
+    *reinterpret_cast<short4_t*>(dst + 4) = out_hi;
+  }
+}
+
 void launch_pool(hipStream_t s, const ushort* x, ushort* y, bool is_max,
                  int N, int H, int W, int C, int Ho, int Wo, int kh, int kw,
                  int sh, int sw, int pt, int pl) {
+  if (C % 8 == 0) {
+    int64_t n_grp = (int64_t)N * Ho * Wo * (C / 8);
+    hipLaunchKernelGGL(k_pool_v8, dim3(grid_for(n_grp)), dim3(TPB), 0, s,
+                       x, y, is_max, N, H, W, C, Ho, Wo, kh, kw, sh, sw,
+                       pt, pl);
+    return;
+  }
   int64_t n_out = (int64_t)N * Ho * Wo * C;
   hipLaunchKernelGGL(k_pool, dim3(grid_for(n_out)), dim3(TPB), 0, s,
                      x, y, is_max, N, H, W, C, Ho, Wo, kh, kw, sh, sw,
