@@ -292,3 +292,37 @@ def test_version_labels(stack):
         inputs={"x": numpy_to_tensorproto(
             np.array([2.0], dtype=np.float32))}))
     assert presp.model_spec.version.value == 123
+
+
+def test_pool_byte_budget(tmp_path):
+    """engine.hbmPoolBytes caps resident bytes, not just model count."""
+    repo = tmp_path / "r2"
+    write_model_repo(str(repo), [("m_a", 1, "mlp"), ("m_b", 1, "mlp"),
+                                 ("m_c", 1, "mlp")])
+    provider = DiskModelProvider(str(repo))
+    cache = LRUCache(str(tmp_path / "c2"), 10 ** 8)
+    sizes = {}
+
+    def size_hint(name, version):
+        e = cache.get(name, version)
+        return e.size_on_disk if e else 0
+
+    pool = ModelPool(make_cpu_loader(cache), max_concurrent_models=10,
+                     max_bytes=None, size_hint=size_hint)
+    cm = CacheManager(provider, cache, pool)
+    handler = LocalServingHandler(cm)
+    x = np.zeros((1, 16), dtype=np.float32)
+    handler.predict(m.PredictRequest(
+        model_spec=m.ModelSpec(name="m_a", version=m.Int64Value(value=1)),
+        inputs={"x": numpy_to_tensorproto(x)}))
+    one_size = cache.get("m_a", 1).size_on_disk
+    # now budget for ~1.5 models
+    pool.max_bytes = int(one_size * 1.5)
+    for name in ("m_b", "m_c"):
+        handler.predict(m.PredictRequest(
+            model_spec=m.ModelSpec(name=name, version=m.Int64Value(value=1)),
+            inputs={"x": numpy_to_tensorproto(x)}))
+    time.sleep(0.2)
+    avail = [mid for mid, s in cm.pool.model_states().items()
+             if s == m.STATE_AVAILABLE]
+    assert len(avail) == 1          # byte budget, not count, limited it

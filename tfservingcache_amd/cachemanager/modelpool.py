@@ -49,10 +49,21 @@ class ModelPool:
 
     def __init__(self, loader: Callable[[str, int], object],
                  max_concurrent_models: int = 2,
-                 device: str = "cpu"):
-        """loader(name, version) -> LoadedModel (blocking compile+upload)."""
+                 device: str = "cpu",
+                 max_bytes: Optional[int] = None,
+                 size_hint: Optional[Callable[[str, int], int]] = None):
+        """loader(name, version) -> LoadedModel (blocking compile+upload).
+
+        max_bytes: optional HBM budget for resident models (the per-GPU
+        pool is sized for 288 GB of HBM3E; `serving.maxConcurrentModels`
+        caps the COUNT like the reference, `engine.hbmPoolBytes` caps the
+        BYTES). size_hint(name, version) estimates a model's device
+        footprint before it is loaded (e.g. 3x its on-disk size for
+        fp32->bf16 masters + transformed GEMM layouts)."""
         self._loader = loader
         self.max_concurrent = max_concurrent_models
+        self.max_bytes = max_bytes
+        self.size_hint = size_hint
         self.device = device
         self._lock = threading.Lock()
         self._cond = threading.Condition(self._lock)
@@ -95,6 +106,19 @@ class ModelPool:
         unloaded; new ones load on background threads, signalled via the
         condition variable (replaces the reference's 500 ms status poll)."""
         desired = list(desired)[: self.max_concurrent]
+        if self.max_bytes is not None and self.size_hint is not None:
+            budgeted = []
+            total = 0
+            for mid in desired:
+                try:
+                    est = self.size_hint(*mid)
+                except Exception:       # noqa: BLE001
+                    est = 0
+                if budgeted and total + est > self.max_bytes:
+                    break
+                budgeted.append(mid)
+                total += est
+            desired = budgeted
         want = set(desired)
         to_load: List[Tuple[str, int]] = []
         with self._lock:

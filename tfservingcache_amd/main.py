@@ -146,8 +146,17 @@ def create_cache_manager(cfg: Config) -> CacheManager:
     else:
         loader = make_cpu_loader(cache)
         device = "cpu"
+    hbm_budget = cfg.get_int("engine.hbmPoolBytes") or None
+
+    def size_hint(name: str, version: int) -> int:
+        # device footprint estimate: ~3x on-disk (fp32 master + bf16 +
+        # transformed GEMM layouts)
+        entry = cache.get(name, version)
+        return 3 * entry.size_on_disk if entry else 0
+
     pool = ModelPool(loader, max_concurrent_models=max_models,
-                     device=device)
+                     device=device, max_bytes=hbm_budget,
+                     size_hint=size_hint if hbm_budget else None)
     fetch_timeout = cfg.get_float("serving.grpcConfigTimeout") or 10.0
     return CacheManager(provider, cache, pool,
                         model_fetch_timeout=fetch_timeout,
