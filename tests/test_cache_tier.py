@@ -326,3 +326,32 @@ def test_pool_byte_budget(tmp_path):
     avail = [mid for mid, s in cm.pool.model_states().items()
              if s == m.STATE_AVAILABLE]
     assert len(avail) == 1          # byte budget, not count, limited it
+
+
+def test_warmup_requests_executed(tmp_path):
+    """assets.extra/tf_serving_warmup_requests run at load (TF Serving
+    warmup parity)."""
+    import os
+    from tfservingcache_amd.engine import warmup as wu
+    from tfservingcache_amd.engine.model import load_model_from_dir
+    from tfservingcache_amd.engine.savedmodel import write_saved_model
+    from tfservingcache_amd.models import build_mlp
+
+    vdir = tmp_path / "wm" / "1"
+    write_saved_model(build_mlp(seed=4), str(vdir))
+    req = m.PredictRequest(
+        model_spec=m.ModelSpec(name="wm", version=m.Int64Value(value=1)),
+        inputs={"x": numpy_to_tensorproto(
+            np.zeros((2, 16), dtype=np.float32))})
+    rec = wu.PredictionLog(predict_log=wu.PredictLog(request=req)).encode()
+    wu.write_tfrecord(str(vdir / wu.WARMUP_PATH), [rec, rec])
+
+    lm = load_model_from_dir(str(vdir), "wm", 1)
+    calls = []
+    orig = lm.predict
+    lm.predict = lambda *a, **k: (calls.append(1), orig(*a, **k))[1]
+    n = wu.run_warmup(lm, str(vdir))
+    assert n == 2
+    assert len(calls) == 2
+    # roundtrip of the tfrecord reader
+    assert len(list(wu.read_tfrecords(str(vdir / wu.WARMUP_PATH)))) == 2

@@ -179,6 +179,7 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
 
     def loader(name: str, version: int) -> LoadedModel:
         from ..engine.gpu import GpuModel
+        from ..engine.warmup import run_warmup
         vdir = os.path.join(cache.base_dir, name, str(version))
         lm = load_model_from_dir(vdir, name, version)
         dev = pick_device(name, version)
@@ -189,5 +190,8 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
         if batching:
             lm.enable_batching(max_batch=max_batch,
                                timeout_s=batch_timeout_s)
+        # TF Serving warmup files (assets.extra/tf_serving_warmup_requests):
+        # contexts build + hipGraphs capture before AVAILABLE
+        run_warmup(lm, vdir)
         return lm
     return loader
