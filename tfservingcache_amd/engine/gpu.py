@@ -71,6 +71,20 @@ def gpu_available() -> bool:
 
 ALIGN = 256  # byte alignment of workspace slices
 
+# shared pinned staging for weight uploads (64 MiB bf16), reused across
+# model loads under a lock — page-locking memory per load is slower than
+# the copies it serves
+_staging_lock = threading.Lock()
+_staging_buf = None
+
+
+def _get_staging(torch):
+    global _staging_buf
+    if _staging_buf is None:
+        _staging_buf = torch.empty(32 * 1024 * 1024, dtype=torch.bfloat16,
+                                   pin_memory=True)
+    return _staging_buf
+
 
 def _pad64(k: int) -> int:
     return (k + 63) // 64 * 64
@@ -655,12 +669,39 @@ class GpuModel:
                 total += (n + 127) // 128 * 128   # keep 256B alignment
         if not float_ws:
             return
-        pin = torch.empty(total, dtype=torch.bfloat16, pin_memory=True)
-        for _idx, w, off, n in float_ws:
-            src = torch.from_numpy(np.ascontiguousarray(
-                np.asarray(w, dtype=np.float32))).view(-1)
-            pin[off:off + n].copy_(src)       # CPU f32 -> bf16 convert
-        blob = pin.to(self.device, non_blocking=False)
+        blob = torch.empty(total, dtype=torch.bfloat16, device=self.device)
+        # stage through a SHARED reusable pinned buffer: per-load pinned
+        # allocation (page-locking ~100 MB) costs more than the copy and
+        # thrashes badly under LRU churn
+        with _staging_lock:
+            stage = _get_staging(torch)
+            cap = stage.numel()
+            batch_items = []        # (stage_off, blob_off, n)
+            stage_off = 0
+
+            def flush():
+                nonlocal stage_off
+                for s_off, b_off, n_ in batch_items:
+                    blob[b_off:b_off + n_].copy_(
+                        stage[s_off:s_off + n_], non_blocking=True)
+                torch.cuda.synchronize(self.device)
+                batch_items.clear()
+                stage_off = 0
+
+            for _idx, w, off, n in float_ws:
+                src = torch.from_numpy(np.ascontiguousarray(
+                    np.asarray(w, dtype=np.float32))).view(-1)
+                done = 0
+                while done < n:
+                    if stage_off >= cap:
+                        flush()
+                    take = min(n - done, cap - stage_off)
+                    stage[stage_off:stage_off + take].copy_(
+                        src[done:done + take])   # CPU f32->bf16 convert
+                    batch_items.append((stage_off, off + done, take))
+                    stage_off += take
+                    done += take
+            flush()
         self._weight_blob = blob              # keep the allocation alive
         for idx, w, off, n in float_ws:
             self._weights[idx] = blob[off:off + n].view(tuple(w.shape))
