@@ -219,8 +219,13 @@ class Server:
             def pick_grpc(model, version):
                 return self.cluster.node_for_key(model, version).grpc_addr
 
+            from .tfservingproxy import GrpcForwarder
+            fwd = GrpcForwarder(
+                max_msg=self.cfg.get_int("serving.grpcMaxMsgSize") or
+                16 * 1024 * 1024,
+                timeout_s=self.cfg.get_float("proxy.grpcTimeout") or 10.0)
             proxy_grpc, _, self._fwd = make_proxy_grpc_server(
-                pick_grpc, health=self.health,
+                pick_grpc, forwarder=fwd, health=self.health,
                 max_msg=self.cfg.get_int("serving.grpcMaxMsgSize") or
                 16 * 1024 * 1024)
             proxy_grpc.add_insecure_port(f"[::]:{self.proxy_grpc_port}")

@@ -10,6 +10,7 @@ protocol fronts shared one CacheManager (cachemanager.go:268-309).
 from __future__ import annotations
 
 import logging
+import os
 from typing import Dict, List, Optional, Tuple
 
 import numpy as np
@@ -256,6 +257,20 @@ class LocalServingHandler:
                 versions: List[int] = []
                 if pol is not None and pol.specific is not None:
                     versions = [int(v) for v in pol.specific.versions]
+                elif pol is not None and pol.latest is not None and \
+                        pol.latest.num_versions > 1 and \
+                        hasattr(self.cm.provider, "latest_version"):
+                    # latest{num_versions: N}: the N newest versions
+                    base = os.path.join(self.cm.provider.base_dir, mc.name) \
+                        if hasattr(self.cm.provider, "base_dir") else None
+                    found = []
+                    if base and os.path.isdir(base):
+                        for entry in os.listdir(base):
+                            try:
+                                found.append(int(entry))
+                            except ValueError:
+                                pass
+                    versions = sorted(found)[-pol.latest.num_versions:]
                 if not versions:
                     v = self._resolve_version(mc.name, 0)
                     versions = [v]
