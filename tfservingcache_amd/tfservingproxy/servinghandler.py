@@ -176,8 +176,15 @@ class LocalServingHandler:
             feats = {"": next(iter(feats.values()))}
         outputs, version = self.predict_arrays(spec.name,
                                                spec.version_value(), feats)
-        # scores: first output, rows = examples
-        scores = next(iter(outputs.values()))
+        # TF classification signatures name their outputs 'scores' (and
+        # 'classes'); fall back to the first output otherwise
+        scores = None
+        for key in ("scores", "probabilities", "probs"):
+            if key in outputs:
+                scores = outputs[key]
+                break
+        if scores is None:
+            scores = next(iter(outputs.values()))
         scores = np.atleast_2d(np.asarray(scores, dtype=np.float32))
         result = m.ClassificationResult()
         for row in scores:
