@@ -256,6 +256,19 @@ def main() -> int:
         def pick(i):
             return names[0]
 
+    lat_lock = threading.Lock()
+    latencies = []          # per-request wall seconds (timed section)
+    timing_on = [False]
+
+    def timed(fn):
+        def run(i):
+            t0 = time.monotonic()
+            fn(i)
+            if timing_on[0]:
+                with lat_lock:
+                    latencies.append(time.monotonic() - t0)
+        return run
+
     grpc_server = None
     if args.transport == "grpc":
         import grpc as grpc_mod
@@ -280,6 +293,8 @@ def main() -> int:
         def one_request(i: int) -> None:
             handler.predict_bytes(req_cache[pick(i)])
 
+    one_request = timed(one_request)
+
     # initial load (timed -> cold-load sample even in warm mode)
     t0 = time.monotonic()
     one_request(0)
@@ -303,9 +318,11 @@ def main() -> int:
         step(w * REQS_PER_STEP)
 
     barrier_sync()
+    timing_on[0] = True
     t_start = time.monotonic()
     for k in range(args.steps):
         step((args.warmup + k) * REQS_PER_STEP)
+    timing_on[0] = False
     barrier_sync()
     elapsed = time.monotonic() - t_start
 
@@ -355,6 +372,14 @@ def main() -> int:
                 "n_models": n_models,
                 "pool_size": pool_cap,
                 "cold_load_p50_ms": round(cold_p50, 1),
+                "latency_ms": {
+                    "p50": round(float(np.percentile(latencies, 50)) * 1e3,
+                                 2) if latencies else None,
+                    "p95": round(float(np.percentile(latencies, 95)) * 1e3,
+                                 2) if latencies else None,
+                    "p99": round(float(np.percentile(latencies, 99)) * 1e3,
+                                 2) if latencies else None,
+                },
                 "n_cold_loads": len(loads),
                 "transport": ("real gRPC server+client over TCP loopback"
                               if args.transport == "grpc" else

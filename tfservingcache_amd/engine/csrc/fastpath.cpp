@@ -497,7 +497,8 @@ void register_fastpath(py::module_& mod) {
       .def("predict", [](FastModel& fm, py::bytes data) {
         char* buf = nullptr;
         Py_ssize_t len = 0;
-        PyBytes_AsStringAndSize(data.ptr(), &buf, &len);
+        if (PyBytes_AsStringAndSize(data.ptr(), &buf, &len) != 0)
+          throw py::error_already_set();
         std::string out;
         {
           py::gil_scoped_release rel;

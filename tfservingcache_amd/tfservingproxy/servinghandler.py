@@ -341,8 +341,12 @@ def _predict_bytes(self, data: bytes) -> bytes:
     if gpu is not None and model._batcher is None:
         fb = _fast_fallback_cls()
         if fb is not None:
+            from ..utils import metrics as mt
             try:
-                return gpu.fast_predict(data)
+                with mt.engine_predict_duration.labels(
+                        model.name, str(model.version),
+                        model.device).time():
+                    return gpu.fast_predict(data)
             except fb:
                 pass
             except Exception as e:      # noqa: BLE001
