@@ -188,8 +188,14 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
                            model_name=name, model_version=version)
         lm.device = dev
         if batching:
-            lm.enable_batching(max_batch=max_batch,
-                               timeout_s=batch_timeout_s)
+            # merging happens inside the C++ fast path (leader-follower
+            # on raw requests); the Python batcher stays off so
+            # predict_bytes keeps taking the fast path. Requests that
+            # fall back to the Python path run unmerged.
+            lm._gpu._fast.enable_batching(
+                max_batch, int(batch_timeout_s * 1e6))
+            # the merged bucket must exist before the C++ path can merge
+            lm._gpu.prewarm(max_batch)
         # TF Serving warmup files (assets.extra/tf_serving_warmup_requests):
         # contexts build + hipGraphs capture before AVAILABLE
         run_warmup(lm, vdir)

@@ -19,6 +19,8 @@
 
 #include <cstring>
 #include <atomic>
+#include <chrono>
+#include <condition_variable>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -243,8 +245,21 @@ class FastModel {
     ctx->ins = std::move(ins);
     ctx->outs = std::move(outs);
     std::lock_guard<std::mutex> g(mu_);
+    if (specs_in_.empty()) {
+      specs_in_ = ctx->ins;
+      specs_out_ = ctx->outs;
+    }
     ctxs_.push_back(std::move(ctx));
     return int(ctxs_.size()) - 1;
+  }
+
+  // server-side dynamic batching inside the fast path: requests whose
+  // batch is below `merge_cap` wait up to `timeout_us` to merge with
+  // concurrent requests into one plan execution (leader-follower)
+  void enable_batching(int merge_cap, int64_t timeout_us) {
+    merge_cap_ = merge_cap;
+    timeout_us_ = timeout_us;
+    batching_ = merge_cap > 1;
   }
 
   // shared-lock interface: the Python execution path locks the same
@@ -270,10 +285,34 @@ class FastModel {
     return false;
   }
 
+  struct Entry {
+    const ParsedRequest* req;
+    int64_t rows = 0;
+    std::string out;
+    int err = -1;                 // -1 pending, 0 ok, 1 fallback, 2 error
+    std::string err_msg;
+    bool promoted = false;        // follower promoted to leader
+  };
+
   std::string predict(const uint8_t* data, size_t len) {
     ParsedRequest req = parse_request(data, len);
+    int64_t rows = validate(req);
+    Entry me;
+    me.req = &req;
+    me.rows = rows;
+    if (!batching_ || rows >= merge_cap_) {
+      Entry* one[1] = {&me};
+      run_batch(one, 1);
+    } else {
+      grouped_predict(me);
+    }
+    if (me.err == 1) throw FastFallback(me.err_msg);
+    if (me.err == 2) throw std::runtime_error(me.err_msg);
+    return std::move(me.out);
+  }
+
+  int64_t validate(const ParsedRequest& req) {
     if (req.inputs.empty()) throw FastFallback("no inputs");
-    // batch from the first input's leading dim
     int64_t batch = -1;
     for (auto& kv : req.inputs) {
       if (kv.second.has_typed_vals && !kv.second.content)
@@ -284,18 +323,12 @@ class FastModel {
       else if (batch != kv.second.dims[0])
         throw FastFallback("inconsistent batch");
     }
-
-    FastContext* ctx = acquire(int(batch));
-    std::lock_guard<std::mutex> g2(ctx->mu, std::adopt_lock);
-    if (ctx->disabled) throw FastFallback("model released");
-
-    hipStream_t s = ctx->stream;
-    // inputs
-    for (auto& io : ctx->ins) {
+    std::lock_guard<std::mutex> g(mu_);
+    if (specs_in_.empty()) throw FastFallback("no contexts yet");
+    for (auto& io : specs_in_) {
       auto it = req.inputs.find(io.alias);
       if (it == req.inputs.end()) throw FastFallback("missing input");
       const ParsedTensor& t = it->second;
-      // validate trailing dims
       if (int64_t(t.dims.size()) != int64_t(io.tail_dims.size()) + 1)
         throw FastFallback("rank mismatch");
       int64_t row = 1;
@@ -304,16 +337,115 @@ class FastModel {
           throw FastFallback("dim mismatch");
         row *= t.dims[d + 1];
       }
-      size_t esz = 4;            // f32 or i32 on the wire
-      int dt = t.dtype;
       if (io.is_int) {
-        if (dt != 3) throw FastFallback("want DT_INT32");
-      } else if (dt != 1) {
+        if (t.dtype != 3) throw FastFallback("want DT_INT32");
+      } else if (t.dtype != 1) {
         throw FastFallback("want DT_FLOAT");
       }
-      size_t want = size_t(batch) * row * esz;
-      if (t.content_len != want) throw FastFallback("content size");
-      std::memcpy(reinterpret_cast<void*>(io.pin), t.content, want);
+      if (t.content_len != size_t(batch) * row * 4)
+        throw FastFallback("content size");
+    }
+    return batch;
+  }
+
+  void grouped_predict(Entry& me) {
+    // one merge group per output_filter signature
+    std::string key;
+    for (auto& f : me.req->output_filter) {
+      key += f;
+      key.push_back(0);
+    }
+    BatchGroup* g;
+    {
+      std::lock_guard<std::mutex> gm(groups_mu_);
+      auto& slot = groups_[key];
+      if (!slot) slot = std::make_unique<BatchGroup>();
+      g = slot.get();
+    }
+    std::unique_lock<std::mutex> gl(g->m);
+    g->q.push_back(&me);
+    g->cv.notify_all();
+    if (g->leader_active) {
+      g->cv.wait(gl, [&] { return me.err != -1 || me.promoted; });
+      if (me.err != -1) return;       // served (or failed) by a leader
+      // promoted: fall through as the new leader
+    } else {
+      g->leader_active = true;
+    }
+    // leader: wait for followers until the window closes or full
+    auto deadline = std::chrono::steady_clock::now() +
+                    std::chrono::microseconds(timeout_us_);
+    while (std::chrono::steady_clock::now() < deadline) {
+      int64_t total = 0;
+      for (auto* e : g->q) total += e->rows;
+      if (total >= merge_cap_) break;
+      if (g->cv.wait_until(gl, deadline) == std::cv_status::timeout)
+        break;
+    }
+    // take a batch (cap by merge_cap_), leave the rest to a promoted
+    // leader
+    std::vector<Entry*> take;
+    int64_t total = 0;
+    size_t i = 0;
+    for (; i < g->q.size(); ++i) {
+      Entry* e = g->q[i];
+      if (!take.empty() && total + e->rows > merge_cap_) break;
+      take.push_back(e);
+      total += e->rows;
+    }
+    g->q.erase(g->q.begin(), g->q.begin() + i);
+    if (!g->q.empty()) {
+      g->q.front()->promoted = true;
+    } else {
+      g->leader_active = false;
+    }
+    g->cv.notify_all();
+    gl.unlock();
+
+    run_batch(take.data(), take.size());
+
+    gl.lock();
+    g->cv.notify_all();
+  }
+
+  void run_batch(Entry** batch, size_t n) {
+    try {
+      run_batch_inner(batch, n);
+      for (size_t i = 0; i < n; ++i)
+        if (batch[i]->err == -1) batch[i]->err = 0;
+    } catch (const FastFallback& e) {
+      for (size_t i = 0; i < n; ++i) {
+        batch[i]->err = 1;
+        batch[i]->err_msg = e.what();
+      }
+    } catch (const std::exception& e) {
+      for (size_t i = 0; i < n; ++i) {
+        batch[i]->err = 2;
+        batch[i]->err_msg = e.what();
+      }
+    }
+  }
+
+  void run_batch_inner(Entry** batch, size_t n) {
+    int64_t total = 0;
+    for (size_t i = 0; i < n; ++i) total += batch[i]->rows;
+
+    FastContext* ctx = acquire(int(total));
+    std::lock_guard<std::mutex> g2(ctx->mu, std::adopt_lock);
+    if (ctx->disabled) throw FastFallback("model released");
+
+    hipStream_t s = ctx->stream;
+    for (auto& io : ctx->ins) {
+      int64_t row = io.row_elems;
+      size_t row_off = 0;
+      for (size_t i = 0; i < n; ++i) {
+        const ParsedTensor& t = batch[i]->req->inputs.at(io.alias);
+        size_t nb = size_t(batch[i]->rows) * row * 4;
+        std::memcpy(reinterpret_cast<char*>(io.pin) + row_off * row * 4,
+                    t.content, nb);
+        row_off += size_t(batch[i]->rows);
+      }
+      size_t want = size_t(total) * row * 4;
       if (io.is_int) {
         HIPCHK(hipMemcpyAsync(reinterpret_cast<void*>(io.dev),
                               reinterpret_cast<void*>(io.pin), want,
@@ -329,9 +461,9 @@ class FastModel {
         launch_f32_to_bf16(s,
                            reinterpret_cast<const float*>(io.dev_stage),
                            reinterpret_cast<ushort*>(io.dev),
-                           batch * row);
+                           total * row);
         size_t cap2 = size_t(ctx->bucket) * row * 2;
-        size_t used2 = size_t(batch) * row * 2;
+        size_t used2 = size_t(total) * row * 2;
         if (used2 < cap2)
           HIPCHK(hipMemsetAsync(
               reinterpret_cast<char*>(io.dev) + used2, 0, cap2 - used2,
@@ -341,12 +473,12 @@ class FastModel {
 
     fast_run_plan(ctx->exec_plan, s);
 
-    // outputs: cast + D2H
+    const auto& filt = batch[0]->req->output_filter;
     std::vector<const FastIO*> wanted;
     for (auto& io : ctx->outs) {
-      if (!req.output_filter.empty()) {
+      if (!filt.empty()) {
         bool keep = false;
-        for (auto& f : req.output_filter)
+        for (auto& f : filt)
           if (f == io.alias) keep = true;
         if (!keep) continue;
       }
@@ -355,10 +487,10 @@ class FastModel {
     for (auto* io : wanted) {
       launch_bf16_to_f32(s, reinterpret_cast<const ushort*>(io->dev),
                          reinterpret_cast<float*>(io->dev_stage),
-                         batch * io->row_elems);
+                         total * io->row_elems);
       HIPCHK(hipMemcpyAsync(reinterpret_cast<void*>(io->pin),
                             reinterpret_cast<void*>(io->dev_stage),
-                            size_t(batch) * io->row_elems * 4,
+                            size_t(total) * io->row_elems * 4,
                             hipMemcpyDeviceToHost, s));
     }
     HIPCHK(hipStreamSynchronize(s));
@@ -367,53 +499,58 @@ class FastModel {
       throw std::runtime_error(std::string("fastpath kernel error: ") +
                                hipGetErrorString(ke));
 
-    // serialize PredictResponse
-    std::string out;
-    out.reserve(wanted.size() * 64 + 1024);
-    for (auto* io : wanted) {
-      std::string tp;
-      w_tag(tp, 1, 0);                 // dtype = DT_FLOAT
-      w_varint(tp, 1);
-      {
-        std::string shape;
+    // per-request responses from row slices of the pinned outputs
+    size_t row_off = 0;
+    for (size_t i = 0; i < n; ++i) {
+      int64_t rows = batch[i]->rows;
+      std::string& out = batch[i]->out;
+      out.reserve(wanted.size() * 64 + 1024);
+      for (auto* io : wanted) {
+        std::string tp;
+        w_tag(tp, 1, 0);
+        w_varint(tp, 1);
         {
-          std::string dim;
-          w_tag(dim, 1, 0);
-          w_varint(dim, uint64_t(batch));
-          w_len_prefixed(shape, 2, dim);
+          std::string shape;
+          {
+            std::string dim;
+            w_tag(dim, 1, 0);
+            w_varint(dim, uint64_t(rows));
+            w_len_prefixed(shape, 2, dim);
+          }
+          for (int64_t d : io->tail_dims) {
+            std::string dim;
+            w_tag(dim, 1, 0);
+            w_varint(dim, uint64_t(d));
+            w_len_prefixed(shape, 2, dim);
+          }
+          w_len_prefixed(tp, 2, shape);
         }
-        for (int64_t d : io->tail_dims) {
-          std::string dim;
-          w_tag(dim, 1, 0);
-          w_varint(dim, uint64_t(d));
-          w_len_prefixed(shape, 2, dim);
-        }
-        w_len_prefixed(tp, 2, shape);
+        size_t nbytes = size_t(rows) * io->row_elems * 4;
+        w_tag(tp, 4, 2);
+        w_varint(tp, nbytes);
+        tp.append(reinterpret_cast<const char*>(io->pin) +
+                      row_off * io->row_elems * 4,
+                  nbytes);
+        std::string entry;
+        w_tag(entry, 1, 2);
+        w_varint(entry, io->alias.size());
+        entry += io->alias;
+        w_len_prefixed(entry, 2, tp);
+        w_len_prefixed(out, 1, entry);
       }
-      size_t nbytes = size_t(batch) * io->row_elems * 4;
-      w_tag(tp, 4, 2);
-      w_varint(tp, nbytes);
-      tp.append(reinterpret_cast<const char*>(io->pin), nbytes);
-
-      std::string entry;
-      w_tag(entry, 1, 2);
-      w_varint(entry, io->alias.size());
-      entry += io->alias;
-      w_len_prefixed(entry, 2, tp);
-      w_len_prefixed(out, 1, entry);   // outputs map
+      {
+        std::string spec;
+        w_tag(spec, 1, 2);
+        w_varint(spec, name_.size());
+        spec += name_;
+        std::string ver;
+        w_tag(ver, 1, 0);
+        w_varint(ver, uint64_t(version_));
+        w_len_prefixed(spec, 2, ver);
+        w_len_prefixed(out, 2, spec);
+      }
+      row_off += size_t(rows);
     }
-    {
-      std::string spec;
-      w_tag(spec, 1, 2);
-      w_varint(spec, name_.size());
-      spec += name_;
-      std::string ver;
-      w_tag(ver, 1, 0);
-      w_varint(ver, uint64_t(version_));
-      w_len_prefixed(spec, 2, ver);
-      w_len_prefixed(out, 2, spec);    // model_spec
-    }
-    return out;
   }
 
  private:
@@ -451,6 +588,13 @@ class FastModel {
     return c;
   }
 
+  struct BatchGroup {
+    std::mutex m;
+    std::condition_variable cv;
+    std::vector<Entry*> q;
+    bool leader_active = false;
+  };
+
   std::string name_;
   int64_t version_;
   int target_;
@@ -458,6 +602,12 @@ class FastModel {
   std::atomic<unsigned> rr_{0};
   std::mutex mu_;
   std::vector<std::unique_ptr<FastContext>> ctxs_;
+  std::vector<FastIO> specs_in_, specs_out_;
+  bool batching_ = false;
+  int merge_cap_ = 1;
+  int64_t timeout_us_ = 2000;
+  std::mutex groups_mu_;
+  std::map<std::string, std::unique_ptr<BatchGroup>> groups_;
 };
 
 }  // namespace tfsc
@@ -489,6 +639,7 @@ void register_fastpath(py::module_& mod) {
       .def(py::init<std::string, int64_t, int>())
       .def("add_context", &FastModel::add_context)
       .def("has_bucket", &FastModel::has_bucket)
+      .def("enable_batching", &FastModel::enable_batching)
       .def("lock_ctx", &FastModel::lock_ctx,
            py::call_guard<py::gil_scoped_release>())
       .def("unlock_ctx", &FastModel::unlock_ctx)

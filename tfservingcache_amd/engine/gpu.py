@@ -900,6 +900,21 @@ class GpuModel:
             ctx.batch, self.streams_for(ctx.batch), ctx.exec_plan.ptr(),
             ctx.stream.cuda_stream, ins, outs)
 
+    def prewarm(self, batch: int) -> None:
+        """Build + capture + fast-register the context(s) for `batch`
+        with zero feeds (used when server-side batching needs the merged
+        bucket ready, and by loaders that want first-request latency
+        paid at load time)."""
+        plan = self.plan
+        feeds = {}
+        for _alias, idx in plan.sig_inputs.items():
+            shape = plan.resolve_shape(plan.tensors[idx].shape, batch)
+            dt = np.int32 if plan.tensors[idx].dtype == "i32" \
+                else np.float32
+            feeds[idx] = np.zeros(shape, dtype=dt)
+        fetch = list(plan.sig_outputs.values())
+        self.run(feeds, batch, fetch)
+
     def release(self) -> None:
         # mark released (new runs fail fast; the cache manager
         # re-fetches), then wait for each in-flight context before
