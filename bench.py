@@ -368,6 +368,7 @@ def main() -> int:
                 "mode": args.mode,
                 "requests_per_step": REQS_PER_STEP,
                 "batch_per_request": args.batch,
+                "threads": args.threads,
                 "images_per_sec": round(req_per_sec * args.batch, 1),
                 "n_models": n_models,
                 "pool_size": pool_cap,
