@@ -430,7 +430,7 @@ class FastModel {
     int64_t total = 0;
     for (size_t i = 0; i < n; ++i) total += batch[i]->rows;
 
-    FastContext* ctx = acquire(int(total));
+    FastContext* ctx = acquire(int(total), n > 1);
     std::lock_guard<std::mutex> g2(ctx->mu, std::adopt_lock);
     if (ctx->disabled) throw FastFallback("model released");
 
@@ -554,13 +554,15 @@ class FastModel {
   }
 
  private:
-  FastContext* acquire(int batch) {
+  FastContext* acquire(int batch, bool block_if_warming = false) {
     // smallest registered bucket >= batch; prefer an idle context.
     // While fewer than `target_` contexts exist for that bucket and all
     // are busy, raise FastFallback so the Python path runs instead —
     // its contention handling builds (and then registers) the
     // remaining multi-stream contexts. At steady state, block
-    // round-robin.
+    // round-robin. Merged batches pass block_if_warming: a fallback
+    // would explode the merge into per-request Python runs, and the
+    // merge bucket is prewarmed anyway.
     int bucket = -1;
     std::vector<FastContext*> cand;
     {
@@ -581,7 +583,7 @@ class FastModel {
       auto it = bucket_target_.find(bucket);
       if (it != bucket_target_.end()) want = it->second;
     }
-    if (int(cand.size()) < want)
+    if (!block_if_warming && int(cand.size()) < want)
       throw FastFallback("contexts warming");
     FastContext* c = cand[rr_++ % cand.size()];
     c->mu.lock();

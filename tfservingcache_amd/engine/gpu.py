@@ -901,19 +901,33 @@ class GpuModel:
             ctx.stream.cuda_stream, ins, outs)
 
     def prewarm(self, batch: int) -> None:
-        """Build + capture + fast-register the context(s) for `batch`
-        with zero feeds (used when server-side batching needs the merged
-        bucket ready, and by loaders that want first-request latency
-        paid at load time)."""
+        """Build + capture + fast-register ALL of the bucket's stream
+        contexts with zero feeds. Server-side batching needs this: the
+        merged bucket is only ever used by the C++ fast path, so the
+        Python path's contention-driven context building never reaches
+        it — every context must exist before merging starts."""
         plan = self.plan
+        b = self._bucket(batch)
         feeds = {}
         for _alias, idx in plan.sig_inputs.items():
-            shape = plan.resolve_shape(plan.tensors[idx].shape, batch)
+            shape = plan.resolve_shape(plan.tensors[idx].shape, b)
             dt = np.int32 if plan.tensors[idx].dtype == "i32" \
                 else np.float32
             feeds[idx] = np.zeros(shape, dtype=dt)
         fetch = list(plan.sig_outputs.values())
-        self.run(feeds, batch, fetch)
+        with self._lock:
+            if self._released:
+                raise ModelReleasedError(
+                    "model was evicted from the GPU pool")
+            ctxs = self._contexts.setdefault(b, [])
+            while len(ctxs) < self.streams_for(b):
+                ctxs.append(ExecContext(self, b))
+            todo = [c for c in ctxs if c.fast_id is None]
+        for ctx in todo:
+            with ctx.lock:
+                ctx.run(feeds, fetch)
+                if ctx.fast_id is None and ctx.captured:
+                    self._register_fast(ctx)
 
     def release(self) -> None:
         # mark released (new runs fail fast; the cache manager
