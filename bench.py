@@ -128,6 +128,8 @@ def main() -> int:
                     help="CPU engine (CI smoke only; not a benchmark)")
     ap.add_argument("--dyn-batch", action="store_true",
                     help="enable server-side dynamic batching")
+    ap.add_argument("--batch-timeout-ms", type=float, default=2.0,
+                    help="dynamic-batching merge window")
     ap.add_argument("--streams", type=int, default=6,
                     help="execution contexts (HIP streams) per model")
     ap.add_argument("--transport", choices=["inproc", "grpc"],
@@ -180,6 +182,8 @@ def main() -> int:
                                  max_batch=max(args.batch, 64)
                                  if args.dyn_batch else max(args.batch, 1),
                                  batching=args.dyn_batch,
+                                 batch_timeout_s=args.batch_timeout_ms
+                                 / 1e3,
                                  n_streams=args.streams)
     pool = ModelPool(loader, max_concurrent_models=pool_cap, device=device)
     cm = CacheManager(provider, cache, pool, model_fetch_timeout=300.0)

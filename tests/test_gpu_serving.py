@@ -94,6 +94,69 @@ def test_server_gpu_end_to_end(tmp_path):
         server.stop()
 
 
+def test_dynamic_batching_merges_and_matches(tmp_path):
+    """Concurrent batch-1 Predicts with server-side batching on: every
+    response matches the unbatched result AND the C++ fast path actually
+    merged (plan executions < requests served)."""
+    import threading
+    from tfservingcache_amd.cachemanager import (CacheManager, LRUCache,
+                                                 ModelPool, make_gpu_loader)
+    from tfservingcache_amd.cachemanager.providers import DiskModelProvider
+    from tfservingcache_amd.tfservingproxy import LocalServingHandler
+
+    repo = tmp_path / "repo3"
+    write_model_repo(str(repo), [("mlp", 1, "mlp")])
+    provider = DiskModelProvider(str(repo))
+    cache = LRUCache(str(tmp_path / "cache3"), 10 ** 9)
+    loader = make_gpu_loader(cache, device="cuda:0", max_batch=8,
+                             batching=True, batch_timeout_s=0.002,
+                             n_streams=2)
+    pool = ModelPool(loader, max_concurrent_models=2)
+    cm = CacheManager(provider, cache, pool, model_fetch_timeout=60.0)
+    handler = LocalServingHandler(cm)
+
+    rng = np.random.default_rng(1)
+    xs = [rng.standard_normal((1, 16)).astype(np.float32)
+          for _ in range(16)]
+    reqs = [m.PredictRequest(
+        model_spec=m.ModelSpec(name="mlp", version=m.Int64Value(value=1)),
+        inputs={"x": numpy_to_tensorproto(x)}).encode() for x in xs]
+    # serial pass (merging may or may not kick in) = expected values
+    expected = [tensorproto_to_numpy(m.PredictResponse.decode(
+        handler.predict_bytes(r)).outputs["probs"]) for r in reqs]
+
+    lm = pool.get_model("mlp", 1)
+    runs0, reqs0, _rows0 = lm._gpu._fast.stats()
+    errors, mismatches = [], []
+
+    def worker(i):
+        for it in range(25):
+            j = (i * 25 + it) % len(reqs)
+            try:
+                out = tensorproto_to_numpy(m.PredictResponse.decode(
+                    handler.predict_bytes(reqs[j])).outputs["probs"])
+                if not np.allclose(out, expected[j], rtol=1e-2,
+                                   atol=1e-3):
+                    mismatches.append((i, it))
+            except Exception as e:      # noqa: BLE001
+                errors.append((i, it, repr(e)))
+
+    threads = [threading.Thread(target=worker, args=(i,))
+               for i in range(12)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors, errors[:5]
+    assert not mismatches, mismatches[:5]
+    runs1, reqs1, rows1 = lm._gpu._fast.stats()
+    served = reqs1 - reqs0
+    assert served > 0
+    # 12 threads x 25 batch-1 requests against a 2 ms window: plan
+    # executions must be well below request count if merging works
+    assert runs1 - runs0 < served, (runs1 - runs0, served)
+
+
 def test_eviction_under_concurrent_load(tmp_path):
     """Hammer models from many threads while the pool evicts/reloads:
     no crashes, no wrong results, every request eventually served."""

@@ -278,6 +278,10 @@ class FastModel {
     }
   }
 
+  std::tuple<long long, long long, long long> stats() const {
+    return {n_runs_.load(), n_requests_.load(), n_rows_.load()};
+  }
+
   bool has_bucket(int batch) {
     std::lock_guard<std::mutex> g(mu_);
     for (auto& c : ctxs_)
@@ -429,6 +433,9 @@ class FastModel {
   void run_batch_inner(Entry** batch, size_t n) {
     int64_t total = 0;
     for (size_t i = 0; i < n; ++i) total += batch[i]->rows;
+    n_runs_.fetch_add(1, std::memory_order_relaxed);
+    n_requests_.fetch_add(static_cast<long long>(n), std::memory_order_relaxed);
+    n_rows_.fetch_add(total, std::memory_order_relaxed);
 
     FastContext* ctx = acquire(int(total), n > 1);
     std::lock_guard<std::mutex> g2(ctx->mu, std::adopt_lock);
@@ -602,6 +609,9 @@ class FastModel {
   int target_;
   std::map<int, int> bucket_target_;
   std::atomic<unsigned> rr_{0};
+  std::atomic<long long> n_runs_{0};       // plan executions
+  std::atomic<long long> n_requests_{0};   // requests served
+  std::atomic<long long> n_rows_{0};       // total rows executed
   std::mutex mu_;
   std::vector<std::unique_ptr<FastContext>> ctxs_;
   std::vector<FastIO> specs_in_, specs_out_;
@@ -616,11 +626,72 @@ class FastModel {
 
 namespace py = pybind11;
 
+// fast partial parse of PredictRequest.model_spec (field 1): the
+// Python routing layer calls this per request instead of its own
+// wire-level peek (tfservingcache_amd/wire/messages.py peek_model_spec)
+static py::tuple peek_spec(const uint8_t* p, size_t len) {
+  using tfsc::Reader;
+  std::string name, label;
+  int64_t version = 0;
+  bool has_version = false;
+  Reader r{p, p + len};
+  while (r.p < r.end) {
+    uint64_t tag = r.varint();
+    if ((tag >> 3) == 1 && (tag & 7) == 2) {
+      uint64_t n = r.varint();
+      Reader rs{r.p, r.p + n};
+      r.p += n;
+      while (rs.p < rs.end) {
+        uint64_t stag = rs.varint();
+        int fno = int(stag >> 3), wt = int(stag & 7);
+        if (fno == 1 && wt == 2) {
+          uint64_t kn = rs.varint();
+          name.assign(reinterpret_cast<const char*>(rs.p), kn);
+          rs.p += kn;
+        } else if (fno == 2 && wt == 2) {          // Int64Value version
+          uint64_t vn = rs.varint();
+          Reader rv{rs.p, rs.p + vn};
+          rs.p += vn;
+          while (rv.p < rv.end) {
+            uint64_t vtag = rv.varint();
+            if ((vtag >> 3) == 1 && (vtag & 7) == 0) {
+              version = int64_t(rv.varint());
+              has_version = true;
+            } else {
+              rv.skip(int(vtag & 7));
+            }
+          }
+        } else if (fno == 4 && wt == 2) {
+          uint64_t kn = rs.varint();
+          label.assign(reinterpret_cast<const char*>(rs.p), kn);
+          rs.p += kn;
+        } else {
+          rs.skip(wt);
+        }
+      }
+      break;
+    }
+    r.skip(int(tag & 7));
+  }
+  return py::make_tuple(py::str(name),
+                        has_version ? py::cast(version) : py::none(),
+                        py::str(label));
+}
+
 void register_fastpath(py::module_& mod) {
   using tfsc::FastIO;
   using tfsc::FastModel;
 
   py::register_exception<tfsc::FastFallback>(mod, "FastFallback");
+
+  mod.def("peek_spec", [](py::bytes data) {
+    char* buf = nullptr;
+    Py_ssize_t blen = 0;
+    if (PyBytes_AsStringAndSize(data.ptr(), &buf, &blen) != 0)
+      throw py::error_already_set();
+    return peek_spec(reinterpret_cast<const uint8_t*>(buf),
+                     size_t(blen));
+  });
 
   py::class_<FastIO>(mod, "FastIO")
       .def(py::init([](std::string alias, bool is_int, uintptr_t pin,
@@ -641,6 +712,7 @@ void register_fastpath(py::module_& mod) {
       .def(py::init<std::string, int64_t, int>())
       .def("add_context", &FastModel::add_context)
       .def("has_bucket", &FastModel::has_bucket)
+      .def("stats", &FastModel::stats)
       .def("enable_batching", &FastModel::enable_batching)
       .def("lock_ctx", &FastModel::lock_ctx,
            py::call_guard<py::gil_scoped_release>())

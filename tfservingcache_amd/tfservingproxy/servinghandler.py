@@ -329,14 +329,30 @@ def _fast_fallback_cls():
         return None
 
 
+def _peek_fn():
+    """C++ model_spec peek (~0.4 us vs ~11 us for the Python wire
+    peek) when the engine extension is importable."""
+    try:
+        from ..engine import _tfsc_engine as ext
+        return ext.peek_spec
+    except Exception:       # noqa: BLE001
+        return None
+
+
 def _predict_bytes(self, data: bytes) -> bytes:
     """Raw-request Predict: C++ end-to-end fast path when the model's
     engine has a registered fast context for the request's batch bucket;
     Python decode/execute/encode otherwise (also the warm-up path that
     builds and registers the context)."""
-    spec = m.peek_model_spec(data)
-    model, _version = self.get_model(spec.name, spec.version_value(),
-                                     spec.version_label)
+    peek = _peek_fn()
+    if peek is not None:
+        name, ver, label = peek(data)
+        ver = int(ver) if ver else 0
+    else:
+        spec = m.peek_model_spec(data)
+        name, ver, label = (spec.name, spec.version_value(),
+                            spec.version_label)
+    model, _version = self.get_model(name, ver, label)
     gpu = getattr(model, "_gpu", None)
     if gpu is not None and model._batcher is None:
         fb = _fast_fallback_cls()
@@ -353,8 +369,7 @@ def _predict_bytes(self, data: bytes) -> bytes:
                 from ..engine.gpu import ModelReleasedError
                 if not isinstance(e, ModelReleasedError):
                     raise
-                model, _version = self.get_model(spec.name,
-                                                 spec.version_value())
+                model, _version = self.get_model(name, ver)
                 gpu = getattr(model, "_gpu", None)
                 if gpu is not None:
                     try:
