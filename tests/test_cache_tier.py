@@ -355,3 +355,35 @@ def test_warmup_requests_executed(tmp_path):
     assert len(calls) == 2
     # roundtrip of the tfrecord reader
     assert len(list(wu.read_tfrecords(str(vdir / wu.WARMUP_PATH)))) == 2
+
+
+def test_plan_cache_shares_hardlinked_models(tmp_path):
+    """Hardlinked copies of one SavedModel compile once (inode-keyed
+    plan cache); a rewritten file gets a fresh plan."""
+    import os
+    import shutil
+    from tfservingcache_amd.engine.model import load_model_from_dir
+    from tfservingcache_amd.models import write_model_repo
+
+    write_model_repo(str(tmp_path), [("a", 1, "half_plus_two")])
+    src = tmp_path / "a" / "1"
+    dst = tmp_path / "b" / "1"
+    os.makedirs(dst.parent, exist_ok=True)
+    shutil.copytree(src, dst, copy_function=os.link)
+
+    lm_a = load_model_from_dir(str(src), "a", 1)
+    lm_b = load_model_from_dir(str(dst), "b", 1)
+    assert lm_a.plan is lm_b.plan       # same inode -> shared plan
+
+    # distinct content (fresh copy, new inode) -> fresh compile
+    dst2 = tmp_path / "c" / "1"
+    os.makedirs(dst2.parent, exist_ok=True)
+    shutil.copytree(src, dst2)
+    lm_c = load_model_from_dir(str(dst2), "c", 1)
+    assert lm_c.plan is not lm_a.plan
+    # both still serve correctly
+    import numpy as np
+    out_a = lm_a.predict({"x": np.array([2.0], np.float32)})
+    out_c = lm_c.predict({"x": np.array([2.0], np.float32)})
+    np.testing.assert_allclose(list(out_a.values())[0],
+                               list(out_c.values())[0])

@@ -5,7 +5,9 @@ Mirrors the reference's DiskModelProvider behavior
   * version directories are matched NUMERICALLY, so "000000042" serves
     version 42 (findSrcPathForModel, diskmodelprovider.go:46-69);
   * load_model recursively copies baseDir/<model>/<version-dir>/ into the
-    cache dir;
+    cache dir — by HARDLINK when the cache shares a filesystem with the
+    repo (eviction unlinks, refcounts keep the repo intact), byte copy
+    otherwise;
   * model_size is the RECURSIVE content size (the reference stat'ed the
     directory inode — a known bug, SURVEY.md §2.3 — fixed here).
 """
@@ -16,6 +18,13 @@ import shutil
 
 from ..lrucache import Model, dir_size
 from ..modelprovider import ModelNotFoundError, ModelProvider
+
+
+def _link_or_copy(src: str, dst: str) -> None:
+    try:
+        os.link(src, dst)
+    except OSError:                    # cross-device / FS without links
+        shutil.copy2(src, dst)
 
 
 class DiskModelProvider(ModelProvider):
@@ -44,7 +53,7 @@ class DiskModelProvider(ModelProvider):
         dst = os.path.join(dest_base_dir, rel)
         if os.path.exists(dst):
             shutil.rmtree(dst)
-        shutil.copytree(src, dst)
+        shutil.copytree(src, dst, copy_function=_link_or_copy)
         return Model(name=model_name, version=version, path=rel,
                      size_on_disk=dir_size(dst))
 

@@ -7,6 +7,8 @@ facade the cache tier talks to.
 """
 from __future__ import annotations
 
+import collections
+import os
 import threading
 from typing import Dict, Optional
 
@@ -163,9 +165,34 @@ class LoadedModel:
         return {name: vals[idx] for name, idx in zip(fetch_names, fetch)}
 
 
-def load_model_from_dir(version_dir: str, name: str, version: int,
-                        signature_name: str = g.DEFAULT_SERVING_SIGNATURE
-                        ) -> LoadedModel:
+# content-identity plan cache: a fleet serving many copies/versions of
+# one architecture (hardlinked model repos, blue/green rollouts of the
+# same graph) should not recompile the identical SavedModel per name.
+# Keyed by (dev, inode, size, mtime_ns) of saved_model.pb — with the
+# disk provider's hardlink fetch a cache copy shares the repo's inode,
+# so hits skip reading the file entirely. Plans are immutable after
+# compile (executors only read them), so sharing is safe.
+_PLAN_CACHE_CAP = 32
+_plan_cache_lock = threading.Lock()
+_plan_cache: "collections.OrderedDict[tuple, Plan]" = \
+    collections.OrderedDict()
+
+
+def _compile_cached(version_dir: str, signature_name: str) -> Plan:
+    from .savedmodel import SAVED_MODEL_FILENAME
+    key = None
+    try:
+        st = os.stat(os.path.join(version_dir, SAVED_MODEL_FILENAME))
+        key = (st.st_dev, st.st_ino, st.st_size, st.st_mtime_ns,
+               signature_name)
+    except OSError:
+        pass
+    if key is not None:
+        with _plan_cache_lock:
+            plan = _plan_cache.get(key)
+            if plan is not None:
+                _plan_cache.move_to_end(key)
+                return plan
     graph_def, signatures = read_saved_model(version_dir)
     sig = signatures.get(signature_name)
     if sig is None and signatures:
@@ -173,4 +200,16 @@ def load_model_from_dir(version_dir: str, name: str, version: int,
     if sig is None:
         raise ModelExecError(f"no signatures in {version_dir}")
     plan = compile_graph(graph_def, sig)
+    if key is not None:
+        with _plan_cache_lock:
+            _plan_cache[key] = plan
+            while len(_plan_cache) > _PLAN_CACHE_CAP:
+                _plan_cache.popitem(last=False)
+    return plan
+
+
+def load_model_from_dir(version_dir: str, name: str, version: int,
+                        signature_name: str = g.DEFAULT_SERVING_SIGNATURE
+                        ) -> LoadedModel:
+    plan = _compile_cached(version_dir, signature_name)
     return LoadedModel(name, version, plan)
