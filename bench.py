@@ -375,6 +375,8 @@ def main() -> int:
                 "threads": args.threads,
                 "images_per_sec": round(req_per_sec * args.batch, 1),
                 "n_models": n_models,
+                "model_repo": "hardlinked copies of one SavedModel "
+                              "per family (content-dedup applies)",
                 "pool_size": pool_cap,
                 "cold_load_p50_ms": round(cold_p50, 1),
                 "latency_ms": {
