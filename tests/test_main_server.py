@@ -154,3 +154,49 @@ def test_config_env_override(tmp_path, monkeypatch):
     assert cfg.get_int("modelCache.size") == 12345
     assert cfg.get_string("healthProbe.modelName") == \
         "__TFSERVINGCACHE_PROBE_CHECK__"
+
+
+def test_warm_handoff_on_ring_change(tmp_path):
+    """When membership shifts a cached model's ownership away from this
+    node, the old owner nudges the new owner (GetModelMetadata over the
+    cache gRPC surface) so it loads warm instead of missing cold."""
+    from tfservingcache_amd.taskhandler.discovery.base import ServingService
+
+    def mock_cfg(name):
+        cfg = make_cfg(tmp_path, name=name, with_discovery=False)
+        values = dict(cfg._values)      # noqa: SLF001
+        values["serviceDiscovery"] = {"type": "mock"}
+        return Config(values)
+
+    a = Server(mock_cfg("hand_a"))
+    b = Server(mock_cfg("hand_b"))
+    a.start()
+    b.start()
+    try:
+        wait_http(a.cache_rest_port)
+        wait_http(b.cache_rest_port)
+        a_svc = a._self_service         # noqa: SLF001
+        b_svc = ServingService("127.0.0.1", b.cache_rest_port,
+                               b.cache_grpc_port)
+        # seed: A alone owns everything
+        a.discovery.push([a_svc])
+        # load the model on A
+        r = requests.post(
+            f"http://127.0.0.1:{a.cache_rest_port}"
+            "/v1/models/half_plus_two:predict",
+            json={"instances": [1.0]}, timeout=30)
+        assert r.status_code == 200, r.text
+        assert a.cm.pool.get_model("half_plus_two", 1) is not None
+        assert b.cm.pool.get_model("half_plus_two", 1) is None
+
+        # membership flips: only B in the ring -> A hands the model off
+        a.discovery.push([b_svc])
+        deadline = time.time() + 15
+        while time.time() < deadline:
+            if b.cm.pool.get_model("half_plus_two", 1) is not None:
+                break
+            time.sleep(0.1)
+        assert b.cm.pool.get_model("half_plus_two", 1) is not None
+    finally:
+        a.stop()
+        b.stop()

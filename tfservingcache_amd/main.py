@@ -72,6 +72,9 @@ def create_discovery_service(cfg: Config, health_check) -> Optional[DiscoverySer
     ttl = cfg.get_float("serviceDiscovery.heartbeatTTL") or 5.0
     if not dtype:
         return None
+    if dtype == "mock":
+        from .taskhandler.discovery.base import MockDiscovery
+        return MockDiscovery()
     if dtype == "static":
         return StaticDiscovery(
             [str(m) for m in cfg.get_list("serviceDiscovery.static.members")])
@@ -173,6 +176,9 @@ class Server:
         self.discovery: Optional[DiscoveryService] = None
         self._stop = threading.Event()
         self._grpc_servers = []
+        self._self_service = None
+        self._last_member_ids = None
+        self._handoff_fwd = None
         self._runners = []
         self._loop = None
 
@@ -210,8 +216,18 @@ class Server:
             self.cluster = ClusterConnection(self.discovery, replicas)
             host = self.cfg.get_string("proxy.advertiseHost") or \
                 socket.gethostbyname(socket.gethostname())
-            self.cluster.connect(ServingService(
-                host, self.cache_rest_port, self.cache_grpc_port))
+            self._self_service = ServingService(
+                host, self.cache_rest_port, self.cache_grpc_port)
+            self.cluster.connect(self._self_service)
+            # warm handoff on ring changes: when membership shifts and a
+            # locally-cached model's ownership moves away from this node,
+            # nudge a new owner to load it NOW (a GetModelMetadata call —
+            # existing wire surface, triggers its ensure_loaded) instead
+            # of serving the first request cold. The reference relies on
+            # the natural cache miss. Disable: proxy.warmHandoff: false.
+            raw = (self.cfg.get_string("proxy.warmHandoff") or "").strip()
+            if raw.lower() not in ("false", "0", "no"):
+                self.discovery.add_listener(self._on_membership_change)
 
             def pick_rest(model, version):
                 return self.cluster.node_for_key(model, version).rest_addr
@@ -265,6 +281,46 @@ class Server:
                 self._runners.append(runner)
         self._loop.run_until_complete(boot())
         self._loop.run_forever()
+
+    # -- warm handoff ------------------------------------------------------
+    def _on_membership_change(self, members) -> None:
+        ids = sorted(mm.serialize() for mm in members)
+        prev, self._last_member_ids = self._last_member_ids, ids
+        if prev is None or prev == ids or self._stop.is_set():
+            return                         # initial seed / no-op update
+        threading.Thread(target=self._warm_handoff, daemon=True).start()
+
+    def _warm_handoff(self) -> None:
+        from .taskhandler.cluster import model_key
+        from .wire import messages as wm
+        if self._handoff_fwd is None:
+            from .tfservingproxy import GrpcForwarder
+            self._handoff_fwd = GrpcForwarder(
+                timeout_s=self.cfg.get_float("proxy.grpcTimeout") or 10.0)
+        self_id = self._self_service.serialize()
+        for entry in self.cm.list_cached():
+            if self._stop.is_set():
+                return
+            try:
+                owners = self.cluster.find_nodes_for_key(
+                    model_key(entry.name, entry.version))
+                if not owners or self_id in [o.serialize()
+                                             for o in owners]:
+                    continue
+                req = wm.GetModelMetadataRequest(
+                    model_spec=wm.ModelSpec(
+                        name=entry.name,
+                        version=wm.Int64Value(value=entry.version)),
+                    metadata_field=["signature_def"])
+                self._handoff_fwd.call(
+                    owners[0].grpc_addr,
+                    f"/{wm.PREDICTION_SERVICE}/GetModelMetadata",
+                    req.encode())
+                log.info("warm handoff: %s:%s -> %s", entry.name,
+                         entry.version, owners[0].grpc_addr)
+            except Exception:       # noqa: BLE001
+                log.warning("warm handoff failed for %s:%s",
+                            entry.name, entry.version, exc_info=True)
 
     def _health_loop(self) -> None:
         while not self._stop.wait(30.0):
