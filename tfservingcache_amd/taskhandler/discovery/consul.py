@@ -6,8 +6,11 @@ Behavior mirrors the reference's consul integration
     DeregisterCriticalServiceAfter = 100*TTL (consul.go:49-68);
   * heartbeat the TTL check every TTL/2, pass/fail from the health
     callback (consul.go:138-160);
-  * poll /v1/health/service/<name>?passing every 5s and push full member
-    lists (consul.go:70-117);
+  * watch /v1/health/service/<name>?passing with BLOCKING QUERIES
+    (index + wait=55s long poll): membership changes propagate the
+    moment consul sees them, instead of on the reference's 5s poll tick
+    (consul.go:70-117). Falls back to plain polling on errors or
+    non-indexed responses;
   * ports are encoded as tags "rest:<p>" / "grpc:<p>" (consul.go:54-57)
     plus "slot:<gpuN>" for per-GPU ring slots.
 """
@@ -99,21 +102,43 @@ class ConsulDiscovery(DiscoveryService):
 
     def _poll_loop(self) -> None:
         last = None
+        index = None
         while not self._stop.is_set():
             try:
-                members = self.fetch_members()
+                members, index = self.fetch_members(index)
                 if members != last:
                     last = members
                     self._notify(members)
+                if index is None:       # server doesn't support blocking
+                    self._stop.wait(self.poll_interval)
             except requests.RequestException:
-                log.warning("consul health poll failed", exc_info=True)
-            self._stop.wait(self.poll_interval)
+                log.warning("consul health watch failed", exc_info=True)
+                index = None
+                self._stop.wait(self.poll_interval)
 
-    def fetch_members(self) -> List[ServingService]:
+    def fetch_members(self, index: Optional[int] = None
+                      ) -> "tuple[List[ServingService], Optional[int]]":
+        """One health query. With `index`, a blocking query: consul
+        holds the request (wait=55s) until the service list changes past
+        that index, so the loop is change-driven."""
+        params = {"passing": "true"}
+        timeout = 10.0
+        if index is not None:
+            params["index"] = str(index)
+            params["wait"] = "55s"
+            timeout = 70.0
         r = self._session.get(
             f"{self.base}/v1/health/service/{self.name}",
-            params={"passing": "true"}, timeout=10)
+            params=params, timeout=timeout)
         r.raise_for_status()
+        new_index = None
+        try:
+            new_index = int(r.headers.get("X-Consul-Index", ""))
+            # consul docs: reset on non-monotonic or absurd indexes
+            if new_index < 1 or (index is not None and new_index < index):
+                new_index = None
+        except ValueError:
+            pass
         members = []
         for entry in r.json():
             svc = entry.get("Service", {})
@@ -128,4 +153,4 @@ class ConsulDiscovery(DiscoveryService):
                     slot = tag[5:]
             if host:
                 members.append(ServingService(host, rest, grpc_, slot))
-        return sorted(members, key=lambda s: s.serialize())
+        return sorted(members, key=lambda s: s.serialize()), new_index

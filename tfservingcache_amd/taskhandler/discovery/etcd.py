@@ -6,13 +6,15 @@ Behavior mirrors the reference's etcd integration
     (etcd.go:53, cluster.go:142-144);
   * every TTL/2: grant a fresh lease of TTL seconds and Put the key
     with it, so the key expires if the node dies (etcd.go:134-148);
-  * membership from the key prefix; the reference watches the prefix,
-    here the prefix is polled every TTL/2 (same freshness bound as the
-    lease cadence);
+  * membership from the key prefix, WATCHED (etcdserverpb.Watch bidi
+    stream over the prefix, like the reference's clientv3 watch,
+    etcd.go:150-170): changes push immediately; falls back to TTL/2
+    polling while the watch stream is unavailable;
   * optional username/password auth (etcd.go:29-47).
 
 No protoc in this environment, so the etcdserverpb messages used
-(LeaseGrant / Put / Range / Authenticate) are declared on the wire codec.
+(LeaseGrant / Put / Range / Watch / Authenticate) are declared on the
+wire codec.
 """
 from __future__ import annotations
 
@@ -60,6 +62,28 @@ class KeyValue(Message):
 class RangeResponse(Message):
     FIELDS = [("kvs", 2, "message", dict(msg_cls=KeyValue, repeated=True)),
               ("more", 3, "bool"), ("count", 4, "int64")]
+
+
+class WatchCreateRequest(Message):
+    FIELDS = [("key", 1, "bytes"), ("range_end", 2, "bytes"),
+              ("start_revision", 3, "int64")]
+
+
+class WatchRequest(Message):
+    FIELDS = [("create_request", 1, "message",
+               dict(msg_cls=WatchCreateRequest))]
+
+
+class Event(Message):
+    FIELDS = [("type", 1, "int64"),
+              ("kv", 2, "message", dict(msg_cls=KeyValue))]
+
+
+class WatchResponse(Message):
+    FIELDS = [("watch_id", 2, "int64"), ("created", 3, "bool"),
+              ("canceled", 4, "bool"),
+              ("events", 11, "message", dict(msg_cls=Event,
+                                             repeated=True))]
 
 
 class AuthenticateRequest(Message):
@@ -172,13 +196,50 @@ class EtcdDiscovery(DiscoveryService):
         return sorted(members, key=lambda s: s.serialize())
 
     def _poll_loop(self) -> None:
+        """Watch-driven: open a Watch stream on the prefix; every
+        created/event response triggers a Range re-fetch + notify. While
+        the stream can't be established, degrade to TTL/2 polling."""
         last = None
+
+        def refresh():
+            nonlocal last
+            members = self.fetch_members()
+            if members != last:
+                last = members
+                self._notify(members)
+
+        prefix = f"/service/{self.name}/".encode()
         while not self._stop.is_set():
             try:
-                members = self.fetch_members()
-                if members != last:
-                    last = members
-                    self._notify(members)
+                refresh()
             except grpc.RpcError:
-                log.warning("etcd member poll failed", exc_info=True)
-            self._stop.wait(self.ttl / 2)
+                log.warning("etcd member fetch failed", exc_info=True)
+                self._stop.wait(self.ttl / 2)
+                continue
+            try:
+                stream = self._chan().stream_stream(
+                    "/etcdserverpb.Watch/Watch",
+                    request_serializer=lambda r: r.encode(),
+                    response_deserializer=WatchResponse.decode)
+                metadata = [("token", self._token)] if self._token \
+                    else None
+
+                def requests_iter():
+                    yield WatchRequest(create_request=WatchCreateRequest(
+                        key=prefix,
+                        range_end=_prefix_range_end(prefix)))
+                    self._stop.wait()          # hold the send side open
+
+                for resp in stream(requests_iter(), metadata=metadata):
+                    if self._stop.is_set():
+                        return
+                    if resp.canceled:
+                        break
+                    if resp.events or resp.created:
+                        refresh()
+            except grpc.RpcError:
+                if self._stop.is_set():
+                    return
+                log.warning("etcd watch unavailable; polling",
+                            exc_info=True)
+                self._stop.wait(self.ttl / 2)
