@@ -180,11 +180,15 @@ _plan_cache: "collections.OrderedDict[tuple, Plan]" = \
 
 def _compile_cached(version_dir: str, signature_name: str) -> Plan:
     from .savedmodel import SAVED_MODEL_FILENAME
+    var_prefix = os.path.join(version_dir, "variables", "variables")
     key = None
     try:
         st = os.stat(os.path.join(version_dir, SAVED_MODEL_FILENAME))
         key = (st.st_dev, st.st_ino, st.st_size, st.st_mtime_ns,
                signature_name)
+        if os.path.exists(var_prefix + ".index"):
+            sv = os.stat(var_prefix + ".index")
+            key += (sv.st_dev, sv.st_ino, sv.st_size, sv.st_mtime_ns)
     except OSError:
         pass
     if key is not None:
@@ -199,7 +203,11 @@ def _compile_cached(version_dir: str, signature_name: str) -> Plan:
         sig = next(iter(signatures.values()))
     if sig is None:
         raise ModelExecError(f"no signatures in {version_dir}")
-    plan = compile_graph(graph_def, sig)
+    variables = None
+    if os.path.exists(var_prefix + ".index"):
+        from .tensor_bundle import read_bundle
+        variables = read_bundle(var_prefix)
+    plan = compile_graph(graph_def, sig, variables)
     if key is not None:
         with _plan_cache_lock:
             _plan_cache[key] = plan

@@ -152,13 +152,33 @@ class _Lowerer:
     """One-pass topological lowering with peephole fusion."""
 
     def __init__(self, graph_def: g.GraphDef,
-                 signature: m.SignatureDef):
+                 signature: m.SignatureDef,
+                 variables: Optional[Dict[str, np.ndarray]] = None):
         self.nodes: Dict[str, g.NodeDef] = {}
         self.order: List[g.NodeDef] = []
         for nd in graph_def.node:
             self.nodes[nd.name] = nd
             self.order.append(nd)
         self.signature = signature
+        self.variables = variables or {}
+        if self.variables:
+            # non-frozen SavedModels carry a save/restore subgraph
+            # (SaveV2/RestoreV2/Assign/string Consts) that never executes
+            # at serving time — lower only ancestors of the signature
+            # outputs (+ inputs), like TF Serving's session pruning
+            keep = set()
+            stack = [_node_of(ti.name)
+                     for ti in signature.outputs.values()]
+            stack += [_node_of(ti.name)
+                      for ti in signature.inputs.values()]
+            while stack:
+                name = stack.pop()
+                if name in keep or name not in self.nodes:
+                    continue
+                keep.add(name)
+                for ref in self.nodes[name].input:
+                    stack.append(_node_of(ref.lstrip("^")))
+            self.order = [nd for nd in self.order if nd.name in keep]
         self.consumers: Dict[str, List[g.NodeDef]] = {}
         for nd in self.order:
             for ref in nd.input:
@@ -204,12 +224,19 @@ class _Lowerer:
         # Const nodes are lowered); follow Identity chains
         nd = self.nodes.get(node)
         seen = 0
-        while nd is not None and nd.op in ("Identity",) and seen < 8:
+        while nd is not None and nd.op in ("Identity", "ReadVariableOp") \
+                and seen < 8:
             nd = self.nodes.get(_node_of(nd.input[0]))
             seen += 1
         if nd is not None and nd.op == "Const":
             arr = tensorproto_to_numpy(nd.attr["value"].tensor)
             self.consts[node] = arr
+            return arr
+        if nd is not None and nd.op in ("VariableV2", "Variable",
+                                        "VarHandleOp"):
+            arr = self.variables.get(nd.name)
+            if arr is not None:
+                self.consts[node] = arr
             return arr
         return None
 
@@ -289,6 +316,28 @@ class _Lowerer:
                 arr = arr.astype(np.float32)
             self.new_tensor(tuple(arr.shape), dtype, "weight", out,
                             weight=np.asarray(arr))
+            return
+        if op in ("VariableV2", "Variable", "VarHandleOp"):
+            arr = self.variables.get(nd.name)
+            if arr is None:
+                raise PlanError(
+                    f"variable {nd.name} not found in the checkpoint "
+                    f"bundle ({len(self.variables)} tensors loaded)")
+            if arr.dtype == np.float64:
+                arr = arr.astype(np.float32)
+            self.consts[nd.name] = arr
+            dtype = "i32" if arr.dtype in (np.int32, np.int64) else "f32"
+            self.new_tensor(tuple(arr.shape), dtype, "weight", out,
+                            weight=np.asarray(arr))
+            return
+        if op in ("ReadVariableOp",):
+            src = self.tid(nd.input[0])
+            root = self.tensors[src]
+            self.new_tensor(root.shape, root.dtype, root.kind, out,
+                            weight=root.weight, alias_of=src)
+            cv = self.const_value(nd.input[0])
+            if cv is not None:
+                self.consts[nd.name] = cv
             return
         if op in ("Identity", "StopGradient", "PreventGradient", "Snapshot"):
             src = self.tid(nd.input[0])
@@ -968,5 +1017,11 @@ def fuse_attention(plan: Plan) -> Plan:
 
 
 def compile_graph(graph_def: g.GraphDef,
-                  signature: m.SignatureDef) -> Plan:
-    return fuse_attention(_Lowerer(graph_def, signature).run())
+                  signature: m.SignatureDef,
+                  variables: Optional[Dict[str, np.ndarray]] = None
+                  ) -> Plan:
+    """variables: checkpoint tensors (engine/tensor_bundle.py) for
+    non-frozen SavedModels — VariableV2/VarHandleOp nodes resolve to
+    these by node name, exactly as TF Serving's restore would."""
+    return fuse_attention(
+        _Lowerer(graph_def, signature, variables).run())

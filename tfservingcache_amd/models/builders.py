@@ -56,6 +56,44 @@ def build_mlp(d_in=16, d_hidden=32, d_out=8, seed=0):
 
 
 # ---------------------------------------------------------------------------
+# same MLP, NON-frozen (TF1 Saver style): weights are VariableV2 nodes
+# resolved from a variables/ tensor_bundle checkpoint at load, plus a
+# save/restore subgraph the planner must prune (engine/tensor_bundle.py)
+# ---------------------------------------------------------------------------
+def build_mlp_vars(d_in=16, d_hidden=32, d_out=8, seed=0):
+    rng = np.random.default_rng(seed)
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x = gb.placeholder("x", np.float32, [-1, d_in], signature_name="x")
+    weights = {
+        "w1": rng.standard_normal((d_in, d_hidden),
+                                  dtype=np.float32) * 0.3,
+        "b1": rng.standard_normal(d_hidden, dtype=np.float32) * 0.1,
+        "w2": rng.standard_normal((d_hidden, d_out),
+                                  dtype=np.float32) * 0.3,
+        "b2": rng.standard_normal(d_out, dtype=np.float32) * 0.1,
+    }
+    reads = {}
+    for name, arr in weights.items():
+        gb.node("VariableV2", name, [], dtype=f32,
+                shape=gb.a_shape(arr.shape))
+        reads[name] = gb.node("Identity", f"{name}/read", [f"{name}:0"],
+                              T=f32)
+        # restore subgraph (never reachable from the outputs): the
+        # planner prunes it, like TF Serving's session does
+        gb.node("Assign", f"{name}/Assign",
+                [f"{name}:0", f"{name}/read:0"], T=f32)
+    mm1 = gb.node("MatMul", "mm1", [x, reads["w1"]], T=f32)
+    ba1 = gb.node("BiasAdd", "ba1", [mm1, reads["b1"]], T=f32)
+    r1 = gb.node("Relu", "relu1", [ba1], T=f32)
+    mm2 = gb.node("MatMul", "mm2", [r1, reads["w2"]], T=f32)
+    ba2 = gb.node("BiasAdd", "ba2", [mm2, reads["b2"]], T=f32)
+    sm = gb.node("Softmax", "probs", [ba2], T=f32)
+    gb.mark_output("probs", sm)
+    return gb.build(), weights
+
+
+# ---------------------------------------------------------------------------
 # ResNet-50 v1.5 (NHWC, inference): conv stem -> 4 stages of bottlenecks ->
 # global mean -> fc -> softmax. BN emitted as FusedBatchNormV3 (inference),
 # which the planner folds into the conv weights.
@@ -274,6 +312,7 @@ def build_bert(seq_len=128, hidden=768, layers=12, heads=12,
 _BUILDERS = {
     "half_plus_two": build_half_plus_two,
     "mlp": build_mlp,
+    "mlp_vars": build_mlp_vars,
     "resnet50": build_resnet50,
     "bert_base": build_bert,
 }
@@ -284,5 +323,13 @@ def write_model_repo(base_dir: str, models: Sequence[tuple],
     """models: list of (model_name, version, builder_name)."""
     for name, version, builder in models:
         kw = (builder_kwargs or {}).get(name, {})
-        sm = _BUILDERS[builder](**kw)
-        write_saved_model(sm, os.path.join(base_dir, name, str(version)))
+        built = _BUILDERS[builder](**kw)
+        vdir = os.path.join(base_dir, name, str(version))
+        if isinstance(built, tuple):         # (SavedModel, variables)
+            sm, variables = built
+            write_saved_model(sm, vdir)
+            from ..engine.tensor_bundle import write_bundle
+            write_bundle(os.path.join(vdir, "variables", "variables"),
+                         variables)
+        else:
+            write_saved_model(built, vdir)

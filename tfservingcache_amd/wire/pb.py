@@ -129,6 +129,9 @@ def _encode_scalar(buf: bytearray, kind: str, number: int, value: Any) -> None:
     elif kind == "float":
         write_tag(buf, number, WIRE_FIXED32)
         buf += _f32.pack(value)
+    elif kind == "fixed32":
+        write_tag(buf, number, WIRE_FIXED32)
+        buf += _u32.pack(int(value) & 0xFFFFFFFF)
     elif kind == "double":
         write_tag(buf, number, WIRE_FIXED64)
         buf += _f64.pack(value)
