@@ -132,10 +132,12 @@ def main() -> int:
                     help="dynamic-batching merge window")
     ap.add_argument("--streams", type=int, default=6,
                     help="execution contexts (HIP streams) per model")
-    ap.add_argument("--transport", choices=["inproc", "grpc"],
+    ap.add_argument("--transport", choices=["inproc", "grpc", "native"],
                     default="inproc",
                     help="inproc: gRPC message path without sockets; "
-                         "grpc: real gRPC server+client over loopback")
+                         "grpc: Python grpcio server over TCP loopback; "
+                         "native: C++ nghttp2 gRPC front-end over TCP "
+                         "loopback (registered Predicts bypass Python)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -274,14 +276,33 @@ def main() -> int:
         return run
 
     grpc_server = None
-    if args.transport == "grpc":
+    if args.transport in ("grpc", "native"):
         import grpc as grpc_mod
-        from tfservingcache_amd.tfservingproxy import make_cache_grpc_server
-        grpc_server, _health = make_cache_grpc_server(
-            handler, max_workers=max(args.threads * 2, 8),
-            max_msg=256 * 1024 * 1024)
-        gport = grpc_server.add_insecure_port("127.0.0.1:0")
-        grpc_server.start()
+        if args.transport == "native":
+            from tfservingcache_amd.tfservingproxy.native_frontend import \
+                NativeGrpcServer
+            grpc_server = NativeGrpcServer(
+                handler, workers=max(args.threads, 8))
+            grpc_server.add_insecure_port("127.0.0.1:0")
+            grpc_server.start()
+            gport = grpc_server.port
+            # pool -> registry so warmed Predicts run fully in C++
+            def _on_avail(name, version, model):
+                fast = getattr(getattr(model, "_gpu", None), "_fast",
+                               None)
+                if fast is not None:
+                    grpc_server.register_model(name, version, fast)
+            pool.on_available = _on_avail
+            pool.on_unload = (lambda name, version, model:
+                              grpc_server.unregister_model(name, version))
+        else:
+            from tfservingcache_amd.tfservingproxy import \
+                make_cache_grpc_server
+            grpc_server, _health = make_cache_grpc_server(
+                handler, max_workers=max(args.threads * 2, 8),
+                max_msg=256 * 1024 * 1024)
+            gport = grpc_server.add_insecure_port("127.0.0.1:0")
+            grpc_server.start()
         channel = grpc_mod.insecure_channel(
             f"127.0.0.1:{gport}",
             options=[("grpc.max_receive_message_length", 256 * 1024 * 1024),

@@ -24,7 +24,13 @@ sources = [
     os.path.join(CSRC, "ops", "conv.hip"),
     os.path.join(CSRC, "ops", "attention.hip"),
     os.path.join(CSRC, "fastpath.cpp"),
+    os.path.join(CSRC, "frontend.cpp"),
 ]
+
+# nghttp2 (HTTP/2 framing/HPACK for the native gRPC front-end) ships in
+# this image's conda tree; same path exists on the GPU boxes (same image)
+NGHTTP2_INC = "/opt/conda/include"
+NGHTTP2_LIB = "/opt/conda/lib"
 
 setup(
     name="tfsc_engine",
@@ -33,9 +39,12 @@ setup(
             name="tfservingcache_amd.engine._tfsc_engine",
             sources=sources,
             extra_compile_args={
-                "cxx": ["-O3", "-std=c++17"],
+                "cxx": ["-O3", "-std=c++17", f"-I{NGHTTP2_INC}"],
                 "nvcc": ["-O3", "-std=c++17"],
             },
+            library_dirs=[NGHTTP2_LIB],
+            libraries=["nghttp2"],
+            extra_link_args=[f"-Wl,-rpath,{NGHTTP2_LIB}"],
         )
     ],
     cmdclass={"build_ext": cpp_extension.BuildExtension.with_options(

@@ -157,6 +157,73 @@ def test_dynamic_batching_merges_and_matches(tmp_path):
     assert runs1 - runs0 < served, (runs1 - runs0, served)
 
 
+def test_native_frontend_gpu_registry(tmp_path):
+    """Registered GPU Predicts bypass Python entirely: native_hits
+    must advance once the model's fast contexts are warm."""
+    import grpc
+    from concurrent.futures import ThreadPoolExecutor
+    from tfservingcache_amd.cachemanager import (CacheManager, LRUCache,
+                                                 ModelPool,
+                                                 make_gpu_loader)
+    from tfservingcache_amd.cachemanager.providers import \
+        DiskModelProvider
+    from tfservingcache_amd.tfservingproxy import LocalServingHandler
+    from tfservingcache_amd.tfservingproxy.native_frontend import \
+        NativeGrpcServer
+
+    repo = tmp_path / "repo_nf"
+    write_model_repo(str(repo), [("mlp", 1, "mlp")])
+    cache = LRUCache(str(tmp_path / "cache_nf"), 10 ** 9)
+    pool = ModelPool(make_gpu_loader(cache, device="cuda:0", max_batch=8,
+                                     n_streams=2),
+                     max_concurrent_models=2)
+    cm = CacheManager(DiskModelProvider(str(repo)), cache, pool,
+                      model_fetch_timeout=60.0)
+    handler = LocalServingHandler(cm)
+    srv = NativeGrpcServer(handler, workers=4)
+    srv.add_insecure_port("[::]:0")
+    srv.start()
+
+    def on_avail(name, version, model):
+        fast = getattr(getattr(model, "_gpu", None), "_fast", None)
+        if fast is not None:
+            srv.register_model(name, version, fast)
+    pool.on_available = on_avail
+    pool.on_unload = (lambda name, version, model:
+                      srv.unregister_model(name, version))
+    try:
+        ch = grpc.insecure_channel(f"127.0.0.1:{srv.port}")
+        predict = ch.unary_unary(
+            "/tensorflow.serving.PredictionService/Predict",
+            request_serializer=lambda r: r.encode(),
+            response_deserializer=m.PredictResponse.decode)
+        x = np.random.default_rng(0).standard_normal((4, 16)).astype(
+            np.float32)
+        req = m.PredictRequest(
+            model_spec=m.ModelSpec(name="mlp",
+                                   version=m.Int64Value(value=1)),
+            inputs={"x": numpy_to_tensorproto(x)})
+        first = tensorproto_to_numpy(
+            predict(req, timeout=60).outputs["probs"])
+        # hammer until the fast contexts warm and register
+        with ThreadPoolExecutor(8) as ex:
+            for _ in range(5):
+                outs = [tensorproto_to_numpy(r.outputs["probs"])
+                        for r in ex.map(
+                            lambda i: predict(req, timeout=60),
+                            range(40))]
+                for o in outs:
+                    np.testing.assert_allclose(o, first, rtol=1e-3,
+                                               atol=1e-4)
+                if srv.native_hits() > 0:
+                    break
+        assert srv.native_hits() > 0, (srv.native_hits(),
+                                       srv.fallback_calls())
+        ch.close()
+    finally:
+        srv.stop()
+
+
 def test_eviction_under_concurrent_load(tmp_path):
     """Hammer models from many threads while the pool evicts/reloads:
     no crashes, no wrong results, every request eventually served."""

@@ -70,6 +70,13 @@ class ModelPool:
         self._entries: Dict[ModelId, PoolEntry] = {}
         # seconds per completed load — benchmark/observability hook
         self.load_durations: List[float] = []
+        # lifecycle hooks (e.g. the native front-end's FastModel
+        # registry). on_available runs outside the pool lock;
+        # on_unload runs under it (before release) — keep it light
+        self.on_available: Optional[Callable[[str, int, object], None]] \
+            = None
+        self.on_unload: Optional[Callable[[str, int, object], None]] \
+            = None
 
     # -- introspection (GetModelStatus semantics) --------------------------
     def get_status(self, name: str, version: Optional[int] = None
@@ -147,6 +154,11 @@ class ModelPool:
         model, e.model = e.model, None
         e.state = END
         del self._entries[mid]
+        if model is not None and self.on_unload is not None:
+            try:
+                self.on_unload(mid[0], mid[1], model)
+            except Exception:       # noqa: BLE001
+                log.exception("on_unload hook failed for %s", mid)
         if model is not None and hasattr(model, "release"):
             try:
                 model.release()
@@ -193,6 +205,11 @@ class ModelPool:
             except Exception:       # noqa: BLE001
                 pass
             self._cond.notify_all()
+        if model is not None and self.on_available is not None:
+            try:
+                self.on_available(name, version, model)
+            except Exception:       # noqa: BLE001
+                log.exception("on_available hook failed for %s", mid)
 
     # -- event-driven wait (replaces the 500 ms poll loop) -----------------
     def wait_available(self, name: str, version: int,

@@ -227,6 +227,7 @@ void fast_run_plan(void* plan, hipStream_t s) {
 namespace py = pybind11;
 
 void register_fastpath(py::module_& mod);
+void register_frontend(py::module_& mod);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   using tfsc::Call;
@@ -277,6 +278,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("K_ATTENTION") = int(tfsc::K_ATTENTION);
 
   register_fastpath(mod);
+  register_frontend(mod);
 
   // elementwise fn codes
   mod.attr("ELT_ADD") = int(tfsc::ELT_ADD);

@@ -29,15 +29,12 @@
 #include <vector>
 
 #include "kernels.h"
+#include "fastpath_api.h"
 
 namespace tfsc {
 
 class ExecPlan;  // from executor.cpp
 void fast_run_plan(void* plan, hipStream_t s);  // defined in executor.cpp
-
-struct FastFallback : std::runtime_error {
-  using std::runtime_error::runtime_error;
-};
 
 #define HIPCHK(x)                                                        \
   do {                                                                   \
@@ -622,17 +619,18 @@ class FastModel {
   std::map<std::string, std::unique_ptr<BatchGroup>> groups_;
 };
 
-}  // namespace tfsc
-
-namespace py = pybind11;
+// frontend.cpp entry points (fastpath_api.h)
+std::string fastmodel_predict(FastModel* fm, const uint8_t* data,
+                              size_t len) {
+  return fm->predict(data, len);
+}
 
 // fast partial parse of PredictRequest.model_spec (field 1): the
-// Python routing layer calls this per request instead of its own
-// wire-level peek (tfservingcache_amd/wire/messages.py peek_model_spec)
-static py::tuple peek_spec(const uint8_t* p, size_t len) {
-  using tfsc::Reader;
-  std::string name, label;
-  int64_t version = 0;
+// Python routing layer (and the native front-end's registry lookup)
+// use this instead of a full decode
+bool peek_spec_raw(const uint8_t* p, size_t len, std::string* name,
+                   long long* version, std::string* label) {
+  int64_t ver = 0;
   bool has_version = false;
   Reader r{p, p + len};
   while (r.p < r.end) {
@@ -646,7 +644,7 @@ static py::tuple peek_spec(const uint8_t* p, size_t len) {
         int fno = int(stag >> 3), wt = int(stag & 7);
         if (fno == 1 && wt == 2) {
           uint64_t kn = rs.varint();
-          name.assign(reinterpret_cast<const char*>(rs.p), kn);
+          name->assign(reinterpret_cast<const char*>(rs.p), kn);
           rs.p += kn;
         } else if (fno == 2 && wt == 2) {          // Int64Value version
           uint64_t vn = rs.varint();
@@ -655,7 +653,7 @@ static py::tuple peek_spec(const uint8_t* p, size_t len) {
           while (rv.p < rv.end) {
             uint64_t vtag = rv.varint();
             if ((vtag >> 3) == 1 && (vtag & 7) == 0) {
-              version = int64_t(rv.varint());
+              ver = int64_t(rv.varint());
               has_version = true;
             } else {
               rv.skip(int(vtag & 7));
@@ -663,7 +661,7 @@ static py::tuple peek_spec(const uint8_t* p, size_t len) {
           }
         } else if (fno == 4 && wt == 2) {
           uint64_t kn = rs.varint();
-          label.assign(reinterpret_cast<const char*>(rs.p), kn);
+          label->assign(reinterpret_cast<const char*>(rs.p), kn);
           rs.p += kn;
         } else {
           rs.skip(wt);
@@ -673,10 +671,13 @@ static py::tuple peek_spec(const uint8_t* p, size_t len) {
     }
     r.skip(int(tag & 7));
   }
-  return py::make_tuple(py::str(name),
-                        has_version ? py::cast(version) : py::none(),
-                        py::str(label));
+  *version = ver;
+  return has_version;
 }
+
+}  // namespace tfsc
+
+namespace py = pybind11;
 
 void register_fastpath(py::module_& mod) {
   using tfsc::FastIO;
@@ -689,8 +690,14 @@ void register_fastpath(py::module_& mod) {
     Py_ssize_t blen = 0;
     if (PyBytes_AsStringAndSize(data.ptr(), &buf, &blen) != 0)
       throw py::error_already_set();
-    return peek_spec(reinterpret_cast<const uint8_t*>(buf),
-                     size_t(blen));
+    std::string name, label;
+    long long version = 0;
+    bool has = tfsc::peek_spec_raw(
+        reinterpret_cast<const uint8_t*>(buf), size_t(blen), &name,
+        &version, &label);
+    return py::make_tuple(py::str(name),
+                          has ? py::cast(version) : py::none(),
+                          py::str(label));
   });
 
   py::class_<FastIO>(mod, "FastIO")
@@ -713,6 +720,9 @@ void register_fastpath(py::module_& mod) {
       .def("add_context", &FastModel::add_context)
       .def("has_bucket", &FastModel::has_bucket)
       .def("stats", &FastModel::stats)
+      .def("_ptr", [](FastModel& fm) {
+        return reinterpret_cast<uintptr_t>(&fm);
+      })
       .def("enable_batching", &FastModel::enable_batching)
       .def("lock_ctx", &FastModel::lock_ctx,
            py::call_guard<py::gil_scoped_release>())

@@ -199,14 +199,26 @@ class Server:
             self.cfg.get_string("serving.metricsScrapeUrl") or None,
             timeout=self.cfg.get_float("metrics.timeout") or 3.0)
 
-        # cache tier gRPC
-        cache_grpc, _ = make_cache_grpc_server(
-            self.handler, health=self.health,
-            max_msg=self.cfg.get_int("serving.grpcMaxMsgSize") or
-            16 * 1024 * 1024)
-        cache_grpc.add_insecure_port(f"[::]:{self.cache_grpc_port}")
-        cache_grpc.start()
-        self._grpc_servers.append(cache_grpc)
+        # cache tier gRPC: grpcio server, or the native nghttp2
+        # front-end (serving.nativeFrontend) whose registered Predicts
+        # never touch Python
+        if self.cfg.get_bool("serving.nativeFrontend"):
+            from .tfservingproxy.native_frontend import NativeGrpcServer
+            cache_grpc = NativeGrpcServer(
+                self.handler, health=self.health,
+                workers=self.cfg.get_int("serving.nativeWorkers") or 16)
+            cache_grpc.add_insecure_port(f"[::]:{self.cache_grpc_port}")
+            cache_grpc.start()
+            self._grpc_servers.append(cache_grpc)
+            self._wire_native_registry(cache_grpc)
+        else:
+            cache_grpc, _ = make_cache_grpc_server(
+                self.handler, health=self.health,
+                max_msg=self.cfg.get_int("serving.grpcMaxMsgSize") or
+                16 * 1024 * 1024)
+            cache_grpc.add_insecure_port(f"[::]:{self.cache_grpc_port}")
+            cache_grpc.start()
+            self._grpc_servers.append(cache_grpc)
 
         # proxy tier (only with service discovery — main.go:103-105)
         self.discovery = create_discovery_service(self.cfg, self.is_healthy)
@@ -281,6 +293,21 @@ class Server:
                 self._runners.append(runner)
         self._loop.run_until_complete(boot())
         self._loop.run_forever()
+
+    def _wire_native_registry(self, native) -> None:
+        """Pool lifecycle -> C++ front-end FastModel registry."""
+        pool = self.cm.pool
+
+        def on_available(name, version, model):
+            fast = getattr(getattr(model, "_gpu", None), "_fast", None)
+            if fast is not None:
+                native.register_model(name, version, fast)
+
+        def on_unload(name, version, model):
+            native.unregister_model(name, version)
+
+        pool.on_available = on_available
+        pool.on_unload = on_unload
 
     # -- warm handoff ------------------------------------------------------
     def _on_membership_change(self, members) -> None:
