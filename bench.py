@@ -132,6 +132,8 @@ def main() -> int:
                     help="dynamic-batching merge window")
     ap.add_argument("--streams", type=int, default=6,
                     help="execution contexts (HIP streams) per model")
+    ap.add_argument("--channels", type=int, default=1,
+                    help="client gRPC channels (grpc/native transports)")
     ap.add_argument("--transport", choices=["inproc", "grpc", "native"],
                     default="inproc",
                     help="inproc: gRPC message path without sockets; "
@@ -303,17 +305,23 @@ def main() -> int:
                 max_msg=256 * 1024 * 1024)
             gport = grpc_server.add_insecure_port("127.0.0.1:0")
             grpc_server.start()
-        channel = grpc_mod.insecure_channel(
-            f"127.0.0.1:{gport}",
-            options=[("grpc.max_receive_message_length", 256 * 1024 * 1024),
-                     ("grpc.max_send_message_length", 256 * 1024 * 1024)])
-        predict_rpc = channel.unary_unary(
-            "/tensorflow.serving.PredictionService/Predict",
-            request_serializer=lambda b: b,
-            response_deserializer=m.PredictResponse.decode)
+        rpcs = []
+        for ci in range(max(1, args.channels)):
+            channel = grpc_mod.insecure_channel(
+                f"127.0.0.1:{gport}",
+                options=[("grpc.max_receive_message_length",
+                          256 * 1024 * 1024),
+                         ("grpc.max_send_message_length",
+                          256 * 1024 * 1024),
+                         # distinct arg -> distinct subchannel per channel
+                         ("tfsc.channel_id", ci)])
+            rpcs.append(channel.unary_unary(
+                "/tensorflow.serving.PredictionService/Predict",
+                request_serializer=lambda b: b,
+                response_deserializer=m.PredictResponse.decode))
 
         def one_request(i: int) -> None:
-            predict_rpc(req_cache[pick(i)], timeout=300)
+            rpcs[i % len(rpcs)](req_cache[pick(i)], timeout=300)
     else:
         def one_request(i: int) -> None:
             handler.predict_bytes(req_cache[pick(i)])
