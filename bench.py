@@ -417,10 +417,15 @@ def main() -> int:
                                  2) if latencies else None,
                 },
                 "n_cold_loads": len(loads),
-                "transport": ("real gRPC server+client over TCP loopback"
-                              if args.transport == "grpc" else
-                              "in-process gRPC message path "
-                              "(protobuf decode/encode included)"),
+                "transport": {
+                    "grpc": "real gRPC (Python grpcio server) over TCP "
+                            "loopback",
+                    "native": "real gRPC (native nghttp2 front-end) "
+                              "over TCP loopback",
+                    "inproc": "in-process gRPC message path (protobuf "
+                              "decode/encode included)",
+                }[args.transport],
+                "client_channels": args.channels,
                 "hipgraph": _graph_status(pool),
                 "dynamic_batching": bool(args.dyn_batch),
             },
