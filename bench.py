@@ -56,6 +56,59 @@ from tfservingcache_amd.wire.tensor import numpy_to_tensorproto  # noqa: E402
 REQS_PER_STEP = 50
 
 
+def _client_proc_main(conn, port, model_name, model_kind, batch,
+                      image_size, seq_len, threads, channels):
+    """Spawned gRPC client worker (warm mode): builds its own request
+    bytes, opens its own channels, executes 'run N' commands from the
+    parent over the pipe, returns per-request latencies."""
+    import grpc as grpc_mod
+    import numpy as _np
+    import time as _time
+    from concurrent.futures import ThreadPoolExecutor as _TPE
+    from tfservingcache_amd.wire import messages as _m
+    from tfservingcache_amd.wire.tensor import numpy_to_tensorproto as _n2t
+
+    rng = _np.random.default_rng(os.getpid())
+    if model_kind == "bert_base":
+        inputs = {"input_ids": _n2t(rng.integers(
+            0, 30000, (batch, seq_len)).astype(_np.int32))}
+    else:
+        inputs = {"input": _n2t((rng.standard_normal(
+            (batch, image_size, image_size, 3)) * 0.5
+        ).astype(_np.float32))}
+    req = _m.PredictRequest(
+        model_spec=_m.ModelSpec(name=model_name,
+                                version=_m.Int64Value(value=1)),
+        inputs=inputs).encode()
+    rpcs = []
+    for ci in range(max(1, channels)):
+        ch = grpc_mod.insecure_channel(
+            f"127.0.0.1:{port}",
+            options=[("grpc.max_receive_message_length", 256 << 20),
+                     ("grpc.max_send_message_length", 256 << 20),
+                     ("tfsc.channel_id", ci)])
+        rpcs.append(ch.unary_unary(
+            "/tensorflow.serving.PredictionService/Predict",
+            request_serializer=lambda b: b,
+            response_deserializer=lambda b: b))
+
+    pool = _TPE(max_workers=threads)
+
+    def one(i):
+        t0 = _time.monotonic()
+        rpcs[i % len(rpcs)](req, timeout=300)
+        return _time.monotonic() - t0
+
+    while True:
+        cmd = conn.recv()
+        if cmd[0] == "stop":
+            conn.close()
+            return
+        _, n = cmd
+        lats = list(pool.map(one, range(n)))
+        conn.send(lats)
+
+
 def _link_tree(src: str, dst: str) -> None:
     for root, _dirs, files in os.walk(src):
         rel = os.path.relpath(root, src)
@@ -134,6 +187,11 @@ def main() -> int:
                     help="execution contexts (HIP streams) per model")
     ap.add_argument("--channels", type=int, default=1,
                     help="client gRPC channels (grpc/native transports)")
+    ap.add_argument("--client-procs", type=int, default=0,
+                    help="run gRPC clients in N separate PROCESSES "
+                         "(warm mode, grpc/native transports): measures "
+                         "the server without the client's GIL in the "
+                         "way")
     ap.add_argument("--transport", choices=["inproc", "grpc", "native"],
                     default="inproc",
                     help="inproc: gRPC message path without sockets; "
@@ -335,7 +393,40 @@ def main() -> int:
 
     pool_executor = ThreadPoolExecutor(max_workers=args.threads)
 
+    client_conns = []
+    client_procs = []
+    if args.client_procs > 0:
+        if args.transport not in ("grpc", "native") or \
+                args.mode != "warm":
+            raise SystemExit("--client-procs needs warm mode and a "
+                             "socket transport")
+        import multiprocessing as mp
+        ctx = mp.get_context("spawn")
+        per = max(1, args.threads // args.client_procs)
+        for _ in range(args.client_procs):
+            parent, child = ctx.Pipe()
+            proc = ctx.Process(
+                target=_client_proc_main,
+                args=(child, gport, names[0], kind_of[names[0]],
+                      args.batch, args.image_size, args.seq_len, per,
+                      max(1, args.channels // args.client_procs)),
+                daemon=True)
+            proc.start()
+            client_conns.append(parent)
+            client_procs.append(proc)
+
     def step(base: int) -> None:
+        if client_conns:
+            n = REQS_PER_STEP // len(client_conns)
+            extra = REQS_PER_STEP - n * len(client_conns)
+            for i, conn in enumerate(client_conns):
+                conn.send(("run", n + (extra if i == 0 else 0)))
+            for conn in client_conns:
+                lats = conn.recv()
+                if timing_on[0]:
+                    with lat_lock:
+                        latencies.extend(lats)
+            return
         futs = [pool_executor.submit(one_request, base + j)
                 for j in range(REQS_PER_STEP)]
         for f in futs:
@@ -358,6 +449,14 @@ def main() -> int:
     timing_on[0] = False
     barrier_sync()
     elapsed = time.monotonic() - t_start
+
+    for conn in client_conns:
+        try:
+            conn.send(("stop",))
+        except (BrokenPipeError, OSError):
+            pass
+    for proc in client_procs:
+        proc.join(timeout=5)
 
     # MAX over ranks
     if dist is not None:
