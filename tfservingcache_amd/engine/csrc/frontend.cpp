@@ -568,7 +568,16 @@ class Frontend {
   std::atomic<long long> fallback_calls_{0};
 
  public:
-  ~Frontend() { stop(); }
+  ~Frontend() {
+    // destructor runs with the GIL held (pybind); workers may be
+    // blocked acquiring it for a fallback call — release while joining
+    if (PyGILState_Check()) {
+      py::gil_scoped_release rel;
+      stop();
+    } else {
+      stop();
+    }
+  }
 };
 
 }  // namespace tfsc_fe
