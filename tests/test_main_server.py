@@ -200,3 +200,24 @@ def test_warm_handoff_on_ring_change(tmp_path):
     finally:
         a.stop()
         b.stop()
+
+
+def test_logging_config(capsys):
+    """logging.{level,format} config surface (cfg.go:28-60 analog)."""
+    import json as json_mod
+    import logging
+    from tfservingcache_amd.utils.logsetup import setup_logging
+
+    setup_logging(Config({"logging": {"level": "debug",
+                                      "format": "json"}}))
+    assert logging.getLogger().level == logging.DEBUG
+    logging.getLogger("tfsc.test").warning("hello %s", "world")
+    line = capsys.readouterr().err.strip().splitlines()[-1]
+    entry = json_mod.loads(line)
+    assert entry["msg"] == "hello world"
+    assert entry["level"] == "warning"
+
+    setup_logging(Config({"logging": {"level": "warning"}}))
+    assert logging.getLogger().level == logging.WARNING
+    # restore defaults for other tests
+    setup_logging(Config({}))

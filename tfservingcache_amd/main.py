@@ -377,6 +377,8 @@ class Server:
 
 def main() -> int:
     cfg = Config.load()
+    from .utils.logsetup import setup_logging
+    setup_logging(cfg)
     server = Server(cfg)
     server.start()
     log.info("tfservingcache-amd serving: cacheRest=%d cacheGrpc=%d",
