@@ -84,11 +84,12 @@ struct Conn {
   std::atomic<bool> dead{false};
 
   void enqueue(std::unique_ptr<Response> r) {
-    {
-      std::lock_guard<std::mutex> g(out_mu);
-      if (dead.load()) return;
-      outbox.push_back(std::move(r));
-    }
+    // the wake write stays under out_mu: the conn thread sets `dead`
+    // under the same mutex BEFORE closing wake_fd, so no writer can
+    // race the close into a reused fd
+    std::lock_guard<std::mutex> g(out_mu);
+    if (dead.load()) return;
+    outbox.push_back(std::move(r));
     uint64_t one = 1;
     ssize_t n = write(wake_fd, &one, sizeof(one));
     (void)n;
@@ -349,7 +350,10 @@ class Frontend {
         if (rv < 0) break;
       }
     }
-    conn->dead.store(true);
+    {
+      std::lock_guard<std::mutex> g(conn->out_mu);
+      conn->dead.store(true);
+    }
     {
       std::lock_guard<std::mutex> g(conns_mu_);
       conns_.erase(conn);
