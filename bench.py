@@ -181,7 +181,7 @@ def main() -> int:
                     help="CPU engine (CI smoke only; not a benchmark)")
     ap.add_argument("--dyn-batch", action="store_true",
                     help="enable server-side dynamic batching")
-    ap.add_argument("--batch-timeout-ms", type=float, default=2.0,
+    ap.add_argument("--batch-timeout-ms", type=float, default=1.0,
                     help="dynamic-batching merge window")
     ap.add_argument("--streams", type=int, default=6,
                     help="execution contexts (HIP streams) per model")

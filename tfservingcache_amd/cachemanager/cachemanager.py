@@ -158,7 +158,7 @@ def make_cpu_loader(cache: LRUCache) -> Callable[[str, int], LoadedModel]:
 def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
                     max_batch: int = 64, use_graphs: bool = True,
                     batching: bool = False,
-                    batch_timeout_s: float = 0.002,
+                    batch_timeout_s: float = 0.001,
                     devices: Optional[List[str]] = None,
                     n_streams: int = 6
                     ) -> Callable[[str, int], LoadedModel]:
