@@ -19,7 +19,7 @@ import logging
 import os
 import threading
 import time
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Callable, Dict, List, Optional
 
 from ..engine.model import LoadedModel, load_model_from_dir
 from ..utils import metrics as mt

@@ -21,7 +21,7 @@ import os
 import shutil
 import threading
 from collections import OrderedDict
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Callable, List, Optional, Tuple
 
 log = logging.getLogger("tfsc.lru")

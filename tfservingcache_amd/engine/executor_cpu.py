@@ -13,7 +13,7 @@ from typing import Dict, List, Optional
 
 import numpy as np
 
-from .planner import Plan, PlanOp, resolve_dim
+from .planner import Plan, PlanOp
 
 
 def _act(x: np.ndarray, act: str) -> np.ndarray:

@@ -17,7 +17,6 @@ GPU box; no silent eager fallback).
 from __future__ import annotations
 
 import logging
-import math
 import threading
 from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple

@@ -15,7 +15,7 @@ from typing import Dict, Optional
 import numpy as np
 
 from .executor_cpu import CpuExecutor
-from .planner import Plan, compile_graph, is_sym, resolve_dim
+from .planner import Plan, compile_graph, is_sym
 from .savedmodel import read_saved_model
 from ..wire import graph as g
 from ..wire import messages as m

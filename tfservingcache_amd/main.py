@@ -37,7 +37,6 @@ from .tfservingproxy import (LocalServingHandler, make_cache_grpc_server,
                              make_cache_rest_app, make_proxy_grpc_server,
                              make_proxy_rest_app)
 from .tfservingproxy.grpc_server import HealthState
-from .utils import metrics as mt
 
 log = logging.getLogger("tfsc.main")
 

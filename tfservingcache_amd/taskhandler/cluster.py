@@ -15,7 +15,7 @@ from __future__ import annotations
 import logging
 import random
 import threading
-from typing import List, Optional
+from typing import List
 
 from .discovery.base import DiscoveryService, ServingService
 from .ring import ConsistentHashRing

@@ -19,7 +19,7 @@ import os
 import threading
 import time
 from dataclasses import dataclass
-from typing import Callable, Dict, List, Optional
+from typing import Callable, List, Optional
 
 log = logging.getLogger("tfsc.discovery")
 

@@ -10,7 +10,7 @@ reference proxies through unchanged):
 from __future__ import annotations
 
 import base64
-from typing import Any, Dict, List, Tuple
+from typing import Dict, List, Tuple
 
 import numpy as np
 

@@ -13,7 +13,7 @@ format is not parsed in this round.
 from __future__ import annotations
 
 from .pb import Message
-from .messages import (Any, SignatureDef, TensorProto, TensorShapeProto)
+from .messages import (SignatureDef, TensorProto, TensorShapeProto)
 
 
 class AttrListValue(Message):

@@ -18,7 +18,7 @@ Design goals:
 from __future__ import annotations
 
 import struct
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, List, Optional, Tuple
 
 WIRE_VARINT = 0
 WIRE_FIXED64 = 1
