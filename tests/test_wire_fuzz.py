@@ -1,0 +1,99 @@
+"""Property-based robustness for the hand-rolled protobuf codec:
+round-trips over randomized messages, resilience of the partial peek
+and the full decoder against arbitrary byte garbage (servers parse
+untrusted request bytes), and tensor codec round-trips."""
+import numpy as np
+import pytest
+from hypothesis import given, settings, strategies as st
+
+from tfservingcache_amd.wire import messages as m
+from tfservingcache_amd.wire.pb import Message
+from tfservingcache_amd.wire.tensor import (numpy_to_tensorproto,
+                                            tensorproto_to_numpy)
+
+names = st.text(
+    alphabet=st.characters(min_codepoint=32, max_codepoint=126),
+    max_size=40)
+
+
+@settings(deadline=None, max_examples=150)
+@given(name=names, version=st.integers(0, 2 ** 62),
+       label=names, sig=names)
+def test_model_spec_round_trip(name, version, label, sig):
+    spec = m.ModelSpec(name=name, version=m.Int64Value(value=version),
+                       signature_name=sig, version_label=label)
+    got = m.ModelSpec.decode(spec.encode())
+    assert got.name == name
+    assert got.version_value() == version
+    assert got.version_label == label
+    assert got.signature_name == sig
+
+
+@settings(deadline=None, max_examples=100)
+@given(data=st.binary(max_size=400))
+def test_decoders_never_crash_on_garbage(data):
+    """Arbitrary bytes: decode/peek either succeed or raise a clean
+    ValueError/IndexError — never hang or segfault."""
+    for op in (m.PredictRequest.decode, m.peek_model_spec,
+               m.GetModelStatusRequest.decode):
+        try:
+            op(data)
+        except (ValueError, IndexError, KeyError):
+            pass
+
+
+@settings(deadline=None, max_examples=60)
+@given(shape=st.lists(st.integers(1, 6), min_size=0, max_size=4),
+       dtype=st.sampled_from([np.float32, np.int32, np.int64,
+                              np.float64]),
+       splat=st.booleans())
+def test_tensor_round_trip(shape, dtype, splat):
+    rng = np.random.default_rng(0)
+    if np.issubdtype(dtype, np.floating):
+        arr = rng.standard_normal(shape).astype(dtype)
+    else:
+        arr = rng.integers(-1000, 1000, size=shape).astype(dtype)
+    tp = numpy_to_tensorproto(arr)
+    out = tensorproto_to_numpy(m.TensorProto.decode(tp.encode()))
+    np.testing.assert_array_equal(out, arr)
+
+
+@settings(deadline=None, max_examples=80)
+@given(inputs=st.dictionaries(
+    names.filter(bool),
+    st.lists(st.floats(-1e6, 1e6, width=32), min_size=1, max_size=8),
+    min_size=0, max_size=4),
+    name=names)
+def test_predict_request_round_trip(inputs, name):
+    req = m.PredictRequest(
+        model_spec=m.ModelSpec(name=name),
+        inputs={k: numpy_to_tensorproto(np.array(v, np.float32))
+                for k, v in inputs.items()})
+    data = req.encode()
+    got = m.PredictRequest.decode(data)
+    assert set(got.inputs) == set(inputs)
+    for k, v in inputs.items():
+        np.testing.assert_allclose(
+            tensorproto_to_numpy(got.inputs[k]),
+            np.array(v, np.float32))
+    peek = m.peek_model_spec(data)
+    assert peek.name == name
+
+
+@settings(deadline=None, max_examples=60)
+@given(data=st.binary(max_size=200))
+def test_unknown_fields_preserved(data):
+    """A message with unknown trailing fields re-encodes them verbatim
+    (the proxy tier forwards messages it only partially understands)."""
+    spec = m.ModelSpec(name="m").encode()
+    # append a syntactically valid unknown field (tag 1000, bytes)
+    from tfservingcache_amd.wire.pb import write_tag, write_varint
+    buf = bytearray(spec)
+    write_tag(buf, 1000, 2)
+    write_varint(buf, len(data))
+    buf += data
+    decoded = m.ModelSpec.decode(bytes(buf))
+    assert decoded.name == "m"
+    re = decoded.encode()
+    # unknown payload bytes survive the round trip
+    assert bytes(data) in bytes(re)
