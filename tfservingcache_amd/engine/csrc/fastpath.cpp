@@ -279,6 +279,10 @@ class FastModel {
     return {n_runs_.load(), n_requests_.load(), n_rows_.load()};
   }
 
+  std::tuple<long long, long long, long long> stage_ns() const {
+    return {ns_stage_in_.load(), ns_gpu_.load(), ns_serialize_.load()};
+  }
+
   bool has_bucket(int batch) {
     std::lock_guard<std::mutex> g(mu_);
     for (auto& c : ctxs_)
@@ -438,6 +442,7 @@ class FastModel {
     std::lock_guard<std::mutex> g2(ctx->mu, std::adopt_lock);
     if (ctx->disabled) throw FastFallback("model released");
 
+    auto t0 = std::chrono::steady_clock::now();
     hipStream_t s = ctx->stream;
     for (auto& io : ctx->ins) {
       int64_t row = io.row_elems;
@@ -475,6 +480,7 @@ class FastModel {
       }
     }
 
+    auto t1 = std::chrono::steady_clock::now();
     fast_run_plan(ctx->exec_plan, s);
 
     const auto& filt = batch[0]->req->output_filter;
@@ -502,6 +508,13 @@ class FastModel {
     if (ke != hipSuccess)
       throw std::runtime_error(std::string("fastpath kernel error: ") +
                                hipGetErrorString(ke));
+    auto t2 = std::chrono::steady_clock::now();
+    using std::chrono::nanoseconds, std::chrono::duration_cast;
+    ns_stage_in_.fetch_add(
+        duration_cast<nanoseconds>(t1 - t0).count(),
+        std::memory_order_relaxed);
+    ns_gpu_.fetch_add(duration_cast<nanoseconds>(t2 - t1).count(),
+                      std::memory_order_relaxed);
 
     // per-request responses from row slices of the pinned outputs
     size_t row_off = 0;
@@ -555,6 +568,10 @@ class FastModel {
       }
       row_off += size_t(rows);
     }
+    ns_serialize_.fetch_add(
+        duration_cast<nanoseconds>(
+            std::chrono::steady_clock::now() - t2).count(),
+        std::memory_order_relaxed);
   }
 
  private:
@@ -609,6 +626,11 @@ class FastModel {
   std::atomic<long long> n_runs_{0};       // plan executions
   std::atomic<long long> n_requests_{0};   // requests served
   std::atomic<long long> n_rows_{0};       // total rows executed
+  // per-stage wall time (ns) — queue->H2D staging, GPU (H2D DMA +
+  // kernels + D2H, stream-synchronized), response serialization
+  std::atomic<long long> ns_stage_in_{0};
+  std::atomic<long long> ns_gpu_{0};
+  std::atomic<long long> ns_serialize_{0};
   std::mutex mu_;
   std::vector<std::unique_ptr<FastContext>> ctxs_;
   std::vector<FastIO> specs_in_, specs_out_;
@@ -720,6 +742,7 @@ void register_fastpath(py::module_& mod) {
       .def("add_context", &FastModel::add_context)
       .def("has_bucket", &FastModel::has_bucket)
       .def("stats", &FastModel::stats)
+      .def("stage_ns", &FastModel::stage_ns)
       .def("_ptr", [](FastModel& fm) {
         return reinterpret_cast<uintptr_t>(&fm);
       })

@@ -169,6 +169,8 @@ class Server:
     def __init__(self, cfg: Config):
         self.cfg = cfg
         self.cm = create_cache_manager(cfg)
+        from .utils import metrics as mt
+        mt.engine_stages.add_pool(self.cm.pool)
         self.handler = LocalServingHandler(self.cm)
         self.health = HealthState()
         self.cluster: Optional[ClusterConnection] = None

@@ -60,6 +60,54 @@ engine_pool_models = Gauge(
     "Number of models resident in the pool", ["device"], registry=REGISTRY)
 
 
+# per-stage fast-path time: queue->pinned staging, GPU (H2D DMA +
+# kernels + D2H, stream-synchronized), response serialization —
+# aggregated at scrape time from every resident model's C++ counters
+# (SURVEY.md §5 "per-stage timing" tracing suggestion)
+class _EngineStageCollector:
+    def __init__(self):
+        import weakref
+        self._pools = []
+        self._weakref = weakref
+
+    def add_pool(self, pool) -> None:
+        self._pools.append(self._weakref.ref(pool))
+
+    def collect(self):
+        from prometheus_client.core import CounterMetricFamily
+        fam = CounterMetricFamily(
+            "tfservingcache_engine_stage_seconds",
+            "Cumulative fast-path time per stage across resident models",
+            labels=["stage"])
+        totals = {"stage_in": 0, "gpu": 0, "serialize": 0}
+        alive = []
+        for ref in self._pools:
+            pool = ref()
+            if pool is None:
+                continue
+            alive.append(ref)
+            try:
+                for entry in list(pool._entries.values()):  # noqa: SLF001
+                    fast = getattr(getattr(entry.model, "_gpu", None),
+                                   "_fast", None)
+                    if fast is None:
+                        continue
+                    si, gp, se = fast.stage_ns()
+                    totals["stage_in"] += si
+                    totals["gpu"] += gp
+                    totals["serialize"] += se
+            except Exception:       # noqa: BLE001
+                pass
+        self._pools = alive
+        for stage, ns in totals.items():
+            fam.add_metric([stage], ns / 1e9)
+        yield fam
+
+
+engine_stages = _EngineStageCollector()
+REGISTRY.register(engine_stages)
+
+
 def render() -> bytes:
     return generate_latest(REGISTRY)
 

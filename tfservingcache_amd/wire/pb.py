@@ -159,10 +159,14 @@ def _decode_scalar(kind: str, data: bytes, pos: int, wire_type: int):
             v = bool(v)
         return v, pos
     if wire_type == WIRE_FIXED32:
+        if pos + 4 > len(data):
+            raise ValueError("truncated fixed32 field")
         if kind == "float":
             return _f32.unpack_from(data, pos)[0], pos + 4
         return _u32.unpack_from(data, pos)[0], pos + 4
     if wire_type == WIRE_FIXED64:
+        if pos + 8 > len(data):
+            raise ValueError("truncated fixed64 field")
         if kind == "double":
             return _f64.unpack_from(data, pos)[0], pos + 8
         return _u64.unpack_from(data, pos)[0], pos + 8
