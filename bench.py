@@ -474,6 +474,20 @@ def main() -> int:
     loads = sorted(pool.load_durations) or [first_load_s]
     cold_p50 = loads[len(loads) // 2] * 1000.0
 
+    # fast-path stage breakdown (seconds, cumulative over the run)
+    stages = {"stage_in": 0, "gpu": 0, "serialize": 0}
+    try:
+        for e in pool._entries.values():      # noqa: SLF001
+            fast = getattr(getattr(e.model, "_gpu", None), "_fast", None)
+            if fast is not None:
+                si, gp, se = fast.stage_ns()
+                stages["stage_in"] += si
+                stages["gpu"] += gp
+                stages["serialize"] += se
+    except Exception:       # noqa: BLE001
+        pass
+    stages = {k: round(v / 1e9, 3) for k, v in stages.items()}
+
     if rank == 0:
         result = {
             "metric": "predict req/sec (whole node)",
@@ -525,6 +539,7 @@ def main() -> int:
                               "decode/encode included)",
                 }[args.transport],
                 "client_channels": args.channels,
+                "fastpath_stage_seconds": stages,
                 "hipgraph": _graph_status(pool),
                 "dynamic_batching": bool(args.dyn_batch),
             },

@@ -89,6 +89,18 @@ def test_server_gpu_end_to_end(tmp_path):
         np.testing.assert_allclose(first.sum(-1), np.ones(4), rtol=1e-2)
         mlp = server.cm.pool.get_model("mlp", 1)
         assert mlp._gpu._fast.has_bucket(4)   # C++ path registered
+        # per-stage engine metrics appear on the scrape endpoint
+        for _ in range(10):
+            predict(req, timeout=60)
+        scrape = requests.get(
+            f"http://127.0.0.1:{server.cache_rest_port}"
+            "/monitoring/prometheus/metrics", timeout=10).text
+        assert "tfservingcache_engine_stage_seconds" in scrape
+        gpu_line = [ln for ln in scrape.splitlines()
+                    if ln.startswith(
+                        "tfservingcache_engine_stage_seconds_total"
+                        '{stage="gpu"}')]
+        assert gpu_line and float(gpu_line[0].split()[-1]) > 0
         ch.close()
     finally:
         server.stop()
