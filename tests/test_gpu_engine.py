@@ -159,6 +159,26 @@ def test_bert_fused_attention_gpu_vs_cpu(tmp_path):
              {"sequence_output": c["sequence_output"]}, rtol=0.2, atol=0.15)
 
 
+def test_fused_attention_long_sequences(tmp_path):
+    """The flash-attention kernel's kv-tile loop + tail masking at
+    BERT-large-ish sequence lengths (S=384, 512; S%64 != 0 case)."""
+    for seq, hidden, heads in [(384, 128, 2), (512, 128, 2),
+                               (200, 128, 2)]:
+        sm = build_bert(seq_len=seq, hidden=hidden, layers=1,
+                        heads=heads, intermediate=256, vocab=300,
+                        seed=seq)
+        gm = _gpu_model(tmp_path, sm, name=f"bl{seq}")
+        cm = _cpu_model(tmp_path, sm, name=f"bl{seq}_cpu")
+        assert any(op.kind == "attention" for op in gm.plan.ops)
+        ids = np.random.default_rng(seq).integers(
+            0, 300, (2, seq)).astype(np.int32)
+        g = gm.predict({"input_ids": ids})
+        c = cm.predict({"input_ids": ids})
+        _compare({"sequence_output": g["sequence_output"]},
+                 {"sequence_output": c["sequence_output"]},
+                 rtol=0.2, atol=0.15)
+
+
 def test_fast_predict_path_matches_python(tmp_path):
     """C++ fast predict (bytes->bytes) must match the Python path."""
     from tfservingcache_amd.wire import messages as m
