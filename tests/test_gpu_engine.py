@@ -179,6 +179,28 @@ def test_fused_attention_long_sequences(tmp_path):
                  rtol=0.2, atol=0.15)
 
 
+def test_variables_model_on_gpu(tmp_path):
+    """Non-frozen SavedModel (variables/ tensor_bundle) compiles onto
+    the GPU engine and matches the frozen equivalent."""
+    import os
+    from tfservingcache_amd.engine.model import load_model_from_dir
+    from tfservingcache_amd.engine.gpu import GpuModel
+    from tfservingcache_amd.models import write_model_repo
+
+    write_model_repo(str(tmp_path / "vr"), [("mv", 1, "mlp_vars"),
+                                            ("mf", 1, "mlp")])
+    lm_v = load_model_from_dir(
+        os.path.join(str(tmp_path / "vr"), "mv", "1"), "mv", 1)
+    lm_v._gpu = GpuModel(lm_v.plan, max_batch=8, n_streams=1)
+    lm_f = load_model_from_dir(
+        os.path.join(str(tmp_path / "vr"), "mf", "1"), "mf", 1)
+    x = np.random.default_rng(9).standard_normal((4, 16)).astype(
+        np.float32)
+    g = lm_v.predict({"x": x})["probs"]        # GPU (vars)
+    c = lm_f.predict({"x": x})["probs"]        # CPU (frozen, same seed)
+    np.testing.assert_allclose(g, c, rtol=5e-2, atol=1e-2)
+
+
 def test_fast_predict_path_matches_python(tmp_path):
     """C++ fast predict (bytes->bytes) must match the Python path."""
     from tfservingcache_amd.wire import messages as m
