@@ -215,3 +215,78 @@ def test_etcd_watch_discovery():
     finally:
         d.unregister()
         server.stop(grace=0.5)
+
+
+# -- kubernetes mock --------------------------------------------------------
+def test_k8s_endpoints_watch_discovery():
+    """KubernetesDiscovery against a mock API server streaming watch
+    events: readiness-gated members, port matching by NAME, live
+    updates, re-watch after channel breakage."""
+    from tfservingcache_amd.taskhandler.discovery.kubernetes import \
+        KubernetesDiscovery
+
+    events = queue.Queue()
+    watch_count = [0]
+
+    def endpoints_obj(ips):
+        return {"type": "MODIFIED", "object": {"subsets": [{
+            "addresses": [{"ip": ip} for ip in ips],
+            "ports": [{"name": "grpccache", "port": 8100},
+                      {"name": "httpcache", "port": 8093},
+                      {"name": "other", "port": 9999}],
+        }]}}
+
+    class Handler(BaseHTTPRequestHandler):
+        def log_message(self, *a):      # noqa: D102
+            pass
+
+        def do_GET(self):
+            assert "watch=true" in self.path
+            assert "fieldSelector=metadata.name%3Dtfsc" in self.path
+            watch_count[0] += 1
+            self.send_response(200)
+            self.send_header("Transfer-Encoding", "chunked")
+            self.end_headers()
+            # stream events until the test closes the server
+            sent = 0
+            while sent < 10:
+                try:
+                    ev = events.get(timeout=3)
+                except queue.Empty:
+                    break
+                body = json.dumps(ev).encode() + b"\n"
+                self.wfile.write(f"{len(body):x}\r\n".encode() + body +
+                                 b"\r\n")
+                self.wfile.flush()
+                sent += 1
+            self.wfile.write(b"0\r\n\r\n")
+
+    httpd = ThreadingHTTPServer(("127.0.0.1", 0), Handler)
+    port = httpd.server_address[1]
+    threading.Thread(target=httpd.serve_forever, daemon=True).start()
+
+    seen = queue.Queue()
+    d = KubernetesDiscovery(
+        field_selector={"metadata.name": "tfsc"},
+        api_base=f"http://127.0.0.1:{port}", namespace="ns",
+        token="tok", verify=False)
+    d.add_listener(seen.put)
+    try:
+        events.put(endpoints_obj(["10.1.0.1"]))
+        d.register(ServingService("10.1.0.1", 8093, 8100))
+        members = seen.get(timeout=5)
+        assert [(m.host, m.rest_port, m.grpc_port) for m in members] == \
+            [("10.1.0.1", 8093, 8100)]
+
+        events.put(endpoints_obj(["10.1.0.1", "10.1.0.2"]))
+        members = seen.get(timeout=5)
+        assert len(members) == 2
+
+        # channel breakage: the stream ends after the queued events;
+        # the watcher must reconnect and pick up new state
+        events.put(endpoints_obj(["10.1.0.3"]))
+        members = seen.get(timeout=10)
+        assert [m.host for m in members] == ["10.1.0.3"]
+    finally:
+        d.unregister()
+        httpd.shutdown()
