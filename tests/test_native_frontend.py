@@ -134,6 +134,27 @@ def test_native_concurrent_and_large(served):
     assert srv.fallback_calls() >= 150   # CPU models: all via Python
 
 
+def test_forwarder_interop_with_native_server(served):
+    """The proxy tier's GrpcForwarder (bytes-level grpcio client, the
+    node-to-node hop) speaks to the native front-end."""
+    from tfservingcache_amd.tfservingproxy.grpc_server import \
+        GrpcForwarder
+    srv, _ch = served
+    fwd = GrpcForwarder(timeout_s=30.0)
+    x = np.random.default_rng(2).standard_normal((2, 16)).astype(
+        np.float32)
+    req = m.PredictRequest(
+        model_spec=m.ModelSpec(name="mlp"),
+        inputs={"x": numpy_to_tensorproto(x)})
+    raw = fwd.call(f"127.0.0.1:{srv.port}",
+                   "/tensorflow.serving.PredictionService/Predict",
+                   req.encode())
+    out = tensorproto_to_numpy(
+        m.PredictResponse.decode(raw).outputs["probs"])
+    np.testing.assert_allclose(out.sum(-1), np.ones(2), rtol=1e-4)
+    fwd.close()
+
+
 def test_server_native_frontend_config(tmp_path):
     """serving.nativeFrontend switches the cache gRPC implementation;
     the pool registry hooks are attached."""
