@@ -85,13 +85,11 @@ def _get_staging(torch):
     return _staging_buf
 
 
-# converted-weight cache: bf16 CPU blobs keyed by Plan identity.
+# converted-weight cache: pinned bf16 blobs keyed by Plan identity.
 # Plans are already content-deduplicated (engine/model.py inode-keyed
 # plan cache), so models sharing a SavedModel's bytes share a Plan —
 # their cold loads skip the CPU f32->bf16 convert and become one DMA.
-# Bounded: ~51 MB per ResNet-50-sized plan. Deliberately PAGEABLE, not
-# pinned: hipHostMalloc during another model's hipGraph capture poisons
-# the capture on ROCm, and loads run concurrently with captures.
+# Bounded: ~51 MB pinned per ResNet-50-sized plan.
 _BLOB_CACHE_CAP = 4
 _blob_cache_lock = threading.Lock()
 _blob_cache: "Dict[int, tuple]" = {}       # id(plan) -> (plan, blob)
@@ -717,10 +715,11 @@ class GpuModel:
             # next load of this plan skips the CPU convert entirely.
             cpu_blob = None
             try:
-                cpu_blob = torch.empty(total, dtype=torch.bfloat16)
+                cpu_blob = torch.empty(total, dtype=torch.bfloat16,
+                                       pin_memory=True)
             except RuntimeError:
-                log.warning("weight-blob allocation failed; skipping "
-                            "the converted-weight cache")
+                log.warning("pinned weight-blob allocation failed; "
+                            "skipping the converted-weight cache")
             with _staging_lock:
                 stage = _get_staging(torch)
                 cap = stage.numel()
