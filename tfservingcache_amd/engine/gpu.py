@@ -85,39 +85,6 @@ def _get_staging(torch):
     return _staging_buf
 
 
-# converted-weight cache: pinned bf16 blobs keyed by Plan identity.
-# Plans are already content-deduplicated (engine/model.py inode-keyed
-# plan cache), so models sharing a SavedModel's bytes share a Plan —
-# their cold loads skip the CPU f32->bf16 convert and become one DMA.
-# Bounded: ~51 MB pinned per ResNet-50-sized plan.
-_BLOB_CACHE_CAP = 4
-_blob_cache_lock = threading.Lock()
-_blob_cache: "Dict[int, tuple]" = {}       # id(plan) -> (plan, blob)
-_blob_cache_order: List[int] = []
-
-
-def _blob_cache_get(plan):
-    with _blob_cache_lock:
-        ent = _blob_cache.get(id(plan))
-        if ent is None or ent[0] is not plan:
-            return None
-        _blob_cache_order.remove(id(plan))
-        _blob_cache_order.append(id(plan))
-        return ent[1]
-
-
-def _blob_cache_put(plan, blob) -> None:
-    with _blob_cache_lock:
-        key = id(plan)
-        if key not in _blob_cache:
-            _blob_cache_order.append(key)
-        # keep the Plan object referenced so id() stays unambiguous
-        _blob_cache[key] = (plan, blob)
-        while len(_blob_cache_order) > _BLOB_CACHE_CAP:
-            old = _blob_cache_order.pop(0)
-            _blob_cache.pop(old, None)
-
-
 def _pad64(k: int) -> int:
     return (k + 63) // 64 * 64
 
@@ -702,58 +669,38 @@ class GpuModel:
         if not float_ws:
             return
         blob = torch.empty(total, dtype=torch.bfloat16, device=self.device)
-        cached = _blob_cache_get(self.plan)
-        if cached is not None and cached.numel() == total:
-            # converted blob already pinned in CPU RAM: one DMA
-            blob.copy_(cached, non_blocking=True)
-            torch.cuda.synchronize(self.device)
-        else:
-            # stage through a SHARED reusable pinned buffer: per-load
-            # pinned allocation (page-locking ~100 MB) costs more than
-            # the copy and thrashes badly under LRU churn. The converted
-            # result is ALSO written into a cacheable pinned blob so the
-            # next load of this plan skips the CPU convert entirely.
-            cpu_blob = None
-            try:
-                cpu_blob = torch.empty(total, dtype=torch.bfloat16,
-                                       pin_memory=True)
-            except RuntimeError:
-                log.warning("pinned weight-blob allocation failed; "
-                            "skipping the converted-weight cache")
-            with _staging_lock:
-                stage = _get_staging(torch)
-                cap = stage.numel()
-                batch_items = []        # (stage_off, blob_off, n)
+        # stage through a SHARED reusable pinned buffer: per-load pinned
+        # allocation (page-locking ~100 MB) costs more than the copy and
+        # thrashes badly under LRU churn
+        with _staging_lock:
+            stage = _get_staging(torch)
+            cap = stage.numel()
+            batch_items = []        # (stage_off, blob_off, n)
+            stage_off = 0
+
+            def flush():
+                nonlocal stage_off
+                for s_off, b_off, n_ in batch_items:
+                    blob[b_off:b_off + n_].copy_(
+                        stage[s_off:s_off + n_], non_blocking=True)
+                torch.cuda.synchronize(self.device)
+                batch_items.clear()
                 stage_off = 0
 
-                def flush():
-                    nonlocal stage_off
-                    for s_off, b_off, n_ in batch_items:
-                        blob[b_off:b_off + n_].copy_(
-                            stage[s_off:s_off + n_], non_blocking=True)
-                    torch.cuda.synchronize(self.device)
-                    batch_items.clear()
-                    stage_off = 0
-
-                for _idx, w, off, n in float_ws:
-                    src = torch.from_numpy(np.ascontiguousarray(
-                        np.asarray(w, dtype=np.float32))).view(-1)
-                    done = 0
-                    while done < n:
-                        if stage_off >= cap:
-                            flush()
-                        take = min(n - done, cap - stage_off)
-                        stage[stage_off:stage_off + take].copy_(
-                            src[done:done + take])  # CPU f32->bf16
-                        if cpu_blob is not None:
-                            cpu_blob[off + done:off + done + take].copy_(
-                                stage[stage_off:stage_off + take])
-                        batch_items.append((stage_off, off + done, take))
-                        stage_off += take
-                        done += take
-                flush()
-            if cpu_blob is not None:
-                _blob_cache_put(self.plan, cpu_blob)
+            for _idx, w, off, n in float_ws:
+                src = torch.from_numpy(np.ascontiguousarray(
+                    np.asarray(w, dtype=np.float32))).view(-1)
+                done = 0
+                while done < n:
+                    if stage_off >= cap:
+                        flush()
+                    take = min(n - done, cap - stage_off)
+                    stage[stage_off:stage_off + take].copy_(
+                        src[done:done + take])   # CPU f32->bf16 convert
+                    batch_items.append((stage_off, off + done, take))
+                    stage_off += take
+                    done += take
+            flush()
         self._weight_blob = blob              # keep the allocation alive
         for idx, w, off, n in float_ws:
             self._weights[idx] = blob[off:off + n].view(tuple(w.shape))
