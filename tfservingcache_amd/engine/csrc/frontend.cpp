@@ -237,9 +237,12 @@ class Frontend {
     for (auto& t : workers_)
       if (t.joinable()) t.join();
     workers_.clear();
-    for (auto& t : conn_threads_)
-      if (t.joinable()) t.join();
-    conn_threads_.clear();
+    {
+      std::lock_guard<std::mutex> g(threads_mu_);
+      for (auto& t : conn_threads_)
+        if (t.second.joinable()) t.second.join();
+      conn_threads_.clear();
+    }
   }
 
   void register_model(const std::string& name, long long version,
@@ -284,8 +287,22 @@ class Frontend {
         std::lock_guard<std::mutex> g(conns_mu_);
         conns_.insert(conn);
       }
-      conn_threads_.emplace_back(
-          [this, conn] { conn_loop(conn); });
+      {
+        // reap threads of connections that have since closed (a
+        // long-lived server must not accumulate unjoined stacks)
+        std::lock_guard<std::mutex> g(threads_mu_);
+        for (auto it = conn_threads_.begin();
+             it != conn_threads_.end();) {
+          if (it->first->dead.load() && it->second.joinable()) {
+            it->second.join();
+            it = conn_threads_.erase(it);
+          } else {
+            ++it;
+          }
+        }
+        conn_threads_.emplace_back(
+            conn, std::thread([this, conn] { conn_loop(conn); }));
+      }
     }
   }
 
@@ -559,7 +576,10 @@ class Frontend {
   int bound_port_ = 0;
   std::thread accept_thread_;
   std::vector<std::thread> workers_;
-  std::vector<std::thread> conn_threads_;
+  std::mutex threads_mu_;
+  // shared_ptr key: the Conn must outlive the reaper's dead-check
+  std::vector<std::pair<std::shared_ptr<Conn>, std::thread>>
+      conn_threads_;
   std::mutex conns_mu_;
   std::set<std::shared_ptr<Conn>> conns_;
   std::mutex jobs_mu_;
