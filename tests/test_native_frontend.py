@@ -134,6 +134,22 @@ def test_native_concurrent_and_large(served):
     assert srv.fallback_calls() >= 150   # CPU models: all via Python
 
 
+def test_connection_churn(served):
+    """Many short-lived connections: threads are reaped, serving keeps
+    working (long-lived-server hygiene)."""
+    srv, _ch = served
+    req = m.PredictRequest(
+        model_spec=m.ModelSpec(name="mlp"),
+        inputs={"x": numpy_to_tensorproto(
+            np.zeros((1, 16), np.float32))})
+    for i in range(30):
+        ch = grpc.insecure_channel(f"127.0.0.1:{srv.port}",
+                                   options=[("tfsc.conn", i)])
+        out = _predict_rpc(ch)(req, timeout=30)
+        assert out.outputs["probs"].tensor_shape.dim[0].size == 1
+        ch.close()
+
+
 def test_forwarder_interop_with_native_server(served):
     """The proxy tier's GrpcForwarder (bytes-level grpcio client, the
     node-to-node hop) speaks to the native front-end."""
