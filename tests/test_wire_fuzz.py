@@ -97,3 +97,44 @@ def test_unknown_fields_preserved(data):
     re = decoded.encode()
     # unknown payload bytes survive the round trip
     assert bytes(data) in bytes(re)
+
+
+@settings(deadline=None, max_examples=80)
+@given(body=st.recursive(
+    st.one_of(st.none(), st.booleans(), st.integers(-10, 10),
+              st.floats(-100, 100, allow_nan=False), st.text(max_size=8)),
+    lambda children: st.one_of(
+        st.lists(children, max_size=4),
+        st.dictionaries(st.text(max_size=8), children, max_size=4)),
+    max_leaves=12))
+def test_rest_codec_never_crashes_on_garbage(body):
+    """Arbitrary JSON-shaped bodies: parse either succeeds or raises
+    RestCodecError — the REST handler's 400 path — never an unhandled
+    TypeError deep in numpy."""
+    from tfservingcache_amd.tfservingproxy.json_codec import (
+        RestCodecError, parse_predict_body)
+    if not isinstance(body, dict):
+        return
+    try:
+        parse_predict_body(body)
+    except RestCodecError:
+        pass
+
+
+@settings(deadline=None, max_examples=60)
+@given(rows=st.integers(1, 5), cols=st.integers(1, 4),
+       columnar=st.booleans())
+def test_rest_codec_round_trip(rows, cols, columnar):
+    from tfservingcache_amd.tfservingproxy.json_codec import (
+        parse_predict_body, render_predict_response)
+    rng = np.random.default_rng(rows * 10 + cols)
+    arr = rng.standard_normal((rows, cols)).astype(np.float32)
+    if columnar:
+        body = {"inputs": {"x": arr.tolist()}}
+    else:
+        body = {"instances": [{"x": row.tolist()} for row in arr]}
+    feeds, fmt, _sig = parse_predict_body(body)
+    np.testing.assert_allclose(np.asarray(feeds["x"], np.float32)
+                               .reshape(arr.shape), arr, rtol=1e-6)
+    out = render_predict_response({"y": arr}, fmt)
+    assert ("outputs" if columnar else "predictions") in out
