@@ -387,3 +387,26 @@ def test_plan_cache_shares_hardlinked_models(tmp_path):
     out_c = lm_c.predict({"x": np.array([2.0], np.float32)})
     np.testing.assert_allclose(list(out_a.values())[0],
                                list(out_c.values())[0])
+
+
+def test_rebuild_from_disk_after_restart(tmp_path):
+    """modelCache.rebuildFromDisk re-indexes an existing cache dir: a
+    restarted node serves what is already on disk instead of leaking
+    the files (SURVEY.md §2.3 — the reference leaves stale files and
+    an empty map)."""
+    from tfservingcache_amd.cachemanager import LRUCache
+    from tfservingcache_amd.models import write_model_repo
+
+    cache_dir = tmp_path / "c"
+    write_model_repo(str(cache_dir), [("m1", 1, "half_plus_two"),
+                                      ("m1", 2, "half_plus_two"),
+                                      ("m2", 7, "mlp")])
+    # fresh process, same dir
+    cache = LRUCache(str(cache_dir), 10 ** 9, rebuild_from_disk=True)
+    ids = {(e.name, e.version) for e in cache.list_models()}
+    assert ids == {("m1", 1), ("m1", 2), ("m2", 7)}
+    assert cache.current_size > 0
+    assert all(e.size_on_disk > 0 for e in cache.list_models())
+    # without the flag the index starts empty (reference behavior)
+    cache2 = LRUCache(str(cache_dir), 10 ** 9)
+    assert cache2.list_models() == []
