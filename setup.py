@@ -28,9 +28,23 @@ sources = [
 ]
 
 # nghttp2 (HTTP/2 framing/HPACK for the native gRPC front-end) ships in
-# this image's conda tree; same path exists on the GPU boxes (same image)
-NGHTTP2_INC = "/opt/conda/include"
-NGHTTP2_LIB = "/opt/conda/lib"
+# this image's conda tree; probe the usual prefixes so other ROCm bases
+# (system libnghttp2-dev) work too
+def _find_nghttp2():
+    for inc, lib in [("/opt/conda/include", "/opt/conda/lib"),
+                     ("/usr/include", "/usr/lib/x86_64-linux-gnu"),
+                     ("/usr/local/include", "/usr/local/lib")]:
+        if os.path.exists(os.path.join(inc, "nghttp2", "nghttp2.h")):
+            return inc, lib
+    raise RuntimeError(
+        "nghttp2 headers not found (needed by the native gRPC "
+        "front-end, engine/csrc/frontend.cpp) — install libnghttp2-dev "
+        "or point NGHTTP2_INC/NGHTTP2_LIB at a prefix")
+
+
+NGHTTP2_INC, NGHTTP2_LIB = (
+    (os.environ.get("NGHTTP2_INC"), os.environ.get("NGHTTP2_LIB"))
+    if os.environ.get("NGHTTP2_INC") else _find_nghttp2())
 
 setup(
     name="tfsc_engine",
