@@ -255,3 +255,34 @@ def test_file_discovery_member_expiry(tmp_path):
     finally:
         d1.unregister()
         d2.unregister()
+
+
+def test_ring_load_balance():
+    """20 vnodes/member should spread keys reasonably evenly: over 10k
+    model##version keys on 8 members, no member owns more than 2.2x or
+    less than 0.35x its fair share (the reference's ring lib uses the
+    same CRC32 + 20-replica construction)."""
+    from collections import Counter
+    ring = ConsistentHashRing()
+    members = [f"10.0.0.{i}:8094:8095" for i in range(8)]
+    ring.set_members(members)
+    counts = Counter(ring.get_n(f"model_{i:05d}##1", 1)[0]
+                     for i in range(10000))
+    fair = 10000 / 8
+    assert set(counts) == set(members)           # everyone owns keys
+    for m, c in counts.items():
+        assert 0.35 * fair < c < 2.2 * fair, (m, c)
+
+
+def test_ring_replica_distinctness_and_spread():
+    """get_n returns N DISTINCT members, and replica PAIRS vary (a
+    degenerate ring would co-locate every pair)."""
+    ring = ConsistentHashRing()
+    members = [f"n{i}" for i in range(6)]
+    ring.set_members(members)
+    pairs = set()
+    for i in range(2000):
+        owners = ring.get_n(f"m{i}##1", 2)
+        assert len(owners) == 2 and len(set(owners)) == 2
+        pairs.add(tuple(sorted(owners)))
+    assert len(pairs) >= 10     # many distinct replica pairs in use
