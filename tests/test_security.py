@@ -1,0 +1,157 @@
+"""Hardening regressions: model-name path traversal, truncated wire
+bytes, and single-flight lock lifetime (round-1 advisor findings)."""
+import threading
+import time
+
+import pytest
+
+from tfservingcache_amd.cachemanager import (CacheManager, LRUCache,
+                                             ModelPool, make_cpu_loader)
+from tfservingcache_amd.cachemanager.modelprovider import (
+    InvalidModelNameError, validate_model_name)
+from tfservingcache_amd.cachemanager.providers import DiskModelProvider
+from tfservingcache_amd.models import write_model_repo
+from tfservingcache_amd.wire import messages as m
+
+
+# -- model-name validation ---------------------------------------------------
+
+@pytest.mark.parametrize("bad", [
+    "", ".", "..", "../x", "a/b", "a/../b", "\\..\\x", "a\\b",
+    "x\x00y", "m" * 513,
+])
+def test_invalid_model_names_rejected(bad):
+    with pytest.raises(InvalidModelNameError):
+        validate_model_name(bad)
+
+
+@pytest.mark.parametrize("ok", ["m", "half_plus_two", "model-1.2_x",
+                                "UPPER", "00042"])
+def test_valid_model_names_accepted(ok):
+    assert validate_model_name(ok) == ok
+
+
+def test_traversal_name_cannot_delete_repo(tmp_path):
+    """A gRPC-style name '../model_repo/<m>' must be rejected before it
+    reaches the disk provider's rmtree+copytree (advisor high #2)."""
+    repo = tmp_path / "repo"
+    write_model_repo(str(repo), [("victim", 1, "half_plus_two")])
+    cache = LRUCache(str(tmp_path / "cache"), 10 ** 8)
+    pool = ModelPool(make_cpu_loader(cache), max_concurrent_models=2)
+    cm = CacheManager(DiskModelProvider(str(repo)), cache, pool)
+    with pytest.raises(InvalidModelNameError):
+        cm.ensure_loaded("../repo/victim", 1)
+    assert (repo / "victim" / "1").is_dir()     # untouched
+
+
+def test_provider_rejects_traversal_directly(tmp_path):
+    prov = DiskModelProvider(str(tmp_path))
+    with pytest.raises(InvalidModelNameError):
+        prov.load_model("../evil", 1, str(tmp_path / "cache"))
+    with pytest.raises(InvalidModelNameError):
+        prov.model_size("a/b", 1)
+    with pytest.raises(InvalidModelNameError):
+        prov.latest_version("..")
+
+
+# -- truncated protobuf bytes ------------------------------------------------
+
+def _sample_request() -> bytes:
+    import numpy as np
+    from tfservingcache_amd.wire.tensor import numpy_to_tensorproto
+    req = m.PredictRequest(
+        model_spec=m.ModelSpec(name="mm", version=m.Int64Value(value=3)),
+        inputs={"x": numpy_to_tensorproto(
+            np.arange(64, dtype=np.float32).reshape(4, 16))})
+    return req.encode()
+
+
+def test_truncated_request_raises_cleanly():
+    data = _sample_request()
+    for cut in range(1, len(data)):
+        try:
+            m.PredictRequest.decode(data[:cut])
+        except ValueError:
+            continue
+        # a successful decode of a prefix is only OK if the prefix
+        # happens to end exactly on a field boundary — re-encode must
+        # then be a prefix-consistent message, never garbage memory
+    # full message still decodes
+    got = m.PredictRequest.decode(data)
+    assert got.model_spec.name == "mm"
+
+
+def test_truncated_tensor_content_rejected():
+    """Declared tensor_content length beyond the buffer must raise, not
+    silently truncate (advisor low #5)."""
+    data = bytearray(_sample_request())
+    # chop the last 10 bytes: tensor_content's declared length now
+    # exceeds the remaining buffer
+    with pytest.raises(ValueError):
+        m.PredictRequest.decode(bytes(data[:-10]))
+
+
+def test_peek_spec_truncated():
+    data = _sample_request()
+    spec = m.peek_model_spec(data)
+    assert spec.name == "mm"
+    with pytest.raises((ValueError, IndexError)):
+        m.peek_model_spec(data[:3])
+
+
+def test_cpp_peek_spec_truncated_no_crash():
+    pytest.importorskip("torch")        # extension links torch libs
+    ext = pytest.importorskip(
+        "tfservingcache_amd.engine._tfsc_engine")
+    data = _sample_request()
+    name, ver, _label = ext.peek_spec(data)
+    assert name == "mm" and ver == 3
+    for cut in range(len(data)):
+        try:
+            ext.peek_spec(data[:cut])
+        except Exception:       # FastFallback — clean python exception
+            pass
+
+
+# -- single-flight lock lifetime ---------------------------------------------
+
+class _SlowProvider(DiskModelProvider):
+    def __init__(self, base_dir, delay=0.15):
+        super().__init__(base_dir)
+        self.delay = delay
+        self.loads = 0
+        self._mu = threading.Lock()
+
+    def load_model(self, name, version, dest):
+        with self._mu:
+            self.loads += 1
+        time.sleep(self.delay)
+        return super().load_model(name, version, dest)
+
+
+def test_single_flight_no_duplicate_fetch_under_waiters(tmp_path):
+    """Waiters queued on the in-flight lock must not allow a later
+    thread to start a second concurrent fetch (advisor medium #3)."""
+    repo = tmp_path / "repo"
+    write_model_repo(str(repo), [("m", 1, "half_plus_two")])
+    cache = LRUCache(str(tmp_path / "cache"), 10 ** 8)
+    pool = ModelPool(make_cpu_loader(cache), max_concurrent_models=4)
+    prov = _SlowProvider(str(repo))
+    cm = CacheManager(prov, cache, pool)
+
+    errs = []
+
+    def worker():
+        try:
+            cm.ensure_loaded("m", 1)
+        except Exception as e:      # noqa: BLE001
+            errs.append(e)
+
+    threads = [threading.Thread(target=worker) for _ in range(12)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errs
+    assert prov.loads == 1
+    assert cm._in_flight == {}

@@ -25,7 +25,7 @@ from ..engine.model import LoadedModel, load_model_from_dir
 from ..utils import metrics as mt
 from .lrucache import LRUCache, Model, ModelId
 from .modelpool import ModelPool
-from .modelprovider import ModelProvider
+from .modelprovider import ModelProvider, validate_model_name
 
 log = logging.getLogger("tfsc.cache")
 
@@ -48,11 +48,16 @@ class CacheManager:
         self.fetch_timeout = model_fetch_timeout
         self.model_labels = model_labels
         self._flight_lock = threading.Lock()
-        self._in_flight: Dict[ModelId, threading.Lock] = {}
+        # mid -> [lock, refcount]; the entry stays registered while any
+        # thread holds OR waits on the lock (a pop while waiters were
+        # still queued on a stale lock object would let a later thread
+        # create a fresh lock and fetch the same model concurrently)
+        self._in_flight: Dict[ModelId, list] = {}
 
     # -- the hot path ------------------------------------------------------
     def ensure_loaded(self, name: str, version: int) -> LoadedModel:
         """Returns the AVAILABLE model, fetching/loading on miss."""
+        validate_model_name(name)
         labels = mt.model_labels(self.model_labels, name, version)
         start = time.monotonic()
         mt.cache_total.labels(*labels).inc()
@@ -73,7 +78,12 @@ class CacheManager:
         mid: ModelId = (name, version)
         # single-flight: one concurrent fetch per model-version
         with self._flight_lock:
-            lock = self._in_flight.setdefault(mid, threading.Lock())
+            flight = self._in_flight.get(mid)
+            if flight is None:
+                flight = [threading.Lock(), 0]
+                self._in_flight[mid] = flight
+            flight[1] += 1
+            lock = flight[0]
         with lock:
             try:
                 # re-check after winning the lock
@@ -111,7 +121,9 @@ class CacheManager:
                             raise
             finally:
                 with self._flight_lock:
-                    self._in_flight.pop(mid, None)
+                    flight[1] -= 1
+                    if flight[1] == 0:
+                        self._in_flight.pop(mid, None)
 
     def _reload_pool(self) -> None:
         """Declarative reload: pool gets the MRU-first prefix of the disk

@@ -27,7 +27,7 @@ import requests
 
 from ..lrucache import Model, dir_size
 from ..modelprovider import ModelNotFoundError, ModelProvider, \
-    ModelProviderError
+    ModelProviderError, validate_model_name
 
 
 class AZBlobModelProvider(ModelProvider):
@@ -96,6 +96,7 @@ class AZBlobModelProvider(ModelProvider):
     # -- ModelProvider -----------------------------------------------------
     def load_model(self, model_name: str, version: int,
                    dest_base_dir: str) -> Model:
+        validate_model_name(model_name)
         prefix = self._prefix(model_name, version)
         rel = os.path.join(model_name, str(version))
         dst_root = os.path.join(dest_base_dir, rel)
@@ -104,6 +105,9 @@ class AZBlobModelProvider(ModelProvider):
             sub = name[len(prefix):]
             if not sub or sub.endswith("/"):
                 continue
+            if ".." in sub.split("/") or sub.startswith("/"):
+                raise ModelProviderError(
+                    f"refusing traversal in blob name: {name!r}")
             dst = os.path.join(dst_root, sub)
             os.makedirs(os.path.dirname(dst), exist_ok=True)
             url = f"{self.endpoint}/{self.container}/" + \

@@ -17,7 +17,8 @@ import os
 import shutil
 
 from ..lrucache import Model, dir_size
-from ..modelprovider import ModelNotFoundError, ModelProvider
+from ..modelprovider import (ModelNotFoundError, ModelProvider,
+                             validate_model_name)
 
 
 def _link_or_copy(src: str, dst: str) -> None:
@@ -32,6 +33,7 @@ class DiskModelProvider(ModelProvider):
         self.base_dir = base_dir
 
     def _find_src_dir(self, model_name: str, version: int) -> str:
+        validate_model_name(model_name)
         model_dir = os.path.join(self.base_dir, model_name)
         if not os.path.isdir(model_dir):
             raise ModelNotFoundError(f"model dir not found: {model_dir}")
@@ -48,6 +50,7 @@ class DiskModelProvider(ModelProvider):
             f"version {version} of model {model_name} not found")
 
     def load_model(self, model_name: str, version: int, dest_base_dir: str) -> Model:
+        validate_model_name(model_name)
         src = self._find_src_dir(model_name, version)
         rel = os.path.join(model_name, str(version))
         dst = os.path.join(dest_base_dir, rel)
@@ -64,6 +67,7 @@ class DiskModelProvider(ModelProvider):
         return os.path.isdir(self.base_dir)
 
     def latest_version(self, model_name: str):
+        validate_model_name(model_name)
         model_dir = os.path.join(self.base_dir, model_name)
         if not os.path.isdir(model_dir):
             return None

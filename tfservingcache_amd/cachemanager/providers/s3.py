@@ -28,7 +28,7 @@ import requests
 
 from ..lrucache import Model, dir_size
 from ..modelprovider import ModelNotFoundError, ModelProvider, \
-    ModelProviderError
+    ModelProviderError, validate_model_name
 
 
 def _sha256(data: bytes) -> str:
@@ -157,6 +157,7 @@ class S3ModelProvider(ModelProvider):
     # -- ModelProvider -----------------------------------------------------
     def load_model(self, model_name: str, version: int,
                    dest_base_dir: str) -> Model:
+        validate_model_name(model_name)
         prefix = self._key_prefix(model_name, version)
         rel = os.path.join(model_name, str(version))
         dst_root = os.path.join(dest_base_dir, rel)
@@ -165,6 +166,9 @@ class S3ModelProvider(ModelProvider):
             sub = key[len(prefix):]
             if not sub or sub.endswith("/"):
                 continue
+            if ".." in sub.split("/") or sub.startswith("/"):
+                raise ModelProviderError(
+                    f"refusing traversal in object key: {key!r}")
             dst = os.path.join(dst_root, sub)
             os.makedirs(os.path.dirname(dst), exist_ok=True)
             r = self._get(self._url(key), stream=True)

@@ -64,11 +64,19 @@ struct Reader {
     throw FastFallback("bad varint");
   }
 
+  // bounds check before consuming n length-delimited bytes: a
+  // truncated-but-internally-consistent request must throw here, never
+  // read past the buffer (remote OOB read otherwise)
+  uint64_t need(uint64_t n) {
+    if (n > uint64_t(end - p)) throw FastFallback("truncated field");
+    return n;
+  }
+
   void skip(int wt) {
     switch (wt) {
       case 0: varint(); break;
       case 1: p += 8; break;
-      case 2: { uint64_t n = varint(); p += n; break; }
+      case 2: { uint64_t n = need(varint()); p += n; break; }
       case 5: p += 4; break;
       default: throw FastFallback("bad wire type");
     }
@@ -93,13 +101,13 @@ static ParsedTensor parse_tensor(const uint8_t* p, const uint8_t* end) {
     if (fno == 1 && wt == 0) {
       t.dtype = int(r.varint());
     } else if (fno == 2 && wt == 2) {            // tensor_shape
-      uint64_t n = r.varint();
+      uint64_t n = r.need(r.varint());
       Reader rs{r.p, r.p + n};
       r.p += n;
       while (rs.p < rs.end) {
         uint64_t stag = rs.varint();
         if ((stag >> 3) == 2 && (stag & 7) == 2) {   // dim
-          uint64_t dn = rs.varint();
+          uint64_t dn = rs.need(rs.varint());
           Reader rd{rs.p, rs.p + dn};
           rs.p += dn;
           int64_t size = 0;
@@ -116,7 +124,7 @@ static ParsedTensor parse_tensor(const uint8_t* p, const uint8_t* end) {
         }
       }
     } else if (fno == 4 && wt == 2) {            // tensor_content
-      uint64_t n = r.varint();
+      uint64_t n = r.need(r.varint());
       t.content = r.p;
       t.content_len = size_t(n);
       r.p += n;
@@ -142,7 +150,7 @@ static ParsedRequest parse_request(const uint8_t* p, size_t len) {
     uint64_t tag = r.varint();
     int fno = int(tag >> 3), wt = int(tag & 7);
     if (fno == 2 && wt == 2) {        // inputs map entry
-      uint64_t n = r.varint();
+      uint64_t n = r.need(r.varint());
       Reader re{r.p, r.p + n};
       r.p += n;
       std::string key;
@@ -151,11 +159,11 @@ static ParsedRequest parse_request(const uint8_t* p, size_t len) {
       while (re.p < re.end) {
         uint64_t etag = re.varint();
         if ((etag >> 3) == 1 && (etag & 7) == 2) {
-          uint64_t kn = re.varint();
+          uint64_t kn = re.need(re.varint());
           key.assign(reinterpret_cast<const char*>(re.p), kn);
           re.p += kn;
         } else if ((etag >> 3) == 2 && (etag & 7) == 2) {
-          uint64_t vn = re.varint();
+          uint64_t vn = re.need(re.varint());
           vptr = re.p;
           vlen = size_t(vn);
           re.p += vn;
@@ -165,7 +173,7 @@ static ParsedRequest parse_request(const uint8_t* p, size_t len) {
       }
       if (vptr) req.inputs[key] = parse_tensor(vptr, vptr + vlen);
     } else if (fno == 3 && wt == 2) {  // output_filter
-      uint64_t n = r.varint();
+      uint64_t n = r.need(r.varint());
       req.output_filter.emplace_back(
           reinterpret_cast<const char*>(r.p), n);
       r.p += n;
@@ -328,6 +336,12 @@ class FastModel {
       else if (batch != kv.second.dims[0])
         throw FastFallback("inconsistent batch");
     }
+    if (batch < 0 || batch > (1 << 20))
+      throw FastFallback("batch out of range");
+    for (auto& kv : req.inputs)
+      for (int64_t d : kv.second.dims)
+        if (d < 0 || d > (1 << 24))
+          throw FastFallback("dim out of range");
     std::lock_guard<std::mutex> g(mu_);
     if (specs_in_.empty()) throw FastFallback("no contexts yet");
     for (auto& io : specs_in_) {
@@ -658,18 +672,18 @@ bool peek_spec_raw(const uint8_t* p, size_t len, std::string* name,
   while (r.p < r.end) {
     uint64_t tag = r.varint();
     if ((tag >> 3) == 1 && (tag & 7) == 2) {
-      uint64_t n = r.varint();
+      uint64_t n = r.need(r.varint());
       Reader rs{r.p, r.p + n};
       r.p += n;
       while (rs.p < rs.end) {
         uint64_t stag = rs.varint();
         int fno = int(stag >> 3), wt = int(stag & 7);
         if (fno == 1 && wt == 2) {
-          uint64_t kn = rs.varint();
+          uint64_t kn = rs.need(rs.varint());
           name->assign(reinterpret_cast<const char*>(rs.p), kn);
           rs.p += kn;
         } else if (fno == 2 && wt == 2) {          // Int64Value version
-          uint64_t vn = rs.varint();
+          uint64_t vn = rs.need(rs.varint());
           Reader rv{rs.p, rs.p + vn};
           rs.p += vn;
           while (rv.p < rv.end) {
@@ -682,7 +696,7 @@ bool peek_spec_raw(const uint8_t* p, size_t len, std::string* name,
             }
           }
         } else if (fno == 4 && wt == 2) {
-          uint64_t kn = rs.varint();
+          uint64_t kn = rs.need(rs.varint());
           label->assign(reinterpret_cast<const char*>(rs.p), kn);
           rs.p += kn;
         } else {

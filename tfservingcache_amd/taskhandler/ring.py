@@ -42,7 +42,10 @@ class ConsistentHashRing:
             at = {}
             for mem in set(members):
                 for i in range(self.vnodes):
-                    h = _hash(f"{mem}{i}")
+                    # '#' separator so members differing only in trailing
+                    # digits can't produce colliding vnode keys
+                    # (e.g. 'gpu1'+'12' vs 'gpu11'+'2')
+                    h = _hash(f"{mem}#{i}")
                     at[h] = mem
                     hashes.append(h)
             hashes.sort()

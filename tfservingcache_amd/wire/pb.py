@@ -48,6 +48,8 @@ def read_varint(data: bytes, pos: int) -> Tuple[int, int]:
     result = 0
     shift = 0
     while True:
+        if pos >= len(data):
+            raise ValueError("truncated varint")
         b = data[pos]
         pos += 1
         result |= (b & 0x7F) << shift
@@ -76,11 +78,15 @@ def skip_field(data: bytes, pos: int, wire_type: int) -> int:
         pos += 8
     elif wire_type == WIRE_LEN:
         n, pos = read_varint(data, pos)
+        if pos + n > len(data):
+            raise ValueError("truncated length-delimited field")
         pos += n
     elif wire_type == WIRE_FIXED32:
         pos += 4
     else:
         raise ValueError(f"unsupported wire type {wire_type}")
+    if pos > len(data):
+        raise ValueError("truncated field")
     return pos
 
 
@@ -172,6 +178,8 @@ def _decode_scalar(kind: str, data: bytes, pos: int, wire_type: int):
         return _u64.unpack_from(data, pos)[0], pos + 8
     if wire_type == WIRE_LEN:
         n, pos = read_varint(data, pos)
+        if pos + n > len(data):
+            raise ValueError("truncated length-delimited field")
         if kind == "string":
             raw = data[pos:pos + n]
             pos += n
@@ -315,6 +323,8 @@ class Message(metaclass=MessageMeta):
                 continue
             if f.kind == "map":
                 n, pos = read_varint(data, pos)
+                if pos + n > end:
+                    raise ValueError("truncated map entry")
                 entry_end = pos + n
                 k, v = "", None
                 epos = pos
@@ -336,6 +346,8 @@ class Message(metaclass=MessageMeta):
                 pos = entry_end
             elif f.kind == "message":
                 n, pos = read_varint(data, pos)
+                if pos + n > end:
+                    raise ValueError("truncated nested message")
                 sub = f.msg_cls.decode(data, pos, pos + n)
                 pos += n
                 if f.repeated:
