@@ -13,7 +13,16 @@ decode -> CacheManager -> engine -> protobuf encode; TCP loopback
 excluded). A "step" is REQS_PER_STEP completed requests per rank.
 
 Modes (--mode):
-  warm  ResNet-50 warm-cache predict loop (BASELINE configs[1]; default)
+  headline  (default) the BASELINE.json headline workload: 1000
+        hardlinked ResNet-50 models, per-GPU pool capped at 10, Zipf-1.1
+        access, served over the NATIVE socket front-end (real gRPC over
+        TCP loopback). Emits BOTH metric halves: LRU predict req/sec
+        (the timed region) and p50 cold-load latency, plus a warm-cache
+        req/sec measured separately BEFORE the timed region. With
+        world>1 it becomes the ring workload (configs[3]): ownership by
+        consistent hash over ranks, replicas=2, replica fan-out over
+        the RCCL/xGMI plane at preload.
+  warm  ResNet-50 warm-cache predict loop (BASELINE configs[1])
   lru   --models N ResNet-50s, pool capped at --pool-size, Zipf access:
         the evict/reload path (BASELINE configs[2]); cold-load latencies
         recorded from the pool.
@@ -159,18 +168,21 @@ def main() -> int:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--mode", choices=["warm", "lru", "ring"],
-                    default="warm")
+    ap.add_argument("--mode",
+                    choices=["headline", "warm", "lru", "ring"],
+                    default="headline")
     ap.add_argument("--model", choices=["resnet50", "bert_base", "mixed"],
                     default="resnet50")
     ap.add_argument("--replicas", type=int, default=2,
                     help="replicasPerModel (ring mode)")
     ap.add_argument("--seq-len", type=int, default=128,
                     help="BERT sequence length")
-    ap.add_argument("--batch", type=int, default=16,
-                    help="images per predict request")
-    ap.add_argument("--models", type=int, default=100,
-                    help="models in the repo (lru mode)")
+    ap.add_argument("--batch", type=int, default=None,
+                    help="images per predict request "
+                         "(default: 8 headline, 16 otherwise)")
+    ap.add_argument("--models", type=int, default=None,
+                    help="models in the repo (default: 1000 headline, "
+                         "100 lru/ring)")
     ap.add_argument("--pool-size", type=int, default=10,
                     help="models resident per GPU (lru mode)")
     ap.add_argument("--image-size", type=int, default=224)
@@ -193,7 +205,7 @@ def main() -> int:
                          "the server without the client's GIL in the "
                          "way")
     ap.add_argument("--transport", choices=["inproc", "grpc", "native"],
-                    default="inproc",
+                    default=None,
                     help="inproc: gRPC message path without sockets; "
                          "grpc: Python grpcio server over TCP loopback; "
                          "native: C++ nghttp2 gRPC front-end over TCP "
@@ -203,6 +215,19 @@ def main() -> int:
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    # headline = the BASELINE workload; resolves to the LRU-churn path
+    # on one rank and the ring/replica path on many, over real sockets
+    headline = args.mode == "headline"
+    if args.batch is None:
+        args.batch = 8 if headline else 16
+    if args.models is None:
+        args.models = 1000 if headline else 100
+    if args.transport is None:
+        args.transport = "native" if headline else "inproc"
+    eff_mode = args.mode
+    if headline:
+        eff_mode = "lru" if world == 1 else "ring"
 
     dist = None
     torch = None
@@ -228,7 +253,7 @@ def main() -> int:
     tmp = tempfile.mkdtemp(prefix=f"tfsc_bench_r{rank}_")
     repo = os.path.join(tmp, "repo")
     cache_dir = os.path.join(tmp, "cache")
-    n_models = args.models if args.mode in ("lru", "ring") else 1
+    n_models = args.models if eff_mode in ("lru", "ring") else 1
     named = build_repo(repo, n_models, args.image_size, args.model,
                        args.seq_len)
     names = [n for n, _k in named]
@@ -236,7 +261,7 @@ def main() -> int:
 
     provider = DiskModelProvider(repo)
     cache = LRUCache(cache_dir, max_size_bytes=200 * 10 ** 9)
-    pool_cap = args.pool_size if args.mode in ("lru", "ring") else 4
+    pool_cap = args.pool_size if eff_mode in ("lru", "ring") else 4
     if args.cpu:
         loader = make_cpu_loader(cache)
     else:
@@ -276,7 +301,8 @@ def main() -> int:
     probs_all /= probs_all.sum()
     order = rng.permutation(n_models)
 
-    if args.mode == "ring":
+    warm_name = None
+    if eff_mode == "ring":
         # ring over ranks; this rank serves the models it owns
         from tfservingcache_amd.taskhandler import (ConsistentHashRing,
                                                     model_key)
@@ -313,14 +339,17 @@ def main() -> int:
 
         def pick(i):
             return owned_names[sched[i % 65536]]
-    elif args.mode == "lru":
+        warm_name = owned_names[0]
+    elif eff_mode == "lru":
         sched = rng.choice(np.arange(n_models), size=65536, p=probs_all)
 
         def pick(i):
             return names[order[sched[i % 65536]]]
+        warm_name = names[order[sched[0]]]
     else:
         def pick(i):
             return names[0]
+        warm_name = names[0]
 
     lat_lock = threading.Lock()
     latencies = []          # per-request wall seconds (timed section)
@@ -393,11 +422,31 @@ def main() -> int:
 
     pool_executor = ThreadPoolExecutor(max_workers=args.threads)
 
+    # headline: measure the WARM-cache half first (one resident model,
+    # same transport), strictly outside the LRU timed region
+    warm_req_per_sec = None
+    if headline:
+        if args.transport in ("grpc", "native"):
+            def warm_one(i: int) -> None:
+                rpcs[i % len(rpcs)](req_cache[warm_name], timeout=300)
+        else:
+            def warm_one(i: int) -> None:
+                handler.predict_bytes(req_cache[warm_name])
+        warm_one(0)                       # ensure resident + contexts
+        for _ in range(3):                # context warm-up passes
+            list(pool_executor.map(warm_one, range(50)))
+        wt0 = time.monotonic()
+        wn = 0
+        while time.monotonic() - wt0 < 2.0:
+            list(pool_executor.map(warm_one, range(100)))
+            wn += 100
+        warm_req_per_sec = wn / (time.monotonic() - wt0)
+
     client_conns = []
     client_procs = []
     if args.client_procs > 0:
         if args.transport not in ("grpc", "native") or \
-                args.mode != "warm":
+                eff_mode != "warm":
             raise SystemExit("--client-procs needs warm mode and a "
                              "socket transport")
         import multiprocessing as mp
@@ -510,8 +559,14 @@ def main() -> int:
                 "seq_len": args.seq_len if args.model != "resnet50" else args.image_size,
                 "parallelism": f"ring-sharded serving x{world}" +
                                (f", replicas={args.replicas}"
-                                if args.mode == "ring" else ""),
-                "mode": args.mode,
+                                if eff_mode == "ring" else ""),
+                "mode": (f"headline ({eff_mode})" if headline
+                         else args.mode),
+                "warm_req_per_sec": (round(warm_req_per_sec, 2)
+                                     if warm_req_per_sec else None),
+                "warm_images_per_sec": (
+                    round(warm_req_per_sec * args.batch, 1)
+                    if warm_req_per_sec else None),
                 "requests_per_step": REQS_PER_STEP,
                 "batch_per_request": args.batch,
                 "threads": args.threads,
