@@ -53,6 +53,10 @@ class CacheManager:
         # still queued on a stale lock object would let a later thread
         # create a fresh lock and fetch the same model concurrently)
         self._in_flight: Dict[ModelId, list] = {}
+        # fired after a provider fetch lands on disk (before the pool
+        # load) — main.py uses it to push the bytes to the model's other
+        # replica slots over the RCCL plane
+        self.on_cold_load: Optional[Callable[[str, int], None]] = None
 
     # -- the hot path ------------------------------------------------------
     def ensure_loaded(self, name: str, version: int) -> LoadedModel:
@@ -96,6 +100,11 @@ class CacheManager:
                     entry = self.provider.load_model(name, version,
                                                      self.cache.base_dir)
                     self.cache.put(entry)
+                    if self.on_cold_load is not None:
+                        try:
+                            self.on_cold_load(name, version)
+                        except Exception:       # noqa: BLE001
+                            log.exception("on_cold_load hook failed")
                 # a concurrent reload may evict this model between our
                 # reload and the wait — retry a bounded number of times
                 deadline = time.monotonic() + self.fetch_timeout

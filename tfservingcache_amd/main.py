@@ -19,6 +19,7 @@ from __future__ import annotations
 
 import asyncio
 import logging
+import os
 import socket
 import threading
 import time
@@ -124,14 +125,17 @@ def _gpu_devices(cfg: Config) -> List[str]:
     return [f"cuda:{i}" for i in range(n)]
 
 
-def create_cache_manager(cfg: Config) -> CacheManager:
+def create_cache_manager(cfg: Config,
+                         devices: Optional[List[str]] = None
+                         ) -> CacheManager:
     provider = create_model_provider(cfg)
     cache = LRUCache(
         cfg.get_string("modelCache.hostModelPath") or "./models",
         max_size_bytes=cfg.get_int("modelCache.size") or 10 ** 9,
         rebuild_from_disk=cfg.get_bool("modelCache.rebuildFromDisk"))
     max_models = cfg.get_int("serving.maxConcurrentModels") or 2
-    devices = _gpu_devices(cfg)
+    if devices is None:
+        devices = _gpu_devices(cfg)
     if devices:
         # models are hash-sharded over the node's visible GPUs (the
         # single-process analog of per-GPU ring slots; for one process
@@ -166,9 +170,31 @@ def create_cache_manager(cfg: Config) -> CacheManager:
 
 
 class Server:
+    """One serving process. In multi-GPU production each GPU gets its
+    own process (torchrun / torch.distributed.run, one rank per GPU):
+    every rank registers a SLOT-TAGGED ring member (gpu<local_rank>)
+    with its own 4 ports (base + local_rank * engine.portStride), so
+    the ring's keys land on (node, GPU) slots — SURVEY §2.4. When
+    proxy.replicasPerModel > 1 the cold-load path pushes model bytes
+    to the other replica slots over the RCCL/xGMI plane instead of
+    each replica re-fetching from the model store."""
+
     def __init__(self, cfg: Config):
         self.cfg = cfg
-        self.cm = create_cache_manager(cfg)
+        self._plane = None
+        self._rank = int(os.environ.get("RANK", "0"))
+        self._world = int(os.environ.get("WORLD_SIZE", "1"))
+        self._local_rank = int(os.environ.get("LOCAL_RANK",
+                                              str(self._rank)))
+        self._slot = ""
+        devices = None
+        plane_cfg = (cfg.get_string("engine.replicaPlane") or "").lower()
+        self._plane_enabled = (self._world > 1 and
+                               plane_cfg not in ("false", "0", "no"))
+        if self._plane_enabled:
+            self._slot = f"gpu{self._local_rank}"
+            devices = self._init_distributed(cfg)
+        self.cm = create_cache_manager(cfg, devices=devices)
         from .utils import metrics as mt
         mt.engine_stages.add_pool(self.cm.pool)
         self.handler = LocalServingHandler(self.cm)
@@ -187,6 +213,29 @@ class Server:
         self.cache_grpc_port = cfg.get_int("cacheGrpcPort") or 8095
         self.proxy_rest_port = cfg.get_int("proxyRestPort") or 8093
         self.proxy_grpc_port = cfg.get_int("proxyGrpcPort") or 8100
+        if self._plane_enabled:
+            # per-GPU process: distinct ports per local rank
+            stride = cfg.get_int("engine.portStride") or 10
+            off = self._local_rank * stride
+            self.cache_rest_port += off
+            self.cache_grpc_port += off
+            self.proxy_rest_port += off
+            self.proxy_grpc_port += off
+
+    def _init_distributed(self, cfg: Config) -> Optional[List[str]]:
+        """One-process-per-GPU: init torch.distributed (nccl == RCCL on
+        ROCm when GPUs are visible, gloo otherwise) and pin this process
+        to its GPU. Returns the device list override for the loader."""
+        import torch
+        import torch.distributed as dist
+        if not dist.is_initialized():
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            dist.init_process_group(backend=backend)
+        if torch.cuda.is_available():
+            dev = self._local_rank % torch.cuda.device_count()
+            torch.cuda.set_device(dev)
+            return [f"cuda:{dev}"]
+        return []
 
     def is_healthy(self) -> bool:
         probe = self.cfg.get_string("healthProbe.modelName")
@@ -230,8 +279,11 @@ class Server:
             host = self.cfg.get_string("proxy.advertiseHost") or \
                 socket.gethostbyname(socket.gethostname())
             self._self_service = ServingService(
-                host, self.cache_rest_port, self.cache_grpc_port)
+                host, self.cache_rest_port, self.cache_grpc_port,
+                slot=self._slot)
             self.cluster.connect(self._self_service)
+            if self._plane_enabled:
+                self._wire_replica_plane(replicas)
             # warm handoff on ring changes: when membership shifts and a
             # locally-cached model's ownership moves away from this node,
             # nudge a new owner to load it NOW (a GetModelMetadata call —
@@ -294,6 +346,50 @@ class Server:
                 self._runners.append(runner)
         self._loop.run_until_complete(boot())
         self._loop.run_forever()
+
+    def _wire_replica_plane(self, replicas: int) -> None:
+        """Cold-load fan-out: after this rank fetches a model from the
+        provider, push its bytes over the RCCL/xGMI plane to the other
+        owner slots of the ring; receivers register the files in their
+        LRU so their first request skips the provider."""
+        from .cachemanager.lrucache import Model as CacheModel
+        from .parallel.plane_service import PlaneService
+        from .taskhandler.cluster import model_key
+
+        cache = self.cm.cache
+
+        def on_receive(name, version, vdir, total):
+            if not cache.contains(name, version):
+                cache.put(CacheModel(
+                    name=name, version=version,
+                    path=os.path.join(name, str(version)),
+                    size_on_disk=total))
+
+        self._plane = PlaneService(cache.base_dir,
+                                   on_receive=on_receive)
+        self._plane.announce_member(self._self_service.serialize())
+        if replicas <= 1:
+            return
+        self_id = self._self_service.serialize()
+
+        def on_cold_load(name, version):
+            owners = self.cluster.find_nodes_for_key(
+                model_key(name, version))
+            ids = [o.serialize() for o in owners]
+            if self_id not in ids:
+                return                  # serving off-ring; no fan-out
+            dsts = []
+            for oid in ids:
+                if oid == self_id:
+                    continue
+                r = self._plane.rank_of_member(oid)
+                if r is not None:
+                    dsts.append(r)
+            if dsts:
+                vdir = os.path.join(cache.base_dir, name, str(version))
+                self._plane.push_files_async(name, version, vdir, dsts)
+
+        self.cm.on_cold_load = on_cold_load
 
     def _wire_native_registry(self, native) -> None:
         """Pool lifecycle -> C++ front-end FastModel registry."""
@@ -359,6 +455,8 @@ class Server:
 
     def stop(self) -> None:
         self._stop.set()
+        if self._plane is not None:
+            self._plane.stop()
         if self.cluster is not None:
             self.cluster.disconnect()
         for s in self._grpc_servers:
