@@ -16,6 +16,7 @@ GPU box; no silent eager fallback).
 """
 from __future__ import annotations
 
+import contextlib
 import logging
 import threading
 from dataclasses import dataclass
@@ -83,6 +84,105 @@ def _get_staging(torch):
         _staging_buf = torch.empty(32 * 1024 * 1024, dtype=torch.bfloat16,
                                    pin_memory=True)
     return _staging_buf
+
+
+# ---------------------------------------------------------------------------
+# capture/upload coordination.
+#
+# Round-1 finding (ROADMAP item 8): pinned host allocation (hipHostMalloc)
+# and DMAs sourced from freshly-allocated host blobs, issued WHILE another
+# thread is inside hipStreamBeginCapture..EndCapture, poison the capture —
+# replay then fails with "previous error during capture" — even though
+# captures use hipStreamCaptureModeThreadLocal. Loads and captures overlap
+# by design under LRU churn, so the two operation classes are serialized
+# process-wide: any number of concurrent captures OR any number of
+# concurrent unsafe host-memory ops, never both. Captures are ~ms and
+# uploads ~10-100 ms, so the serialization cost is noise next to the
+# ~50-100 ms saved per duplicate-content cold load by the blob cache.
+# ---------------------------------------------------------------------------
+class _CaptureGuard:
+    def __init__(self):
+        self._cv = threading.Condition()
+        self._captures = 0
+        self._unsafe = 0
+
+    @contextlib.contextmanager
+    def capture(self):
+        with self._cv:
+            while self._unsafe:
+                self._cv.wait()
+            self._captures += 1
+        try:
+            yield
+        finally:
+            with self._cv:
+                self._captures -= 1
+                self._cv.notify_all()
+
+    @contextlib.contextmanager
+    def unsafe_host_op(self):
+        with self._cv:
+            while self._captures:
+                self._cv.wait()
+            self._unsafe += 1
+        try:
+            yield
+        finally:
+            with self._cv:
+                self._unsafe -= 1
+                self._cv.notify_all()
+
+
+capture_guard = _CaptureGuard()
+
+
+# converted-weight cache: pinned bf16 blobs keyed by Plan identity.
+# Plans are content-deduplicated (engine/model.py inode-keyed plan
+# cache), so models sharing a SavedModel's bytes share a Plan — their
+# cold loads skip the CPU f32->bf16 convert and become one DMA.
+# Bounded: ~51 MB pinned per ResNet-50-sized plan.
+_BLOB_CACHE_CAP = 4
+_blob_cache_lock = threading.Lock()
+_blob_cache: Dict[int, tuple] = {}       # id(plan) -> (plan, blob)
+_blob_cache_order: List[int] = []
+
+
+def _blob_cache_get(plan):
+    with _blob_cache_lock:
+        ent = _blob_cache.get(id(plan))
+        if ent is None or ent[0] is not plan:
+            return None
+        _blob_cache_order.remove(id(plan))
+        _blob_cache_order.append(id(plan))
+        return ent[1]
+
+
+def _blob_cache_put(plan, blob) -> None:
+    with _blob_cache_lock:
+        key = id(plan)
+        if key not in _blob_cache:
+            _blob_cache_order.append(key)
+        # keep the Plan object referenced so id() stays unambiguous
+        _blob_cache[key] = (plan, blob)
+        while len(_blob_cache_order) > _BLOB_CACHE_CAP:
+            old = _blob_cache_order.pop(0)
+            _blob_cache.pop(old, None)
+
+
+# dedicated per-device upload streams: the cached-blob DMA must not run
+# on the default (legacy) stream, whose implicit cross-stream semantics
+# interact with capturing streams
+_upload_streams: Dict[str, object] = {}
+_upload_streams_lock = threading.Lock()
+
+
+def _get_upload_stream(torch, device: str):
+    with _upload_streams_lock:
+        s = _upload_streams.get(device)
+        if s is None:
+            s = torch.cuda.Stream(device=device)
+            _upload_streams[device] = s
+        return s
 
 
 def _pad64(k: int) -> int:
@@ -587,10 +687,13 @@ class ExecContext:
                 v.copy_(pin, non_blocking=True)
             if self.gm.use_graphs and not self.captured:
                 # warm-up eager run, then capture on this stream
+                # (serialized against unsafe host-memory ops — see
+                # capture_guard)
                 self.exec_plan.run()
                 self.stream.synchronize()
                 try:
-                    self.exec_plan.capture()
+                    with capture_guard.capture():
+                        self.exec_plan.capture()
                 except Exception:       # noqa: BLE001
                     log.exception("hipGraph capture failed; staying eager")
                 self.captured = True
@@ -669,38 +772,63 @@ class GpuModel:
         if not float_ws:
             return
         blob = torch.empty(total, dtype=torch.bfloat16, device=self.device)
-        # stage through a SHARED reusable pinned buffer: per-load pinned
-        # allocation (page-locking ~100 MB) costs more than the copy and
-        # thrashes badly under LRU churn
-        with _staging_lock:
-            stage = _get_staging(torch)
-            cap = stage.numel()
-            batch_items = []        # (stage_off, blob_off, n)
-            stage_off = 0
-
-            def flush():
-                nonlocal stage_off
-                for s_off, b_off, n_ in batch_items:
-                    blob[b_off:b_off + n_].copy_(
-                        stage[s_off:s_off + n_], non_blocking=True)
-                torch.cuda.synchronize(self.device)
-                batch_items.clear()
+        cached = _blob_cache_get(self.plan)
+        if cached is not None and cached.numel() == total:
+            # converted blob already pinned in CPU RAM: ONE DMA on the
+            # dedicated upload stream, serialized against captures
+            up = _get_upload_stream(torch, self.device)
+            with capture_guard.unsafe_host_op():
+                with torch.cuda.stream(up):
+                    blob.copy_(cached, non_blocking=True)
+                up.synchronize()
+        else:
+            # stage through a SHARED reusable pinned buffer: per-load
+            # pinned allocation (page-locking ~100 MB) costs more than
+            # the copy and thrashes badly under LRU churn. The converted
+            # result is ALSO written to a cacheable pinned blob so the
+            # next load of this plan skips the CPU convert entirely.
+            cpu_blob = None
+            try:
+                with capture_guard.unsafe_host_op():
+                    cpu_blob = torch.empty(total, dtype=torch.bfloat16,
+                                           pin_memory=True)
+            except RuntimeError:
+                log.warning("pinned weight-blob allocation failed; "
+                            "skipping the converted-weight cache")
+            with _staging_lock:
+                stage = _get_staging(torch)
+                cap = stage.numel()
+                batch_items = []        # (stage_off, blob_off, n)
                 stage_off = 0
 
-            for _idx, w, off, n in float_ws:
-                src = torch.from_numpy(np.ascontiguousarray(
-                    np.asarray(w, dtype=np.float32))).view(-1)
-                done = 0
-                while done < n:
-                    if stage_off >= cap:
-                        flush()
-                    take = min(n - done, cap - stage_off)
-                    stage[stage_off:stage_off + take].copy_(
-                        src[done:done + take])   # CPU f32->bf16 convert
-                    batch_items.append((stage_off, off + done, take))
-                    stage_off += take
-                    done += take
-            flush()
+                def flush():
+                    nonlocal stage_off
+                    for s_off, b_off, n_ in batch_items:
+                        blob[b_off:b_off + n_].copy_(
+                            stage[s_off:s_off + n_], non_blocking=True)
+                    torch.cuda.synchronize(self.device)
+                    batch_items.clear()
+                    stage_off = 0
+
+                for _idx, w, off, n in float_ws:
+                    src = torch.from_numpy(np.ascontiguousarray(
+                        np.asarray(w, dtype=np.float32))).view(-1)
+                    done = 0
+                    while done < n:
+                        if stage_off >= cap:
+                            flush()
+                        take = min(n - done, cap - stage_off)
+                        stage[stage_off:stage_off + take].copy_(
+                            src[done:done + take])  # CPU f32->bf16
+                        if cpu_blob is not None:
+                            cpu_blob[off + done:off + done + take].copy_(
+                                stage[stage_off:stage_off + take])
+                        batch_items.append((stage_off, off + done, take))
+                        stage_off += take
+                        done += take
+                flush()
+            if cpu_blob is not None:
+                _blob_cache_put(self.plan, cpu_blob)
         self._weight_blob = blob              # keep the allocation alive
         for idx, w, off, n in float_ws:
             self._weights[idx] = blob[off:off + n].view(tuple(w.shape))
