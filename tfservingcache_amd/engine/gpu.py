@@ -102,16 +102,33 @@ def _get_staging(torch):
 # ---------------------------------------------------------------------------
 class _CaptureGuard:
     def __init__(self):
+        import os as _os
         self._cv = threading.Condition()
         self._captures = 0
         self._unsafe = 0
+        self.enabled = _os.environ.get(
+            "TFSC_CAPTURE_GUARD", "on").lower() not in ("off", "0", "no")
+
+    def _waited(self, t0: float, what: str) -> None:
+        import time as _time
+        dt = _time.monotonic() - t0
+        if dt > 0.25:
+            log.warning("capture-guard: %s waited %.2fs "
+                        "(captures=%d unsafe=%d)", what, dt,
+                        self._captures, self._unsafe)
 
     @contextlib.contextmanager
     def capture(self):
+        if not self.enabled:
+            yield
+            return
+        import time as _time
+        t0 = _time.monotonic()
         with self._cv:
             while self._unsafe:
                 self._cv.wait()
             self._captures += 1
+        self._waited(t0, "capture")
         try:
             yield
         finally:
@@ -121,10 +138,16 @@ class _CaptureGuard:
 
     @contextlib.contextmanager
     def unsafe_host_op(self):
+        if not self.enabled:
+            yield
+            return
+        import time as _time
+        t0 = _time.monotonic()
         with self._cv:
             while self._captures:
                 self._cv.wait()
             self._unsafe += 1
+        self._waited(t0, "unsafe-host-op")
         try:
             yield
         finally:
