@@ -106,9 +106,12 @@ class CacheManager:
                         except Exception:       # noqa: BLE001
                             log.exception("on_cold_load hook failed")
                 # a concurrent reload may evict this model between our
-                # reload and the wait — retry a bounded number of times
+                # reload and the wait, or its MRU-prefix truncation may
+                # exclude it entirely — wait in SHORT chunks and re-front
+                # the model in the MRU between chunks, so an eviction
+                # race costs ~1s instead of a full fetch-timeout chunk
                 deadline = time.monotonic() + self.fetch_timeout
-                attempt = 0
+                failures = 0
                 while True:
                     self.cache.get(name, version)     # keep it MRU
                     self._reload_pool()
@@ -119,14 +122,17 @@ class CacheManager:
                             f"{self.fetch_timeout}s")
                     try:
                         return self.pool.wait_available(
-                            name, version, min(remaining,
-                                               self.fetch_timeout / 4 + 1))
-                    except (TimeoutError, RuntimeError):
-                        # RuntimeError: the pool load failed (e.g. a
-                        # ReloadConfig raced the disk fetch) — the END
-                        # entry is re-created by the next reload
-                        attempt += 1
-                        if attempt >= 4:
+                            name, version, min(remaining, 1.0))
+                    except TimeoutError:
+                        continue        # re-front + reload, bounded by
+                                        # the deadline above
+                    except RuntimeError:
+                        # the pool load FAILED (e.g. a ReloadConfig
+                        # raced the disk fetch) — the END entry is
+                        # re-created by the next reload; genuine load
+                        # errors shouldn't loop forever
+                        failures += 1
+                        if failures >= 4:
                             raise
             finally:
                 with self._flight_lock:

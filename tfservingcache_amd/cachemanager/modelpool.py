@@ -217,15 +217,25 @@ class ModelPool:
         """Block until (name, version) is AVAILABLE; returns the model or
         raises TimeoutError / RuntimeError on failed load."""
         deadline = time.monotonic() + timeout
+        seen = False
         with self._lock:
             while True:
                 e = self._entries.get((name, version))
                 if e is not None:
+                    seen = True
                     if e.state == AVAILABLE:
                         return e.model
                     if e.state == END and e.error:
                         raise RuntimeError(
                             f"load of {name}:{version} failed: {e.error}")
+                elif seen:
+                    # the entry existed and vanished: it was unloaded
+                    # between becoming AVAILABLE and this waiter waking
+                    # (a concurrent reload's MRU truncation). Waiting
+                    # longer is pointless — time out NOW so the caller
+                    # re-fronts the model in the MRU and reloads.
+                    raise TimeoutError(
+                        f"model {name}:{version} evicted before use")
                 remaining = deadline - time.monotonic()
                 if remaining <= 0:
                     raise TimeoutError(
