@@ -246,3 +246,59 @@ def test_fast_predict_path_matches_python(tmp_path):
             float_val=[0.0] * 64)}).encode()
     with _pytest.raises(ext.FastFallback):
         gm._gpu.fast_predict(bad)
+
+
+def test_depthwise_gpu_vs_cpu(tmp_path):
+    """Depthwise kernel (vector + scalar channel paths) vs CPU fp32."""
+    from tfservingcache_amd.engine.savedmodel import GraphBuilder
+    rng = np.random.default_rng(11)
+    for C, stride in ((32, 1), (32, 2), (6, 1)):   # 6: scalar tail path
+        gb = GraphBuilder()
+        f32 = gb.a_type(1)
+        x_ph = gb.placeholder("input", np.float32, [-1, 14, 14, C],
+                              signature_name="input")
+        w = (rng.standard_normal((3, 3, C, 1)) * 0.3).astype(np.float32)
+        d = gb.node("DepthwiseConv2dNative", "dw", [x_ph, gb.const("w", w)],
+                    T=f32, strides=gb.a_ints([1, stride, stride, 1]),
+                    padding=gb.a_str("SAME"), data_format=gb.a_str("NHWC"))
+        r6 = gb.node("Relu6", "r6", [d], T=f32)
+        gb.mark_output("y", r6)
+        sm = gb.build()
+        name = f"dw{C}s{stride}"
+        gm = _gpu_model(tmp_path, sm, name=name)
+        cm = _cpu_model(tmp_path, sm, name=name + "cpu")
+        x = (rng.standard_normal((4, 14, 14, C)) * 0.8).astype(np.float32)
+        _compare(gm.predict({"input": x}), cm.predict({"input": x}),
+                 rtol=0.05, atol=0.03)
+
+
+def test_mobilenet_v2_gpu_vs_cpu(tmp_path):
+    from tfservingcache_amd.models.builders import build_mobilenet_v2
+    sm = build_mobilenet_v2(image_size=64, num_classes=100)
+    gm = _gpu_model(tmp_path, sm, name="mnv2")
+    cm = _cpu_model(tmp_path, sm, name="mnv2cpu")
+    x = (np.random.default_rng(2).standard_normal((4, 64, 64, 3))
+         * 0.5).astype(np.float32)
+    gout, cout = gm.predict({"input": x}), cm.predict({"input": x})
+    # probabilities compare tightly; argmax must agree
+    np.testing.assert_allclose(gout["probs"], cout["probs"], atol=0.04)
+    assert (gout["probs"].argmax(1) == cout["probs"].argmax(1)).all()
+
+
+def test_einsum_gpu_vs_cpu(tmp_path):
+    from tfservingcache_amd.engine.savedmodel import GraphBuilder
+    rng = np.random.default_rng(13)
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, [-1, 12, 256],
+                          signature_name="x")
+    w = (rng.standard_normal((256, 2, 64)) * 0.1).astype(np.float32)
+    e = gb.node("Einsum", "ein", [x_ph, gb.const("w", w)], T=f32,
+                equation=gb.a_str("abc,cde->abde"))
+    gb.mark_output("y", e)
+    sm = gb.build()
+    gm = _gpu_model(tmp_path, sm, name="ein")
+    cm = _cpu_model(tmp_path, sm, name="eincpu")
+    x = (rng.standard_normal((4, 12, 256)) * 0.5).astype(np.float32)
+    _compare(gm.predict({"x": x}), cm.predict({"x": x}),
+             rtol=0.05, atol=0.03)

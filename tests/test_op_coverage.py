@@ -1,0 +1,257 @@
+"""Round-2 op-coverage widening: DepthwiseConv2dNative
+(MobileNetV2 family), Einsum (transformer exports), Relu6, and
+DT_STRING handling. CPU numerics compare the plan execution against
+independent naive numpy computations (not the executor's own code
+paths)."""
+import numpy as np
+import pytest
+
+from tfservingcache_amd.engine.savedmodel import (GraphBuilder,
+                                                  write_saved_model)
+from tfservingcache_amd.engine.model import load_model_from_dir
+from tfservingcache_amd.models.builders import build_mobilenet_v2
+
+
+def _load(tmp_path, sm, name="m"):
+    d = str(tmp_path / name / "1")
+    write_saved_model(sm, d)
+    return load_model_from_dir(d, name, 1)
+
+
+# -- depthwise ---------------------------------------------------------------
+
+def _naive_depthwise(x, w, stride, padding):
+    """Independent reference: direct loops (no im2col, no executor code)."""
+    N, H, W_, C = x.shape
+    R, S, C2, M = w.shape
+    assert C2 == C and M == 1
+    if padding == "SAME":
+        Ho = -(-H // stride)
+        Wo = -(-W_ // stride)
+        ph = max((Ho - 1) * stride + R - H, 0)
+        pw = max((Wo - 1) * stride + S - W_, 0)
+        pt, pl = ph // 2, pw // 2
+    else:
+        Ho, Wo = (H - R) // stride + 1, (W_ - S) // stride + 1
+        pt = pl = 0
+    y = np.zeros((N, Ho, Wo, C), dtype=np.float64)
+    for n in range(N):
+        for ho in range(Ho):
+            for wo in range(Wo):
+                for r in range(R):
+                    for s in range(S):
+                        hi = ho * stride - pt + r
+                        wi = wo * stride - pl + s
+                        if 0 <= hi < H and 0 <= wi < W_:
+                            y[n, ho, wo] += x[n, hi, wi] * w[r, s, :, 0]
+    return y.astype(np.float32)
+
+
+@pytest.mark.parametrize("stride,padding", [(1, "SAME"), (2, "SAME"),
+                                            (1, "VALID")])
+def test_depthwise_numerics_vs_naive(tmp_path, stride, padding):
+    rng = np.random.default_rng(3)
+    C = 8
+    w = (rng.standard_normal((3, 3, C, 1)) * 0.3).astype(np.float32)
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("input", np.float32, [-1, 10, 10, C],
+                          signature_name="input")
+    wc = gb.const("w", w)
+    d = gb.node("DepthwiseConv2dNative", "dw", [x_ph, wc], T=f32,
+                strides=gb.a_ints([1, stride, stride, 1]),
+                padding=gb.a_str(padding),
+                data_format=gb.a_str("NHWC"))
+    gb.mark_output("y", d)
+    model = _load(tmp_path, gb.build())
+    assert [op.kind for op in model.plan.ops] == ["depthwise_conv"]
+    x = (rng.standard_normal((2, 10, 10, C)) * 0.5).astype(np.float32)
+    got = model.predict({"input": x})["y"]
+    want = _naive_depthwise(x, w, stride, padding)
+    np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)
+
+
+def test_depthwise_bn_relu6_folds(tmp_path):
+    rng = np.random.default_rng(5)
+    C = 8
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("input", np.float32, [-1, 6, 6, C],
+                          signature_name="input")
+    w = (rng.standard_normal((3, 3, C, 1)) * 0.5).astype(np.float32)
+    wc = gb.const("w", w)
+    d = gb.node("DepthwiseConv2dNative", "dw", [x_ph, wc], T=f32,
+                strides=gb.a_ints([1, 1, 1, 1]), padding=gb.a_str("SAME"),
+                data_format=gb.a_str("NHWC"))
+    scale = np.abs(rng.standard_normal(C)).astype(np.float32) + 0.5
+    offset = (rng.standard_normal(C) * 0.1).astype(np.float32)
+    mean = (rng.standard_normal(C) * 0.1).astype(np.float32)
+    var = np.abs(rng.standard_normal(C)).astype(np.float32) + 0.5
+    bn = gb.node("FusedBatchNormV3", "bn",
+                 [d, gb.const("g", scale), gb.const("b", offset),
+                  gb.const("m", mean), gb.const("v", var)],
+                 T=f32, U=f32, epsilon=gb.a_float(1e-3),
+                 is_training=gb.a_bool(False),
+                 data_format=gb.a_str("NHWC"))
+    r6 = gb.node("Relu6", "r6", [bn], T=f32)
+    gb.mark_output("y", r6)
+    model = _load(tmp_path, gb.build())
+    # BN + Relu6 folded into the depthwise op
+    kinds = [op.kind for op in model.plan.ops]
+    assert kinds == ["depthwise_conv"]
+    assert model.plan.ops[0].params["act"] == "relu6"
+    x = (rng.standard_normal((2, 6, 6, C)) * 2.0).astype(np.float32)
+    got = model.predict({"input": x})["y"]
+    raw = _naive_depthwise(x, w, 1, "SAME")
+    want = np.clip((raw - mean) / np.sqrt(var + 1e-3) * scale + offset,
+                   0.0, 6.0)
+    np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)
+    assert got.max() <= 6.0
+
+
+def test_mobilenet_v2_serves_cpu(tmp_path):
+    sm = build_mobilenet_v2(image_size=32, num_classes=10)
+    model = _load(tmp_path, sm)
+    kinds = [op.kind for op in model.plan.ops]
+    assert kinds.count("depthwise_conv") == 17
+    assert "bn_act" not in kinds          # all BN folded
+    x = (np.random.default_rng(0).standard_normal((2, 32, 32, 3))
+         * 0.5).astype(np.float32)
+    out = model.predict({"input": x})
+    assert out["probs"].shape == (2, 10)
+    np.testing.assert_allclose(out["probs"].sum(-1), [1, 1], rtol=1e-4)
+
+
+# -- einsum ------------------------------------------------------------------
+
+@pytest.mark.parametrize("eq,ashape,wshape", [
+    ("ij,jk->ik", (4, 6), (6, 5)),
+    ("abc,cd->abd", (-1, 3, 8), (8, 5)),
+    ("abc,cde->abde", (-1, 3, 8), (8, 2, 5)),
+    ("abcd,cde->abe", (-1, 3, 4, 6), (4, 6, 7)),
+])
+def test_einsum_dense_family(tmp_path, eq, ashape, wshape):
+    rng = np.random.default_rng(7)
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, list(ashape),
+                          signature_name="x")
+    w = (rng.standard_normal(wshape) * 0.3).astype(np.float32)
+    wc = gb.const("w", w)
+    e = gb.node("Einsum", "ein", [x_ph, wc], T=f32,
+                equation=gb.a_str(eq), N=gb.a_ints([2]))
+    gb.mark_output("y", e)
+    model = _load(tmp_path, gb.build())
+    assert [op.kind for op in model.plan.ops] == ["gemm"]
+    concrete = tuple(2 if d == -1 else d for d in ashape)
+    x = (rng.standard_normal(concrete) * 0.5).astype(np.float32)
+    got = model.predict({"x": x})["y"]
+    want = np.einsum(eq, x, w)
+    assert got.shape == want.shape
+    np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)
+
+
+def test_einsum_with_bias_fuses(tmp_path):
+    rng = np.random.default_rng(9)
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, [-1, 4, 8], signature_name="x")
+    w = (rng.standard_normal((8, 5)) * 0.3).astype(np.float32)
+    b = (rng.standard_normal(5) * 0.1).astype(np.float32)
+    e = gb.node("Einsum", "ein", [x_ph, gb.const("w", w)], T=f32,
+                equation=gb.a_str("abc,cd->abd"))
+    ba = gb.node("BiasAdd", "ba", [e, gb.const("b", b)], T=f32)
+    r = gb.node("Relu", "r", [ba], T=f32)
+    gb.mark_output("y", r)
+    model = _load(tmp_path, gb.build())
+    ops = model.plan.ops
+    assert [op.kind for op in ops] == ["gemm"]
+    assert ops[0].params.get("has_bias") and ops[0].params["act"] == "relu"
+    x = (rng.standard_normal((2, 4, 8)) * 0.5).astype(np.float32)
+    got = model.predict({"x": x})["y"]
+    want = np.maximum(np.einsum("abc,cd->abd", x, w) + b, 0.0)
+    np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)
+
+
+def test_einsum_unsupported_is_loud(tmp_path):
+    from tfservingcache_amd.engine.planner import PlanError
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, [-1, 4, 8], signature_name="x")
+    w = np.zeros((4, 8), dtype=np.float32)
+    e = gb.node("Einsum", "ein", [x_ph, gb.const("w", w)], T=f32,
+                equation=gb.a_str("abc,bc->ab"))   # K not trailing-only
+    gb.mark_output("y", e)
+    with pytest.raises(PlanError):
+        _load(tmp_path, gb.build())
+
+
+# -- relu6 standalone --------------------------------------------------------
+
+def test_relu6_eltwise(tmp_path):
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, [-1, 8], signature_name="x")
+    sq = gb.node("Square", "sq", [x_ph], T=f32)   # block act fusion
+    r6 = gb.node("Relu6", "r6", [sq], T=f32)
+    gb.mark_output("y", r6)
+    model = _load(tmp_path, gb.build())
+    x = np.linspace(-4, 4, 16, dtype=np.float32).reshape(2, 8)
+    got = model.predict({"x": x})["y"]
+    np.testing.assert_allclose(got, np.clip(x * x, 0, 6), rtol=1e-5)
+
+
+# -- DT_STRING ---------------------------------------------------------------
+
+def test_string_tensorproto_round_trip():
+    from tfservingcache_amd.wire import messages as m
+    from tfservingcache_amd.wire.tensor import (numpy_to_tensorproto,
+                                                tensorproto_to_numpy)
+    arr = np.empty((2, 2), dtype=object)
+    arr[0, 0], arr[0, 1] = b"hello", b"world"
+    arr[1, 0], arr[1, 1] = b"", bytes(range(256))
+    tp = numpy_to_tensorproto(arr)
+    assert tp.dtype == m.DT_STRING
+    enc = tp.encode()
+    got = tensorproto_to_numpy(m.TensorProto.decode(enc))
+    assert got.shape == (2, 2)
+    assert got[1, 1] == bytes(range(256))
+
+
+def test_string_predict_clean_error(tmp_path):
+    from tfservingcache_amd.cachemanager import (CacheManager, LRUCache,
+                                                 ModelPool,
+                                                 make_cpu_loader)
+    from tfservingcache_amd.cachemanager.providers import DiskModelProvider
+    from tfservingcache_amd.models import write_model_repo
+    from tfservingcache_amd.tfservingproxy import LocalServingHandler
+    from tfservingcache_amd.tfservingproxy.servinghandler import \
+        ServingError
+    from tfservingcache_amd.wire import messages as m
+    from tfservingcache_amd.wire.tensor import numpy_to_tensorproto
+
+    write_model_repo(str(tmp_path / "repo"), [("mlp", 1, "mlp")])
+    cache = LRUCache(str(tmp_path / "cache"), 10 ** 8)
+    pool = ModelPool(make_cpu_loader(cache), max_concurrent_models=2)
+    cm = CacheManager(DiskModelProvider(str(tmp_path / "repo")), cache,
+                      pool)
+    handler = LocalServingHandler(cm)
+    s = np.empty((1,), dtype=object)
+    s[0] = b"some-bytes"
+    req = m.PredictRequest(
+        model_spec=m.ModelSpec(name="mlp", version=m.Int64Value(value=1)),
+        inputs={"x": numpy_to_tensorproto(s)})
+    with pytest.raises(ServingError) as ei:
+        handler.predict(req)
+    assert ei.value.code == m.ERROR_INVALID_ARGUMENT
+    assert "DT_STRING" in str(ei.value)
+
+
+def test_rest_json_string_inputs_decode():
+    from tfservingcache_amd.tfservingproxy.json_codec import \
+        parse_predict_body
+    inputs, fmt, _sig = parse_predict_body(
+        {"instances": [{"text": "abc"}, {"text": {"b64": "aGk="}}]})
+    arr = inputs["text"]
+    assert arr.dtype == object
+    assert arr[0] == b"abc" and arr[1] == b"hi"

@@ -40,6 +40,7 @@ enum CallKind : int {
   K_CONV,
   K_PAD_LAST,
   K_ATTENTION,
+  K_DEPTHWISE,
 };
 
 struct Call {
@@ -140,6 +141,15 @@ static void launch_call(const Call& c, hipStream_t s) {
       // ptrs: [q, k, v, out]; ints: [B, S, H, D]; floats: [scale]
       launch_attention(s, cp(0), cp(1), cp(2), p(3), int(I[0]), int(I[1]),
                        int(I[2]), int(I[3]), c.floats[0]);
+      break;
+    case K_DEPTHWISE:
+      // ptrs: [x, w, bias, y]
+      // ints: [N,H,W,C,R,S,sh,sw,pt,pl,Ho,Wo,act]
+      launch_depthwise_conv(s, cp(0), cp(1), cp(2), p(3), int(I[0]),
+                            int(I[1]), int(I[2]), int(I[3]), int(I[4]),
+                            int(I[5]), int(I[6]), int(I[7]), int(I[8]),
+                            int(I[9]), int(I[10]), int(I[11]),
+                            int(I[12]));
       break;
     default:
       throw std::runtime_error("unknown call kind " +
@@ -276,6 +286,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("K_CONV") = int(tfsc::K_CONV);
   mod.attr("K_PAD_LAST") = int(tfsc::K_PAD_LAST);
   mod.attr("K_ATTENTION") = int(tfsc::K_ATTENTION);
+  mod.attr("K_DEPTHWISE") = int(tfsc::K_DEPTHWISE);
 
   register_fastpath(mod);
   register_frontend(mod);
@@ -298,10 +309,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("ELT_NEG") = int(tfsc::ELT_NEG);
   mod.attr("ELT_SQUARE") = int(tfsc::ELT_SQUARE);
   mod.attr("ELT_GELU") = int(tfsc::ELT_GELU);
+  mod.attr("ELT_RELU6") = int(tfsc::ELT_RELU6);
 
   mod.attr("ACT_NONE") = int(tfsc::ACT_NONE);
   mod.attr("ACT_RELU") = int(tfsc::ACT_RELU);
   mod.attr("ACT_TANH") = int(tfsc::ACT_TANH);
   mod.attr("ACT_SIGMOID") = int(tfsc::ACT_SIGMOID);
   mod.attr("ACT_GELU") = int(tfsc::ACT_GELU);
+  mod.attr("ACT_RELU6") = int(tfsc::ACT_RELU6);
 }

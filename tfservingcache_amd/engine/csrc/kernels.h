@@ -12,11 +12,12 @@ namespace tfsc {
 enum EltFn : int {
   ELT_ADD = 0, ELT_SUB, ELT_MUL, ELT_DIV, ELT_MAX, ELT_MIN, ELT_SQDIFF,
   ELT_RELU, ELT_TANH, ELT_SIGMOID, ELT_ERF, ELT_SQRT, ELT_RSQRT, ELT_EXP,
-  ELT_NEG, ELT_SQUARE, ELT_GELU,
+  ELT_NEG, ELT_SQUARE, ELT_GELU, ELT_RELU6,
 };
 
 // activation codes for fused epilogues
-enum Act : int { ACT_NONE = 0, ACT_RELU, ACT_TANH, ACT_SIGMOID, ACT_GELU };
+enum Act : int { ACT_NONE = 0, ACT_RELU, ACT_TANH, ACT_SIGMOID, ACT_GELU,
+                 ACT_RELU6 };
 
 constexpr int MAX_DIMS = 6;
 
@@ -102,6 +103,18 @@ void launch_im2col(hipStream_t s, const ushort* x, ushort* y,
                    int N, int H, int W, int C, int R, int S,
                    int sh, int sw, int pt, int pl, int Ho, int Wo,
                    int k_pad);
+
+// Depthwise NHWC conv (depth_multiplier == 1, MobileNet-class):
+//   y[n,ho,wo,c] = act(sum_{r,s} x[n, ho*sh-pt+r, wo*sw-pl+s, c] * w[r,s,c]
+//                      + bias[c])
+// Memory-bound (no MFMA shape): vectorized 4x bf16 over the contiguous
+// channel dim, grid-stride over N*Ho*Wo*C/4. w is the [R,S,C,1] master
+// read as flat [R*S*C].
+void launch_depthwise_conv(hipStream_t s, const ushort* x, const ushort* w,
+                           const ushort* bias, ushort* y,
+                           int N, int H, int W, int C, int R, int S,
+                           int sh, int sw, int pt, int pl, int Ho, int Wo,
+                           int act);
 
 // Fused multi-head attention over the natural [B*S, H*D] QKV layout
 // (flash-style online softmax; D must be 64). See ops/attention.hip.

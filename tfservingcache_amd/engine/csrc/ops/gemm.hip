@@ -124,6 +124,7 @@ TFSC_DEV float act_apply(float v, int act) {
     case ACT_TANH: return tanhf(v);
     case ACT_SIGMOID: return 1.f / (1.f + __expf(-v));
     case ACT_GELU: return 0.5f * v * (1.f + erff(v * 0.70710678f));
+    case ACT_RELU6: return v < 0.f ? 0.f : (v > 6.f ? 6.f : v);
     default: return v;
   }
 }

@@ -35,6 +35,7 @@ TFSC_DEV float apply_unary(float v, int fn) {
     case ELT_NEG: return -v;
     case ELT_SQUARE: return v * v;
     case ELT_GELU: return 0.5f * v * (1.f + erff(v * 0.70710678f));
+    case ELT_RELU6: return v < 0.f ? 0.f : (v > 6.f ? 6.f : v);
     default: return v;
   }
 }
@@ -153,6 +154,7 @@ __global__ void k_bn_act(const ushort* __restrict__ x,
       case ACT_TANH: v = tanhf(v); break;
       case ACT_SIGMOID: v = 1.f / (1.f + __expf(-v)); break;
       case ACT_GELU: v = 0.5f * v * (1.f + erff(v * 0.70710678f)); break;
+      case ACT_RELU6: v = v < 0.f ? 0.f : (v > 6.f ? 6.f : v); break;
     }
     y[i] = f2bf(v);
   }
@@ -655,6 +657,118 @@ void launch_pad_nhwc(hipStream_t s, const ushort* x, ushort* y,
   int64_t n_out = (int64_t)N * Ho * Wo * C;
   hipLaunchKernelGGL(k_pad_nhwc, dim3(grid_for(n_out)), dim3(TPB), 0, s,
                      x, y, N, H, W, C, pt, pl, Ho, Wo);
+}
+
+// ---------------------------------------------------------------------------
+// depthwise NHWC conv (depth_multiplier == 1): per-channel R x S taps,
+// memory-bound — channels are contiguous in NHWC so the channel dim is
+// the vector dim (4x bf16). w is [R,S,C] flat; per-tap weight loads hit
+// L2 (w is tiny and every pixel of the image reuses it).
+// ---------------------------------------------------------------------------
+TFSC_DEV float dw_act(float v, int act) {
+  switch (act) {
+    case ACT_RELU: return v > 0.f ? v : 0.f;
+    case ACT_TANH: return tanhf(v);
+    case ACT_SIGMOID: return 1.f / (1.f + __expf(-v));
+    case ACT_GELU: return 0.5f * v * (1.f + erff(v * 0.70710678f));
+    case ACT_RELU6: return v < 0.f ? 0.f : (v > 6.f ? 6.f : v);
+    default: return v;
+  }
+}
+
+__global__ void k_depthwise_v4(const ushort* __restrict__ x,
+                               const ushort* __restrict__ w,
+                               const ushort* __restrict__ bias,
+                               ushort* __restrict__ y,
+                               int N, int H, int W, int C, int R, int S,
+                               int sh, int sw, int pt, int pl,
+                               int Ho, int Wo, int act) {
+  int c4 = C / 4;
+  int64_t total = (int64_t)N * Ho * Wo * c4;
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < total; i += stride) {
+    int64_t t = i;
+    int cv = int(t % c4); t /= c4;
+    int wo = int(t % Wo); t /= Wo;
+    int ho = int(t % Ho); int n = int(t / Ho);
+    int c = cv * 4;
+    float acc[4];
+    short4_t bv = reinterpret_cast<const short4_t*>(bias)[cv];
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) acc[j] = bf2f((ushort)bv[j]);
+    int hi0 = ho * sh - pt, wi0 = wo * sw - pl;
+    for (int r = 0; r < R; ++r) {
+      int hi = hi0 + r;
+      if (hi < 0 || hi >= H) continue;
+      for (int sx = 0; sx < S; ++sx) {
+        int wi = wi0 + sx;
+        if (wi < 0 || wi >= W) continue;
+        short4_t xv = reinterpret_cast<const short4_t*>(
+            x + ((((int64_t)n * H + hi) * W + wi) * C))[cv];
+        short4_t wv = reinterpret_cast<const short4_t*>(
+            w + ((r * S + sx) * C))[cv];
+        #pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[j] += bf2f((ushort)xv[j]) * bf2f((ushort)wv[j]);
+      }
+    }
+    short4_t out;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      out[j] = (short)f2bf(dw_act(acc[j], act));
+    reinterpret_cast<short4_t*>(
+        y + ((((int64_t)n * Ho + ho) * Wo + wo) * C))[cv] = out;
+  }
+}
+
+__global__ void k_depthwise_scalar(const ushort* __restrict__ x,
+                                   const ushort* __restrict__ w,
+                                   const ushort* __restrict__ bias,
+                                   ushort* __restrict__ y,
+                                   int N, int H, int W, int C, int R, int S,
+                                   int sh, int sw, int pt, int pl,
+                                   int Ho, int Wo, int act) {
+  int64_t total = (int64_t)N * Ho * Wo * C;
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < total; i += stride) {
+    int64_t t = i;
+    int c = int(t % C); t /= C;
+    int wo = int(t % Wo); t /= Wo;
+    int ho = int(t % Ho); int n = int(t / Ho);
+    float acc = bf2f(bias[c]);
+    int hi0 = ho * sh - pt, wi0 = wo * sw - pl;
+    for (int r = 0; r < R; ++r) {
+      int hi = hi0 + r;
+      if (hi < 0 || hi >= H) continue;
+      for (int sx = 0; sx < S; ++sx) {
+        int wi = wi0 + sx;
+        if (wi < 0 || wi >= W) continue;
+        acc += bf2f(x[(((int64_t)n * H + hi) * W + wi) * C + c]) *
+               bf2f(w[(r * S + sx) * C + c]);
+      }
+    }
+    y[i] = f2bf(dw_act(acc, act));
+  }
+}
+
+void launch_depthwise_conv(hipStream_t s, const ushort* x, const ushort* w,
+                           const ushort* bias, ushort* y,
+                           int N, int H, int W, int C, int R, int S,
+                           int sh, int sw, int pt, int pl, int Ho, int Wo,
+                           int act) {
+  if (C % 4 == 0) {
+    int64_t total = (int64_t)N * Ho * Wo * (C / 4);
+    hipLaunchKernelGGL(k_depthwise_v4, dim3(grid_for(total)), dim3(TPB),
+                       0, s, x, w, bias, y, N, H, W, C, R, S, sh, sw,
+                       pt, pl, Ho, Wo, act);
+  } else {
+    int64_t total = (int64_t)N * Ho * Wo * C;
+    hipLaunchKernelGGL(k_depthwise_scalar, dim3(grid_for(total)),
+                       dim3(TPB), 0, s, x, w, bias, y, N, H, W, C, R, S,
+                       sh, sw, pt, pl, Ho, Wo, act);
+  }
 }
 
 }  // namespace tfsc

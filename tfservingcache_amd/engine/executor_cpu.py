@@ -21,6 +21,8 @@ def _act(x: np.ndarray, act: str) -> np.ndarray:
         return x
     if act == "relu":
         return np.maximum(x, 0.0)
+    if act == "relu6":
+        return np.clip(x, 0.0, 6.0)
     if act == "tanh":
         return np.tanh(x)
     if act == "sigmoid":
@@ -33,6 +35,7 @@ def _act(x: np.ndarray, act: str) -> np.ndarray:
 
 _UNARY = {
     "relu": lambda x: np.maximum(x, 0.0),
+    "relu6": lambda x: np.clip(x, 0.0, 6.0),
     "tanh": np.tanh,
     "sigmoid": lambda x: 1.0 / (1.0 + np.exp(-x)),
     "erf": None,  # filled lazily from scipy
@@ -114,7 +117,10 @@ class CpuExecutor:
                 ni += 1
             if p.get("residual"):
                 y = y + _get(op.inputs[ni])
-            vals[op.outputs[0]] = _act(y, p.get("act", "none")).astype(np.float32)
+            y = _act(y, p.get("act", "none")).astype(np.float32)
+            out_shape = self.plan.resolve_shape(
+                self.plan.tensors[op.outputs[0]].shape, batch)
+            vals[op.outputs[0]] = y.reshape(out_shape)
         elif k == "batched_gemm":
             a, b = _get(op.inputs[0]), _get(op.inputs[1])
             if p.get("trans_a"):
@@ -124,6 +130,8 @@ class CpuExecutor:
             vals[op.outputs[0]] = (a @ b).astype(np.float32)
         elif k == "conv2d":
             self._conv2d(op, _get, vals)
+        elif k == "depthwise_conv":
+            self._depthwise(op, _get, vals)
         elif k == "pool":
             self._pool(op, _get, vals)
         elif k == "global_mean":
@@ -209,6 +217,26 @@ class CpuExecutor:
         if p.get("residual"):
             y = y + _get(op.inputs[ni])
         vals[op.outputs[0]] = _act(y, p.get("act", "none")).astype(np.float32)
+
+    def _depthwise(self, op: PlanOp, _get, vals) -> None:
+        x = _get(op.inputs[0])          # [N,H,W,C]
+        w = _get(op.inputs[1])          # [R,S,C]
+        b = _get(op.inputs[2])          # [C]
+        p = op.params
+        sh, sw = p["stride"]
+        pt, pb, pl, pr = p["pads"]
+        R, S, C = w.shape
+        N = x.shape[0]
+        Ho, Wo = p["out_hw"]
+        xp = np.pad(x, ((0, 0), (pt, pb), (pl, pr), (0, 0)))
+        y = np.zeros((N, Ho, Wo, C), dtype=np.float32)
+        for r in range(R):
+            for s_ in range(S):
+                patch = xp[:, r:r + sh * Ho:sh, s_:s_ + sw * Wo:sw, :]
+                y += patch * w[r, s_]
+        y = y + b
+        vals[op.outputs[0]] = _act(y, p.get("act", "none")).astype(
+            np.float32)
 
     def _pool(self, op: PlanOp, _get, vals) -> None:
         x = _get(op.inputs[0])

@@ -212,13 +212,14 @@ def _pad64(k: int) -> int:
     return (k + 63) // 64 * 64
 
 
-_ACT_CODE = {"none": 0, "relu": 1, "tanh": 2, "sigmoid": 3, "gelu": 4}
+_ACT_CODE = {"none": 0, "relu": 1, "tanh": 2, "sigmoid": 3, "gelu": 4,
+             "relu6": 5}
 
 _ELT_CODE = {
     "add": 0, "sub": 1, "mul": 2, "div": 3, "max": 4, "min": 5,
     "sqdiff": 6, "relu": 7, "tanh": 8, "sigmoid": 9, "erf": 10,
     "sqrt": 11, "rsqrt": 12, "exp": 13, "neg": 14, "square": 15,
-    "gelu": 16,
+    "gelu": 16, "relu6": 17,
 }
 
 
@@ -390,6 +391,20 @@ class ExecContext:
                 calls.extend(self._c_gemm(op, oi, ext))
             elif k == "conv2d":
                 calls.extend(self._c_conv(op, oi, ext))
+            elif k == "depthwise_conv":
+                n, h, w_, c = self.shapes[op.inputs[0]]
+                ho, wo = p["out_hw"]
+                R, S, _C = p["rsc"]
+                sh, sw = p["stride"]
+                pt, _pb, pl, _pr = p["pads"]
+                calls.append((ext.K_DEPTHWISE,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.inputs[1]),
+                               self._ptr(op.inputs[2]),
+                               self._ptr(op.outputs[0])],
+                              [n, h, w_, c, R, S, sh, sw, pt, pl,
+                               ho, wo, _ACT_CODE[p.get("act", "none")]],
+                              []))
             elif k == "batched_gemm":
                 calls.extend(self._c_bgemm(op, ext))
             elif k == "attention":

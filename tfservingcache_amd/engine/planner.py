@@ -137,10 +137,11 @@ def numel(shape: Sequence[Dim], batch: int = 1) -> int:
     return n
 
 
-ACT_NONE, ACT_RELU, ACT_TANH, ACT_SIGMOID, ACT_GELU = \
-    "none", "relu", "tanh", "sigmoid", "gelu"
+ACT_NONE, ACT_RELU, ACT_TANH, ACT_SIGMOID, ACT_GELU, ACT_RELU6 = \
+    "none", "relu", "tanh", "sigmoid", "gelu", "relu6"
 
-_ELTWISE_UNARY = {"Relu": "relu", "Tanh": "tanh", "Sigmoid": "sigmoid",
+_ELTWISE_UNARY = {"Relu": "relu", "Relu6": "relu6", "Tanh": "tanh",
+                  "Sigmoid": "sigmoid",
                   "Erf": "erf", "Sqrt": "sqrt", "Rsqrt": "rsqrt",
                   "Exp": "exp", "Neg": "neg", "Square": "square"}
 _ELTWISE_BINARY = {"Add": "add", "AddV2": "add", "Sub": "sub", "Mul": "mul",
@@ -358,6 +359,12 @@ class _Lowerer:
         if op == "Conv2D":
             self.lower_conv(nd, out)
             return
+        if op == "DepthwiseConv2dNative":
+            self.lower_depthwise(nd, out)
+            return
+        if op == "Einsum":
+            self.lower_einsum(nd, out)
+            return
         if op == "MatMul":
             self.lower_matmul(nd, out)
             return
@@ -510,6 +517,8 @@ class _Lowerer:
                 residual = other[0]
             elif nxt.op == "Relu" and act == ACT_NONE:
                 act = ACT_RELU
+            elif nxt.op == "Relu6" and act == ACT_NONE:
+                act = ACT_RELU6
             elif nxt.op == "Tanh" and act == ACT_NONE and residual is None:
                 act = ACT_TANH
             elif nxt.op == "Sigmoid" and act == ACT_NONE and residual is None:
@@ -575,6 +584,151 @@ class _Lowerer:
         self.ops.append(PlanOp("conv2d", inputs, [y], params))
         if out_name != out:
             self.by_name[out] = y   # conv node's own tensor name -> fused out
+
+    def lower_depthwise(self, nd: g.NodeDef, out: str) -> None:
+        """DepthwiseConv2dNative (MobileNet-class), depth_multiplier==1.
+        Per-channel R x S taps — no MFMA shape, lowered to the dedicated
+        memory-bound kernel (csrc/ops/ops_memory.hip depthwise). BN and
+        Relu/Relu6 consumers fold into the kernel like Conv2D."""
+        x_ref, w_ref = nd.input[0], nd.input[1]
+        x = self.tid(x_ref)
+        if _attr_s(nd, "data_format", "NHWC") != "NHWC":
+            raise PlanError("only NHWC DepthwiseConv2dNative supported")
+        w = np.asarray(self.weight_of(w_ref), dtype=np.float32)  # [R,S,C,M]
+        R, S, Cin, M = w.shape
+        if M != 1:
+            raise PlanError(
+                f"depth_multiplier {M} unsupported (only 1)")
+        strides = _attr_ints(nd, "strides") or [1, 1, 1, 1]
+        padding = _attr_s(nd, "padding", "SAME")
+        xs = self.tensors[x].shape
+        H, W = int(xs[1]), int(xs[2])
+        sh, sw = strides[1], strides[2]
+        if padding == "SAME":
+            Ho, Wo = math.ceil(H / sh), math.ceil(W / sw)
+            pad_h = max((Ho - 1) * sh + R - H, 0)
+            pad_w = max((Wo - 1) * sw + S - W, 0)
+            pads = (pad_h // 2, pad_h - pad_h // 2,
+                    pad_w // 2, pad_w - pad_w // 2)
+        elif padding == "VALID":
+            Ho, Wo = (H - R) // sh + 1, (W - S) // sw + 1
+            pads = (0, 0, 0, 0)
+        else:
+            raise PlanError(f"depthwise padding {padding}")
+
+        bias, bn, residual, act, final = self._absorb_act_chain(nd)
+        if bn is not None:
+            scale, offset, mean, var, eps = bn
+            g_ = scale / np.sqrt(var + eps)
+            w = w * g_.reshape(1, 1, Cin, 1)
+            base = bias if bias is not None else np.zeros(Cin, np.float32)
+            bias = (base - mean) * g_ + offset
+        if bias is None:
+            bias = np.zeros(Cin, np.float32)
+        w_id = self.new_tensor((R, S, Cin), "f32", "weight",
+                               f"{nd.name}/fused_w",
+                               weight=w.reshape(R, S, Cin))
+        b_id = self.new_tensor((Cin,), "f32", "weight",
+                               f"{nd.name}/fused_b", weight=bias)
+        out_name = f"{final.name}:0"
+        dw_act = act if residual is None else ACT_NONE
+        y = self.new_tensor((B, Ho, Wo, Cin), "f32", "activation",
+                            out_name if residual is None
+                            else f"{nd.name}/dw:0")
+        self.ops.append(PlanOp("depthwise_conv", [x, w_id, b_id], [y],
+                               {"stride": (sh, sw), "pads": pads,
+                                "act": dw_act, "rsc": (R, S, Cin),
+                                "hw": (H, W), "out_hw": (Ho, Wo)}))
+        if residual is not None:
+            # rare shape (MobileNet residuals connect 1x1 convs): emit
+            # the absorbed add/act as separate eltwise ops
+            y2 = self.new_tensor((B, Ho, Wo, Cin), "f32", "activation",
+                                 out_name if act == ACT_NONE
+                                 else f"{nd.name}/res:0")
+            self.ops.append(PlanOp("eltwise", [y, self.tid(residual)],
+                                   [y2], {"fn": "add"}))
+            y = y2
+            if act != ACT_NONE:
+                y3 = self.new_tensor((B, Ho, Wo, Cin), "f32",
+                                     "activation", out_name)
+                self.ops.append(PlanOp("eltwise", [y], [y3], {"fn": act}))
+                y = y3
+        if out_name != out:
+            self.by_name[out] = y
+
+    def lower_einsum(self, nd: g.NodeDef, out: str) -> None:
+        """Einsum (transformer SavedModel exports). Supported family:
+        2-operand contractions where the contracted indices are the
+        TRAILING indices of the lhs and the LEADING indices of the rhs
+        (in the same order), the rhs is a constant weight, and the
+        output is lhs-prefix + rhs-suffix — i.e. every Dense-layer
+        einsum ("ij,jk->ik", "abc,cd->abd", "abc,cde->abde",
+        "abcd,cde->abe", ...). Lowered to the MFMA GEMM on a reshaped
+        view. Anything else raises PlanError (loudly unsupported)."""
+        eq = _attr_s(nd, "equation", "").replace(" ", "")
+        if len(nd.input) != 2 or "->" not in eq or "." in eq:
+            raise PlanError(f"unsupported einsum {eq!r}")
+        lhs_rhs, outs = eq.split("->", 1)
+        if "," not in lhs_rhs:
+            raise PlanError(f"unsupported einsum {eq!r}")
+        lhs, rhs = lhs_rhs.split(",", 1)
+        if len(set(lhs)) != len(lhs) or len(set(rhs)) != len(rhs):
+            raise PlanError(f"repeated index in einsum {eq!r}")
+        kchars = [ch for ch in lhs if ch in rhs and ch not in outs]
+        nk = len(kchars)
+        if nk == 0 or lhs[-nk:] != rhs[:nk] or \
+                outs != lhs[:-nk] + rhs[nk:]:
+            raise PlanError(f"einsum {eq!r} is not a trailing-K "
+                            "contraction (unsupported)")
+        a = self.tid(nd.input[0])
+        sa = self.tensors[a].shape
+        w = self.const_value(nd.input[1])
+        if w is None:
+            raise PlanError(
+                f"einsum {eq!r}: rhs must be a constant weight")
+        w = np.asarray(w, dtype=np.float32)
+        kflat = int(np.prod(w.shape[:nk]))
+        suffix = tuple(int(d) for d in w.shape[nk:])
+        nflat = int(np.prod(suffix)) if suffix else 1
+        w2 = w.reshape(kflat, nflat)
+        # reshape lhs to [..., kflat] via an alias (K dims contiguous)
+        a2 = a
+        if nk > 1:
+            a2 = self.new_tensor(tuple(sa[:-nk]) + (kflat,), "f32",
+                                 "activation", f"{nd.name}/a2d",
+                                 alias_of=a)
+        act = ACT_NONE
+        bias = None
+        residual = None
+        final = nd
+        if len(suffix) <= 1:
+            bias, bn, residual, act, final = self._absorb_act_chain(nd)
+            if bn is not None:
+                raise PlanError("BN after Einsum unsupported")
+            if act == ACT_NONE and residual is None:
+                gl = self.try_match_gelu(final)
+                if gl is not None:
+                    act = ACT_GELU
+                    final = gl
+        w_id = self.new_tensor((kflat, nflat), "f32", "weight",
+                               f"{nd.name}/w2d", weight=w2)
+        out_name = f"{final.name}:0"
+        y = self.new_tensor(tuple(sa[:-nk]) + suffix, "f32",
+                            "activation", out_name)
+        inputs = [a2, w_id]
+        params = {"trans_a": False, "trans_b": False, "act": act}
+        if bias is not None:
+            bias_id = self.new_tensor(bias.shape, "f32", "weight",
+                                      f"{nd.name}/fused_bias",
+                                      weight=bias)
+            inputs.append(bias_id)
+            params["has_bias"] = True
+        if residual is not None:
+            inputs.append(self.tid(residual))
+            params["residual"] = True
+        self.ops.append(PlanOp("gemm", inputs, [y], params))
+        if out_name != out:
+            self.by_name[out] = y
 
     def lower_matmul(self, nd: g.NodeDef, out: str) -> None:
         a_ref, b_ref = nd.input[0], nd.input[1]

@@ -36,8 +36,23 @@ def _to_array(value) -> np.ndarray:
         arr = arr.astype(np.float32)
     elif arr.dtype == np.int64:
         arr = arr.astype(np.int32)
+    elif arr.dtype.kind in ("U", "S"):
+        # JSON strings / b64 bytes -> DT_STRING object array (decoded at
+        # this tier, rejected with a clean error by the native engine;
+        # proxied untouched at the routing tier)
+        flat = arr.reshape(-1)
+        out = np.empty(flat.shape, dtype=object)
+        for i, v in enumerate(flat):
+            out[i] = v.encode() if isinstance(v, str) else bytes(v)
+        return out.reshape(arr.shape)
     elif arr.dtype == object:
-        raise RestCodecError("ragged or string inputs unsupported")
+        flat = arr.reshape(-1)
+        if all(isinstance(v, (str, bytes)) for v in flat):
+            out = np.empty(flat.shape, dtype=object)
+            for i, v in enumerate(flat):
+                out[i] = v.encode() if isinstance(v, str) else v
+            return out.reshape(arr.shape)
+        raise RestCodecError("ragged inputs unsupported")
     return arr
 
 

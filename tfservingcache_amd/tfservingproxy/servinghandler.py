@@ -102,6 +102,17 @@ class LocalServingHandler:
             else:
                 raise ServingError(
                     "anonymous input given but model has multiple inputs")
+        for alias, arr in inputs.items():
+            if getattr(arr, "dtype", None) is not None and \
+                    arr.dtype == object:
+                # DT_STRING decodes fine (wire/tensor.py) but the native
+                # engine has no string ops — reject cleanly instead of
+                # failing deep in the planner (reference passes strings
+                # through to TF Serving untouched: tfservingproxy.go:201)
+                raise ServingError(
+                    f"input {alias!r} is DT_STRING: string tensors are "
+                    "proxied at the routing tier but not executable by "
+                    "the native engine", m.ERROR_INVALID_ARGUMENT)
         try:
             out = model.predict(inputs, output_filter)
         except ModelExecError as e:
