@@ -1,0 +1,337 @@
+"""External numerics cross-checks (VERDICT round-1 item 7).
+
+The GPU tests compare HIP kernels against the in-repo numpy fp32
+executor — but a planner-level semantic bug shared by both executors
+would be invisible. These tests compare the planner + CPU executor
+against INDEPENDENT implementations of the same math:
+
+  * HuggingFace transformers BertModel (random-init config, its weights
+    exported into our GraphBuilder graph): catches attention/LayerNorm/
+    GELU/embedding semantics bugs;
+  * plain torch.nn convolutions/batchnorm (PyTorch's CPU kernels):
+    catches conv padding/stride/BN-folding/depthwise semantics bugs.
+
+All fp32 on CPU, tight tolerances.
+"""
+import numpy as np
+import pytest
+
+torch = pytest.importorskip("torch")
+
+from tfservingcache_amd.engine.model import load_model_from_dir  # noqa: E402
+from tfservingcache_amd.engine.savedmodel import (GraphBuilder,  # noqa: E402
+                                                  write_saved_model)
+
+
+def _load(tmp_path, sm, name="m"):
+    d = str(tmp_path / name / "1")
+    write_saved_model(sm, d)
+    return load_model_from_dir(d, name, 1)
+
+
+# ---------------------------------------------------------------------------
+# BERT vs HuggingFace transformers
+# ---------------------------------------------------------------------------
+
+def _export_hf_bert(hf, seq_len):
+    """Emit our frozen-graph BERT from a transformers BertModel's
+    weights (nn.Linear weights are [out, in] -> transposed for TF
+    MatMul)."""
+    cfg = hf.config
+    hidden, heads = cfg.hidden_size, cfg.num_attention_heads
+    dh = hidden // heads
+    sd = {k: v.detach().numpy() for k, v in hf.state_dict().items()}
+
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    i32 = gb.a_type(3)
+
+    def layernorm(name, x, gamma, beta, eps):
+        axes = gb.const(f"{name}/axes", np.array([-1], dtype=np.int32))
+        axes2 = gb.const(f"{name}/axes2", np.array([-1], dtype=np.int32))
+        mean = gb.node("Mean", f"{name}/mean", [x, axes], T=f32,
+                       keep_dims=gb.a_bool(True))
+        sqd = gb.node("SquaredDifference", f"{name}/sqd", [x, mean], T=f32)
+        var = gb.node("Mean", f"{name}/var", [sqd, axes2], T=f32,
+                      keep_dims=gb.a_bool(True))
+        addeps = gb.node("AddV2", f"{name}/addeps",
+                         [var, gb.const(f"{name}/eps", np.float32(eps))],
+                         T=f32)
+        rstd = gb.node("Rsqrt", f"{name}/rsqrt", [addeps], T=f32)
+        sub = gb.node("Sub", f"{name}/sub", [x, mean], T=f32)
+        mul = gb.node("Mul", f"{name}/mul", [sub, rstd], T=f32)
+        mulg = gb.node("Mul", f"{name}/mulg",
+                       [mul, gb.const(f"{name}/gamma",
+                                      gamma.astype(np.float32))], T=f32)
+        return gb.node("AddV2", f"{name}/out",
+                       [mulg, gb.const(f"{name}/beta",
+                                       beta.astype(np.float32))], T=f32)
+
+    def gelu(name, x):
+        m1 = gb.node("Mul", f"{name}/m1",
+                     [x, gb.const(f"{name}/rsqrt2",
+                                  np.float32(0.7071067811865476))], T=f32)
+        erf = gb.node("Erf", f"{name}/erf", [m1], T=f32)
+        a1 = gb.node("AddV2", f"{name}/a1",
+                     [erf, gb.const(f"{name}/one", np.float32(1.0))],
+                     T=f32)
+        m2 = gb.node("Mul", f"{name}/m2",
+                     [x, gb.const(f"{name}/half", np.float32(0.5))], T=f32)
+        return gb.node("Mul", f"{name}/out", [m2, a1], T=f32)
+
+    def dense(name, x, wkey, bkey):
+        w = sd[wkey].T.astype(np.float32)       # [in, out]
+        b = sd[bkey].astype(np.float32)
+        mm = gb.node("MatMul", f"{name}/mm",
+                     [x, gb.const(f"{name}/w", w)], T=f32)
+        return gb.node("BiasAdd", f"{name}/out",
+                       [mm, gb.const(f"{name}/b", b)], T=f32)
+
+    ids = gb.placeholder("input_ids", np.int32, [-1, seq_len],
+                         signature_name="input_ids")
+    # embeddings: word[ids] + position[0..S-1] + token_type[0], then LN
+    word = sd["embeddings.word_embeddings.weight"].astype(np.float32)
+    pos = sd["embeddings.position_embeddings.weight"][:seq_len].astype(
+        np.float32)
+    tok0 = sd["embeddings.token_type_embeddings.weight"][0].astype(
+        np.float32)
+    pos = pos + tok0                           # token_type all-zero fold
+    ids_flat = gb.node("Reshape", "ids_flat",
+                       [ids, gb.const("flat", np.array([-1], np.int32))])
+    emb = gb.node("GatherV2", "embed",
+                  [gb.const("word_emb", word), ids_flat,
+                   gb.const("gax", np.array(0, np.int32))],
+                  Tparams=f32, Tindices=i32)
+    emb3 = gb.node("Reshape", "embed3",
+                   [emb, gb.const("to3", np.array([-1, seq_len,
+                                                   word.shape[1]],
+                                                  np.int32))])
+    pos3 = gb.node("Reshape", "pos3",
+                   [gb.const("pos_emb", pos),
+                    gb.const("pos_shape",
+                             np.array([1, seq_len, word.shape[1]],
+                                      np.int32))])
+    h3 = gb.node("AddV2", "embed_sum", [emb3, pos3], T=f32)
+    h = gb.node("Reshape", "embed2d",
+                [h3, gb.const("to2", np.array([-1, word.shape[1]],
+                                              np.int32))])
+    h = layernorm("embeddings/ln", h,
+                  sd["embeddings.LayerNorm.weight"],
+                  sd["embeddings.LayerNorm.bias"],
+                  cfg.layer_norm_eps)
+
+    to_heads = gb.const("to_heads",
+                        np.array([-1, seq_len, heads, dh], np.int32))
+    perm = gb.const("perm0213", np.array([0, 2, 1, 3], np.int32))
+    to_ctx = gb.const("to_ctx", np.array([-1, hidden], np.int32))
+
+    for li in range(cfg.num_hidden_layers):
+        p = f"encoder.layer.{li}"
+        pre = f"layer{li}"
+        q = dense(f"{pre}/q", h, f"{p}.attention.self.query.weight",
+                  f"{p}.attention.self.query.bias")
+        k = dense(f"{pre}/k", h, f"{p}.attention.self.key.weight",
+                  f"{p}.attention.self.key.bias")
+        v = dense(f"{pre}/v", h, f"{p}.attention.self.value.weight",
+                  f"{p}.attention.self.value.bias")
+
+        def split(name, t):
+            r = gb.node("Reshape", f"{name}/r", [t, to_heads])
+            return gb.node("Transpose", f"{name}/t", [r, perm], T=f32)
+
+        qh, kh, vh = (split(f"{pre}/qh", q), split(f"{pre}/kh", k),
+                      split(f"{pre}/vh", v))
+        scores = gb.node("BatchMatMulV2", f"{pre}/scores", [qh, kh],
+                         T=f32, adj_x=gb.a_bool(False),
+                         adj_y=gb.a_bool(True))
+        scaled = gb.node("Mul", f"{pre}/scaled",
+                         [scores, gb.const(f"{pre}/scale",
+                                           np.float32(1 / np.sqrt(dh)))],
+                         T=f32)
+        probs = gb.node("Softmax", f"{pre}/probs", [scaled], T=f32)
+        ctx = gb.node("BatchMatMulV2", f"{pre}/ctx", [probs, vh], T=f32)
+        ctx_t = gb.node("Transpose", f"{pre}/ctx_t", [ctx, perm], T=f32)
+        ctx2 = gb.node("Reshape", f"{pre}/ctx2", [ctx_t, to_ctx])
+        att = dense(f"{pre}/att_out", ctx2,
+                    f"{p}.attention.output.dense.weight",
+                    f"{p}.attention.output.dense.bias")
+        res1 = gb.node("AddV2", f"{pre}/res1", [att, h], T=f32)
+        h1 = layernorm(f"{pre}/ln1", res1,
+                       sd[f"{p}.attention.output.LayerNorm.weight"],
+                       sd[f"{p}.attention.output.LayerNorm.bias"],
+                       cfg.layer_norm_eps)
+        ffn1 = dense(f"{pre}/ffn1", h1, f"{p}.intermediate.dense.weight",
+                     f"{p}.intermediate.dense.bias")
+        act = gelu(f"{pre}/gelu", ffn1)
+        ffn2 = dense(f"{pre}/ffn2", act, f"{p}.output.dense.weight",
+                     f"{p}.output.dense.bias")
+        res2 = gb.node("AddV2", f"{pre}/res2", [ffn2, h1], T=f32)
+        h = layernorm(f"{pre}/ln2", res2,
+                      sd[f"{p}.output.LayerNorm.weight"],
+                      sd[f"{p}.output.LayerNorm.bias"],
+                      cfg.layer_norm_eps)
+
+    h3out = gb.node("Reshape", "seq_out3",
+                    [h, gb.const("to3o", np.array([-1, seq_len, hidden],
+                                                  np.int32))])
+    gb.mark_output("sequence_output", h3out)
+    return gb.build()
+
+
+def test_bert_vs_transformers(tmp_path):
+    from transformers import BertConfig, BertModel
+    torch.manual_seed(0)
+    cfg = BertConfig(vocab_size=97, hidden_size=64,
+                     num_hidden_layers=2, num_attention_heads=4,
+                     intermediate_size=128, max_position_embeddings=32,
+                     hidden_act="gelu", attention_probs_dropout_prob=0.0,
+                     hidden_dropout_prob=0.0)
+    hf = BertModel(cfg).eval()
+    seq_len = 16
+    sm = _export_hf_bert(hf, seq_len)
+    model = _load(tmp_path, sm, "bert_hf")
+
+    rng = np.random.default_rng(1)
+    ids = rng.integers(0, 97, (3, seq_len)).astype(np.int32)
+    with torch.no_grad():
+        want = hf(input_ids=torch.from_numpy(ids.astype(np.int64))
+                  ).last_hidden_state.numpy()
+    got = model.predict({"input_ids": ids})["sequence_output"]
+    assert got.shape == want.shape
+    np.testing.assert_allclose(got, want, rtol=1e-3, atol=2e-4)
+
+
+# ---------------------------------------------------------------------------
+# CNN family vs torch.nn (conv / BN folding / depthwise / pool semantics)
+# ---------------------------------------------------------------------------
+
+def _tf_same_pads(h, w, k, s):
+    ho, wo = -(-h // s), -(-w // s)
+    ph = max((ho - 1) * s + k - h, 0)
+    pw = max((wo - 1) * s + k - w, 0)
+    return ph // 2, ph - ph // 2, pw // 2, pw - pw // 2
+
+
+class _TorchCnn(torch.nn.Module):
+    """conv3x3/2+BN+ReLU -> depthwise3x3/1+BN+ReLU6 -> conv1x1+BN
+    + residual -> maxpool -> global mean -> fc -> softmax, with TF
+    SAME padding replicated via explicit F.pad."""
+
+    def __init__(self, cin=3, c=16, classes=7):
+        super().__init__()
+        self.conv1 = torch.nn.Conv2d(cin, c, 3, 2, 0)
+        self.bn1 = torch.nn.BatchNorm2d(c)
+        self.dw = torch.nn.Conv2d(c, c, 3, 1, 0, groups=c)
+        self.bn2 = torch.nn.BatchNorm2d(c)
+        self.pw = torch.nn.Conv2d(c, c, 1, 1, 0)
+        self.bn3 = torch.nn.BatchNorm2d(c)
+        self.fc = torch.nn.Linear(c, classes)
+        # make BN stats non-trivial
+        for bn in (self.bn1, self.bn2, self.bn3):
+            torch.nn.init.normal_(bn.running_mean, 0, 0.2)
+            bn.running_var.uniform_(0.5, 1.5)
+            torch.nn.init.normal_(bn.weight, 1.0, 0.2)
+            torch.nn.init.normal_(bn.bias, 0, 0.1)
+
+    def forward(self, x):                        # x NCHW
+        import torch.nn.functional as F
+        pt, pb, pl, pr = _tf_same_pads(x.shape[2], x.shape[3], 3, 2)
+        h = F.relu(self.bn1(self.conv1(F.pad(x, (pl, pr, pt, pb)))))
+        pt, pb, pl, pr = _tf_same_pads(h.shape[2], h.shape[3], 3, 1)
+        d = F.relu6(self.bn2(self.dw(F.pad(h, (pl, pr, pt, pb)))))
+        p = self.bn3(self.pw(d)) + h             # residual
+        pt, pb, pl, pr = _tf_same_pads(p.shape[2], p.shape[3], 2, 2)
+        p = F.max_pool2d(F.pad(p, (pl, pr, pt, pb), value=-1e30), 2, 2)
+        g = p.mean(dim=(2, 3))
+        return F.softmax(self.fc(g), dim=-1)
+
+
+def _export_torch_cnn(net, image_size):
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+
+    def conv_w(conv):      # torch [K,C,R,S] -> TF [R,S,C,K]
+        return conv.weight.detach().permute(2, 3, 1, 0).numpy().astype(
+            np.float32)
+
+    def dw_w(conv):        # torch [C,1,R,S] -> TF [R,S,C,1]
+        return conv.weight.detach().permute(2, 3, 0, 1).numpy().astype(
+            np.float32)
+
+    def bn_nodes(name, x, conv, bn, relu=None):
+        c = gb.node("FusedBatchNormV3", f"{name}/bn",
+                    [x,
+                     gb.const(f"{name}/g", bn.weight.detach().numpy()),
+                     gb.const(f"{name}/b", bn.bias.detach().numpy()),
+                     gb.const(f"{name}/m",
+                              bn.running_mean.detach().numpy()),
+                     gb.const(f"{name}/v",
+                              bn.running_var.detach().numpy())],
+                    T=f32, U=f32, epsilon=gb.a_float(bn.eps),
+                    is_training=gb.a_bool(False),
+                    data_format=gb.a_str("NHWC"))
+        if relu:
+            c = gb.node(relu, f"{name}/{relu.lower()}", [c], T=f32)
+        return c
+
+    x = gb.placeholder("input", np.float32,
+                       [-1, image_size, image_size, 3],
+                       signature_name="input")
+    c1 = gb.node("Conv2D", "c1", [x, gb.const("c1/w", conv_w(net.conv1))],
+                 T=f32, strides=gb.a_ints([1, 2, 2, 1]),
+                 padding=gb.a_str("SAME"), data_format=gb.a_str("NHWC"))
+    # conv1 has a bias in torch
+    c1 = gb.node("BiasAdd", "c1/bias",
+                 [c1, gb.const("c1/b",
+                               net.conv1.bias.detach().numpy())], T=f32)
+    h = bn_nodes("bn1", c1, net.conv1, net.bn1, "Relu")
+    d = gb.node("DepthwiseConv2dNative", "dw",
+                [h, gb.const("dw/w", dw_w(net.dw))], T=f32,
+                strides=gb.a_ints([1, 1, 1, 1]), padding=gb.a_str("SAME"),
+                data_format=gb.a_str("NHWC"))
+    d = gb.node("BiasAdd", "dw/bias",
+                [d, gb.const("dw/b", net.dw.bias.detach().numpy())],
+                T=f32)
+    d = bn_nodes("bn2", d, net.dw, net.bn2, "Relu6")
+    p = gb.node("Conv2D", "pw", [d, gb.const("pw/w", conv_w(net.pw))],
+                T=f32, strides=gb.a_ints([1, 1, 1, 1]),
+                padding=gb.a_str("SAME"), data_format=gb.a_str("NHWC"))
+    p = gb.node("BiasAdd", "pw/bias",
+                [p, gb.const("pw/b", net.pw.bias.detach().numpy())],
+                T=f32)
+    p = bn_nodes("bn3", p, net.pw, net.bn3)
+    p = gb.node("AddV2", "res", [p, h], T=f32)
+    p = gb.node("MaxPool", "pool", [p], T=f32,
+                ksize=gb.a_ints([1, 2, 2, 1]),
+                strides=gb.a_ints([1, 2, 2, 1]),
+                padding=gb.a_str("SAME"), data_format=gb.a_str("NHWC"))
+    gap = gb.node("Mean", "gap",
+                  [p, gb.const("gax", np.array([1, 2], np.int32))],
+                  T=f32, keep_dims=gb.a_bool(False))
+    mm = gb.node("MatMul", "fc",
+                 [gap, gb.const("fc/w",
+                                net.fc.weight.detach().t().numpy()
+                                .astype(np.float32))], T=f32)
+    lo = gb.node("BiasAdd", "logits",
+                 [mm, gb.const("fc/b", net.fc.bias.detach().numpy())],
+                 T=f32)
+    sm = gb.node("Softmax", "probs", [lo], T=f32)
+    gb.mark_output("probs", sm)
+    return gb.build()
+
+
+def test_cnn_vs_torch_nn(tmp_path):
+    torch.manual_seed(7)
+    net = _TorchCnn().eval()
+    size = 18                               # odd-ish: asymmetric SAME pads
+    sm = _export_torch_cnn(net, size)
+    model = _load(tmp_path, sm, "cnn_torch")
+    rng = np.random.default_rng(4)
+    x = (rng.standard_normal((3, size, size, 3)) * 0.7).astype(np.float32)
+    with torch.no_grad():
+        want = net(torch.from_numpy(
+            x.transpose(0, 3, 1, 2).copy())).numpy()
+    got = model.predict({"input": x})["probs"]
+    assert got.shape == want.shape
+    np.testing.assert_allclose(got, want, rtol=1e-3, atol=1e-4)
+    assert (got.argmax(1) == want.argmax(1)).all()
