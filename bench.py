@@ -204,12 +204,15 @@ def main() -> int:
                          "(warm mode, grpc/native transports): measures "
                          "the server without the client's GIL in the "
                          "way")
-    ap.add_argument("--transport", choices=["inproc", "grpc", "native"],
+    ap.add_argument("--transport",
+                    choices=["inproc", "grpc", "native", "native-rest"],
                     default=None,
                     help="inproc: gRPC message path without sockets; "
                          "grpc: Python grpcio server over TCP loopback; "
                          "native: C++ nghttp2 gRPC front-end over TCP "
-                         "loopback (registered Predicts bypass Python)")
+                         "loopback (registered Predicts bypass Python); "
+                         "native-rest: C++ HTTP/1.1 JSON front-end over "
+                         "TCP loopback")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -365,7 +368,86 @@ def main() -> int:
         return run
 
     grpc_server = None
-    if args.transport in ("grpc", "native"):
+    if args.transport == "native-rest":
+        import socket as socket_mod
+        import threading as threading_mod
+        from tfservingcache_amd.tfservingproxy.native_frontend import \
+            NativeRestServer
+        grpc_server = NativeRestServer(handler)
+        grpc_server.add_insecure_port("127.0.0.1:0")
+        grpc_server.start()
+        rport = grpc_server.port
+
+        def _on_avail_rest(name, version, model):
+            fast = getattr(getattr(model, "_gpu", None), "_fast", None)
+            if fast is not None:
+                grpc_server.register_model(name, version, fast)
+        pool.on_available = _on_avail_rest
+        pool.on_unload = (lambda name, version, model:
+                          grpc_server.unregister_model(name, version))
+
+        # pre-encoded HTTP request bytes per model (one JSON body per
+        # family; only the path differs)
+        import json as json_mod
+        bodies = {}
+
+        def _rest_body(kind: str) -> bytes:
+            if kind not in bodies:
+                if kind == "bert_base":
+                    payload = {"instances": ids.tolist()}
+                else:
+                    payload = {"instances": np.round(img, 5).tolist()}
+                bodies[kind] = json_mod.dumps(payload).encode()
+            return bodies[kind]
+
+        rest_reqs = {}
+        for n in names:
+            body = _rest_body(kind_of[n])
+            head = (f"POST /v1/models/{n}:predict HTTP/1.1\r\n"
+                    f"Host: 127.0.0.1\r\n"
+                    f"Content-Type: application/json\r\n"
+                    f"Content-Length: {len(body)}\r\n\r\n"
+                    ).encode()
+            rest_reqs[n] = head + body
+
+        tl = threading_mod.local()
+
+        def _rest_call(name: str) -> None:
+            sock = getattr(tl, "sock", None)
+            if sock is None:
+                sock = socket_mod.create_connection(
+                    ("127.0.0.1", rport), timeout=300)
+                sock.setsockopt(socket_mod.IPPROTO_TCP,
+                                socket_mod.TCP_NODELAY, 1)
+                tl.sock = sock
+                tl.buf = b""
+            sock.sendall(rest_reqs[name])
+            buf = tl.buf
+            while b"\r\n\r\n" not in buf:
+                chunk = sock.recv(1 << 16)
+                if not chunk:
+                    raise RuntimeError("rest connection closed")
+                buf += chunk
+            head, rest = buf.split(b"\r\n\r\n", 1)
+            status = int(head.split(b" ", 2)[1])
+            clen = 0
+            for line in head.split(b"\r\n"):
+                if line.lower().startswith(b"content-length"):
+                    clen = int(line.split(b":")[1])
+            while len(rest) < clen:
+                chunk = sock.recv(1 << 18)
+                if not chunk:
+                    raise RuntimeError("rest connection closed")
+                rest += chunk
+            tl.buf = rest[clen:]
+            if status != 200:
+                raise RuntimeError(
+                    f"rest predict failed: {status} "
+                    f"{rest[:200].decode(errors='replace')}")
+
+        def one_request(i: int) -> None:
+            _rest_call(pick(i))
+    elif args.transport in ("grpc", "native"):
         import grpc as grpc_mod
         if args.transport == "native":
             from tfservingcache_amd.tfservingproxy.native_frontend import \
@@ -426,7 +508,10 @@ def main() -> int:
     # same transport), strictly outside the LRU timed region
     warm_req_per_sec = None
     if headline:
-        if args.transport in ("grpc", "native"):
+        if args.transport == "native-rest":
+            def warm_one(i: int) -> None:
+                _rest_call(warm_name)
+        elif args.transport in ("grpc", "native"):
             def warm_one(i: int) -> None:
                 rpcs[i % len(rpcs)](req_cache[warm_name], timeout=300)
         else:
@@ -590,6 +675,8 @@ def main() -> int:
                             "loopback",
                     "native": "real gRPC (native nghttp2 front-end) "
                               "over TCP loopback",
+                    "native-rest": "real REST (native C++ HTTP/1.1 JSON "
+                                   "front-end) over TCP loopback",
                     "inproc": "in-process gRPC message path (protobuf "
                               "decode/encode included)",
                 }[args.transport],

@@ -25,6 +25,7 @@ sources = [
     os.path.join(CSRC, "ops", "attention.hip"),
     os.path.join(CSRC, "fastpath.cpp"),
     os.path.join(CSRC, "frontend.cpp"),
+    os.path.join(CSRC, "rest_frontend.cpp"),
 ]
 
 # nghttp2 (HTTP/2 framing/HPACK for the native gRPC front-end) ships in

@@ -238,6 +238,7 @@ namespace py = pybind11;
 
 void register_fastpath(py::module_& mod);
 void register_frontend(py::module_& mod);
+void register_rest_frontend(py::module_& mod);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   using tfsc::Call;
@@ -290,6 +291,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
 
   register_fastpath(mod);
   register_frontend(mod);
+  register_rest_frontend(mod);
 
   // elementwise fn codes
   mod.attr("ELT_ADD") = int(tfsc::ELT_ADD);

@@ -283,6 +283,19 @@ class FastModel {
     }
   }
 
+  std::vector<FastIOSpec> input_specs() {
+    std::lock_guard<std::mutex> g(mu_);
+    std::vector<FastIOSpec> out;
+    for (auto& io : specs_in_) {
+      FastIOSpec sp;
+      sp.alias = io.alias;
+      sp.is_int = io.is_int;
+      for (int64_t d : io.tail_dims) sp.tail.push_back(d);
+      out.push_back(std::move(sp));
+    }
+    return out;
+  }
+
   std::tuple<long long, long long, long long> stats() const {
     return {n_runs_.load(), n_requests_.load(), n_rows_.load()};
   }
@@ -659,6 +672,10 @@ class FastModel {
 std::string fastmodel_predict(FastModel* fm, const uint8_t* data,
                               size_t len) {
   return fm->predict(data, len);
+}
+
+std::vector<FastIOSpec> fastmodel_input_specs(FastModel* fm) {
+  return fm->input_specs();
 }
 
 // fast partial parse of PredictRequest.model_spec (field 1): the

@@ -7,6 +7,7 @@
 #include <cstdint>
 #include <stdexcept>
 #include <string>
+#include <vector>
 
 namespace tfsc {
 
@@ -15,6 +16,15 @@ struct FastFallback : std::runtime_error {
 };
 
 class FastModel;  // full definition in fastpath.cpp
+
+// input-tensor schema of a registered model (for the native REST
+// front-end's JSON -> proto bridge)
+struct FastIOSpec {
+  std::string alias;
+  bool is_int = false;
+  std::vector<long long> tail;     // dims after the batch dim
+};
+std::vector<FastIOSpec> fastmodel_input_specs(FastModel* fm);
 
 // PredictRequest bytes -> PredictResponse bytes entirely in C++
 // (throws FastFallback when the request needs the Python path).

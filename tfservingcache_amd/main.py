@@ -252,6 +252,7 @@ class Server:
         # cache tier gRPC: grpcio server, or the native nghttp2
         # front-end (serving.nativeFrontend) whose registered Predicts
         # never touch Python
+        self._native_frontends = []
         if self.cfg.get_bool("serving.nativeFrontend"):
             from .tfservingproxy.native_frontend import NativeGrpcServer
             cache_grpc = NativeGrpcServer(
@@ -260,7 +261,7 @@ class Server:
             cache_grpc.add_insecure_port(f"[::]:{self.cache_grpc_port}")
             cache_grpc.start()
             self._grpc_servers.append(cache_grpc)
-            self._wire_native_registry(cache_grpc)
+            self._native_frontends.append(cache_grpc)
         else:
             cache_grpc, _ = make_cache_grpc_server(
                 self.handler, health=self.health,
@@ -320,12 +321,27 @@ class Server:
         else:
             log.info("Proxy is disabled (no serviceDiscovery.type)")
 
-        cache_app = make_cache_rest_app(self.handler,
-                                        metrics_path=metrics_path,
-                                        metrics_render=merger.render)
+        # cache-tier REST: native (C++ HTTP/1.1 + JSON, registered
+        # Predicts bypass Python) or the aiohttp app
+        if self.cfg.get_bool("serving.nativeRestFrontend"):
+            from .tfservingproxy.native_frontend import NativeRestServer
+            cache_rest = NativeRestServer(self.handler,
+                                          metrics_path=metrics_path,
+                                          metrics_render=merger.render)
+            cache_rest.add_insecure_port(f"[::]:{self.cache_rest_port}")
+            cache_rest.start()
+            self._grpc_servers.append(cache_rest)    # stop() plumbing
+            self._native_frontends.append(cache_rest)
+            apps = [] + proxy_apps
+        else:
+            cache_app = make_cache_rest_app(self.handler,
+                                            metrics_path=metrics_path,
+                                            metrics_render=merger.render)
+            apps = [(cache_app, self.cache_rest_port)] + proxy_apps
+        if self._native_frontends:
+            self._wire_native_registry(self._native_frontends)
 
         # REST servers on a dedicated asyncio loop thread
-        apps = [(cache_app, self.cache_rest_port)] + proxy_apps
         t = threading.Thread(target=self._run_rest, args=(apps,),
                              daemon=True)
         t.start()
@@ -391,17 +407,20 @@ class Server:
 
         self.cm.on_cold_load = on_cold_load
 
-    def _wire_native_registry(self, native) -> None:
-        """Pool lifecycle -> C++ front-end FastModel registry."""
+    def _wire_native_registry(self, natives) -> None:
+        """Pool lifecycle -> C++ front-end FastModel registries (the
+        native gRPC and native REST servers share one model pool)."""
         pool = self.cm.pool
 
         def on_available(name, version, model):
             fast = getattr(getattr(model, "_gpu", None), "_fast", None)
             if fast is not None:
-                native.register_model(name, version, fast)
+                for fe in natives:
+                    fe.register_model(name, version, fast)
 
         def on_unload(name, version, model):
-            native.unregister_model(name, version)
+            for fe in natives:
+                fe.unregister_model(name, version)
 
         pool.on_available = on_available
         pool.on_unload = on_unload

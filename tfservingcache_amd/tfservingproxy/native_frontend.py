@@ -124,3 +124,47 @@ class NativeGrpcServer:
 
     def fallback_calls(self) -> int:
         return self._fe.fallback_calls()
+
+
+class NativeRestServer:
+    """HTTP/1.1 + JSON front-end (engine/csrc/rest_frontend.cpp):
+    registered numeric Predicts run fully in C++; everything else hits
+    the same sync dispatcher logic as the aiohttp app
+    (rest.make_rest_dispatcher), keeping the two REST servers
+    wire-identical."""
+
+    def __init__(self, handler: LocalServingHandler,
+                 metrics_path: str = "/monitoring/prometheus/metrics",
+                 metrics_render=None):
+        import torch  # noqa: F401
+        from ..engine import _tfsc_engine as ext
+        from .rest import make_rest_dispatcher
+        self._fe = ext.RestFrontendNative(
+            make_rest_dispatcher(handler, metrics_path, metrics_render))
+        self._port: Optional[int] = None
+
+    def add_insecure_port(self, addr: str) -> int:
+        self._port = int(addr.rsplit(":", 1)[1])
+        return self._port
+
+    def start(self) -> None:
+        self._port = self._fe.start(self._port or 0)
+
+    def stop(self, grace: float = 0.0) -> None:     # noqa: ARG002
+        self._fe.stop()
+
+    def register_model(self, name: str, version: int, fast) -> None:
+        self._fe.register_model(name, int(version), fast._ptr(), fast)
+
+    def unregister_model(self, name: str, version: int) -> None:
+        self._fe.unregister_model(name, int(version))
+
+    @property
+    def port(self) -> int:
+        return self._fe.port()
+
+    def native_hits(self) -> int:
+        return self._fe.native_hits()
+
+    def fallback_calls(self) -> int:
+        return self._fe.fallback_calls()

@@ -89,19 +89,19 @@ def make_cache_rest_app(handler: LocalServingHandler,
         try:
             if verb == "predict":
                 body = await request.read()
-                return await loop.run_in_executor(
-                    None, _predict_sync, handler, name, version, body)
+                return web.json_response(await loop.run_in_executor(
+                    None, _predict_sync, handler, name, version, body))
             if verb in ("classify", "regress"):
                 body = await request.read()
-                return await loop.run_in_executor(
+                return web.json_response(await loop.run_in_executor(
                     None, _classify_regress_sync, handler, name, version,
-                    verb, body)
+                    verb, body))
             if verb == "metadata":
-                return await loop.run_in_executor(
-                    None, _metadata_sync, handler, name, version)
+                return web.json_response(await loop.run_in_executor(
+                    None, _metadata_sync, handler, name, version))
             if verb == "":
-                return await loop.run_in_executor(
-                    None, _status_sync, handler, name, version)
+                return web.json_response(await loop.run_in_executor(
+                    None, _status_sync, handler, name, version))
             return _error_response(f"unsupported method :{verb}", 400)
         except ServingError as e:
             mt.proxy_requests_failed.labels("http").inc()
@@ -127,7 +127,7 @@ def make_cache_rest_app(handler: LocalServingHandler,
     return app
 
 
-def _predict_sync(handler, name, version, body: bytes) -> web.Response:
+def _predict_sync(handler, name, version, body: bytes) -> dict:
     mt.proxy_requests_total.labels("http").inc()
     try:
         payload = json.loads(body or b"{}")
@@ -138,11 +138,11 @@ def _predict_sync(handler, name, version, body: bytes) -> web.Response:
     except RestCodecError as e:
         raise ServingError(str(e))
     outputs, _version = handler.predict_arrays(name, version, inputs)
-    return web.json_response(render_predict_response(outputs, fmt))
+    return render_predict_response(outputs, fmt)
 
 
 def _classify_regress_sync(handler, name, version, verb,
-                           body: bytes) -> web.Response:
+                           body: bytes) -> dict:
     mt.proxy_requests_total.labels("http").inc()
     try:
         payload = json.loads(body or b"{}")
@@ -164,30 +164,29 @@ def _classify_regress_sync(handler, name, version, verb,
     outputs, _version = handler.predict_arrays(name, version, feats)
     vals = next(iter(outputs.values()))
     if verb == "regress":
-        return web.json_response(
-            {"results": np.asarray(vals, dtype=np.float64).reshape(-1)
-             .tolist()})
+        return {"results": np.asarray(vals, dtype=np.float64).reshape(-1)
+                .tolist()}
     scores = np.atleast_2d(np.asarray(vals, dtype=np.float64))
-    return web.json_response({"results": [
-        [[str(i), float(s)] for i, s in enumerate(row)] for row in scores]})
+    return {"results": [
+        [[str(i), float(s)] for i, s in enumerate(row)] for row in scores]}
 
 
-def _status_sync(handler, name, version) -> web.Response:
+def _status_sync(handler, name, version) -> dict:
     mt.proxy_requests_total.labels("http").inc()
     req = m.GetModelStatusRequest(model_spec=m.ModelSpec(
         name=name, version=m.Int64Value(value=version) if version else None))
     resp = handler.get_model_status(req)
-    return web.json_response({"model_version_status": [
+    return {"model_version_status": [
         {"version": str(s.version),
          "state": m.STATE_NAMES.get(s.state, "UNKNOWN"),
          "status": {"error_code": "OK" if not s.status or
                     s.status.error_code == 0 else "UNKNOWN",
                     "error_message": s.status.error_message
                     if s.status else ""}}
-        for s in resp.model_version_status]})
+        for s in resp.model_version_status]}
 
 
-def _metadata_sync(handler, name, version) -> web.Response:
+def _metadata_sync(handler, name, version) -> dict:
     mt.proxy_requests_total.labels("http").inc()
     req = m.GetModelMetadataRequest(model_spec=m.ModelSpec(
         name=name, version=m.Int64Value(value=version) if version else None),
@@ -207,11 +206,11 @@ def _metadata_sync(handler, name, version) -> web.Response:
             "inputs": {k: ti_json(v) for k, v in sig.inputs.items()},
             "outputs": {k: ti_json(v) for k, v in sig.outputs.items()},
             "method_name": sig.method_name}}
-    return web.json_response({
+    return {
         "model_spec": {"name": name, "version": str(
             resp.model_spec.version_value() if resp.model_spec else 0),
             "signature_name": ""},
-        "metadata": {"signature_def": {"signature_def": sig_json}}})
+        "metadata": {"signature_def": {"signature_def": sig_json}}}
 
 
 # ---------------------------------------------------------------------------
@@ -273,3 +272,50 @@ def make_proxy_rest_app(pick_node: Callable[[str, int], str],
             return web.Response(body=data, content_type="text/plain")
         app.router.add_get(metrics_path, metrics_handler)
     return app
+
+
+# ---------------------------------------------------------------------------
+# Sync dispatcher for the NATIVE REST front-end (engine/csrc/
+# rest_frontend.cpp): (method, path, body) -> (status, content_type,
+# body bytes). Routes exactly like make_cache_rest_app.
+# ---------------------------------------------------------------------------
+def make_rest_dispatcher(handler: LocalServingHandler,
+                         metrics_path: str =
+                         "/monitoring/prometheus/metrics",
+                         metrics_render: Optional[Callable[[], bytes]]
+                         = None):
+    def dispatch(method: str, path: str, body: bytes):
+        try:
+            if path == "/healthz":
+                return 200, "text/plain", b"ok"
+            if metrics_render is not None and path == metrics_path:
+                return 200, "text/plain", metrics_render()
+            parsed = parse_model_url(path)
+            if parsed is None:
+                mt.proxy_requests_failed.labels("http").inc()
+                return (404, "application/json", json.dumps(
+                    {"error": f"Malformed url: {path}"}).encode())
+            name, version, verb = parsed
+            if verb == "predict":
+                payload = _predict_sync(handler, name, version, body)
+            elif verb in ("classify", "regress"):
+                payload = _classify_regress_sync(handler, name, version,
+                                                 verb, body)
+            elif verb == "metadata":
+                payload = _metadata_sync(handler, name, version)
+            elif verb == "":
+                payload = _status_sync(handler, name, version)
+            else:
+                return (400, "application/json", json.dumps(
+                    {"error": f"unsupported method :{verb}"}).encode())
+            return 200, "application/json", json.dumps(payload).encode()
+        except ServingError as e:
+            mt.proxy_requests_failed.labels("http").inc()
+            return (_CODE_TO_HTTP.get(e.code, 500), "application/json",
+                    json.dumps({"error": str(e)}).encode())
+        except Exception as e:      # noqa: BLE001
+            log.exception("REST dispatcher error")
+            mt.proxy_requests_failed.labels("http").inc()
+            return (500, "application/json",
+                    json.dumps({"error": str(e)}).encode())
+    return dispatch
