@@ -198,3 +198,31 @@ def test_native_rest_fast_path_gpu(tmp_path):
                                    rtol=0.05, atol=0.02)
     finally:
         srv.stop()
+
+
+def test_cpp_json_parser_probe():
+    """The C++ dense-JSON parser (fast REST path) against known arrays
+    and ragged rejects — CPU-runnable via the test binding."""
+    import json
+    import torch  # noqa: F401
+    from tfservingcache_amd.engine import _tfsc_engine as ext
+    x = np.random.default_rng(0).standard_normal((3, 5, 2)).astype(
+        np.float32)
+    dims, vals = ext._rest_parse_probe(json.dumps(x.tolist()).encode(),
+                                       False)
+    assert list(dims) == [3, 5, 2]
+    np.testing.assert_allclose(np.array(vals).reshape(x.shape), x,
+                               rtol=1e-6)
+    ids = np.array([[1, 2, 3], [4, 5, 6]], dtype=np.int32)
+    dims, vals = ext._rest_parse_probe(json.dumps(ids.tolist()).encode(),
+                                       True)
+    assert list(dims) == [2, 3] and list(vals) == [1, 2, 3, 4, 5, 6]
+    for bad in (b"[[1,2],[3]]", b"[[1,2],3]", b"[3,[1,2]]", b'["x"]'):
+        with pytest.raises(Exception):
+            ext._rest_parse_probe(bad, False)
+    # exponent + long-mantissa fallbacks agree with python floats
+    dims, vals = ext._rest_parse_probe(
+        b"[1e-5, -2.5E3, 0.12345678901234567890, -7]", False)
+    np.testing.assert_allclose(
+        vals, np.array([1e-5, -2.5e3, 0.12345678901234568, -7.0],
+                       dtype=np.float32), rtol=1e-6)

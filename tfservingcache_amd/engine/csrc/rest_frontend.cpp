@@ -268,8 +268,13 @@ struct JParser {
   }
 };
 
-// parse a nested numeric array; dims discovered from the first element
-// at each depth and enforced for the rest
+// parse a nested numeric array; dims[d] is the (uniform) length of the
+// arrays at depth d. Inner arrays complete before outer ones, so dims
+// slots are filled innermost-first: grow with -1 sentinels and assign
+// on first completion at each depth, enforce equality afterwards.
+// Rectangularity of the element COUNT vs the dims product is
+// re-checked downstream (FastModel::validate), so rare shapes that
+// slip through here still fall back cleanly.
 static void parse_dense(JParser& j, int depth,
                         std::vector<int64_t>& dims,
                         std::vector<float>& fdata,
@@ -290,14 +295,21 @@ static void parse_dense(JParser& j, int depth,
       }
       if (!j.eat(']')) throw RestFallback("expected ]");
     }
-    if (int(dims.size()) <= depth) dims.push_back(count);
+    if (int(dims.size()) <= depth) dims.resize(depth + 1, -1);
+    if (dims[depth] == -1) dims[depth] = count;
     else if (dims[depth] != count) throw RestFallback("ragged");
     return;
   }
   double v = j.number();
+  // a number at depth d is ragged if any ARRAY exists at depth >= d
   if (int(dims.size()) > depth) throw RestFallback("ragged depth");
   if (is_int) idata.push_back(int32_t(llround(v)));
   else fdata.push_back(float(v));
+}
+
+static void check_dims_complete(const std::vector<int64_t>& dims) {
+  for (int64_t d : dims)
+    if (d < 0) throw RestFallback("incomplete dims");
 }
 
 struct ParsedInput {
@@ -694,6 +706,7 @@ class RestFrontend {
     }
     j.expect('}');
     if (!saw_payload || parsed.empty()) throw RestFallback("no payload");
+    for (auto& kv : parsed) check_dims_complete(kv.second.dims);
 
     // build the PredictRequest
     std::string reqpb;
@@ -925,4 +938,23 @@ void register_rest_frontend(py::module_& mod) {
                  std::move(keep));
            })
       .def("unregister_model", &RestFrontend::unregister_model);
+
+  // test-only: run the dense-JSON array parser and return (dims, data)
+  mod.def("_rest_parse_probe", [](py::bytes body, bool is_int) {
+    std::string b = body;
+    tfsc_rest::JParser j{b.data(), b.data() + b.size()};
+    std::vector<int64_t> dims;
+    std::vector<float> f;
+    std::vector<int32_t> iv;
+    tfsc_rest::parse_dense(j, 0, dims, f, iv, is_int);
+    tfsc_rest::check_dims_complete(dims);
+    py::list pd;
+    for (auto d : dims) pd.append(d);
+    py::list pv;
+    if (is_int)
+      for (auto v : iv) pv.append(v);
+    else
+      for (auto v : f) pv.append(v);
+    return py::make_tuple(pd, pv);
+  });
 }
