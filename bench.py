@@ -66,18 +66,77 @@ REQS_PER_STEP = 50
 
 
 def _client_proc_main(conn, port, model_name, model_kind, batch,
-                      image_size, seq_len, threads, channels):
-    """Spawned gRPC client worker (warm mode): builds its own request
-    bytes, opens its own channels, executes 'run N' commands from the
+                      image_size, seq_len, threads, channels,
+                      transport="native"):
+    """Spawned client worker (warm mode): builds its own request bytes,
+    opens its own channels/sockets, executes 'run N' commands from the
     parent over the pipe, returns per-request latencies."""
-    import grpc as grpc_mod
     import numpy as _np
     import time as _time
     from concurrent.futures import ThreadPoolExecutor as _TPE
-    from tfservingcache_amd.wire import messages as _m
-    from tfservingcache_amd.wire.tensor import numpy_to_tensorproto as _n2t
 
     rng = _np.random.default_rng(os.getpid())
+
+    if transport == "native-rest":
+        import json as _json
+        import socket as _socket
+        import threading as _threading
+        if model_kind == "bert_base":
+            payload = {"instances": rng.integers(
+                0, 30000, (batch, seq_len)).astype(int).tolist()}
+        else:
+            payload = {"instances": _np.round(
+                rng.standard_normal(
+                    (batch, image_size, image_size, 3)) * 0.5,
+                5).tolist()}
+        body = _json.dumps(payload).encode()
+        raw = (f"POST /v1/models/{model_name}:predict HTTP/1.1\r\n"
+               f"Host: 127.0.0.1\r\n"
+               f"Content-Type: application/json\r\n"
+               f"Content-Length: {len(body)}\r\n\r\n"
+               ).encode() + body
+        tl = _threading.local()
+
+        def one(i):
+            t0 = _time.monotonic()
+            sock = getattr(tl, "sock", None)
+            if sock is None:
+                sock = _socket.create_connection(("127.0.0.1", port),
+                                                 timeout=300)
+                sock.setsockopt(_socket.IPPROTO_TCP,
+                                _socket.TCP_NODELAY, 1)
+                tl.sock = sock
+                tl.buf = b""
+            sock.sendall(raw)
+            buf = tl.buf
+            while b"\r\n\r\n" not in buf:
+                buf += sock.recv(1 << 16)
+            head, rest = buf.split(b"\r\n\r\n", 1)
+            clen = 0
+            for line in head.split(b"\r\n"):
+                if line.lower().startswith(b"content-length"):
+                    clen = int(line.split(b":")[1])
+            while len(rest) < clen:
+                rest += sock.recv(1 << 18)
+            tl.buf = rest[clen:]
+            if not head.startswith(b"HTTP/1.1 200"):
+                raise RuntimeError(head[:60].decode(errors="replace"))
+            return _time.monotonic() - t0
+
+        pool = _TPE(max_workers=threads)
+        while True:
+            cmd = conn.recv()
+            if cmd[0] == "stop":
+                conn.close()
+                return
+            _, n = cmd
+            lats = list(pool.map(one, range(n)))
+            conn.send(lats)
+        return
+
+    import grpc as grpc_mod
+    from tfservingcache_amd.wire import messages as _m
+    from tfservingcache_amd.wire.tensor import numpy_to_tensorproto as _n2t
     if model_kind == "bert_base":
         inputs = {"input_ids": _n2t(rng.integers(
             0, 30000, (batch, seq_len)).astype(_np.int32))}
@@ -530,7 +589,7 @@ def main() -> int:
     client_conns = []
     client_procs = []
     if args.client_procs > 0:
-        if args.transport not in ("grpc", "native") or \
+        if args.transport not in ("grpc", "native", "native-rest") or \
                 eff_mode != "warm":
             raise SystemExit("--client-procs needs warm mode and a "
                              "socket transport")
@@ -541,9 +600,12 @@ def main() -> int:
             parent, child = ctx.Pipe()
             proc = ctx.Process(
                 target=_client_proc_main,
-                args=(child, gport, names[0], kind_of[names[0]],
+                args=(child, gport if args.transport != "native-rest"
+                      else rport,
+                      names[0], kind_of[names[0]],
                       args.batch, args.image_size, args.seq_len, per,
-                      max(1, args.channels // args.client_procs)),
+                      max(1, args.channels // args.client_procs),
+                      args.transport),
                 daemon=True)
             proc.start()
             client_conns.append(parent)

@@ -217,14 +217,9 @@ def test_cpp_json_parser_probe():
     dims, vals = ext._rest_parse_probe(json.dumps(ids.tolist()).encode(),
                                        True)
     assert list(dims) == [2, 3] and list(vals) == [1, 2, 3, 4, 5, 6]
-    for bad in (b"[[1,2],[3]]", b"[[1,2],3]", b'["x"]'):
+    for bad in (b"[[1,2],[3]]", b"[[1,2],3]", b"[3,[1,2]]", b'["x"]'):
         with pytest.raises(Exception):
             ext._rest_parse_probe(bad, False)
-    # number-before-array raggedness is NOT caught at parse time by
-    # design — the dims product no longer matches the element count, so
-    # FastModel::validate rejects it downstream (clean fallback)
-    dims, vals = ext._rest_parse_probe(b"[3,[1,2]]", False)
-    assert int(np.prod(dims)) != len(vals)
     # exponent + long-mantissa fallbacks agree with python floats
     dims, vals = ext._rest_parse_probe(
         b"[1e-5, -2.5E3, 0.12345678901234567890, -7]", False)

@@ -28,6 +28,7 @@
 
 #include <atomic>
 #include <cctype>
+#include <chrono>
 #include <cmath>
 #include <cstring>
 #include <map>
@@ -285,6 +286,70 @@ static void parse_dense(JParser& j, int depth,
     int64_t count = 0;
     j.ws();
     if (j.p < j.end && *j.p == ']') { ++j.p; }
+    else if (j.p < j.end && *j.p != '[') {
+      // innermost array: tight inline number loop (the bulk of a
+      // multi-MB image body) — no recursion/function calls per leaf
+      static const double kPow10[19] = {
+          1e0, 1e1, 1e2, 1e3, 1e4, 1e5, 1e6, 1e7, 1e8, 1e9, 1e10,
+          1e11, 1e12, 1e13, 1e14, 1e15, 1e16, 1e17, 1e18};
+      const char* p = j.p;
+      const char* end = j.end;
+      while (true) {
+        const char* start = p;
+        bool neg = false;
+        if (p < end && (*p == '-' || *p == '+')) {
+          neg = (*p == '-');
+          ++p;
+        }
+        uint64_t mant = 0;
+        int digits = 0, frac = 0;
+        while (p < end && unsigned(*p - '0') < 10u) {
+          mant = mant * 10 + uint64_t(*p - '0');
+          ++digits;
+          ++p;
+        }
+        if (p < end && *p == '.') {
+          ++p;
+          while (p < end && unsigned(*p - '0') < 10u) {
+            mant = mant * 10 + uint64_t(*p - '0');
+            ++digits;
+            ++frac;
+            ++p;
+          }
+        }
+        double v;
+        if (digits == 0 || digits > 18 ||
+            (p < end && (*p == 'e' || *p == 'E'))) {
+          char* np = nullptr;
+          v = strtod(start, &np);
+          if (np == start) throw RestFallback("expected number");
+          p = np;
+        } else {
+          v = double(mant) / kPow10[frac];
+          if (neg) v = -v;
+        }
+        if (is_int) idata.push_back(int32_t(llround(v)));
+        else fdata.push_back(float(v));
+        ++count;
+        while (p < end && (*p == ' ' || *p == '\t' || *p == '\n' ||
+                           *p == '\r'))
+          ++p;
+        if (p < end && *p == ',') {
+          ++p;
+          while (p < end && (*p == ' ' || *p == '\t' || *p == '\n' ||
+                             *p == '\r'))
+            ++p;
+          continue;
+        }
+        break;
+      }
+      if (p >= end || *p != ']') throw RestFallback("expected ]");
+      ++p;
+      j.p = p;
+      // leaves live one level deeper than this array
+      if (int(dims.size()) > depth + 1)
+        throw RestFallback("ragged depth");
+    }
     else {
       while (true) {
         parse_dense(j, depth + 1, dims, fdata, idata, is_int);
@@ -956,5 +1021,23 @@ void register_rest_frontend(py::module_& mod) {
     else
       for (auto v : f) pv.append(v);
     return py::make_tuple(pd, pv);
+  });
+
+  // test-only: time the raw parse (no python list conversion)
+  mod.def("_rest_parse_bench", [](py::bytes body, int iters) {
+    std::string b = body;
+    double best = 1e30;
+    for (int i = 0; i < iters; ++i) {
+      std::vector<int64_t> dims;
+      std::vector<float> f;
+      std::vector<int32_t> iv;
+      auto t0 = std::chrono::steady_clock::now();
+      tfsc_rest::JParser j{b.data(), b.data() + b.size()};
+      tfsc_rest::parse_dense(j, 0, dims, f, iv, false);
+      auto dt = std::chrono::duration<double>(
+          std::chrono::steady_clock::now() - t0).count();
+      if (dt < best) best = dt;
+    }
+    return best;
   });
 }
