@@ -9,12 +9,15 @@
 // score materialization.
 //
 // Geometry: one workgroup (4 waves) per (b, h, 64-row q-block). Q block
-// kept in registers (2 A-fragments per wave), K and V^T tiles
-// double-buffered in LDS with the gemm.hip XOR swizzle; per kv-tile:
-// QK^T on MFMA -> in-register online softmax (16-lane-group shfl
-// reduction per q-row) -> P bounced through LDS into A-fragment layout
-// -> P@V on MFMA into the O accumulator. Requires head_dim D == 64
-// (BERT-base class); the planner keeps the unfused path otherwise.
+// kept in registers (2 A-fragments per wave). Round-2 revision (the
+// round-1 kernel measured 11% MFMA-busy / 44% wave-parked — barrier-
+// and staging-latency bound): kv tiles are processed TWO per barrier
+// from a 4-deep LDS ring (K and V^T 4 x 8 KB each), so each
+// __syncthreads covers 128 kv columns of MFMA work and the online
+// softmax runs once per 128 columns instead of per 64; the P bounce
+// region overlaps the (dead after the prologue) Q tile. 80 KB LDS ->
+// 2 workgroups/CU. Requires head_dim D == 64 (BERT-base class); the
+// planner keeps the unfused path otherwise.
 #include "../common.h"
 #include "../kernels.h"
 
@@ -29,13 +32,18 @@ using f32x4_t = __attribute__((ext_vector_type(4))) float;
 
 constexpr int D = 64;          // head dim
 constexpr int QBLK = 64;       // q rows per workgroup (16 per wave)
-constexpr int KVBLK = 64;      // kv rows per tile
+constexpr int KVBLK = 64;      // kv rows per staged tile
 constexpr int NW = 4;          // waves
 constexpr int THREADS = NW * WAVE;
 
 // shared 64x64 bf16 tile image with the gemm.hip chunk-XOR swizzle
 TFSC_DEV int t_off(int row, int chunk) {
   return row * 128 + ((chunk ^ (row & 7)) << 4);
+}
+
+// [16][128] bf16 image (P bounce): row stride 256 B, 16 chunks
+TFSC_DEV int p_off(int row, int chunk) {
+  return row * 256 + ((chunk ^ (row & 7)) << 4);
 }
 
 // stage a [64][64] tile from rows of a [rows_total, H*D] matrix
@@ -58,8 +66,8 @@ TFSC_DEV void stage_rows_glds(const ushort* __restrict__ src, int64_t ld,
   }
 }
 
-// stage V^T: [d][kv] image from V rows [kv, H*D] (d-contiguous reads,
-// scattered 2B LDS writes — the bgemm no-trans pattern)
+// stage V^T: [d][kv] image from V rows [kv, H*D] (one vectorized 16 B
+// global read per group, 8 scattered 2 B LDS writes)
 TFSC_DEV void stage_vt(const ushort* __restrict__ src, int64_t ld,
                        int row0, int row_limit, ushort* lds_tile,
                        int wave, int lane) {
@@ -72,12 +80,13 @@ TFSC_DEV void stage_vt(const ushort* __restrict__ src, int64_t ld,
     int gkv = row0 + kv;
     gkv = gkv < row_limit ? gkv : row_limit;
     const ushort* gptr = src + (int64_t)gkv * ld + dg * 8;
+    short8_t v = *reinterpret_cast<const short8_t*>(gptr);
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int d = dg * 8 + j;
       int byte_off = t_off(d, kv >> 3) + (kv & 7) * 2;
       *reinterpret_cast<ushort*>(
-          reinterpret_cast<char*>(lds_tile) + byte_off) = gptr[j];
+          reinterpret_cast<char*>(lds_tile) + byte_off) = ushort(v[j]);
     }
   }
 }
@@ -88,12 +97,14 @@ void attention_kernel(const ushort* __restrict__ Q,
                       const ushort* __restrict__ V,
                       ushort* __restrict__ O,
                       int B, int S, int H, float scale, int n_qblk) {
-  // LDS: Q[64][64] | K dbuf 2x | V^T dbuf 2x | P bounce 4x[16][64]
-  __shared__ __attribute__((aligned(16))) char smem[8192 * 6];
-  char* q_lds = smem;
-  auto k_lds = [&](int buf) -> char* { return smem + 8192 + buf * 8192; };
-  auto vt_lds = [&](int buf) -> char* { return smem + 24576 + buf * 8192; };
-  char* p_lds = smem + 40960;           // 4 waves x 2KB
+  // LDS (80 KB): K ring 4x8KB | V^T ring 4x8KB | q/p overlay 16 KB
+  // (q tile uses the first 8 KB until its registers are loaded; the
+  // per-wave [16][128] P bounce then reuses the whole 16 KB)
+  __shared__ __attribute__((aligned(16))) char smem[8192 * 10];
+  auto k_lds = [&](int buf) -> char* { return smem + buf * 8192; };
+  auto vt_lds = [&](int buf) -> char* { return smem + 32768 + buf * 8192; };
+  char* q_lds = smem + 65536;
+  char* p_lds = smem + 65536;           // 4 waves x 4 KB (after Q)
 
   const int flat = blockIdx.x;
   const int qb = flat % n_qblk;
@@ -121,7 +132,7 @@ void attention_kernel(const ushort* __restrict__ Q,
     qf[ks] = *reinterpret_cast<const bf16x8_t*>(
         q_lds + t_off(row, ks * 4 + kgrp));
   }
-  __syncthreads();   // q_lds no longer needed (could be reused)
+  __syncthreads();   // q_lds consumed; the region becomes the P bounce
 
   // ---- online softmax state: this lane covers rows wave*16 + kgrp*4+r
   float m_run[4], l_run[4];
@@ -133,40 +144,51 @@ void attention_kernel(const ushort* __restrict__ Q,
   }
 
   const int n_kv = (S + KVBLK - 1) / KVBLK;
-  // prologue staging of kv tile 0
+  const int n_pair = (n_kv + 1) / 2;
+  // prologue: stage tile pair 0 into ring slots 0,1 (an absent second
+  // tile stages clamped rows; its columns are masked below)
   stage_rows_glds(Kb, ld, 0, S - 1, k_lds(0), wave, lane);
   stage_vt(Vb, ld, 0, S - 1, reinterpret_cast<ushort*>(vt_lds(0)), wave,
            lane);
+  stage_rows_glds(Kb, ld, KVBLK, S - 1, k_lds(1), wave, lane);
+  stage_vt(Vb, ld, KVBLK, S - 1, reinterpret_cast<ushort*>(vt_lds(1)),
+           wave, lane);
 
-  int cur = 0;
-  for (int t = 0; t < n_kv; ++t) {
-    __syncthreads();          // current tile staged (glds drained here)
-    if (t + 1 < n_kv) {
-      stage_rows_glds(Kb, ld, (t + 1) * KVBLK, S - 1, k_lds(cur ^ 1),
+  for (int t = 0; t < n_pair; ++t) {
+    __syncthreads();          // current pair staged (glds drained here)
+    const int base = (t & 1) * 2;        // ring slots of this pair
+    if (t + 1 < n_pair) {
+      const int nb = ((t + 1) & 1) * 2;
+      stage_rows_glds(Kb, ld, (2 * t + 2) * KVBLK, S - 1, k_lds(nb),
                       wave, lane);
-      stage_vt(Vb, ld, (t + 1) * KVBLK, S - 1,
-               reinterpret_cast<ushort*>(vt_lds(cur ^ 1)), wave, lane);
+      stage_vt(Vb, ld, (2 * t + 2) * KVBLK, S - 1,
+               reinterpret_cast<ushort*>(vt_lds(nb)), wave, lane);
+      stage_rows_glds(Kb, ld, (2 * t + 3) * KVBLK, S - 1, k_lds(nb + 1),
+                      wave, lane);
+      stage_vt(Vb, ld, (2 * t + 3) * KVBLK, S - 1,
+               reinterpret_cast<ushort*>(vt_lds(nb + 1)), wave, lane);
     }
 
-    // ---- QK^T: scores[16 q rows][64 kv] = 1x4 fragments
-    f32x4_t sc[4] = {};
-    {
-      const char* kt = k_lds(cur);
+    // ---- QK^T: scores[16 q rows][128 kv] = 8 x 1x4 fragments
+    f32x4_t sc[8] = {};
+    #pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const char* kt = k_lds(base + half);
       #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         #pragma unroll
         for (int ni = 0; ni < 4; ++ni) {
           bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
               kt + t_off(ni * 16 + frow, ks * 4 + kgrp));
-          sc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              qf[ks], kf, sc[ni], 0, 0, 0);
+          sc[half * 4 + ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qf[ks], kf, sc[half * 4 + ni], 0, 0, 0);
         }
       }
     }
-    // scale + mask the kv tail (rows beyond S were staged as clamps)
-    const int kv_base = t * KVBLK;
+    // scale + mask the kv tail
+    const int kv_base = t * 2 * KVBLK;
     #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
+    for (int ni = 0; ni < 8; ++ni) {
       int kv_col = kv_base + ni * 16 + frow;   // this lane's kv column
       bool valid = kv_col < S;
       #pragma unroll
@@ -174,15 +196,13 @@ void attention_kernel(const ushort* __restrict__ Q,
         sc[ni][r] = valid ? sc[ni][r] * scale : -3.0e38f;
     }
 
-    // ---- per-row max over the 64 kv columns:
-    // row r lives in reg r across the 16 lanes of this lane's quarter
-    // group and across the 4 fragments -> local max over frags, then
-    // 4-step shfl_xor within the 16-lane group
-    float pmax[4], psum[4];
+    // ---- per-row max over the 128 kv columns (one pass per pair)
+    float pmax[4];
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float v = fmaxf(fmaxf(sc[0][r], sc[1][r]),
-                      fmaxf(sc[2][r], sc[3][r]));
+      float v = sc[0][r];
+      #pragma unroll
+      for (int ni = 1; ni < 8; ++ni) v = fmaxf(v, sc[ni][r]);
       #pragma unroll
       for (int off = 8; off > 0; off >>= 1)
         v = fmaxf(v, __shfl_xor(v, off, 16));
@@ -201,14 +221,13 @@ void attention_kernel(const ushort* __restrict__ Q,
     for (int r = 0; r < 4; ++r) {
       float sum = 0.f;
       #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
+      for (int ni = 0; ni < 8; ++ni) {
         sc[ni][r] = __expf(sc[ni][r] - m_run[r]);
         sum += sc[ni][r];
       }
       #pragma unroll
       for (int off = 8; off > 0; off >>= 1)
         sum += __shfl_xor(sum, off, 16);
-      psum[r] = sum;
       l_run[r] = l_run[r] * alpha[r] + sum;
     }
     // rescale O by alpha (rows of O fragments match reg index)
@@ -218,38 +237,37 @@ void attention_kernel(const ushort* __restrict__ Q,
       for (int r = 0; r < 4; ++r)
         o_acc[ni][r] *= alpha[r];
 
-    // ---- P (C-layout) -> LDS -> A-fragment layout
+    // ---- P (C-layout) -> LDS [16][128] -> A-fragment layout
     {
-      ushort* pw = reinterpret_cast<ushort*>(p_lds + wave * 2048);
+      ushort* pw = reinterpret_cast<ushort*>(p_lds + wave * 4096);
       #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
+      for (int ni = 0; ni < 8; ++ni) {
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int row = kgrp * 4 + r;            // 0..15
-          int col = ni * 16 + frow;          // 0..63
-          // swizzled [16][64] image (row stride 128B)
+          int col = ni * 16 + frow;          // 0..127
           *reinterpret_cast<ushort*>(
-              reinterpret_cast<char*>(pw) + t_off(row, col >> 3) +
+              reinterpret_cast<char*>(pw) + p_off(row, col >> 3) +
               (col & 7) * 2) = f2bf(sc[ni][r]);
         }
       }
       // wave-local write->read; compiler orders via lgkmcnt
       const char* pr = reinterpret_cast<const char*>(pw);
-      const char* vt = vt_lds(cur);
       #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
+      for (int ks = 0; ks < 4; ++ks) {       // contraction over 128 kv
         bf16x8_t pf = *reinterpret_cast<const bf16x8_t*>(
-            pr + t_off(frow, ks * 4 + kgrp));
+            pr + p_off(frow, ks * 4 + kgrp));
+        const char* vt = vt_lds(base + (ks >> 1));
+        int ksub = ks & 1;
         #pragma unroll
         for (int ni = 0; ni < 4; ++ni) {
           bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
-              vt + t_off(ni * 16 + frow, ks * 4 + kgrp));
+              vt + t_off(ni * 16 + frow, ksub * 4 + kgrp));
           o_acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               pf, vf, o_acc[ni], 0, 0, 0);
         }
       }
     }
-    cur ^= 1;
   }
 
   // ---- epilogue: O /= l, store rows q0 + wave*16 + kgrp*4 + r
