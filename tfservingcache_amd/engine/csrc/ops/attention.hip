@@ -122,8 +122,16 @@ void attention_kernel(const ushort* __restrict__ Q,
   const int kgrp = lane >> 4;
   const int q0 = qb * QBLK;
 
-  // ---- load Q block to LDS, then this wave's 16 rows into registers
+  // ---- stage Q AND the first kv tile pair concurrently (disjoint
+  // LDS; one barrier covers all five tiles), then pull this wave's 16
+  // q rows into registers
   stage_rows_glds(Qb, ld, q0, S - 1, q_lds, wave, lane);
+  stage_rows_glds(Kb, ld, 0, S - 1, k_lds(0), wave, lane);
+  stage_vt(Vb, ld, 0, S - 1, reinterpret_cast<ushort*>(vt_lds(0)), wave,
+           lane);
+  stage_rows_glds(Kb, ld, KVBLK, S - 1, k_lds(1), wave, lane);
+  stage_vt(Vb, ld, KVBLK, S - 1, reinterpret_cast<ushort*>(vt_lds(1)),
+           wave, lane);
   __syncthreads();
   bf16x8_t qf[2];      // A-fragments for the 2 K=32 steps over D
   #pragma unroll
@@ -145,17 +153,11 @@ void attention_kernel(const ushort* __restrict__ Q,
 
   const int n_kv = (S + KVBLK - 1) / KVBLK;
   const int n_pair = (n_kv + 1) / 2;
-  // prologue: stage tile pair 0 into ring slots 0,1 (an absent second
-  // tile stages clamped rows; its columns are masked below)
-  stage_rows_glds(Kb, ld, 0, S - 1, k_lds(0), wave, lane);
-  stage_vt(Vb, ld, 0, S - 1, reinterpret_cast<ushort*>(vt_lds(0)), wave,
-           lane);
-  stage_rows_glds(Kb, ld, KVBLK, S - 1, k_lds(1), wave, lane);
-  stage_vt(Vb, ld, KVBLK, S - 1, reinterpret_cast<ushort*>(vt_lds(1)),
-           wave, lane);
 
   for (int t = 0; t < n_pair; ++t) {
-    __syncthreads();          // current pair staged (glds drained here)
+    // pair 0 was staged with Q (covered by the prologue barriers); a
+    // barrier here covers the pair t prefetch issued in iteration t-1
+    if (t > 0) __syncthreads();
     const int base = (t & 1) * 2;        // ring slots of this pair
     if (t + 1 < n_pair) {
       const int nb = ((t + 1) & 1) * 2;
