@@ -91,20 +91,26 @@ TFSC_DEV void stage_vt(const ushort* __restrict__ src, int64_t ld,
   }
 }
 
+// RING = staged kv-tile slots: 4 (pair prefetch, 80 KB LDS, 2 WG/CU)
+// for long sequences; 2 (no prefetch needed when n_kv <= 2, 48 KB LDS,
+// 3 WG/CU) for the S <= 128 serving shapes
+template <int RING>
 __global__ __launch_bounds__(THREADS)
 void attention_kernel(const ushort* __restrict__ Q,
                       const ushort* __restrict__ K,
                       const ushort* __restrict__ V,
                       ushort* __restrict__ O,
                       int B, int S, int H, float scale, int n_qblk) {
-  // LDS (80 KB): K ring 4x8KB | V^T ring 4x8KB | q/p overlay 16 KB
+  // LDS: K ring RINGx8KB | V^T ring RINGx8KB | q/p overlay 16 KB
   // (q tile uses the first 8 KB until its registers are loaded; the
   // per-wave [16][128] P bounce then reuses the whole 16 KB)
-  __shared__ __attribute__((aligned(16))) char smem[8192 * 10];
+  __shared__ __attribute__((aligned(16))) char smem[8192 * (2 * RING + 2)];
   auto k_lds = [&](int buf) -> char* { return smem + buf * 8192; };
-  auto vt_lds = [&](int buf) -> char* { return smem + 32768 + buf * 8192; };
-  char* q_lds = smem + 65536;
-  char* p_lds = smem + 65536;           // 4 waves x 4 KB (after Q)
+  auto vt_lds = [&](int buf) -> char* {
+    return smem + RING * 8192 + buf * 8192;
+  };
+  char* q_lds = smem + 2 * RING * 8192;
+  char* p_lds = smem + 2 * RING * 8192;  // 4 waves x 4 KB (after Q)
 
   const int flat = blockIdx.x;
   const int qb = flat % n_qblk;
@@ -158,8 +164,8 @@ void attention_kernel(const ushort* __restrict__ Q,
     // pair 0 was staged with Q (covered by the prologue barriers); a
     // barrier here covers the pair t prefetch issued in iteration t-1
     if (t > 0) __syncthreads();
-    const int base = (t & 1) * 2;        // ring slots of this pair
-    if (t + 1 < n_pair) {
+    const int base = RING >= 4 ? (t & 1) * 2 : 0;  // slots of this pair
+    if (RING >= 4 && t + 1 < n_pair) {
       const int nb = ((t + 1) & 1) * 2;
       stage_rows_glds(Kb, ld, (2 * t + 2) * KVBLK, S - 1, k_lds(nb),
                       wave, lane);
@@ -294,8 +300,14 @@ void launch_attention(hipStream_t s, const ushort* Q, const ushort* K,
     throw std::runtime_error("fused attention requires head_dim == 64");
   int n_qblk = (S + attn::QBLK - 1) / attn::QBLK;
   dim3 grid(B * H * n_qblk);
-  hipLaunchKernelGGL(attn::attention_kernel, grid, dim3(attn::THREADS), 0,
-                     s, Q, K, V, O, B, S, H, scale, n_qblk);
+  if (S <= 2 * attn::KVBLK)
+    hipLaunchKernelGGL(attn::attention_kernel<2>, grid,
+                       dim3(attn::THREADS), 0, s, Q, K, V, O, B, S, H,
+                       scale, n_qblk);
+  else
+    hipLaunchKernelGGL(attn::attention_kernel<4>, grid,
+                       dim3(attn::THREADS), 0, s, Q, K, V, O, B, S, H,
+                       scale, n_qblk);
 }
 
 }  // namespace tfsc
