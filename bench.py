@@ -254,6 +254,9 @@ def main() -> int:
                     help="enable server-side dynamic batching")
     ap.add_argument("--batch-timeout-ms", type=float, default=1.0,
                     help="dynamic-batching merge window")
+    ap.add_argument("--dtype", choices=["bf16", "fp8"], default="bf16",
+                    help="engine compute dtype (fp8: e4m3 GEMMs with "
+                         "rowwise dequant; conv stays bf16)")
     ap.add_argument("--streams", type=int, default=6,
                     help="execution contexts (HIP streams) per model")
     ap.add_argument("--channels", type=int, default=1,
@@ -333,7 +336,8 @@ def main() -> int:
                                  batching=args.dyn_batch,
                                  batch_timeout_s=args.batch_timeout_ms
                                  / 1e3,
-                                 n_streams=args.streams)
+                                 n_streams=args.streams,
+                                 dtype=args.dtype)
     pool = ModelPool(loader, max_concurrent_models=pool_cap, device=device)
     cm = CacheManager(provider, cache, pool, model_fetch_timeout=300.0)
     handler = LocalServingHandler(cm)
@@ -696,7 +700,7 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if not args.cpu else "f32",
+            "dtype": (args.dtype if not args.cpu else "f32"),
             "data": "synthetic",
             "config": {
                 "model": {"resnet50": "resnet50_v1.5",

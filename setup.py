@@ -21,6 +21,7 @@ sources = [
     os.path.join(CSRC, "executor.cpp"),
     os.path.join(CSRC, "ops", "ops_memory.hip"),
     os.path.join(CSRC, "ops", "gemm.hip"),
+    os.path.join(CSRC, "ops", "gemm_fp8.hip"),
     os.path.join(CSRC, "ops", "conv.hip"),
     os.path.join(CSRC, "ops", "attention.hip"),
     os.path.join(CSRC, "fastpath.cpp"),

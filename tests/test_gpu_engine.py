@@ -302,3 +302,57 @@ def test_einsum_gpu_vs_cpu(tmp_path):
     x = (rng.standard_normal((4, 12, 256)) * 0.5).astype(np.float32)
     _compare(gm.predict({"x": x}), cm.predict({"x": x}),
              rtol=0.05, atol=0.03)
+
+
+def test_fp8_gemm_mlp_vs_cpu(tmp_path):
+    """fp8 (e4m3 rowwise) GEMM path vs the CPU fp32 reference — fp8
+    quantization noise bounds the tolerance (~2^-9 relative on amax)."""
+    from tfservingcache_amd.engine.gpu import GpuModel
+    sm = build_mlp(d_in=256, d_hidden=512, d_out=200, seed=7)
+    d = tmp_path / "m8" / "1"
+    write_saved_model(sm, str(d))
+    lm = load_model_from_dir(str(d), "m8", 1)
+    lm._gpu = GpuModel(lm.plan, device="cuda:0", max_batch=16,
+                       model_name="m8", model_version=1, dtype="fp8")
+    cm = _cpu_model(tmp_path, sm, name="m8cpu")
+    x = np.random.default_rng(3).standard_normal((8, 256)).astype(
+        np.float32)
+    g = lm.predict({"x": x})
+    c = cm.predict({"x": x})
+    # softmax output: absolute tolerance dominates
+    np.testing.assert_allclose(g["probs"], c["probs"], atol=0.05)
+    agree = (g["probs"].argmax(1) == c["probs"].argmax(1)).mean()
+    assert agree >= 0.9
+
+
+def test_fp8_accuracy_delta_bert(tmp_path):
+    """Accuracy harness: fp8 vs bf16 on a BERT-tiny —
+    the fp8 engine output must track the bf16 engine closely enough for
+    serving (cosine similarity on hidden states)."""
+    from tfservingcache_amd.engine.gpu import GpuModel
+    sm = build_bert(seq_len=64, hidden=256, layers=2, heads=4,
+                    intermediate=512, vocab=1000, seed=3)
+    d = tmp_path / "bt" / "1"
+    write_saved_model(sm, str(d))
+    ids = np.random.default_rng(0).integers(0, 1000, (4, 64)).astype(
+        np.int32)
+
+    outs = {}
+    for dt in ("bf16", "fp8"):
+        lm = load_model_from_dir(str(d), "bt", 1)
+        lm._gpu = GpuModel(lm.plan, device="cuda:0", max_batch=8,
+                           model_name="bt", model_version=1, dtype=dt)
+        outs[dt] = lm.predict({"input_ids": ids})
+        lm._gpu.release()
+
+    a = outs["bf16"]["sequence_output"].reshape(-1)
+    b = outs["fp8"]["sequence_output"].reshape(-1)
+    cos = float(np.dot(a, b) /
+                (np.linalg.norm(a) * np.linalg.norm(b) + 1e-12))
+    rel = float(np.linalg.norm(a - b) / (np.linalg.norm(a) + 1e-12))
+    print(f"fp8-vs-bf16 BERT: cos={cos:.6f} rel_l2={rel:.4f}")
+    assert cos > 0.999
+    assert rel < 0.05
+    # pooled head too
+    pa, pb = outs["bf16"]["pooled_output"], outs["fp8"]["pooled_output"]
+    np.testing.assert_allclose(pa, pb, atol=0.08)

@@ -187,7 +187,8 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
                     batching: bool = False,
                     batch_timeout_s: float = 0.001,
                     devices: Optional[List[str]] = None,
-                    n_streams: int = 6
+                    n_streams: int = 6,
+                    dtype: str = "bf16"
                     ) -> Callable[[str, int], LoadedModel]:
     """Loader that compiles the SavedModel onto an MI355X: weights land
     in the GPU's HBM pool (bf16, GEMM layouts pre-transformed) and the
@@ -212,7 +213,8 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
         dev = pick_device(name, version)
         lm._gpu = GpuModel(lm.plan, device=dev, max_batch=max_batch,
                            use_graphs=use_graphs, n_streams=n_streams,
-                           model_name=name, model_version=version)
+                           model_name=name, model_version=version,
+                           dtype=dtype)
         lm.device = dev
         if batching:
             # merging happens inside the C++ fast path (leader-follower

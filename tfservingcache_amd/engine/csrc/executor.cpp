@@ -41,6 +41,8 @@ enum CallKind : int {
   K_PAD_LAST,
   K_ATTENTION,
   K_DEPTHWISE,
+  K_GEMM_FP8,
+  K_QUANT_FP8,
 };
 
 struct Call {
@@ -141,6 +143,25 @@ static void launch_call(const Call& c, hipStream_t s) {
       // ptrs: [q, k, v, out]; ints: [B, S, H, D]; floats: [scale]
       launch_attention(s, cp(0), cp(1), cp(2), p(3), int(I[0]), int(I[1]),
                        int(I[2]), int(I[3]), c.floats[0]);
+      break;
+    case K_GEMM_FP8: {
+      // ptrs: [Aq, sa, Bq, sb, bias|0, residual|0, C]
+      // ints: [M, N, Kp, act]
+      const ushort* bias8 = c.ptrs[4] ? cp(4) : nullptr;
+      const ushort* res8 = c.ptrs[5] ? cp(5) : nullptr;
+      launch_gemm_fp8(s, reinterpret_cast<const uint8_t*>(c.ptrs[0]),
+                      reinterpret_cast<const float*>(c.ptrs[1]),
+                      reinterpret_cast<const uint8_t*>(c.ptrs[2]),
+                      reinterpret_cast<const float*>(c.ptrs[3]),
+                      bias8, res8, p(6), I[0], I[1], I[2], int(I[3]));
+      break;
+    }
+    case K_QUANT_FP8:
+      // ptrs: [x(bf16), q(u8), scales(f32)]; ints: [M, K, Kp]
+      launch_quant_rowwise(s, cp(0),
+                           reinterpret_cast<uint8_t*>(c.ptrs[1]),
+                           reinterpret_cast<float*>(c.ptrs[2]),
+                           I[0], I[1], I[2]);
       break;
     case K_DEPTHWISE:
       // ptrs: [x, w, bias, y]
@@ -288,6 +309,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("K_PAD_LAST") = int(tfsc::K_PAD_LAST);
   mod.attr("K_ATTENTION") = int(tfsc::K_ATTENTION);
   mod.attr("K_DEPTHWISE") = int(tfsc::K_DEPTHWISE);
+  mod.attr("K_GEMM_FP8") = int(tfsc::K_GEMM_FP8);
+  mod.attr("K_QUANT_FP8") = int(tfsc::K_QUANT_FP8);
 
   register_fastpath(mod);
   register_frontend(mod);

@@ -116,6 +116,23 @@ void launch_depthwise_conv(hipStream_t s, const ushort* x, const ushort* w,
                            int sh, int sw, int pt, int pl, int Ho, int Wo,
                            int act);
 
+// fp8 (OCP e4m3) GEMM with rowwise dequant in the f32 epilogue:
+// C[M,N] = act(acc(Aq @ Bq^T) * sa[m] * sb[n] + bias [+ residual]).
+// Aq/Bq are e4m3 bytes with K padded to a multiple of 128; B^T layout
+// [N][Kp] like launch_gemm's pre-transposed weights. 2x the bf16 MFMA
+// rate on gfx950 (profiles/fp8_groundwork.md).
+void launch_gemm_fp8(hipStream_t s, const uint8_t* A, const float* sa,
+                     const uint8_t* B, const float* sb,
+                     const ushort* bias, const ushort* residual,
+                     ushort* C, int64_t M, int64_t N, int64_t K,
+                     int act);
+
+// rowwise activation quantization for the fp8 GEMM: bf16 [M,K] ->
+// e4m3 [M,Kp] (zero-padded) + f32 amax/448 scale per row
+void launch_quant_rowwise(hipStream_t s, const ushort* x, uint8_t* q,
+                          float* scales, int64_t M, int64_t K,
+                          int64_t Kp);
+
 // Fused multi-head attention over the natural [B*S, H*D] QKV layout
 // (flash-style online softmax; D must be 64). See ops/attention.hip.
 void launch_attention(hipStream_t s, const ushort* Q, const ushort* K,

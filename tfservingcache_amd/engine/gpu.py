@@ -212,6 +212,14 @@ def _pad64(k: int) -> int:
     return (k + 63) // 64 * 64
 
 
+def _pad128(k: int) -> int:
+    return (k + 127) // 128 * 128
+
+
+def _pad256b(n: int) -> int:
+    return (n + 255) // 256 * 256
+
+
 _ACT_CODE = {"none": 0, "relu": 1, "tanh": 2, "sigmoid": 3, "gelu": 4,
              "relu6": 5}
 
@@ -301,8 +309,12 @@ class ExecContext:
             if op.kind == "gemm":
                 a_shape = self.shapes[op.inputs[0]]
                 K = a_shape[-1]
-                if K % 64 != 0:
-                    M = int(np.prod(a_shape[:-1]))
+                M = int(np.prod(a_shape[:-1]))
+                if self.gm.dtype == "fp8":
+                    # quantized activations (u8 [M, Kp128]) + row scales
+                    Kp = _pad128(K)
+                    scratch_sizes[oi] = _pad256b(M * Kp) + M * 4
+                elif K % 64 != 0:
                     scratch_sizes[oi] = M * _pad64(K) * 2
             if op.kind == "conv2d":
                 R, S, Cin, Kc = op.params["rsck"]
@@ -569,6 +581,8 @@ class ExecContext:
         K = a_shape[-1]
         if p.get("trans_a"):
             raise RuntimeError("gemm trans_a unsupported on GPU")
+        if self.gm.dtype == "fp8":
+            return self._c_gemm_fp8(op, oi, ext)
         wt = self.gm.gemm_weight(w_plan_idx, bool(p.get("trans_b")))
         N = wt.shape[0]
         Kp = wt.shape[1]
@@ -593,6 +607,39 @@ class ExecContext:
                       [M, N, Kp, _ACT_CODE[p.get("act", "none")]],
                       [1.0]))
         return calls
+
+    def _c_gemm_fp8(self, op: PlanOp, oi: int, ext):
+        """fp8 serving path: rowwise-quantize the activation into
+        scratch, then the e4m3 MFMA GEMM with per-row x per-col dequant
+        in the epilogue (engine.dtype: fp8)."""
+        p = op.params
+        a = op.inputs[0]
+        w_plan_idx = op.inputs[1]
+        a_shape = self.shapes[a]
+        M = int(np.prod(a_shape[:-1]))
+        K = a_shape[-1]
+        wq, wscale = self.gm.gemm_weight_fp8(w_plan_idx,
+                                             bool(p.get("trans_b")))
+        N, Kp = wq.shape
+        ni = 2
+        bias_ptr = 0
+        res_ptr = 0
+        if p.get("has_bias"):
+            bias_ptr = self.gm.weight_ptr(op.inputs[ni])
+            ni += 1
+        if p.get("residual"):
+            res_ptr = self._ptr(op.inputs[ni])
+        scratch = self.workspace.data_ptr() + self.scratch_off[oi]
+        q_ptr = scratch
+        sa_ptr = scratch + _pad256b(M * Kp)
+        return [
+            (ext.K_QUANT_FP8, [self._ptr(a), q_ptr, sa_ptr],
+             [M, K, Kp], []),
+            (ext.K_GEMM_FP8,
+             [q_ptr, sa_ptr, wq.data_ptr(), wscale.data_ptr(),
+              bias_ptr, res_ptr, self._ptr(op.outputs[0])],
+             [M, N, Kp, _ACT_CODE[p.get("act", "none")]], []),
+        ]
 
     def _c_conv(self, op: PlanOp, oi: int, ext):
         p = op.params
@@ -761,8 +808,11 @@ class GpuModel:
     def __init__(self, plan: Plan, device: str = "cuda:0",
                  max_batch: int = 64, use_graphs: bool = True,
                  n_streams: int = 6, model_name: str = "",
-                 model_version: int = 0):
+                 model_version: int = 0, dtype: str = "bf16"):
         torch, ext = _load_backend()
+        if dtype not in ("bf16", "fp8"):
+            raise ValueError(f"engine dtype {dtype!r} (bf16|fp8)")
+        self.dtype = dtype
         self.plan = plan
         self.device = device
         self.model_name = model_name
@@ -779,6 +829,7 @@ class GpuModel:
         self.use_graphs = use_graphs
         self._weights: Dict[int, object] = {}
         self._gemm_weights: Dict[Tuple[int, bool], object] = {}
+        self._gemm_weights_fp8: Dict[Tuple[int, bool], tuple] = {}
         self._conv_weights: Dict[int, object] = {}
         self._contexts: Dict[int, List[ExecContext]] = {}
         self._lock = threading.Lock()
@@ -904,6 +955,34 @@ class GpuModel:
         wt = w.to(torch.bfloat16).contiguous()
         self._gemm_weights[key] = wt
         return wt
+
+    def gemm_weight_fp8(self, idx: int, graph_trans_b: bool):
+        """(e4m3 [N][Kp128] bytes, f32 [N] per-output-channel scales)
+        for the fp8 GEMM: amax/448 per row of the pre-transposed weight
+        (the standard rowwise fp8 serving recipe; dequant happens in the
+        kernel's f32 epilogue)."""
+        torch, _ = _load_backend()
+        key = (idx, graph_trans_b)
+        ent = self._gemm_weights_fp8.get(key)
+        if ent is not None:
+            return ent
+        if idx not in self._weights:
+            raise RuntimeError("fp8 gemm requires a constant weight")
+        w = self._weights[idx].float()
+        if not graph_trans_b:
+            w = w.t().contiguous()          # [K,N] -> [N,K]
+        N, K = w.shape
+        amax = w.abs().amax(dim=1).clamp(min=1e-12)
+        scale = (amax / 448.0).to(torch.float32).contiguous()
+        q = (w / scale[:, None]).clamp(-448.0, 448.0).to(
+            torch.float8_e4m3fn).view(torch.uint8)
+        Kp = _pad128(K)
+        if Kp != K:
+            q = torch.nn.functional.pad(q, (0, Kp - K))
+        q = q.contiguous()
+        ent = (q, scale)
+        self._gemm_weights_fp8[key] = ent
+        return ent
 
     def conv_weight(self, idx: int):
         """[Kc][pad64(R*S*C)] bf16 from the [R,S,C,K] master."""
@@ -1113,5 +1192,6 @@ class GpuModel:
         with self._lock:
             self._weights.clear()
             self._gemm_weights.clear()
+            self._gemm_weights_fp8.clear()
             self._conv_weights.clear()
             self._weight_blob = None

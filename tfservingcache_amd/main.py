@@ -146,7 +146,8 @@ def create_cache_manager(cfg: Config,
             batching=cfg.get_bool("serving.batching.enabled"),
             batch_timeout_s=(cfg.get_float(
                 "serving.batching.batchTimeoutMicros") or 2000.0) / 1e6,
-            n_streams=cfg.get_int("engine.streamsPerGpu") or 6)
+            n_streams=cfg.get_int("engine.streamsPerGpu") or 6,
+            dtype=cfg.get_string("engine.dtype") or "bf16")
         device = devices[0] if len(devices) == 1 else \
             f"cuda[0-{len(devices) - 1}]"
     else:
