@@ -271,33 +271,32 @@ TFSC_DEV uint8_t f32_to_e4m3(float f) {
   return sign | uint8_t((e + 7) << 3) | uint8_t(mant);
 }
 
+// one WAVE per row (no block barrier): amax via in-wave shfl, then
+// each lane quantizes its strided elements. Values are re-read from L2
+// for the quantize pass (rows are KB-scale).
 __global__ __launch_bounds__(256)
 void quant_rowwise_kernel(const ushort* __restrict__ x,
                           uint8_t* __restrict__ q,
                           float* __restrict__ scales,
                           int M, int K, int Kp) {
-  __shared__ float red[256 / WAVE];
-  for (int row = blockIdx.x; row < M; row += gridDim.x) {
+  int wv = threadIdx.x / WAVE;
+  int lane = threadIdx.x % WAVE;
+  int waves_total = gridDim.x * (256 / WAVE);
+  for (int row = blockIdx.x * (256 / WAVE) + wv; row < M;
+       row += waves_total) {
     const ushort* xr = x + (int64_t)row * K;
     float amax = 0.f;
-    for (int i = threadIdx.x; i < K; i += 256)
+    for (int i = lane; i < K; i += WAVE)
       amax = fmaxf(amax, fabsf(bf2f(xr[i])));
     #pragma unroll
     for (int off = WAVE / 2; off > 0; off >>= 1)
       amax = fmaxf(amax, __shfl_xor(amax, off, WAVE));
-    int wv = threadIdx.x / WAVE;
-    if (threadIdx.x % WAVE == 0) red[wv] = amax;
-    __syncthreads();
-    amax = red[0];
-    #pragma unroll
-    for (int w = 1; w < 256 / WAVE; ++w) amax = fmaxf(amax, red[w]);
     float scale = amax > 0.f ? amax / 448.f : 1.f;
     float inv = 1.f / scale;
-    if (threadIdx.x == 0) scales[row] = scale;
+    if (lane == 0) scales[row] = scale;
     uint8_t* qr = q + (int64_t)row * Kp;
-    for (int i = threadIdx.x; i < Kp; i += 256)
+    for (int i = lane; i < Kp; i += WAVE)
       qr[i] = i < K ? f32_to_e4m3(bf2f(xr[i]) * inv) : 0;
-    __syncthreads();      // red[] reuse across the row loop
   }
 }
 
@@ -306,7 +305,10 @@ void quant_rowwise_kernel(const ushort* __restrict__ x,
 void launch_quant_rowwise(hipStream_t s, const ushort* x, uint8_t* q,
                           float* scales, int64_t M, int64_t K,
                           int64_t Kp) {
-  int blocks = int(M < 2048 ? M : 2048);
+  int64_t waves_needed = M;
+  int blocks = int(ceil_div(waves_needed, (int64_t)(256 / WAVE)));
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
   hipLaunchKernelGGL(fp8::quant_rowwise_kernel, dim3(blocks), dim3(256),
                      0, s, x, q, scales, int(M), int(K), int(Kp));
 }
