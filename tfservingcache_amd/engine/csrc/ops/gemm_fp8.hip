@@ -247,28 +247,29 @@ void gemm_fp8_kernel(const uint8_t* __restrict__ A,
 // amax-reduce + quantize (the row is re-read from L2, not HBM).
 // ---------------------------------------------------------------------------
 TFSC_DEV uint8_t f32_to_e4m3(float f) {
-  // OCP e4m3fn: bias 7, no inf, max 448 (verified vs HW in the probe)
+  // OCP e4m3fn: bias 7, no inf, max 448. Pure bit manipulation (the
+  // frexpf/roundf version cost 16% of BERT GPU time); round-half-up
+  // on the dropped 20 mantissa bits.
+  uint32_t bits = __float_as_uint(f);
+  uint8_t sign = uint8_t((bits >> 24) & 0x80);
+  float a = fabsf(f);
   if (f != f) return 0x7F;
-  uint8_t sign = f < 0.f ? 0x80 : 0;
-  f = fabsf(f);
-  if (f > 448.f) f = 448.f;
-  if (f < 0.001953125f / 8.f) return sign;
-  int e;
-  float m = frexpf(f, &e);
-  e -= 1;
-  m *= 2.f;
+  if (a >= 448.f) return sign | 0x7E;
+  int e = int((bits >> 23) & 0xFF) - 127;
   if (e < -6) {
-    int mant = int(roundf(f * 512.f));        // 2^-9 steps
+    // subnormal: steps of 2^-9; a < 2^-6 so the product is exact range
+    int mant = int(a * 512.f + 0.5f);
     if (mant > 7) return sign | 0x08;
     return sign | uint8_t(mant);
   }
-  int mant = int(roundf((m - 1.f) * 8.f));
-  if (mant == 8) {
-    mant = 0;
-    e += 1;
+  uint32_t mant23 = bits & 0x7FFFFF;
+  uint32_t mant3 = (mant23 + 0x80000u) >> 20;     // round half up
+  if (mant3 == 8) {
+    mant3 = 0;
+    ++e;
+    if (e > 8) return sign | 0x7E;
   }
-  if (e > 8) return sign | 0x7E;
-  return sign | uint8_t((e + 7) << 3) | uint8_t(mant);
+  return sign | uint8_t((e + 7) << 3) | uint8_t(mant3);
 }
 
 // one WAVE per row (no block barrier): amax via in-wave shfl, then
@@ -286,7 +287,15 @@ void quant_rowwise_kernel(const ushort* __restrict__ x,
        row += waves_total) {
     const ushort* xr = x + (int64_t)row * K;
     float amax = 0.f;
-    for (int i = lane; i < K; i += WAVE)
+    int k4 = K / 4;
+    const short4_t* xr4 = reinterpret_cast<const short4_t*>(xr);
+    for (int i = lane; i < k4; i += WAVE) {
+      short4_t v = xr4[i];
+      #pragma unroll
+      for (int j = 0; j < 4; ++j)
+        amax = fmaxf(amax, fabsf(bf2f((ushort)v[j])));
+    }
+    for (int i = k4 * 4 + lane; i < K; i += WAVE)
       amax = fmaxf(amax, fabsf(bf2f(xr[i])));
     #pragma unroll
     for (int off = WAVE / 2; off > 0; off >>= 1)
@@ -295,7 +304,16 @@ void quant_rowwise_kernel(const ushort* __restrict__ x,
     float inv = 1.f / scale;
     if (lane == 0) scales[row] = scale;
     uint8_t* qr = q + (int64_t)row * Kp;
-    for (int i = lane; i < Kp; i += WAVE)
+    // vector path: 4 bf16 in -> 4 e4m3 bytes out as one u32 store
+    for (int i = lane; i < k4; i += WAVE) {
+      short4_t v = xr4[i];
+      uint32_t outw = 0;
+      #pragma unroll
+      for (int j = 0; j < 4; ++j)
+        outw |= uint32_t(f32_to_e4m3(bf2f((ushort)v[j]) * inv)) << (8 * j);
+      reinterpret_cast<uint32_t*>(qr)[i] = outw;
+    }
+    for (int i = k4 * 4 + lane; i < Kp; i += WAVE)
       qr[i] = i < K ? f32_to_e4m3(bf2f(xr[i]) * inv) : 0;
   }
 }
