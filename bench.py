@@ -384,16 +384,32 @@ def main() -> int:
         owned_idx = [i for i, n in enumerate(names)
                      if rank in owners_of(n)]
         if world > 1:
-            # stage-once + RCCL fan-out of the model files (the stager
-            # fetches from the provider; replicas receive over xGMI)
+            # RCCL/xGMI fan-out of the model BYTES: rank 0 stages each
+            # FAMILY prototype once and broadcasts it over the plane
+            # (nccl == RCCL on GPU nodes); every rank then hardlinks its
+            # owned models from the landed prototype — the same
+            # content-dedup the hardlinked model repo already states in
+            # the bench config. (Fanning out 1000 byte-identical copies
+            # individually would move ~100 GB of duplicate bytes.)
+            from tfservingcache_amd.cachemanager.lrucache import (
+                Model as CacheModel, dir_size)
             plane = ReplicaPlane(device=device if not args.cpu else None)
-
-            def fetch_local(name, version):
-                entry = provider.load_model(name, version, cache.base_dir)
-                cache.put(entry)
-
-            plane.preload_replicated([(n, 1) for n in names], owners_of,
-                                     cache.base_dir, fetch_local)
+            for kind in sorted(set(kind_of.values())):
+                pname = f"_proto_{kind}"
+                if rank == 0:
+                    provider.load_model(pname, 1, cache.base_dir)
+                vdir = os.path.join(cache.base_dir, pname, "1")
+                plane.replicate_files(vdir, 0, list(range(world)))
+                for i in owned_idx:
+                    n = names[i]
+                    if kind_of[n] != kind:
+                        continue
+                    dst = os.path.join(cache.base_dir, n, "1")
+                    if not os.path.isdir(dst):
+                        _link_tree(vdir, dst)
+                    cache.put(CacheModel(name=n, version=1,
+                                         path=os.path.join(n, "1"),
+                                         size_on_disk=dir_size(dst)))
         if not owned_idx:
             owned_idx = [rank % n_models]
         w = probs_all[owned_idx]
