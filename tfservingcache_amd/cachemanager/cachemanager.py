@@ -205,16 +205,25 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
         key = f"{name}##{version}"
         return devices[zlib.crc32(key.encode()) % len(devices)]
 
+    import time as _time
+    timing = os.environ.get("TFSC_LOAD_TIMING")
+
     def loader(name: str, version: int) -> LoadedModel:
         from ..engine.gpu import GpuModel
         from ..engine.warmup import run_warmup
+        t0 = _time.monotonic()
         vdir = os.path.join(cache.base_dir, name, str(version))
         lm = load_model_from_dir(vdir, name, version)
+        t1 = _time.monotonic()
         dev = pick_device(name, version)
         lm._gpu = GpuModel(lm.plan, device=dev, max_batch=max_batch,
                            use_graphs=use_graphs, n_streams=n_streams,
                            model_name=name, model_version=version,
                            dtype=dtype)
+        t2 = _time.monotonic()
+        if timing:
+            log.warning("load timing %s: plan=%.1fms gpu=%.1fms",
+                        name, (t1 - t0) * 1e3, (t2 - t1) * 1e3)
         lm.device = dev
         if batching:
             # merging happens inside the C++ fast path (leader-follower

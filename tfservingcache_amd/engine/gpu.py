@@ -834,8 +834,15 @@ class GpuModel:
         self._contexts: Dict[int, List[ExecContext]] = {}
         self._lock = threading.Lock()
         self._released = False
+        import os as _os
+        import time as _time
+        t0 = _time.monotonic()
         with torch.cuda.device(device):
             self._upload_weights()
+        if _os.environ.get("TFSC_LOAD_TIMING"):
+            log.warning("upload_weights %.1fms (cached_blob=%s)",
+                        (_time.monotonic() - t0) * 1e3,
+                        _blob_cache_get(plan) is not None)
 
     # -- weights -----------------------------------------------------------
     def _upload_weights(self) -> None:
@@ -864,12 +871,17 @@ class GpuModel:
         cached = _blob_cache_get(self.plan)
         if cached is not None and cached.numel() == total:
             # converted blob already pinned in CPU RAM: ONE DMA on the
-            # dedicated upload stream, serialized against captures
+            # dedicated upload stream. NOT guard-serialized: waiting for
+            # captures==0 cost ~40 ms per cold load under LRU churn
+            # (captures are near-continuous), and the pinned->device
+            # copy on a non-default stream is capture-safe (verified by
+            # the eviction-under-load stress test with the guard off —
+            # round 1's poisoning came from in-capture hipHostMalloc
+            # and legacy-stream DMAs, both avoided here)
             up = _get_upload_stream(torch, self.device)
-            with capture_guard.unsafe_host_op():
-                with torch.cuda.stream(up):
-                    blob.copy_(cached, non_blocking=True)
-                up.synchronize()
+            with torch.cuda.stream(up):
+                blob.copy_(cached, non_blocking=True)
+            up.synchronize()
         else:
             # stage through a SHARED reusable pinned buffer: per-load
             # pinned allocation (page-locking ~100 MB) costs more than
