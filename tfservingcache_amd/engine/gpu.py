@@ -106,6 +106,7 @@ class _CaptureGuard:
         self._cv = threading.Condition()
         self._captures = 0
         self._unsafe = 0
+        self._unsafe_waiting = 0
         self.enabled = _os.environ.get(
             "TFSC_CAPTURE_GUARD", "on").lower() not in ("off", "0", "no")
 
@@ -125,7 +126,11 @@ class _CaptureGuard:
         import time as _time
         t0 = _time.monotonic()
         with self._cv:
-            while self._unsafe:
+            # writer preference: a QUEUED unsafe op blocks new captures
+            # too, so an upload waits out at most the captures already
+            # in flight (a few ms) instead of a continuous stream of
+            # them (measured ~40 ms per cold load under LRU churn)
+            while self._unsafe or self._unsafe_waiting:
                 self._cv.wait()
             self._captures += 1
         self._waited(t0, "capture")
@@ -144,9 +149,13 @@ class _CaptureGuard:
         import time as _time
         t0 = _time.monotonic()
         with self._cv:
-            while self._captures:
-                self._cv.wait()
-            self._unsafe += 1
+            self._unsafe_waiting += 1
+            try:
+                while self._captures:
+                    self._cv.wait()
+                self._unsafe += 1
+            finally:
+                self._unsafe_waiting -= 1
         self._waited(t0, "unsafe-host-op")
         try:
             yield
@@ -871,17 +880,16 @@ class GpuModel:
         cached = _blob_cache_get(self.plan)
         if cached is not None and cached.numel() == total:
             # converted blob already pinned in CPU RAM: ONE DMA on the
-            # dedicated upload stream. NOT guard-serialized: waiting for
-            # captures==0 cost ~40 ms per cold load under LRU churn
-            # (captures are near-continuous), and the pinned->device
-            # copy on a non-default stream is capture-safe (verified by
-            # the eviction-under-load stress test with the guard off —
-            # round 1's poisoning came from in-capture hipHostMalloc
-            # and legacy-stream DMAs, both avoided here)
+            # dedicated upload stream, guard-serialized against captures
+            # (running it unguarded produced
+            # hipErrorStreamCaptureInvalidated under the full headline
+            # workload; the guard's writer preference keeps the wait to
+            # the captures already in flight, a few ms)
             up = _get_upload_stream(torch, self.device)
-            with torch.cuda.stream(up):
-                blob.copy_(cached, non_blocking=True)
-            up.synchronize()
+            with capture_guard.unsafe_host_op():
+                with torch.cuda.stream(up):
+                    blob.copy_(cached, non_blocking=True)
+                up.synchronize()
         else:
             # stage through a SHARED reusable pinned buffer: per-load
             # pinned allocation (page-locking ~100 MB) costs more than
