@@ -876,7 +876,12 @@ class GpuModel:
                 total += (n + 127) // 128 * 128   # keep 256B alignment
         if not float_ws:
             return
+        import os as _os
+        import time as _time
+        timing = _os.environ.get("TFSC_LOAD_TIMING")
+        ta = _time.monotonic()
         blob = torch.empty(total, dtype=torch.bfloat16, device=self.device)
+        tb = _time.monotonic()
         cached = _blob_cache_get(self.plan)
         if cached is not None and cached.numel() == total:
             # converted blob already pinned in CPU RAM: ONE DMA on the
@@ -886,10 +891,19 @@ class GpuModel:
             # workload; the guard's writer preference keeps the wait to
             # the captures already in flight, a few ms)
             up = _get_upload_stream(torch, self.device)
+            tc = _time.monotonic()
             with capture_guard.unsafe_host_op():
+                td = _time.monotonic()
                 with torch.cuda.stream(up):
                     blob.copy_(cached, non_blocking=True)
+                te = _time.monotonic()
                 up.synchronize()
+            if timing:
+                tf_ = _time.monotonic()
+                log.warning(
+                    "upload phases: alloc=%.1f guard=%.1f copy=%.1f "
+                    "sync=%.1f ms", (tb - ta) * 1e3, (td - tc) * 1e3,
+                    (te - td) * 1e3, (tf_ - te) * 1e3)
         else:
             # stage through a SHARED reusable pinned buffer: per-load
             # pinned allocation (page-locking ~100 MB) costs more than
