@@ -451,6 +451,30 @@ class Server:
                 if not owners or self_id in [o.serialize()
                                              for o in owners]:
                     continue
+                # byte push first: if the new owner is a replica-plane
+                # member (one-process-per-GPU deployment), send the
+                # model FILES over RCCL/xGMI — it then loads from its
+                # own disk cache without touching the model store
+                # (ROADMAP round-1 item 7's byte-push candidate)
+                pushed = False
+                if self._plane is not None:
+                    dsts = []
+                    for o in owners:
+                        r = self._plane.rank_of_member(o.serialize(),
+                                                       timeout_s=0.5)
+                        if r is not None and r != self._plane.rank:
+                            dsts.append(r)
+                    if dsts:
+                        vdir = os.path.join(
+                            self.cm.cache.base_dir, entry.name,
+                            str(entry.version))
+                        self._plane.push_files_async(
+                            entry.name, entry.version, vdir, dsts)
+                        pushed = True
+                        log.info("warm handoff (plane): %s:%s -> ranks "
+                                 "%s", entry.name, entry.version, dsts)
+                # nudge the owner to LOAD now (from the pushed bytes if
+                # the plane delivered, from the store otherwise)
                 req = wm.GetModelMetadataRequest(
                     model_spec=wm.ModelSpec(
                         name=entry.name,
@@ -460,8 +484,9 @@ class Server:
                     owners[0].grpc_addr,
                     f"/{wm.PREDICTION_SERVICE}/GetModelMetadata",
                     req.encode())
-                log.info("warm handoff: %s:%s -> %s", entry.name,
-                         entry.version, owners[0].grpc_addr)
+                log.info("warm handoff: %s:%s -> %s%s", entry.name,
+                         entry.version, owners[0].grpc_addr,
+                         " (bytes via plane)" if pushed else "")
             except Exception:       # noqa: BLE001
                 log.warning("warm handoff failed for %s:%s",
                             entry.name, entry.version, exc_info=True)
