@@ -337,7 +337,11 @@ def main() -> int:
                                  batch_timeout_s=args.batch_timeout_ms
                                  / 1e3,
                                  n_streams=args.streams,
-                                 dtype=args.dtype)
+                                 dtype=args.dtype,
+                                 prewarm_batch=(args.batch
+                                                if eff_mode in
+                                                ("lru", "ring")
+                                                else None))
     pool = ModelPool(loader, max_concurrent_models=pool_cap, device=device)
     cm = CacheManager(provider, cache, pool, model_fetch_timeout=300.0)
     handler = LocalServingHandler(cm)

@@ -188,7 +188,9 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
                     batch_timeout_s: float = 0.001,
                     devices: Optional[List[str]] = None,
                     n_streams: int = 6,
-                    dtype: str = "bf16"
+                    dtype: str = "bf16",
+                    prewarm_batch: Optional[int] = None,
+                    prewarm_ctxs: int = 1
                     ) -> Callable[[str, int], LoadedModel]:
     """Loader that compiles the SavedModel onto an MI355X: weights land
     in the GPU's HBM pool (bf16, GEMM layouts pre-transformed) and the
@@ -237,5 +239,11 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
         # TF Serving warmup files (assets.extra/tf_serving_warmup_requests):
         # contexts build + hipGraphs capture before AVAILABLE
         run_warmup(lm, vdir)
+        if prewarm_batch and not batching:
+            # build ONE fast context for the expected bucket on THIS
+            # (loader) thread: under LRU churn most requests are a
+            # model's first, and a prewarmed context lets them take the
+            # C++ fast path instead of the Python build path
+            lm._gpu.prewarm(prewarm_batch, prewarm_ctxs)
         return lm
     return loader

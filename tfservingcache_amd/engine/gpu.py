@@ -1178,14 +1178,18 @@ class GpuModel:
             ctx.batch, self.streams_for(ctx.batch), ctx.exec_plan.ptr(),
             ctx.stream.cuda_stream, ins, outs)
 
-    def prewarm(self, batch: int) -> None:
-        """Build + capture + fast-register ALL of the bucket's stream
-        contexts with zero feeds. Server-side batching needs this: the
-        merged bucket is only ever used by the C++ fast path, so the
-        Python path's contention-driven context building never reaches
-        it — every context must exist before merging starts."""
+    def prewarm(self, batch: int, limit: Optional[int] = None) -> None:
+        """Build + capture + fast-register the bucket's stream contexts
+        with zero feeds (all of them by default; `limit` caps the count
+        — LRU-churn loads prewarm ONE context on the loader thread so
+        even a model's FIRST request takes the C++ fast path).
+        Server-side batching needs the full set: the merged bucket is
+        only ever used by the C++ fast path, so the Python path's
+        contention-driven context building never reaches it."""
         plan = self.plan
         b = self._bucket(batch)
+        want = self.streams_for(b) if limit is None \
+            else min(limit, self.streams_for(b))
         feeds = {}
         for _alias, idx in plan.sig_inputs.items():
             shape = plan.resolve_shape(plan.tensors[idx].shape, b)
@@ -1198,7 +1202,7 @@ class GpuModel:
                 raise ModelReleasedError(
                     "model was evicted from the GPU pool")
             ctxs = self._contexts.setdefault(b, [])
-            while len(ctxs) < self.streams_for(b):
+            while len(ctxs) < want:
                 ctxs.append(ExecContext(self, b))
             todo = [c for c in ctxs if c.fast_id is None]
         for ctx in todo:
