@@ -248,6 +248,61 @@ class _BufInfo:
     is_int: bool = False
 
 
+# ---------------------------------------------------------------------------
+# plan-level context templates + transform arenas.
+#
+# Under the 1000-model LRU workload, models sharing a SavedModel's bytes
+# share a Plan (inode-keyed plan cache) — yet every cold load re-ran the
+# Python-heavy buffer planning, call emission and per-layer weight
+# transforms, and under GIL contention with the serving threads that
+# Python work WAS the cold-load cost. Both are plan-deterministic:
+#
+#  * the CONTEXT TEMPLATE caches shapes/roots/buffer offsets/scratch
+#    and the emitted kernel calls with every pointer classified as
+#    (region key, offset); a new context for any model sharing the plan
+#    instantiates by pure pointer relocation;
+#  * the TRANSFORM ARENA caches the transformed weight bytes
+#    (pre-transposed GEMM/conv layouts, fp8 quantized weights) in
+#    pinned host memory; a new model restores them with ONE DMA into a
+#    device arena and views at recorded offsets — no per-layer
+#    transform compute, and every transform lands at a deterministic
+#    arena offset (which is what makes call relocation possible).
+# ---------------------------------------------------------------------------
+_tmpl_lock = threading.Lock()
+_ctx_templates: Dict[tuple, object] = {}    # (id(plan),bucket,dtype) ->
+                                            # (plan, template|None)
+_transform_arenas: Dict[tuple, tuple] = {}  # (id(plan),dtype) ->
+                                            # (plan, entries, pinned)
+
+
+def _get_ctx_template(plan, bucket: int, dtype: str):
+    with _tmpl_lock:
+        ent = _ctx_templates.get((id(plan), bucket, dtype))
+        if ent is None or ent[0] is not plan:
+            return None
+        return ent[1]
+
+
+def _put_ctx_template(plan, bucket: int, dtype: str, tmpl) -> None:
+    with _tmpl_lock:
+        _ctx_templates[(id(plan), bucket, dtype)] = (plan, tmpl)
+
+
+def _get_transform_arena(plan, dtype: str):
+    with _tmpl_lock:
+        ent = _transform_arenas.get((id(plan), dtype))
+        if ent is None or ent[0] is not plan:
+            return None
+        return ent[1], ent[2]
+
+
+def _put_transform_arena(plan, dtype: str, entries, pinned) -> None:
+    with _tmpl_lock:
+        key = (id(plan), dtype)
+        if key not in _transform_arenas:
+            _transform_arenas[key] = (plan, entries, pinned)
+
+
 class ExecContext:
     """Shapes + workspace + calls for one batch bucket."""
 
@@ -258,22 +313,51 @@ class ExecContext:
         plan = gm.plan
         dev = gm.device
 
-        self.shapes: List[Tuple[int, ...]] = [
-            plan.resolve_shape(t.shape, batch) for t in plan.tensors]
-        self.root = [t.alias_of if t.alias_of is not None else t.idx
-                     for t in plan.tensors]
-        # resolve alias chains
-        for i, r in enumerate(self.root):
-            seen = 0
-            while plan.tensors[r].alias_of is not None and seen < 16:
-                r = plan.tensors[r].alias_of
-                seen += 1
-            self.root[i] = r
+        tmpl = _get_ctx_template(plan, batch, gm.dtype)
+        calls = None
+        if tmpl:
+            # relocation fast path: all the Python-heavy planning and
+            # emission was done once for this (plan, bucket); just
+            # allocate the workspace and rebase the pointers
+            self.shapes = tmpl["shapes"]
+            self.root = tmpl["root"]
+            self.bufs = tmpl["bufs"]
+            self.total_bytes = tmpl["total_bytes"]
+            self.scratch_off = tmpl["scratch_off"]
+            self.workspace = torch.empty(self.total_bytes,
+                                         dtype=torch.uint8, device=dev)
+            ws = self.workspace.data_ptr()
+            try:
+                bases = gm.region_bases()
+                calls = [(kind,
+                          [ws + d[1] if d[0] == 0 else
+                           (0 if d[0] == 2 else bases[d[1]] + d[2])
+                           for d in descs],
+                          ints, floats)
+                         for kind, descs, ints, floats in tmpl["calls"]]
+            except KeyError:
+                # this model is missing a region (e.g. the transform
+                # arena failed to publish) — fall through to full emit
+                calls = None
+        if calls is None:
+            self.shapes: List[Tuple[int, ...]] = [
+                plan.resolve_shape(t.shape, batch) for t in plan.tensors]
+            self.root = [t.alias_of if t.alias_of is not None else t.idx
+                         for t in plan.tensors]
+            # resolve alias chains
+            for i, r in enumerate(self.root):
+                seen = 0
+                while plan.tensors[r].alias_of is not None and seen < 16:
+                    r = plan.tensors[r].alias_of
+                    seen += 1
+                self.root[i] = r
 
-        self._plan_buffers()
-        self.workspace = torch.empty(self.total_bytes, dtype=torch.uint8,
-                                     device=dev)
-        calls = self._emit_calls()
+            self._plan_buffers()
+            self.workspace = torch.empty(self.total_bytes,
+                                         dtype=torch.uint8, device=dev)
+            calls = self._emit_calls()
+            if tmpl is None:            # not attempted yet (None = new)
+                self._register_template(calls)
         self.exec_plan = ext.ExecPlan(calls)
         self.captured = False
         self._views: Dict[int, object] = {}
@@ -287,6 +371,47 @@ class ExecContext:
         # pageable f32 copy + on-device convert per request
         self._pinned_in: Dict[int, object] = {}
         self._pinned_out: Dict[int, object] = {}
+
+    def _register_template(self, calls) -> None:
+        """Classify every emitted pointer into (workspace | weight
+        region | null) and cache the relocatable template for this
+        (plan, bucket, dtype). If any pointer can't be classified the
+        template is disabled for this key (correctness first)."""
+        gm = self.gm
+        ws_base = self.workspace.data_ptr()
+        ws_end = ws_base + self.total_bytes
+        regions = gm.region_list()       # [(base, size, key)] sorted
+        tcalls = []
+        try:
+            for kind, ptrs, ints, floats in calls:
+                descs = []
+                for p in ptrs:
+                    if p == 0:
+                        descs.append((2,))
+                    elif ws_base <= p < ws_end:
+                        descs.append((0, p - ws_base))
+                    else:
+                        hit = None
+                        for base, size, key in regions:
+                            if base <= p < base + size:
+                                hit = (1, key, p - base)
+                                break
+                        if hit is None:
+                            raise KeyError(f"unclassifiable ptr {p:#x}")
+                        descs.append(hit)
+                tcalls.append((kind, descs, list(ints), list(floats)))
+        except KeyError as e:
+            log.warning("context template disabled for bucket %d: %s",
+                        self.batch, e)
+            _put_ctx_template(gm.plan, self.batch, gm.dtype, False)
+            return
+        tmpl = {"shapes": self.shapes, "root": self.root,
+                "bufs": self.bufs, "total_bytes": self.total_bytes,
+                "scratch_off": self.scratch_off, "calls": tcalls}
+        _put_ctx_template(gm.plan, self.batch, gm.dtype, tmpl)
+        # the template's region keys must be resolvable by later models
+        # sharing the plan — snapshot the transform bytes now
+        gm.publish_transform_arena()
 
     # -- buffer planning ---------------------------------------------------
     def _buf_bytes(self, idx: int) -> int:
@@ -848,10 +973,11 @@ class GpuModel:
         t0 = _time.monotonic()
         with torch.cuda.device(device):
             self._upload_weights()
+            restored = self.try_restore_transforms()
         if _os.environ.get("TFSC_LOAD_TIMING"):
-            log.warning("upload_weights %.1fms (cached_blob=%s)",
-                        (_time.monotonic() - t0) * 1e3,
-                        _blob_cache_get(plan) is not None)
+            log.warning("upload_weights %.1fms (cached_blob=%s "
+                        "arena=%s)", (_time.monotonic() - t0) * 1e3,
+                        _blob_cache_get(plan) is not None, restored)
 
     # -- weights -----------------------------------------------------------
     def _upload_weights(self) -> None:
@@ -863,6 +989,7 @@ class GpuModel:
         torch, _ = _load_backend()
         float_ws = []
         total = 0
+        self._int_idx = []
         for t in self.plan.tensors:
             if t.kind != "weight" or t.weight is None:
                 continue
@@ -870,6 +997,7 @@ class GpuModel:
             if w.dtype in (np.int32, np.int64):
                 self._weights[t.idx] = torch.from_numpy(
                     np.ascontiguousarray(w.astype(np.int32))).to(self.device)
+                self._int_idx.append(t.idx)
             else:
                 n = int(w.size)
                 float_ws.append((t.idx, w, total, n))
@@ -1053,6 +1181,115 @@ class GpuModel:
         self._conv_weights[key] = wt
         return wt
 
+    # -- transform arena / relocation regions ------------------------------
+    def _region_items(self):
+        """[(key, tensor)] for every device weight region a context's
+        emitted calls can point into (master blob excluded — it has its
+        own entry)."""
+        items = []
+        for idx in getattr(self, "_int_idx", []):
+            items.append((("int", idx), self._weights[idx]))
+        for k, t in self._gemm_weights.items():
+            items.append((("gw",) + tuple(k) if isinstance(k, tuple)
+                          else ("gw", k), t))
+        for k, t in self._conv_weights.items():
+            kk = k if isinstance(k, tuple) else (k,)
+            items.append((("cw",) + kk, t))
+        for k, (q, sc) in self._gemm_weights_fp8.items():
+            items.append((("f8q",) + tuple(k), q))
+            items.append((("f8s",) + tuple(k), sc))
+        return items
+
+    def region_list(self):
+        """[(base, size, key)] for pointer classification."""
+        out = []
+        blob = getattr(self, "_weight_blob", None)
+        if blob is not None:
+            out.append((blob.data_ptr(),
+                        blob.numel() * blob.element_size(), ("blob",)))
+        for key, t in self._region_items():
+            out.append((t.data_ptr(), t.numel() * t.element_size(), key))
+        out.append((self.zeros_ptr(), 64, ("zeros",)))
+        return out
+
+    def region_bases(self):
+        """key -> device base pointer for template instantiation."""
+        bases = {}
+        blob = getattr(self, "_weight_blob", None)
+        if blob is not None:
+            bases[("blob",)] = blob.data_ptr()
+        for key, t in self._region_items():
+            bases[key] = t.data_ptr()
+        bases[("zeros",)] = self.zeros_ptr()
+        return bases
+
+    def publish_transform_arena(self) -> None:
+        """Snapshot this model's transformed weights into a pinned host
+        arena keyed by the (content-deduplicated) plan, so later models
+        restore them with one DMA instead of per-layer transforms."""
+        torch, _ = _load_backend()
+        if _get_transform_arena(self.plan, self.dtype) is not None:
+            return
+        items = self._region_items()
+        entries = []
+        off = 0
+        for key, t in items:
+            nb = t.numel() * t.element_size()
+            entries.append((key, off, nb, t.dtype, tuple(t.shape)))
+            off = _pad256b(off + nb)
+        if off == 0:
+            return
+        try:
+            with capture_guard.unsafe_host_op():
+                pinned = torch.empty(off, dtype=torch.uint8,
+                                     pin_memory=True)
+        except RuntimeError:
+            log.warning("pinned transform-arena alloc failed; later "
+                        "models will re-transform")
+            return
+        for (key, o, nb, dt, shape), (_k, t) in zip(entries, items):
+            pinned[o:o + nb].view(dt)[:t.numel()].copy_(
+                t.reshape(-1), non_blocking=False)
+        _put_transform_arena(self.plan, self.dtype, entries, pinned)
+
+    def try_restore_transforms(self) -> bool:
+        """One DMA restore of the transformed weights from the plan's
+        pinned arena (replaces the per-layer GPU transforms on repeat
+        cold loads of the same content)."""
+        torch, _ = _load_backend()
+        got = _get_transform_arena(self.plan, self.dtype)
+        if got is None:
+            return False
+        entries, pinned = got
+        dev_arena = torch.empty(pinned.numel(), dtype=torch.uint8,
+                                device=self.device)
+        up = _get_upload_stream(torch, self.device)
+        with capture_guard.unsafe_host_op():
+            with torch.cuda.stream(up):
+                dev_arena.copy_(pinned, non_blocking=True)
+            up.synchronize()
+        self._arena_dev = dev_arena
+        for key, off, nb, dt, shape in entries:
+            view = dev_arena[off:off + nb].view(dt).view(shape)
+            kind = key[0]
+            if kind == "int":
+                self._weights[key[1]] = view
+            elif kind == "gw":
+                self._gemm_weights[tuple(key[1:])] = view
+            elif kind == "cw":
+                k = key[1:]
+                self._conv_weights[k[0] if len(k) == 1 else tuple(k)] = \
+                    view
+            elif kind == "f8q":
+                k = tuple(key[1:])
+                q, sc = self._gemm_weights_fp8.get(k, (None, None))
+                self._gemm_weights_fp8[k] = (view, sc)
+            elif kind == "f8s":
+                k = tuple(key[1:])
+                q, sc = self._gemm_weights_fp8.get(k, (None, None))
+                self._gemm_weights_fp8[k] = (q, view)
+        return True
+
     def weight_bytes(self) -> int:
         total = 0
         for d in (self._weights, self._gemm_weights, self._conv_weights):
@@ -1232,4 +1469,5 @@ class GpuModel:
             self._gemm_weights.clear()
             self._gemm_weights_fp8.clear()
             self._conv_weights.clear()
+            self._arena_dev = None
             self._weight_blob = None
