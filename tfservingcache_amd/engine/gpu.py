@@ -971,9 +971,14 @@ class GpuModel:
         import os as _os
         import time as _time
         t0 = _time.monotonic()
+        self._upload_pending = False
         with torch.cuda.device(device):
             self._upload_weights()
             restored = self.try_restore_transforms()
+            if self._upload_pending or restored:
+                # one sync covers both async DMAs (master blob +
+                # transform arena) on the shared upload stream
+                _get_upload_stream(torch, device).synchronize()
         if _os.environ.get("TFSC_LOAD_TIMING"):
             log.warning("upload_weights %.1fms (cached_blob=%s "
                         "arena=%s)", (_time.monotonic() - t0) * 1e3,
@@ -1267,7 +1272,6 @@ class GpuModel:
         with capture_guard.unsafe_host_op():
             with torch.cuda.stream(up):
                 dev_arena.copy_(pinned, non_blocking=True)
-            up.synchronize()
         self._arena_dev = dev_arena
         for key, off, nb, dt, shape in entries:
             view = dev_arena[off:off + nb].view(dt).view(shape)
