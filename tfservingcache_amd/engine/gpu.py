@@ -201,6 +201,56 @@ def _blob_cache_put(plan, blob) -> None:
             _blob_cache.pop(old, None)
 
 
+# ---------------------------------------------------------------------------
+# pinned host-buffer pool. Every cold load used to hipHostMalloc (and
+# every eviction hipHostFree) ~10 MB of per-context staging buffers —
+# milliseconds per call, and host allocs during another thread's
+# capture are exactly the unsafe ops the capture guard serializes.
+# Recycling them removes both costs from the LRU churn path.
+# ---------------------------------------------------------------------------
+class _PinnedPool:
+    KLASS = 1 << 20                 # size classes of 1 MiB
+
+    def __init__(self, cap_bytes: int = 2 << 30):
+        self._lock = threading.Lock()
+        self._free: Dict[int, List[object]] = {}
+        self._pooled = 0
+        self.cap = cap_bytes
+
+    def _klass(self, nbytes: int) -> int:
+        return max(1, (nbytes + self.KLASS - 1) // self.KLASS) * self.KLASS
+
+    def alloc(self, torch, nbytes: int):
+        """Raw pinned uint8 tensor of at least nbytes (class-rounded)."""
+        k = self._klass(nbytes)
+        with self._lock:
+            lst = self._free.get(k)
+            if lst:
+                self._pooled -= k
+                return lst.pop()
+        with capture_guard.unsafe_host_op():
+            return torch.empty(k, dtype=torch.uint8, pin_memory=True)
+
+    def free(self, raw) -> None:
+        if raw is None:
+            return
+        k = raw.numel()
+        with self._lock:
+            if self._pooled + k > self.cap:
+                return              # drop (freed by GC outside the pool)
+            self._free.setdefault(k, []).append(raw)
+            self._pooled += k
+
+
+pinned_pool = _PinnedPool()
+
+
+def _pin_view(torch, raw, shape, dtype):
+    n = int(np.prod(shape)) if shape else 1
+    esz = torch.tensor([], dtype=dtype).element_size()
+    return raw[:n * esz].view(dtype).view(shape)
+
+
 # dedicated per-device upload streams: the cached-blob DMA must not run
 # on the default (legacy) stream, whose implicit cross-stream semantics
 # interact with capturing streams
@@ -371,6 +421,7 @@ class ExecContext:
         # pageable f32 copy + on-device convert per request
         self._pinned_in: Dict[int, object] = {}
         self._pinned_out: Dict[int, object] = {}
+        self._pin_raws: List[object] = []   # pooled pinned backings
 
     def _register_template(self, calls) -> None:
         """Classify every emitted pointer into (workspace | weight
@@ -882,8 +933,10 @@ class ExecContext:
                 v = self.view(idx)
                 pin = self._pinned_in.get(idx)
                 if pin is None or pin.shape != v.shape:
-                    pin = torch.empty(v.shape, dtype=v.dtype,
-                                      pin_memory=True)
+                    raw = pinned_pool.alloc(
+                        torch, int(np.prod(v.shape)) * v.element_size())
+                    self._pin_raws.append(raw)
+                    pin = _pin_view(torch, raw, tuple(v.shape), v.dtype)
                     self._pinned_in[idx] = pin
                 if hasattr(arr, "segments"):
                     # segmented batch: copy each request's rows straight
@@ -924,9 +977,13 @@ class ExecContext:
                 v = self.view(idx)
                 po = self._pinned_out.get(idx)
                 if po is None or po.shape != v.shape:
-                    po = torch.empty(v.shape, dtype=torch.float32
-                                     if v.dtype == torch.bfloat16
-                                     else v.dtype, pin_memory=True)
+                    dt = (torch.float32 if v.dtype == torch.bfloat16
+                          else v.dtype)
+                    raw = pinned_pool.alloc(
+                        torch, int(np.prod(v.shape)) *
+                        torch.tensor([], dtype=dt).element_size())
+                    self._pin_raws.append(raw)
+                    po = _pin_view(torch, raw, tuple(v.shape), dt)
                     self._pinned_out[idx] = po
                 po.copy_(v, non_blocking=True)
             self.stream.synchronize()
@@ -1384,9 +1441,10 @@ class GpuModel:
             is_int = v.dtype == torch.int32
             rows = v.shape[0] if v.ndim else 1
             row_elems = int(np.prod(v.shape[1:])) if v.ndim > 1 else 1
-            pin = torch.empty((rows, row_elems),
-                              dtype=torch.int32 if is_int
-                              else torch.float32, pin_memory=True)
+            raw = pinned_pool.alloc(torch, rows * row_elems * 4)
+            ctx._pin_raws.append(raw)
+            pin = _pin_view(torch, raw, (rows, row_elems),
+                            torch.int32 if is_int else torch.float32)
             if is_int:
                 stage_ptr = 0
             else:
@@ -1406,8 +1464,9 @@ class GpuModel:
                 return
             rows = v.shape[0] if v.ndim else 1
             row_elems = int(np.prod(v.shape[1:])) if v.ndim > 1 else 1
-            pin = torch.empty((rows, row_elems), dtype=torch.float32,
-                              pin_memory=True)
+            raw = pinned_pool.alloc(torch, rows * row_elems * 4)
+            ctx._pin_raws.append(raw)
+            pin = _pin_view(torch, raw, (rows, row_elems), torch.float32)
             stage = torch.empty((rows, row_elems), dtype=torch.float32,
                                 device=self.device)
             keep.extend([pin, stage])
@@ -1468,6 +1527,12 @@ class GpuModel:
             for ctx in ctxs:
                 ctx.lock.acquire()
                 ctx.lock.release()
+                ctx._pinned_in.clear()
+                ctx._pinned_out.clear()
+                ctx._fast_keep = None
+                for raw in ctx._pin_raws:
+                    pinned_pool.free(raw)
+                ctx._pin_raws = []
         with self._lock:
             self._weights.clear()
             self._gemm_weights.clear()
