@@ -22,9 +22,17 @@ import threading
 from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
+import warnings
+
 import numpy as np
 
 from .planner import Plan, PlanOp, is_sym, resolve_dim
+
+# weight arrays from memory-mapped SavedModels are read-only; we only
+# ever READ the torch views of them (H2D staging), so the non-writable
+# warning is noise
+warnings.filterwarnings(
+    "ignore", message="The given NumPy array is not writable")
 
 log = logging.getLogger("tfsc.gpu")
 
