@@ -1042,8 +1042,13 @@ class GpuModel:
             restored = self.try_restore_transforms()
             if self._upload_pending or restored:
                 # one sync covers both async DMAs (master blob +
-                # transform arena) on the shared upload stream
-                _get_upload_stream(torch, device).synchronize()
+                # transform arena) on the shared upload stream. MUST be
+                # guard-serialized: a stream sync concurrent with
+                # another thread's capture invalidates the capture (the
+                # 100-step soak reproduced hipErrorStreamCapture*
+                # within seconds when this ran unguarded)
+                with capture_guard.unsafe_host_op():
+                    _get_upload_stream(torch, device).synchronize()
         if _os.environ.get("TFSC_LOAD_TIMING"):
             log.warning("upload_weights %.1fms (cached_blob=%s "
                         "arena=%s)", (_time.monotonic() - t0) * 1e3,
