@@ -326,39 +326,61 @@ class _BufInfo:
 #    transform compute, and every transform lands at a deterministic
 #    arena offset (which is what makes call relocation possible).
 # ---------------------------------------------------------------------------
+_TMPL_CAP = 64          # templates are ~100 KB of host data each
+_ARENA_CAP = 4          # arenas pin ~50 MB of host RAM each (like the
+                        # blob cache: only the hottest plan contents)
 _tmpl_lock = threading.Lock()
 _ctx_templates: Dict[tuple, object] = {}    # (id(plan),bucket,dtype) ->
                                             # (plan, template|None)
+_tmpl_order: List[tuple] = []
 _transform_arenas: Dict[tuple, tuple] = {}  # (id(plan),dtype) ->
                                             # (plan, entries, pinned)
+_arena_order: List[tuple] = []
 
 
 def _get_ctx_template(plan, bucket: int, dtype: str):
     with _tmpl_lock:
-        ent = _ctx_templates.get((id(plan), bucket, dtype))
+        key = (id(plan), bucket, dtype)
+        ent = _ctx_templates.get(key)
         if ent is None or ent[0] is not plan:
             return None
+        _tmpl_order.remove(key)
+        _tmpl_order.append(key)
         return ent[1]
 
 
 def _put_ctx_template(plan, bucket: int, dtype: str, tmpl) -> None:
     with _tmpl_lock:
-        _ctx_templates[(id(plan), bucket, dtype)] = (plan, tmpl)
+        key = (id(plan), bucket, dtype)
+        if key not in _ctx_templates:
+            _tmpl_order.append(key)
+        _ctx_templates[key] = (plan, tmpl)
+        while len(_tmpl_order) > _TMPL_CAP:
+            old = _tmpl_order.pop(0)
+            _ctx_templates.pop(old, None)
 
 
 def _get_transform_arena(plan, dtype: str):
     with _tmpl_lock:
-        ent = _transform_arenas.get((id(plan), dtype))
+        key = (id(plan), dtype)
+        ent = _transform_arenas.get(key)
         if ent is None or ent[0] is not plan:
             return None
+        _arena_order.remove(key)
+        _arena_order.append(key)
         return ent[1], ent[2]
 
 
 def _put_transform_arena(plan, dtype: str, entries, pinned) -> None:
     with _tmpl_lock:
         key = (id(plan), dtype)
-        if key not in _transform_arenas:
-            _transform_arenas[key] = (plan, entries, pinned)
+        if key in _transform_arenas:
+            return
+        _arena_order.append(key)
+        _transform_arenas[key] = (plan, entries, pinned)
+        while len(_arena_order) > _ARENA_CAP:
+            old = _arena_order.pop(0)
+            _transform_arenas.pop(old, None)
 
 
 class ExecContext:
