@@ -489,7 +489,7 @@ class RestFrontend {
     if (accept_thread_.joinable()) accept_thread_.join();
     std::lock_guard<std::mutex> g(threads_mu_);
     for (auto& t : conn_threads_)
-      if (t.joinable()) t.join();
+      if (t.second.joinable()) t.second.join();
     conn_threads_.clear();
   }
 
@@ -529,11 +529,25 @@ class RestFrontend {
         conn_fds_.insert(fd);
       }
       std::lock_guard<std::mutex> g(threads_mu_);
-      conn_threads_.emplace_back([this, fd] {
+      // reap finished connection threads so a long-lived server does
+      // not accumulate unjoined stacks under connection churn
+      for (auto it = conn_threads_.begin(); it != conn_threads_.end();) {
+        if (it->first->load() && it->second.joinable()) {
+          it->second.join();
+          it = conn_threads_.erase(it);
+        } else {
+          ++it;
+        }
+      }
+      auto done = std::make_shared<std::atomic<bool>>(false);
+      conn_threads_.emplace_back(done, std::thread([this, fd, done] {
         conn_loop(fd);
-        std::lock_guard<std::mutex> g2(fds_mu_);
-        conn_fds_.erase(fd);
-      });
+        {
+          std::lock_guard<std::mutex> g2(fds_mu_);
+          conn_fds_.erase(fd);
+        }
+        done->store(true);
+      }));
     }
   }
 
@@ -961,7 +975,8 @@ class RestFrontend {
   int bound_port_ = 0;
   std::thread accept_thread_;
   std::mutex threads_mu_;
-  std::vector<std::thread> conn_threads_;
+  std::vector<std::pair<std::shared_ptr<std::atomic<bool>>, std::thread>>
+      conn_threads_;
   std::mutex fds_mu_;
   std::set<int> conn_fds_;
   std::shared_mutex reg_mu_;
