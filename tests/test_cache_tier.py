@@ -359,13 +359,16 @@ def test_warmup_requests_executed(tmp_path):
 
 def test_plan_cache_shares_hardlinked_models(tmp_path):
     """Hardlinked copies of one SavedModel compile once (inode-keyed
-    plan cache); a rewritten file gets a fresh plan."""
+    plan cache); byte-identical fresh copies (distinct inodes, e.g.
+    S3 downloads) ALSO share via the content-hash fallback key; a
+    model with different bytes gets its own plan."""
     import os
     import shutil
     from tfservingcache_amd.engine.model import load_model_from_dir
     from tfservingcache_amd.models import write_model_repo
 
-    write_model_repo(str(tmp_path), [("a", 1, "half_plus_two")])
+    write_model_repo(str(tmp_path), [("a", 1, "half_plus_two"),
+                                     ("d", 1, "mlp")])
     src = tmp_path / "a" / "1"
     dst = tmp_path / "b" / "1"
     os.makedirs(dst.parent, exist_ok=True)
@@ -375,13 +378,17 @@ def test_plan_cache_shares_hardlinked_models(tmp_path):
     lm_b = load_model_from_dir(str(dst), "b", 1)
     assert lm_a.plan is lm_b.plan       # same inode -> shared plan
 
-    # distinct content (fresh copy, new inode) -> fresh compile
+    # identical content, fresh inode -> shared via the content key
     dst2 = tmp_path / "c" / "1"
     os.makedirs(dst2.parent, exist_ok=True)
     shutil.copytree(src, dst2)
     lm_c = load_model_from_dir(str(dst2), "c", 1)
-    assert lm_c.plan is not lm_a.plan
-    # both still serve correctly
+    assert lm_c.plan is lm_a.plan
+
+    # genuinely different bytes -> different plan
+    lm_d = load_model_from_dir(str(tmp_path / "d" / "1"), "d", 1)
+    assert lm_d.plan is not lm_a.plan
+
     import numpy as np
     out_a = lm_a.predict({"x": np.array([2.0], np.float32)})
     out_c = lm_c.predict({"x": np.array([2.0], np.float32)})

@@ -170,7 +170,10 @@ class LoadedModel:
 # same graph) should not recompile the identical SavedModel per name.
 # Keyed by (dev, inode, size, mtime_ns) of saved_model.pb — with the
 # disk provider's hardlink fetch a cache copy shares the repo's inode,
-# so hits skip reading the file entirely. Plans are immutable after
+# so hits skip reading the file entirely. When the inode key misses
+# (e.g. S3/AzBlob downloads produce fresh inodes for byte-identical
+# content), a CONTENT key (xxh3 of the file bytes, ~10 GB/s) gives the
+# same dedup at the cost of one hash pass. Plans are immutable after
 # compile (executors only read them), so sharing is safe.
 _PLAN_CACHE_CAP = 32
 _plan_cache_lock = threading.Lock()
@@ -197,6 +200,32 @@ def _compile_cached(version_dir: str, signature_name: str) -> Plan:
             if plan is not None:
                 _plan_cache.move_to_end(key)
                 return plan
+    # content-hash fallback key (distinct inodes, identical bytes)
+    ckey = None
+    try:
+        import xxhash
+        h = xxhash.xxh3_64()
+        with open(os.path.join(version_dir, SAVED_MODEL_FILENAME),
+                  "rb") as f:
+            for chunk in iter(lambda: f.read(1 << 22), b""):
+                h.update(chunk)
+        if os.path.exists(var_prefix + ".index"):
+            for suffix in (".index", ".data-00000-of-00001"):
+                fp = var_prefix + suffix
+                if os.path.exists(fp):
+                    with open(fp, "rb") as f:
+                        for chunk in iter(lambda: f.read(1 << 22), b""):
+                            h.update(chunk)
+        ckey = ("content", h.intdigest(), signature_name)
+        with _plan_cache_lock:
+            plan = _plan_cache.get(ckey)
+            if plan is not None:
+                _plan_cache.move_to_end(ckey)
+                if key is not None:
+                    _plan_cache[key] = plan      # promote the inode key
+                return plan
+    except Exception:       # noqa: BLE001 — hashing is best-effort
+        ckey = None
     graph_def, signatures = read_saved_model(version_dir)
     sig = signatures.get(signature_name)
     if sig is None and signatures:
@@ -208,11 +237,13 @@ def _compile_cached(version_dir: str, signature_name: str) -> Plan:
         from .tensor_bundle import read_bundle
         variables = read_bundle(var_prefix)
     plan = compile_graph(graph_def, sig, variables)
-    if key is not None:
-        with _plan_cache_lock:
+    with _plan_cache_lock:
+        if key is not None:
             _plan_cache[key] = plan
-            while len(_plan_cache) > _PLAN_CACHE_CAP:
-                _plan_cache.popitem(last=False)
+        if ckey is not None:
+            _plan_cache[ckey] = plan
+        while len(_plan_cache) > _PLAN_CACHE_CAP:
+            _plan_cache.popitem(last=False)
     return plan
 
 
