@@ -209,6 +209,80 @@ def build_repo(base: str, n_models: int, image_size: int,
     return names
 
 
+def _start_s3_mock(repo: str):
+    """In-process S3-compatible store over the generated repo: ListV2
+    + streamed GETs served from disk. Cold loads then pay a real
+    per-object HTTP download (the s3Provider regime of BASELINE
+    configs[4]) instead of the disk provider's hardlink fetch."""
+    import http.server
+    import urllib.parse
+    from tfservingcache_amd.cachemanager.providers.s3 import \
+        S3ModelProvider
+
+    index = {}
+    for root, _dirs, files in os.walk(repo):
+        for f in files:
+            full = os.path.join(root, f)
+            key = os.path.relpath(full, repo).replace(os.sep, "/")
+            index[key] = full
+    keys_sorted = sorted(index)
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        protocol_version = "HTTP/1.1"
+
+        def log_message(self, *a):      # noqa: N802
+            pass
+
+        def do_GET(self):               # noqa: N802
+            parsed = urllib.parse.urlparse(self.path)
+            qs = dict(urllib.parse.parse_qsl(parsed.query))
+            if qs.get("list-type") == "2":
+                prefix = qs.get("prefix", "")
+                body = ['<?xml version="1.0"?><ListBucketResult>']
+                import bisect
+                i = bisect.bisect_left(keys_sorted, prefix)
+                while i < len(keys_sorted) and \
+                        keys_sorted[i].startswith(prefix):
+                    k = keys_sorted[i]
+                    body.append(
+                        f"<Contents><Key>{k}</Key>"
+                        f"<Size>{os.path.getsize(index[k])}</Size>"
+                        f"</Contents>")
+                    i += 1
+                body.append("<IsTruncated>false</IsTruncated>"
+                            "</ListBucketResult>")
+                data = "".join(body).encode()
+                self.send_response(200)
+                self.send_header("Content-Length", str(len(data)))
+                self.end_headers()
+                self.wfile.write(data)
+                return
+            path = urllib.parse.unquote(parsed.path).lstrip("/")
+            _bucket, _, key = path.partition("/")
+            full = index.get(key)
+            if full is None:
+                self.send_response(404)
+                self.send_header("Content-Length", "0")
+                self.end_headers()
+                return
+            size = os.path.getsize(full)
+            self.send_response(200)
+            self.send_header("Content-Length", str(size))
+            self.end_headers()
+            with open(full, "rb") as f:
+                while True:
+                    chunk = f.read(1 << 20)
+                    if not chunk:
+                        break
+                    self.wfile.write(chunk)
+
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), Handler)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    return S3ModelProvider(
+        bucket="models", base_path="",
+        endpoint_url=f"http://127.0.0.1:{srv.server_address[1]}")
+
+
 def _graph_status(pool):
     try:
         for e in pool._entries.values():      # noqa: SLF001
@@ -254,6 +328,12 @@ def main() -> int:
                     help="enable server-side dynamic batching")
     ap.add_argument("--batch-timeout-ms", type=float, default=1.0,
                     help="dynamic-batching merge window")
+    ap.add_argument("--provider", choices=["disk", "s3-mock"],
+                    default="disk",
+                    help="model store: local disk (hardlink fetch) or "
+                         "an in-process S3-compatible HTTP server over "
+                         "the same repo (real downloads per cold load "
+                         "- BASELINE configs[4] s3Provider)")
     ap.add_argument("--dtype", choices=["bf16", "fp8"], default="bf16",
                     help="engine compute dtype (fp8: e4m3 GEMMs with "
                          "rowwise dequant; conv stays bf16)")
@@ -324,7 +404,10 @@ def main() -> int:
     names = [n for n, _k in named]
     kind_of = dict(named)
 
-    provider = DiskModelProvider(repo)
+    if args.provider == "s3-mock":
+        provider = _start_s3_mock(repo)
+    else:
+        provider = DiskModelProvider(repo)
     cache = LRUCache(cache_dir, max_size_bytes=200 * 10 ** 9)
     pool_cap = args.pool_size if eff_mode in ("lru", "ring") else 4
     if args.cpu:
@@ -743,8 +826,14 @@ def main() -> int:
                 "threads": args.threads,
                 "images_per_sec": round(req_per_sec * args.batch, 1),
                 "n_models": n_models,
-                "model_repo": "hardlinked copies of one SavedModel "
-                              "per family (content-dedup applies)",
+                "model_repo": ("hardlinked copies of one SavedModel "
+                               "per family (content-dedup applies)"
+                               if args.provider == "disk" else
+                               "in-process S3-compatible store over "
+                               "the same repo (per-object HTTP "
+                               "downloads per cold load; content-dedup "
+                               "via xxh3 plan keys)"),
+                "provider": args.provider,
                 "pool_size": pool_cap,
                 "cold_load_p50_ms": round(cold_p50, 1),
                 "latency_ms": {
