@@ -56,7 +56,8 @@ from tfservingcache_amd.cachemanager import (CacheManager, LRUCache,  # noqa: E4
                                              make_gpu_loader)
 from tfservingcache_amd.cachemanager.providers import DiskModelProvider  # noqa: E402
 from tfservingcache_amd.models.builders import (build_resnet50,  # noqa: E402
-                                                build_bert)
+                                                build_bert,
+                                                build_mobilenet_v2)
 from tfservingcache_amd.engine.savedmodel import write_saved_model  # noqa: E402
 from tfservingcache_amd.tfservingproxy import LocalServingHandler  # noqa: E402
 from tfservingcache_amd.wire import messages as m  # noqa: E402
@@ -196,6 +197,9 @@ def build_repo(base: str, n_models: int, image_size: int,
         proto_dir = os.path.join(base, f"_proto_{kind}", "1")
         if kind == "resnet50":
             sm = build_resnet50(image_size=image_size, num_classes=1000)
+        elif kind == "mobilenet_v2":
+            sm = build_mobilenet_v2(image_size=image_size,
+                                    num_classes=1000)
         else:
             sm = build_bert(seq_len=seq_len)
         write_saved_model(sm, proto_dir)
@@ -304,7 +308,9 @@ def main() -> int:
     ap.add_argument("--mode",
                     choices=["headline", "warm", "lru", "ring"],
                     default="headline")
-    ap.add_argument("--model", choices=["resnet50", "bert_base", "mixed"],
+    ap.add_argument("--model",
+                    choices=["resnet50", "mobilenet_v2", "bert_base",
+                             "mixed"],
                     default="resnet50")
     ap.add_argument("--replicas", type=int, default=2,
                     help="replicasPerModel (ring mode)")
@@ -807,6 +813,7 @@ def main() -> int:
             "data": "synthetic",
             "config": {
                 "model": {"resnet50": "resnet50_v1.5",
+                          "mobilenet_v2": "mobilenet_v2",
                           "bert_base": "bert_base",
                           "mixed": "bert_base+resnet50_v1.5"}[args.model],
                 "global_batch": args.batch * world,
