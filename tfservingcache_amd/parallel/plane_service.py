@@ -180,6 +180,9 @@ class PlaneService:
                 off += int(size)
             for d in dsts:
                 dist.send(buf, dst=d)
+            from ..utils import metrics as mt
+            mt.plane_transfers.labels("send").inc(len(dsts))
+            mt.plane_bytes.labels("send").inc(total * len(dsts))
         else:
             dist.recv(buf, src=src)
             host = buf.cpu().numpy().tobytes()
@@ -195,6 +198,9 @@ class PlaneService:
             log.info("plane: received %s:%s (%d files, %.1f MB) from "
                      "rank %d", name, version, len(entries),
                      total / 1e6, src)
+            from ..utils import metrics as mt
+            mt.plane_transfers.labels("recv").inc()
+            mt.plane_bytes.labels("recv").inc(total)
             if self.on_receive is not None:
                 try:
                     self.on_receive(name, version, vdir, total)

@@ -55,6 +55,12 @@ engine_pool_bytes = Gauge(
     "tfservingcache_engine_pool_bytes",
     "Bytes of model weights resident in the pool", ["device"],
     registry=REGISTRY)
+plane_transfers = Counter(
+    "tfservingcache_replica_plane_transfers_total",
+    "Model byte-pushes over the RCCL/xGMI replica plane", ["role"])
+plane_bytes = Counter(
+    "tfservingcache_replica_plane_bytes_total",
+    "Bytes moved over the RCCL/xGMI replica plane", ["role"])
 engine_pool_models = Gauge(
     "tfservingcache_engine_pool_models",
     "Number of models resident in the pool", ["device"], registry=REGISTRY)
