@@ -174,7 +174,26 @@ def _export_hf_bert(hf, seq_len):
     h3out = gb.node("Reshape", "seq_out3",
                     [h, gb.const("to3o", np.array([-1, seq_len, hidden],
                                                   np.int32))])
+    # HF pooler: dense(tanh) on the CLS token — extracted with the
+    # StridedSlice shape real exports emit (x[:, 0, :])
+    cls = gb.node("StridedSlice", "cls",
+                  [h3out,
+                   gb.const("ssb", np.array([0, 0, 0], np.int32)),
+                   gb.const("sse", np.array([0, 1, 0], np.int32)),
+                   gb.const("sss", np.array([1, 1, 1], np.int32))],
+                  T=f32, Index=i32,
+                  begin_mask=gb.a_int(0b101), end_mask=gb.a_int(0b101),
+                  shrink_axis_mask=gb.a_int(0b010),
+                  ellipsis_mask=gb.a_int(0), new_axis_mask=gb.a_int(0))
+    pw = sd["pooler.dense.weight"].T.astype(np.float32)
+    pb = sd["pooler.dense.bias"].astype(np.float32)
+    pmm = gb.node("MatMul", "pooler/mm",
+                  [cls, gb.const("pooler/w", pw)], T=f32)
+    pba = gb.node("BiasAdd", "pooler/ba",
+                  [pmm, gb.const("pooler/b", pb)], T=f32)
+    pooled = gb.node("Tanh", "pooled", [pba], T=f32)
     gb.mark_output("sequence_output", h3out)
+    gb.mark_output("pooled_output", pooled)
     return gb.build()
 
 
@@ -194,11 +213,16 @@ def test_bert_vs_transformers(tmp_path):
     rng = np.random.default_rng(1)
     ids = rng.integers(0, 97, (3, seq_len)).astype(np.int32)
     with torch.no_grad():
-        want = hf(input_ids=torch.from_numpy(ids.astype(np.int64))
-                  ).last_hidden_state.numpy()
-    got = model.predict({"input_ids": ids})["sequence_output"]
+        hf_out = hf(input_ids=torch.from_numpy(ids.astype(np.int64)))
+        want = hf_out.last_hidden_state.numpy()
+        want_pooled = hf_out.pooler_output.numpy()
+    out = model.predict({"input_ids": ids})
+    got = out["sequence_output"]
     assert got.shape == want.shape
     np.testing.assert_allclose(got, want, rtol=1e-3, atol=2e-4)
+    # CLS pooler via StridedSlice vs HF pooler_output
+    np.testing.assert_allclose(out["pooled_output"], want_pooled,
+                               rtol=1e-3, atol=2e-4)
 
 
 # ---------------------------------------------------------------------------

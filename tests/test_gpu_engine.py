@@ -382,3 +382,49 @@ def test_fp8_accuracy_delta_bert(tmp_path):
     # pooled head too
     pa, pb = outs["bf16"]["pooled_output"], outs["fp8"]["pooled_output"]
     np.testing.assert_allclose(pa, pb, atol=0.08)
+
+
+def test_strided_slice_cast_argmax_gpu(tmp_path):
+    """StridedSlice (CLS extraction shape) + Cast + ArgMax on GPU vs
+    the CPU fp32 reference."""
+    from tfservingcache_amd.engine.savedmodel import GraphBuilder
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    i32 = gb.a_type(3)
+    x_ph = gb.placeholder("x", np.float32, [-1, 9, 64],
+                          signature_name="x")
+    cls = gb.node("StridedSlice", "cls",
+                  [x_ph,
+                   gb.const("b", np.array([0, 0, 0], np.int32)),
+                   gb.const("e", np.array([0, 1, 0], np.int32)),
+                   gb.const("s", np.array([1, 1, 1], np.int32))],
+                  T=f32, Index=i32,
+                  begin_mask=gb.a_int(0b101), end_mask=gb.a_int(0b101),
+                  shrink_axis_mask=gb.a_int(0b010),
+                  ellipsis_mask=gb.a_int(0), new_axis_mask=gb.a_int(0))
+    win = gb.node("StridedSlice", "win",
+                  [x_ph,
+                   gb.const("b2", np.array([0, 1, 2], np.int32)),
+                   gb.const("e2", np.array([0, 8, 62], np.int32)),
+                   gb.const("s2", np.array([1, 2, 3], np.int32))],
+                  T=f32, Index=i32,
+                  begin_mask=gb.a_int(0b001), end_mask=gb.a_int(0b001),
+                  shrink_axis_mask=gb.a_int(0),
+                  ellipsis_mask=gb.a_int(0), new_axis_mask=gb.a_int(0))
+    am = gb.node("ArgMax", "am", [cls, gb.const("ax", np.int32(-1))],
+                 T=f32, output_type=i32)
+    amf = gb.node("Cast", "amf", [am], SrcT=i32, DstT=f32)
+    gb.mark_output("cls", cls)
+    gb.mark_output("win", win)
+    gb.mark_output("idx", am)
+    gb.mark_output("idxf", amf)
+    sm = gb.build()
+    gm = _gpu_model(tmp_path, sm, name="ssc")
+    cm = _cpu_model(tmp_path, sm, name="ssccpu")
+    x = np.random.default_rng(4).standard_normal((4, 9, 64)).astype(
+        np.float32)
+    g, c = gm.predict({"x": x}), cm.predict({"x": x})
+    np.testing.assert_allclose(g["cls"], c["cls"], rtol=0.02, atol=0.02)
+    np.testing.assert_allclose(g["win"], c["win"], rtol=0.02, atol=0.02)
+    np.testing.assert_array_equal(g["idx"], c["idx"])
+    np.testing.assert_allclose(g["idxf"], c["idxf"], rtol=0.01)

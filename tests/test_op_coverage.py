@@ -255,3 +255,89 @@ def test_rest_json_string_inputs_decode():
     arr = inputs["text"]
     assert arr.dtype == object
     assert arr[0] == b"abc" and arr[1] == b"hi"
+
+
+# -- StridedSlice / Slice / Cast / ArgMax ------------------------------------
+
+def _ss_graph(ashape, begin, end, strides, bm=0, em=0, sm=0):
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, list(ashape),
+                          signature_name="x")
+    ss = gb.node("StridedSlice", "ss",
+                 [x_ph,
+                  gb.const("b", np.array(begin, np.int32)),
+                  gb.const("e", np.array(end, np.int32)),
+                  gb.const("s", np.array(strides, np.int32))],
+                 T=f32, Index=gb.a_type(3),
+                 begin_mask=gb.a_int(bm), end_mask=gb.a_int(em),
+                 shrink_axis_mask=gb.a_int(sm),
+                 ellipsis_mask=gb.a_int(0), new_axis_mask=gb.a_int(0))
+    gb.mark_output("y", ss)
+    return gb.build()
+
+
+@pytest.mark.parametrize("begin,end,strides,bm,em,sm,ref", [
+    # x[:, 0, :] — the CLS-token extraction of real BERT exports
+    ([0, 0, 0], [0, 1, 0], [1, 1, 1], 0b101, 0b101, 0b010,
+     lambda x: x[:, 0, :]),
+    # x[:, 1:5, :]
+    ([0, 1, 0], [0, 5, 0], [1, 1, 1], 0b101, 0b101, 0,
+     lambda x: x[:, 1:5, :]),
+    # x[:, ::2, 1:]  (begin_mask must NOT cover the begin=1 dim)
+    ([0, 0, 1], [0, 0, 0], [1, 2, 1], 0b011, 0b111, 0,
+     lambda x: x[:, ::2, 1:]),
+    # negative begin: x[:, -3:, :]
+    ([0, -3, 0], [0, 0, 0], [1, 1, 1], 0b101, 0b111, 0,
+     lambda x: x[:, -3:, :]),
+])
+def test_strided_slice_family(tmp_path, begin, end, strides, bm, em, sm,
+                              ref):
+    sm_ = _ss_graph([-1, 7, 6], begin, end, strides, bm, em, sm)
+    model = _load(tmp_path, sm_)
+    assert [op.kind for op in model.plan.ops] == ["strided_copy"]
+    x = np.random.default_rng(0).standard_normal((3, 7, 6)).astype(
+        np.float32)
+    got = model.predict({"x": x})["y"]
+    want = ref(x)
+    assert got.shape == want.shape, (got.shape, want.shape)
+    np.testing.assert_allclose(got, want, rtol=1e-6)
+
+
+def test_slice_op(tmp_path):
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, [-1, 8, 5],
+                          signature_name="x")
+    sl = gb.node("Slice", "sl",
+                 [x_ph,
+                  gb.const("b", np.array([0, 2, 1], np.int32)),
+                  gb.const("sz", np.array([-1, 4, 3], np.int32))],
+                 T=f32, Index=gb.a_type(3))
+    gb.mark_output("y", sl)
+    model = _load(tmp_path, gb.build())
+    x = np.random.default_rng(1).standard_normal((2, 8, 5)).astype(
+        np.float32)
+    got = model.predict({"x": x})["y"]
+    np.testing.assert_allclose(got, x[:, 2:6, 1:4], rtol=1e-6)
+
+
+def test_cast_and_argmax(tmp_path):
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    i32 = gb.a_type(3)
+    ids = gb.placeholder("ids", np.int32, [-1, 6], signature_name="ids")
+    casted = gb.node("Cast", "c", [ids], SrcT=i32, DstT=f32)
+    sq = gb.node("Square", "sq", [casted], T=f32)
+    am = gb.node("ArgMax", "am",
+                 [sq, gb.const("ax", np.int32(-1))],
+                 T=f32, output_type=i32)
+    gb.mark_output("idx", am)
+    gb.mark_output("sq", sq)
+    model = _load(tmp_path, gb.build())
+    x = np.array([[1, -5, 3, 0, 2, -1], [0, 0, 7, -7, 1, 2]], np.int32)
+    out = model.predict({"ids": x})
+    np.testing.assert_allclose(out["sq"], (x.astype(np.float32)) ** 2)
+    # |-5| and |±7|: argmax of squares; -7 comes after 7 -> first wins
+    np.testing.assert_array_equal(out["idx"], [1, 2])
+    assert out["idx"].dtype == np.int32

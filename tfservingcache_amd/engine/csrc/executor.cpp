@@ -43,6 +43,8 @@ enum CallKind : int {
   K_DEPTHWISE,
   K_GEMM_FP8,
   K_QUANT_FP8,
+  K_CAST,
+  K_ARGMAX_LAST,
 };
 
 struct Call {
@@ -162,6 +164,14 @@ static void launch_call(const Call& c, hipStream_t s) {
                            reinterpret_cast<uint8_t*>(c.ptrs[1]),
                            reinterpret_cast<float*>(c.ptrs[2]),
                            I[0], I[1], I[2]);
+      break;
+    case K_CAST:
+      launch_cast(s, reinterpret_cast<const void*>(c.ptrs[0]),
+                  reinterpret_cast<void*>(c.ptrs[1]), I[0], int(I[1]));
+      break;
+    case K_ARGMAX_LAST:
+      launch_argmax_last(s, cp(0), reinterpret_cast<int*>(c.ptrs[1]),
+                         I[0], I[1]);
       break;
     case K_DEPTHWISE:
       // ptrs: [x, w, bias, y]
@@ -311,6 +321,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("K_DEPTHWISE") = int(tfsc::K_DEPTHWISE);
   mod.attr("K_GEMM_FP8") = int(tfsc::K_GEMM_FP8);
   mod.attr("K_QUANT_FP8") = int(tfsc::K_QUANT_FP8);
+  mod.attr("K_CAST") = int(tfsc::K_CAST);
+  mod.attr("K_ARGMAX_LAST") = int(tfsc::K_ARGMAX_LAST);
 
   register_fastpath(mod);
   register_frontend(mod);

@@ -64,6 +64,14 @@ void launch_pad_last(hipStream_t s, const ushort* x, ushort* y,
                      int64_t rows, int c_in, int c_out);
 void launch_f32_to_bf16(hipStream_t s, const float* x, ushort* y,
                         int64_t n);
+// dtype casts between workspace tensors: 0 = i32 -> bf16,
+// 1 = bf16 -> i32 (truncate toward zero, TF Cast semantics)
+void launch_cast(hipStream_t s, const void* x, void* y, int64_t n,
+                 int mode);
+// rowwise argmax over the last dim: [rows, cols] bf16 -> i32 (first
+// max wins on ties, TF semantics)
+void launch_argmax_last(hipStream_t s, const ushort* x, int* y,
+                        int64_t rows, int64_t cols);
 void launch_bf16_to_f32(hipStream_t s, const ushort* x, float* y,
                         int64_t n);
 

@@ -771,4 +771,77 @@ void launch_depthwise_conv(hipStream_t s, const ushort* x, const ushort* w,
   }
 }
 
+// ---------------------------------------------------------------------------
+// dtype casts + rowwise argmax (classification heads / index ops)
+// ---------------------------------------------------------------------------
+__global__ void k_cast_i2f(const int* __restrict__ x,
+                           ushort* __restrict__ y, int64_t n) {
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n; i += stride)
+    y[i] = f2bf(float(x[i]));
+}
+
+__global__ void k_cast_f2i(const ushort* __restrict__ x,
+                           int* __restrict__ y, int64_t n) {
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n; i += stride)
+    y[i] = int(bf2f(x[i]));      // trunc toward zero (TF Cast)
+}
+
+void launch_cast(hipStream_t s, const void* x, void* y, int64_t n,
+                 int mode) {
+  if (mode == 0)
+    hipLaunchKernelGGL(k_cast_i2f, dim3(grid_for(n)), dim3(TPB), 0, s,
+                       reinterpret_cast<const int*>(x),
+                       reinterpret_cast<ushort*>(y), n);
+  else
+    hipLaunchKernelGGL(k_cast_f2i, dim3(grid_for(n)), dim3(TPB), 0, s,
+                       reinterpret_cast<const ushort*>(x),
+                       reinterpret_cast<int*>(y), n);
+}
+
+// one wave per row; first-max-wins tie-breaking via (value, -index)
+__global__ __launch_bounds__(256)
+void k_argmax_last(const ushort* __restrict__ x, int* __restrict__ y,
+                   int64_t rows, int64_t cols) {
+  int wv = threadIdx.x / WAVE;
+  int lane = threadIdx.x % WAVE;
+  int64_t waves_total = (int64_t)gridDim.x * (256 / WAVE);
+  for (int64_t row = (int64_t)blockIdx.x * (256 / WAVE) + wv;
+       row < rows; row += waves_total) {
+    const ushort* xr = x + row * cols;
+    float best = -3.0e38f;
+    int besti = 0x7FFFFFFF;
+    for (int64_t i = lane; i < cols; i += WAVE) {
+      float v = bf2f(xr[i]);
+      if (v > best || (v == best && int(i) < besti)) {
+        best = v;
+        besti = int(i);
+      }
+    }
+    #pragma unroll
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+      float ov = __shfl_xor(best, off, WAVE);
+      int oi = __shfl_xor(besti, off, WAVE);
+      if (ov > best || (ov == best && oi < besti)) {
+        best = ov;
+        besti = oi;
+      }
+    }
+    if (lane == 0) y[row] = besti;
+  }
+}
+
+void launch_argmax_last(hipStream_t s, const ushort* x, int* y,
+                        int64_t rows, int64_t cols) {
+  int64_t waves = rows;
+  int blocks = int(ceil_div(waves, (int64_t)(256 / WAVE)));
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(k_argmax_last, dim3(blocks), dim3(256), 0, s, x, y,
+                     rows, cols);
+}
+
 }  // namespace tfsc

@@ -191,6 +191,30 @@ class CpuExecutor:
         elif k == "concat":
             parts = [_get(i) for i in op.inputs]
             vals[op.outputs[0]] = np.concatenate(parts, axis=p["axis"])
+        elif k == "strided_copy":
+            x = _get(op.inputs[0])
+            sl = []
+            for d in range(x.ndim):
+                b = p["starts"][d]
+                st = p["steps"][d]
+                if p["shrink"][d]:
+                    sl.append(b)
+                else:
+                    sl.append(slice(b, None, st))
+            y = x[tuple(sl)]
+            out_shape = self.plan.resolve_shape(
+                self.plan.tensors[op.outputs[0]].shape, batch)
+            vals[op.outputs[0]] = np.ascontiguousarray(
+                y)[tuple(slice(0, dd) for dd in out_shape)].copy()
+        elif k == "cast":
+            x = _get(op.inputs[0])
+            if p["mode"] == "i2f":
+                vals[op.outputs[0]] = x.astype(np.float32)
+            else:
+                vals[op.outputs[0]] = x.astype(np.int32)
+        elif k == "argmax_last":
+            x = _get(op.inputs[0])
+            vals[op.outputs[0]] = x.argmax(axis=-1).astype(np.int32)
         else:
             raise ValueError(f"unknown plan op {k}")
 

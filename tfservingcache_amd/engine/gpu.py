@@ -725,6 +725,41 @@ class ExecContext:
                                self._ptr(op.inputs[1]),
                                self._ptr(op.outputs[0])],
                               [n_idx, row], []))
+            elif k == "strided_copy":
+                in_shape = self.shapes[op.inputs[0]]
+                out_shape = self.shapes[op.outputs[0]]
+                istr = [1] * len(in_shape)
+                for d in range(len(in_shape) - 2, -1, -1):
+                    istr[d] = istr[d + 1] * in_shape[d + 1]
+                starts, steps = p["starts"], p["steps"]
+                shrink = p["shrink"]
+                offset = sum(starts[d] * istr[d]
+                             for d in range(len(in_shape)))
+                strides_out = [istr[d] * steps[d]
+                               for d in range(len(in_shape))
+                               if not shrink[d]]
+                n_out = int(np.prod(out_shape)) if out_shape else 1
+                nd_out = max(len(out_shape), 1)
+                od = list(out_shape) or [1]
+                so = strides_out or [1]
+                calls.append((ext.K_TRANSPOSE,
+                              [self._ptr(op.inputs[0]) + offset * 2,
+                               self._ptr(op.outputs[0])],
+                              [nd_out] + od + so + [n_out], []))
+            elif k == "cast":
+                shape = self.shapes[op.inputs[0]]
+                n = int(np.prod(shape)) if shape else 1
+                calls.append((ext.K_CAST,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.outputs[0])],
+                              [n, 0 if p["mode"] == "i2f" else 1], []))
+            elif k == "argmax_last":
+                shape = self.shapes[op.inputs[0]]
+                rows = int(np.prod(shape[:-1])) if len(shape) > 1 else 1
+                calls.append((ext.K_ARGMAX_LAST,
+                              [self._ptr(op.inputs[0]),
+                               self._ptr(op.outputs[0])],
+                              [rows, shape[-1]], []))
             elif k == "pad":
                 shape = self.shapes[op.inputs[0]]
                 pads = p["pads"]

@@ -365,6 +365,15 @@ class _Lowerer:
         if op == "Einsum":
             self.lower_einsum(nd, out)
             return
+        if op in ("StridedSlice", "Slice"):
+            self.lower_strided_slice(nd, out)
+            return
+        if op == "Cast":
+            self.lower_cast(nd, out)
+            return
+        if op == "ArgMax":
+            self.lower_argmax(nd, out)
+            return
         if op == "MatMul":
             self.lower_matmul(nd, out)
             return
@@ -655,6 +664,133 @@ class _Lowerer:
                 y = y3
         if out_name != out:
             self.by_name[out] = y
+
+    def lower_strided_slice(self, nd: g.NodeDef, out: str) -> None:
+        """StridedSlice / Slice with constant begin/end/strides (the
+        common export shapes: CLS-token extraction x[:, 0], windowing,
+        channel slices). Positive strides only; no ellipsis/new-axis
+        masks; the symbolic batch dim must be taken whole. Lowered to a
+        strided-copy (the transpose kernel's arbitrary-stride gather
+        with a baked-in element offset)."""
+        x = self.tid(nd.input[0])
+        xt = self.tensors[x]
+        xs = list(xt.shape)
+        rank = len(xs)
+        begin = self.const_value(nd.input[1])
+        if nd.op == "Slice":
+            sizes = self.const_value(nd.input[2])
+            if begin is None or sizes is None:
+                raise PlanError("dynamic Slice unsupported")
+            begin = [int(b) for b in begin]
+            end = []
+            for d in range(len(begin)):
+                sz = int(sizes[d])
+                if sz == -1:
+                    end.append(None)
+                else:
+                    end.append(begin[d] + sz)
+            strides = [1] * len(begin)
+            bm = em = sm = 0
+        else:
+            end_v = self.const_value(nd.input[2])
+            str_v = self.const_value(nd.input[3])
+            if begin is None or end_v is None or str_v is None:
+                raise PlanError("dynamic StridedSlice unsupported")
+            if _attr_i(nd, "ellipsis_mask", 0) or \
+                    _attr_i(nd, "new_axis_mask", 0):
+                raise PlanError("StridedSlice ellipsis/new-axis masks "
+                                "unsupported")
+            begin = [int(b) for b in begin]
+            end = [int(e) for e in end_v]
+            strides = [int(s_) for s_ in str_v]
+            bm = _attr_i(nd, "begin_mask", 0)
+            em = _attr_i(nd, "end_mask", 0)
+            sm = _attr_i(nd, "shrink_axis_mask", 0)
+        if xt.dtype != "f32":
+            raise PlanError("StridedSlice on non-f32 unsupported")
+
+        starts, steps, shrink, out_dims = [], [], [], []
+        for d in range(rank):
+            if d < len(begin):
+                b = begin[d]
+                e = end[d] if d < len(end) else None
+                st = strides[d] if d < len(strides) else 1
+                if st <= 0:
+                    raise PlanError("negative StridedSlice stride")
+                if (bm >> d) & 1:
+                    b = 0
+                if nd.op == "StridedSlice" and (em >> d) & 1:
+                    e = None
+                sh = bool((sm >> d) & 1)
+            else:
+                b, e, st, sh = 0, None, 1, False
+            if is_sym(xs[d]):
+                if b != 0 or e is not None or st != 1 or sh:
+                    raise PlanError("slicing the batch dim unsupported")
+                starts.append(0)
+                steps.append(1)
+                shrink.append(False)
+                out_dims.append(xs[d])
+                continue
+            size = int(xs[d])
+            if b < 0:
+                b += size
+            b = max(0, min(b, size))
+            if sh:
+                e = b + 1
+            elif e is None:
+                e = size
+            else:
+                if e < 0:
+                    e += size
+                e = max(0, min(e, size))
+            starts.append(b)
+            steps.append(st)
+            shrink.append(sh)
+            if not sh:
+                out_dims.append(max(0, (e - b + st - 1) // st))
+        y = self.new_tensor(tuple(out_dims), "f32", "activation", out)
+        self.ops.append(PlanOp("strided_copy", [x], [y],
+                               {"starts": starts, "steps": steps,
+                                "shrink": shrink}))
+
+    def lower_cast(self, nd: g.NodeDef, out: str) -> None:
+        x = self.tid(nd.input[0])
+        xt = self.tensors[x]
+        dt_attr = nd.attr.get("DstT")
+        dst_t = dt_attr.type if dt_attr is not None else m.DT_FLOAT
+        dst = "i32" if dst_t in (m.DT_INT32, m.DT_INT64) else "f32"
+        cv = self.const_value(nd.input[0])
+        if cv is not None:
+            # constant fold
+            arr = np.asarray(cv)
+            arr = arr.astype(np.int32 if dst == "i32" else np.float32)
+            self.consts[nd.name] = arr
+            self.new_tensor(tuple(arr.shape), dst, "weight", out,
+                            weight=arr)
+            return
+        if xt.dtype == dst:
+            self.new_tensor(xt.shape, dst, xt.kind, out, alias_of=x)
+            return
+        y = self.new_tensor(xt.shape, dst, "activation", out)
+        mode = "i2f" if xt.dtype == "i32" else "f2i"
+        self.ops.append(PlanOp("cast", [x], [y], {"mode": mode}))
+
+    def lower_argmax(self, nd: g.NodeDef, out: str) -> None:
+        """ArgMax over the LAST axis (classification heads). Output is
+        int32 (TF defaults to int64 on the wire; int32 carries the same
+        indices for any realistic class count)."""
+        x = self.tid(nd.input[0])
+        xs = self.tensors[x].shape
+        axis_v = self.const_value(nd.input[1]) if len(nd.input) > 1 \
+            else np.int32(-1)
+        axis = int(np.asarray(axis_v).reshape(-1)[0])
+        if axis < 0:
+            axis += len(xs)
+        if axis != len(xs) - 1:
+            raise PlanError("ArgMax only over the last axis")
+        y = self.new_tensor(tuple(xs[:-1]), "i32", "activation", out)
+        self.ops.append(PlanOp("argmax_last", [x], [y], {}))
 
     def lower_einsum(self, nd: g.NodeDef, out: str) -> None:
         """Einsum (transformer SavedModel exports). Supported family:
