@@ -263,6 +263,43 @@ void fast_run_plan(void* plan, hipStream_t s) {
   reinterpret_cast<ExecPlan*>(plan)->run_on(s);
 }
 
+// Relocatable call template (plan-level context cache): the kernel
+// calls of a (plan, bucket) with every pointer reduced to
+// (tag, region index, offset). Instantiation rebases against a
+// workspace base + region base table — one C++ call per context
+// instead of rebuilding ~10^2 Python tuples per cold load.
+struct PtrDesc {
+  uint8_t tag;          // 0 = workspace+off, 1 = region[reg]+off, 2 = 0
+  int32_t reg = 0;
+  int64_t off = 0;
+};
+
+class CallTemplate {
+ public:
+  std::vector<Call> calls;                  // ptrs left empty
+  std::vector<std::vector<PtrDesc>> descs;
+
+  ExecPlan* instantiate(uintptr_t ws,
+                        const std::vector<uintptr_t>& bases) const {
+    std::vector<Call> out = calls;
+    for (size_t i = 0; i < out.size(); ++i) {
+      auto& ds = descs[i];
+      auto& ptrs = out[i].ptrs;
+      ptrs.resize(ds.size());
+      for (size_t j = 0; j < ds.size(); ++j) {
+        const PtrDesc& d = ds[j];
+        if (d.tag == 0)
+          ptrs[j] = intptr_t(ws) + d.off;
+        else if (d.tag == 1)
+          ptrs[j] = intptr_t(bases.at(size_t(d.reg))) + d.off;
+        else
+          ptrs[j] = 0;
+      }
+    }
+    return new ExecPlan(out);
+  }
+};
+
 }  // namespace tfsc
 
 namespace py = pybind11;
@@ -323,6 +360,42 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("K_QUANT_FP8") = int(tfsc::K_QUANT_FP8);
   mod.attr("K_CAST") = int(tfsc::K_CAST);
   mod.attr("K_ARGMAX_LAST") = int(tfsc::K_ARGMAX_LAST);
+
+  py::class_<tfsc::CallTemplate>(mod, "CallTemplate")
+      .def(py::init([](py::list calls) {
+        auto* t = new tfsc::CallTemplate();
+        t->calls.reserve(calls.size());
+        t->descs.reserve(calls.size());
+        for (auto item : calls) {
+          py::tuple tu = item.cast<py::tuple>();
+          Call c;
+          c.kind = tu[0].cast<int>();
+          std::vector<tfsc::PtrDesc> ds;
+          for (auto dd : tu[1].cast<py::list>()) {
+            py::tuple dt = dd.cast<py::tuple>();
+            tfsc::PtrDesc d;
+            d.tag = uint8_t(dt[0].cast<int>());
+            if (d.tag == 0) {
+              d.off = dt[1].cast<int64_t>();
+            } else if (d.tag == 1) {
+              d.reg = dt[1].cast<int32_t>();
+              d.off = dt[2].cast<int64_t>();
+            }
+            ds.push_back(d);
+          }
+          c.ints = tu[2].cast<std::vector<int64_t>>();
+          c.floats = tu[3].cast<std::vector<float>>();
+          t->calls.push_back(std::move(c));
+          t->descs.push_back(std::move(ds));
+        }
+        return t;
+      }));
+
+  mod.def("instantiate_plan",
+          [](const tfsc::CallTemplate& t, uintptr_t ws,
+             const std::vector<uintptr_t>& bases) {
+            return std::unique_ptr<ExecPlan>(t.instantiate(ws, bases));
+          });
 
   register_fastpath(mod);
   register_frontend(mod);

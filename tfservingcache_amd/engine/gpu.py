@@ -394,11 +394,12 @@ class ExecContext:
         dev = gm.device
 
         tmpl = _get_ctx_template(plan, batch, gm.dtype)
-        calls = None
+        done = False
         if tmpl:
             # relocation fast path: all the Python-heavy planning and
-            # emission was done once for this (plan, bucket); just
-            # allocate the workspace and rebase the pointers
+            # emission was done once for this (plan, bucket); allocate
+            # the workspace and instantiate the C++ call template with
+            # rebased pointers (one pybind call)
             self.shapes = tmpl["shapes"]
             self.root = tmpl["root"]
             self.bufs = tmpl["bufs"]
@@ -406,20 +407,17 @@ class ExecContext:
             self.scratch_off = tmpl["scratch_off"]
             self.workspace = torch.empty(self.total_bytes,
                                          dtype=torch.uint8, device=dev)
-            ws = self.workspace.data_ptr()
             try:
                 bases = gm.region_bases()
-                calls = [(kind,
-                          [ws + d[1] if d[0] == 0 else
-                           (0 if d[0] == 2 else bases[d[1]] + d[2])
-                           for d in descs],
-                          ints, floats)
-                         for kind, descs, ints, floats in tmpl["calls"]]
+                base_list = [bases[k] for k in tmpl["region_keys"]]
+                self.exec_plan = ext.instantiate_plan(
+                    tmpl["ctmpl"], self.workspace.data_ptr(), base_list)
+                done = True
             except KeyError:
                 # this model is missing a region (e.g. the transform
                 # arena failed to publish) — fall through to full emit
-                calls = None
-        if calls is None:
+                done = False
+        if not done:
             self.shapes: List[Tuple[int, ...]] = [
                 plan.resolve_shape(t.shape, batch) for t in plan.tensors]
             self.root = [t.alias_of if t.alias_of is not None else t.idx
@@ -438,7 +436,7 @@ class ExecContext:
             calls = self._emit_calls()
             if tmpl is None:            # not attempted yet (None = new)
                 self._register_template(calls)
-        self.exec_plan = ext.ExecPlan(calls)
+            self.exec_plan = ext.ExecPlan(calls)
         self.captured = False
         self._views: Dict[int, object] = {}
         self.lock = threading.Lock()
@@ -458,10 +456,13 @@ class ExecContext:
         region | null) and cache the relocatable template for this
         (plan, bucket, dtype). If any pointer can't be classified the
         template is disabled for this key (correctness first)."""
+        torch, ext = _load_backend()
         gm = self.gm
         ws_base = self.workspace.data_ptr()
         ws_end = ws_base + self.total_bytes
         regions = gm.region_list()       # [(base, size, key)] sorted
+        region_keys: List = []
+        key_index: Dict = {}
         tcalls = []
         try:
             for kind, ptrs, ints, floats in calls:
@@ -475,7 +476,10 @@ class ExecContext:
                         hit = None
                         for base, size, key in regions:
                             if base <= p < base + size:
-                                hit = (1, key, p - base)
+                                if key not in key_index:
+                                    key_index[key] = len(region_keys)
+                                    region_keys.append(key)
+                                hit = (1, key_index[key], p - base)
                                 break
                         if hit is None:
                             raise KeyError(f"unclassifiable ptr {p:#x}")
@@ -488,7 +492,9 @@ class ExecContext:
             return
         tmpl = {"shapes": self.shapes, "root": self.root,
                 "bufs": self.bufs, "total_bytes": self.total_bytes,
-                "scratch_off": self.scratch_off, "calls": tcalls}
+                "scratch_off": self.scratch_off,
+                "region_keys": region_keys,
+                "ctmpl": ext.CallTemplate(tcalls)}
         _put_ctx_template(gm.plan, self.batch, gm.dtype, tmpl)
         # the template's region keys must be resolvable by later models
         # sharing the plan — snapshot the transform bytes now
