@@ -244,6 +244,10 @@ def make_gpu_loader(cache: LRUCache, device: str = "cuda:0",
             # (loader) thread: under LRU churn most requests are a
             # model's first, and a prewarmed context lets them take the
             # C++ fast path instead of the Python build path
+            t3 = _time.monotonic()
             lm._gpu.prewarm(prewarm_batch, prewarm_ctxs)
+            if timing:
+                log.warning("prewarm %.1fms", (_time.monotonic() - t3)
+                            * 1e3)
         return lm
     return loader
