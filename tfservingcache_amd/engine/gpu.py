@@ -1241,6 +1241,10 @@ class GpuModel:
         wt = self._gemm_weights.get(key)
         if wt is not None:
             return wt
+        av = self._arena_view(("gw",) + key)
+        if av is not None:
+            self._gemm_weights[key] = av
+            return av
         if idx not in self._weights:
             raise RuntimeError(
                 "GPU gemm requires a constant weight operand; "
@@ -1264,8 +1268,14 @@ class GpuModel:
         torch, _ = _load_backend()
         key = (idx, graph_trans_b)
         ent = self._gemm_weights_fp8.get(key)
-        if ent is not None:
+        if ent is not None and ent[0] is not None and \
+                ent[1] is not None:
             return ent
+        q_av = self._arena_view(("f8q",) + key)
+        s_av = self._arena_view(("f8s",) + key)
+        if q_av is not None and s_av is not None:
+            self._gemm_weights_fp8[key] = (q_av, s_av)
+            return q_av, s_av
         if idx not in self._weights:
             raise RuntimeError("fp8 gemm requires a constant weight")
         w = self._weights[idx].float()
@@ -1290,6 +1300,10 @@ class GpuModel:
         wt = self._conv_weights.get(idx)
         if wt is not None:
             return wt
+        av = self._arena_view(("cw", idx))
+        if av is not None:
+            self._conv_weights[idx] = av
+            return av
         w = self._weights[idx].float()          # [R,S,C,K]
         R, S, C, Kc = w.shape
         w = w.reshape(R * S * C, Kc).t().contiguous()   # [K][RSC]
@@ -1308,6 +1322,10 @@ class GpuModel:
         wt = self._conv_weights.get(key)
         if wt is not None:
             return wt
+        av = self._arena_view(("cw",) + key)
+        if av is not None:
+            self._conv_weights[key] = av
+            return av
         w = self._weights[idx].float()          # [R,S,C,K]
         R, S, C, Kc = w.shape
         w = torch.nn.functional.pad(w, (0, 0, 0, c8 - C))  # pad C dim
@@ -1356,6 +1374,12 @@ class GpuModel:
         blob = getattr(self, "_weight_blob", None)
         if blob is not None:
             bases[("blob",)] = blob.data_ptr()
+        arena = getattr(self, "_arena_dev", None)
+        if arena is not None:
+            base = arena.data_ptr()
+            for key, (_k, off, _nb, _dt, _sh) in \
+                    getattr(self, "_arena_entries", {}).items():
+                bases[key] = base + off
         for key, t in self._region_items():
             bases[key] = t.data_ptr()
         bases[("zeros",)] = self.zeros_ptr()
@@ -1406,26 +1430,25 @@ class GpuModel:
             with torch.cuda.stream(up):
                 dev_arena.copy_(pinned, non_blocking=True)
         self._arena_dev = dev_arena
+        # LAZY views: region_bases() computes pointers straight from
+        # the entry offsets; tensor views materialize only if the
+        # (rare) full-emit path asks for a transform. ~300 eager view
+        # creations per cold load were milliseconds of Python.
+        self._arena_entries = {tuple(e[0]): e for e in entries}
+        # int weights ARE needed eagerly (emit reads _weights[idx]) and
+        # are few
         for key, off, nb, dt, shape in entries:
-            view = dev_arena[off:off + nb].view(dt).view(shape)
-            kind = key[0]
-            if kind == "int":
-                self._weights[key[1]] = view
-            elif kind == "gw":
-                self._gemm_weights[tuple(key[1:])] = view
-            elif kind == "cw":
-                k = key[1:]
-                self._conv_weights[k[0] if len(k) == 1 else tuple(k)] = \
-                    view
-            elif kind == "f8q":
-                k = tuple(key[1:])
-                q, sc = self._gemm_weights_fp8.get(k, (None, None))
-                self._gemm_weights_fp8[k] = (view, sc)
-            elif kind == "f8s":
-                k = tuple(key[1:])
-                q, sc = self._gemm_weights_fp8.get(k, (None, None))
-                self._gemm_weights_fp8[k] = (q, view)
+            if key[0] == "int":
+                self._weights[key[1]] = \
+                    dev_arena[off:off + nb].view(dt).view(shape)
         return True
+
+    def _arena_view(self, key: tuple):
+        ent = getattr(self, "_arena_entries", {}).get(tuple(key))
+        if ent is None or self._arena_dev is None:
+            return None
+        _key, off, nb, dt, shape = ent
+        return self._arena_dev[off:off + nb].view(dt).view(shape)
 
     def weight_bytes(self) -> int:
         total = 0
@@ -1615,4 +1638,5 @@ class GpuModel:
             self._gemm_weights_fp8.clear()
             self._conv_weights.clear()
             self._arena_dev = None
+            self._arena_entries = {}
             self._weight_blob = None
