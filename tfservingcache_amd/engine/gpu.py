@@ -413,6 +413,7 @@ class ExecContext:
                 self.exec_plan = ext.instantiate_plan(
                     tmpl["ctmpl"], self.workspace.data_ptr(), base_list)
                 done = True
+                self._from_template = True
             except KeyError:
                 # this model is missing a region (e.g. the transform
                 # arena failed to publish) — fall through to full emit
@@ -437,6 +438,7 @@ class ExecContext:
             if tmpl is None:            # not attempted yet (None = new)
                 self._register_template(calls)
             self.exec_plan = ext.ExecPlan(calls)
+            self._from_template = False
         self.captured = False
         self._views: Dict[int, object] = {}
         self.lock = threading.Lock()
@@ -450,6 +452,22 @@ class ExecContext:
         self._pinned_in: Dict[int, object] = {}
         self._pinned_out: Dict[int, object] = {}
         self._pin_raws: List[object] = []   # pooled pinned backings
+
+    def capture_only(self):
+        """Capture the hipGraph WITHOUT an eager warm-up run: valid
+        for template-instantiated contexts, where every buffer and
+        transform already exists (the warm-up run existed to flush
+        lazy allocations before capture). Saves a full model execution
+        per cold load under LRU churn."""
+        torch, _ = _load_backend()
+        with torch.cuda.device(self.gm.device), \
+                torch.cuda.stream(self.stream):
+            try:
+                with capture_guard.capture():
+                    self.exec_plan.capture()
+            except Exception:       # noqa: BLE001
+                log.exception("hipGraph capture failed; staying eager")
+            self.captured = True
 
     def _register_template(self, calls) -> None:
         """Classify every emitted pointer into (workspace | weight
@@ -1219,11 +1237,38 @@ class GpuModel:
             if cpu_blob is not None:
                 _blob_cache_put(self.plan, cpu_blob)
         self._weight_blob = blob              # keep the allocation alive
-        for idx, w, off, n in float_ws:
-            self._weights[idx] = blob[off:off + n].view(tuple(w.shape))
+        # LAZY views: ~270 torch view creations per cold load cost
+        # milliseconds; weight_ptr/master_weight materialize on demand
+        # (the template fast path needs neither)
+        self._blob_slots = {idx: (off, n, tuple(w.shape))
+                            for idx, w, off, n in float_ws}
+
+    def master_weight(self, idx: int):
+        """bf16 master view of plan weight idx (blob slice or int
+        tensor), materialized lazily."""
+        t = self._weights.get(idx)
+        if t is not None:
+            return t
+        slot = getattr(self, "_blob_slots", {}).get(idx)
+        if slot is None:
+            raise KeyError(f"no master weight for tensor {idx}")
+        off, n, shape = slot
+        t = self._weight_blob[off:off + n].view(shape)
+        self._weights[idx] = t
+        return t
+
+    def has_master_weight(self, idx: int) -> bool:
+        return idx in self._weights or \
+            idx in getattr(self, "_blob_slots", {})
 
     def weight_ptr(self, idx: int) -> int:
-        return self._weights[idx].data_ptr()
+        t = self._weights.get(idx)
+        if t is not None:
+            return t.data_ptr()
+        slot = getattr(self, "_blob_slots", {}).get(idx)
+        if slot is not None:
+            return self._weight_blob.data_ptr() + slot[0] * 2
+        return self._weights[idx].data_ptr()    # raises KeyError
 
     def zeros_ptr(self) -> int:
         """A small zeroed device buffer used as the gather source for
@@ -1245,11 +1290,11 @@ class GpuModel:
         if av is not None:
             self._gemm_weights[key] = av
             return av
-        if idx not in self._weights:
+        if not self.has_master_weight(idx):
             raise RuntimeError(
                 "GPU gemm requires a constant weight operand; "
                 "activation x activation MatMul must be BatchMatMul")
-        w = self._weights[idx].float()
+        w = self.master_weight(idx).float()
         if not graph_trans_b:
             w = w.t().contiguous()      # [K,N] -> [N,K]
         N, K = w.shape
@@ -1276,9 +1321,9 @@ class GpuModel:
         if q_av is not None and s_av is not None:
             self._gemm_weights_fp8[key] = (q_av, s_av)
             return q_av, s_av
-        if idx not in self._weights:
+        if not self.has_master_weight(idx):
             raise RuntimeError("fp8 gemm requires a constant weight")
-        w = self._weights[idx].float()
+        w = self.master_weight(idx).float()
         if not graph_trans_b:
             w = w.t().contiguous()          # [K,N] -> [N,K]
         N, K = w.shape
@@ -1304,7 +1349,7 @@ class GpuModel:
         if av is not None:
             self._conv_weights[idx] = av
             return av
-        w = self._weights[idx].float()          # [R,S,C,K]
+        w = self.master_weight(idx).float()     # [R,S,C,K]
         R, S, C, Kc = w.shape
         w = w.reshape(R * S * C, Kc).t().contiguous()   # [K][RSC]
         Kp = _pad64(R * S * C)
@@ -1326,7 +1371,7 @@ class GpuModel:
         if av is not None:
             self._conv_weights[key] = av
             return av
-        w = self._weights[idx].float()          # [R,S,C,K]
+        w = self.master_weight(idx).float()     # [R,S,C,K]
         R, S, C, Kc = w.shape
         w = torch.nn.functional.pad(w, (0, 0, 0, c8 - C))  # pad C dim
         w = w.reshape(R * S * c8, Kc).t().contiguous()
@@ -1452,9 +1497,17 @@ class GpuModel:
 
     def weight_bytes(self) -> int:
         total = 0
+        blob = getattr(self, "_weight_blob", None)
+        if blob is not None:
+            total += blob.numel() * blob.element_size()
+        arena = getattr(self, "_arena_dev", None)
+        if arena is not None:
+            total += arena.numel() * arena.element_size()
         for d in (self._weights, self._gemm_weights, self._conv_weights):
             for wt in d.values():
-                total += wt.numel() * wt.element_size()
+                if wt is not None and getattr(wt, "untyped_storage",
+                                              None) is not None:
+                    total += wt.numel() * wt.element_size()
         return total
 
     # -- execution ---------------------------------------------------------
@@ -1606,7 +1659,11 @@ class GpuModel:
             todo = [c for c in ctxs if c.fast_id is None]
         for ctx in todo:
             with ctx.lock:
-                ctx.run(feeds, fetch)
+                if getattr(ctx, "_from_template", False) and \
+                        self.use_graphs and not ctx.captured:
+                    ctx.capture_only()
+                else:
+                    ctx.run(feeds, fetch)
                 if ctx.fast_id is None and ctx.captured:
                     self._register_fast(ctx)
 
@@ -1639,4 +1696,5 @@ class GpuModel:
             self._conv_weights.clear()
             self._arena_dev = None
             self._arena_entries = {}
+            self._blob_slots = {}
             self._weight_blob = None
