@@ -295,3 +295,56 @@ def test_eviction_under_concurrent_load(tmp_path):
         t.join()
     assert not errors, errors[:5]
     assert not mismatches, mismatches[:5]
+
+
+@pytest.mark.gpu
+def test_template_capture_only_numerics(tmp_path):
+    """The LRU-churn fast load path end-to-end: content-shared models
+    where the SECOND load takes template instantiation + arena restore
+    + capture WITHOUT an eager warm run (capture_only). Its predictions
+    must match the CPU fp32 reference — this is the exact path the
+    headline bench serves from."""
+    import os
+    import shutil
+    import numpy as np
+    from tfservingcache_amd.cachemanager import (CacheManager, LRUCache,
+                                                 ModelPool,
+                                                 make_gpu_loader)
+    from tfservingcache_amd.cachemanager.providers import \
+        DiskModelProvider
+    from tfservingcache_amd.engine.model import load_model_from_dir
+    from tfservingcache_amd.models import write_model_repo
+
+    repo = tmp_path / "repo"
+    write_model_repo(str(repo), [("m0", 1, "mlp")])
+    # hardlink content-identical copies -> shared plan -> template path
+    for i in (1, 2):
+        dst = repo / f"m{i}" / "1"
+        os.makedirs(dst.parent, exist_ok=True)
+        shutil.copytree(repo / "m0" / "1", dst,
+                        copy_function=os.link)
+
+    cache = LRUCache(str(tmp_path / "cache"), 10 ** 8)
+    pool = ModelPool(
+        make_gpu_loader(cache, max_batch=8, prewarm_batch=8),
+        max_concurrent_models=3)
+    cm = CacheManager(DiskModelProvider(str(repo)), cache, pool)
+
+    x = np.random.default_rng(0).standard_normal((8, 16)).astype(
+        np.float32)
+    outs = []
+    for i in range(3):
+        lm = cm.ensure_loaded(f"m{i}", 1)
+        outs.append(lm.predict({"x": x})["probs"])
+    # CPU fp32 reference on the same content
+    cpu = load_model_from_dir(str(repo / "m0" / "1"), "cpu", 1)
+    want = cpu.predict({"x": x})["probs"]
+    for i, got in enumerate(outs):
+        np.testing.assert_allclose(got, want, rtol=0.05, atol=0.02,
+                                   err_msg=f"model m{i}")
+    # the fast C++ context must actually be registered (prewarm ran)
+    lm2 = cm.ensure_loaded("m2", 1)
+    ctxs = [c for lst in lm2._gpu._contexts.values() for c in lst]
+    assert ctxs and all(c.captured for c in ctxs)
+    assert any(c.fast_id is not None and c.fast_id >= 0 for c in ctxs)
+    assert any(getattr(c, "_from_template", False) for c in ctxs)
