@@ -185,6 +185,21 @@ _BLOB_CACHE_CAP = 4
 _blob_cache_lock = threading.Lock()
 _blob_cache: Dict[int, tuple] = {}       # id(plan) -> (plan, blob)
 _blob_cache_order: List[int] = []
+# single-flight for the CPU convert: concurrent first loads of one
+# content must not both run the (expensive, device-synchronizing)
+# convert path — the loser waits and takes the cached blob
+_convert_locks: Dict[int, threading.Lock] = {}
+
+
+def _convert_lock(plan) -> threading.Lock:
+    with _blob_cache_lock:
+        lk = _convert_locks.get(id(plan))
+        if lk is None:
+            lk = threading.Lock()
+            _convert_locks[id(plan)] = lk
+            while len(_convert_locks) > 4 * _BLOB_CACHE_CAP:
+                _convert_locks.pop(next(iter(_convert_locks)))
+        return lk
 
 
 def _blob_cache_get(plan):
@@ -1167,6 +1182,28 @@ class GpuModel:
         blob = torch.empty(total, dtype=torch.bfloat16, device=self.device)
         tb = _time.monotonic()
         cached = _blob_cache_get(self.plan)
+        conv_held = False
+        conv_lock = None
+        if cached is None:
+            # single-flight the convert; a second concurrent first-load
+            # of the same content waits and takes the cached blob
+            conv_lock = _convert_lock(self.plan)
+            conv_lock.acquire()
+            conv_held = True
+            cached = _blob_cache_get(self.plan)
+            if cached is not None:
+                conv_lock.release()
+                conv_held = False
+        try:
+            self._upload_blob(torch, blob, cached, total, float_ws,
+                              timing, ta, tb)
+        finally:
+            if conv_held:
+                conv_lock.release()
+
+    def _upload_blob(self, torch, blob, cached, total, float_ws,
+                     timing, ta, tb):
+        import time as _time
         if cached is not None and cached.numel() == total:
             # converted blob already pinned in CPU RAM: ONE DMA on the
             # dedicated upload stream, guard-serialized against captures
@@ -1202,7 +1239,12 @@ class GpuModel:
             except RuntimeError:
                 log.warning("pinned weight-blob allocation failed; "
                             "skipping the converted-weight cache")
-            with _staging_lock:
+            # the convert's flush() does DEVICE-WIDE synchronizes —
+            # unguarded, those poison any concurrent hipGraph capture
+            # (reproduced by the 2-rank-on-one-GPU dry run), so the
+            # whole staging loop is an unsafe host op. It runs once per
+            # plan content (single-flighted above).
+            with capture_guard.unsafe_host_op(), _staging_lock:
                 stage = _get_staging(torch)
                 cap = stage.numel()
                 batch_items = []        # (stage_off, blob_off, n)
