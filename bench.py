@@ -63,7 +63,7 @@ from tfservingcache_amd.tfservingproxy import LocalServingHandler  # noqa: E402
 from tfservingcache_amd.wire import messages as m  # noqa: E402
 from tfservingcache_amd.wire.tensor import numpy_to_tensorproto  # noqa: E402
 
-REQS_PER_STEP = 50
+REQS_PER_STEP = 100
 
 
 def _client_proc_main(conn, port, model_name, model_kind, batch,
