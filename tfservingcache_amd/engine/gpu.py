@@ -461,11 +461,6 @@ class ExecContext:
         # dedicated non-default stream (the default stream cannot be
         # hipGraph-captured; copies + kernels + D2H all run here)
         self.stream = torch.cuda.Stream(device=dev)
-        evt = getattr(gm, "_upload_event", None)
-        if evt is not None:
-            # order everything on this context's stream after the
-            # model's weight-upload DMAs (non-blocking)
-            self.stream.wait_event(evt)
         # pinned host staging per feed/fetch tensor: numpy -> pinned
         # (with dtype conversion on CPU) -> one async DMA, instead of a
         # pageable f32 copy + on-device convert per request
@@ -1142,26 +1137,14 @@ class GpuModel:
             self._upload_weights()
             restored = self.try_restore_transforms()
             if self._upload_pending or restored:
-                up = _get_upload_stream(torch, device)
-                if restored:
-                    # arena-restored models never READ weights host-side
-                    # before execution (transforms come from the arena,
-                    # emit passes pointers only), so instead of BLOCKING
-                    # ~5 ms on the DMAs, record an event; every new
-                    # context's stream waits on it (async, us-scale) and
-                    # all execution is ordered after the uploads
-                    evt = torch.cuda.Event()
-                    evt.record(up)
-                    self._upload_event = evt
-                else:
-                    # non-restored loads may materialize transforms from
-                    # the master blob on the default stream — must be
-                    # complete. MUST be guard-serialized: a stream sync
-                    # concurrent with another thread's capture
-                    # invalidates the capture (the 100-step soak
-                    # reproduced hipErrorStreamCapture* when unguarded)
-                    with capture_guard.unsafe_host_op():
-                        up.synchronize()
+                # one sync covers both async DMAs (master blob +
+                # transform arena) on the shared upload stream. MUST be
+                # guard-serialized: a stream sync concurrent with
+                # another thread's capture invalidates the capture (the
+                # 100-step soak reproduced hipErrorStreamCapture*
+                # within seconds when this ran unguarded)
+                with capture_guard.unsafe_host_op():
+                    _get_upload_stream(torch, device).synchronize()
         if _os.environ.get("TFSC_LOAD_TIMING"):
             log.warning("upload_weights %.1fms (cached_blob=%s "
                         "arena=%s)", (_time.monotonic() - t0) * 1e3,
@@ -1756,5 +1739,4 @@ class GpuModel:
             self._arena_dev = None
             self._arena_entries = {}
             self._blob_slots = {}
-            self._upload_event = None
             self._weight_blob = None
