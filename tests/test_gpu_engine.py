@@ -428,3 +428,28 @@ def test_strided_slice_cast_argmax_gpu(tmp_path):
     np.testing.assert_allclose(g["win"], c["win"], rtol=0.02, atol=0.02)
     np.testing.assert_array_equal(g["idx"], c["idx"])
     np.testing.assert_allclose(g["idxf"], c["idxf"], rtol=0.01)
+
+
+def test_concat_pack_gpu(tmp_path):
+    """Concat (scatter kernel) + Pack/Unpack/LeakyRelu on GPU vs CPU."""
+    from tfservingcache_amd.engine.savedmodel import GraphBuilder
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    a = gb.placeholder("a", np.float32, [-1, 32], signature_name="a")
+    b = gb.placeholder("b", np.float32, [-1, 32], signature_name="b")
+    cc = gb.node("ConcatV2", "cc",
+                 [a, b, gb.const("ax", np.int32(1))],
+                 N=gb.a_int(2), T=f32)                   # [B, 64]
+    pk = gb.node("Pack", "pk", [a, b], N=gb.a_int(2), T=f32,
+                 axis=gb.a_int(1))                       # [B, 2, 32]
+    lr = gb.node("LeakyRelu", "lr", [cc], T=f32, alpha=gb.a_float(0.3))
+    gb.mark_output("cat", cc)
+    gb.mark_output("packed", pk)
+    gb.mark_output("lr", lr)
+    sm = gb.build()
+    gm = _gpu_model(tmp_path, sm, name="cat")
+    cm = _cpu_model(tmp_path, sm, name="catcpu")
+    rng = np.random.default_rng(6)
+    feeds = {"a": rng.standard_normal((4, 32)).astype(np.float32),
+             "b": rng.standard_normal((4, 32)).astype(np.float32)}
+    _compare(gm.predict(feeds), cm.predict(feeds), rtol=0.02, atol=0.02)

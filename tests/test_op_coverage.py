@@ -341,3 +341,48 @@ def test_cast_and_argmax(tmp_path):
     # |-5| and |±7|: argmax of squares; -7 comes after 7 -> first wins
     np.testing.assert_array_equal(out["idx"], [1, 2])
     assert out["idx"].dtype == np.int32
+
+
+# -- Pack / Unpack / Concat / LeakyRelu / Pow --------------------------------
+
+def test_pack_unpack_concat(tmp_path):
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    a = gb.placeholder("a", np.float32, [-1, 4], signature_name="a")
+    b = gb.placeholder("b", np.float32, [-1, 4], signature_name="b")
+    pk = gb.node("Pack", "pk", [a, b], N=gb.a_int(2), T=f32,
+                 axis=gb.a_int(1))                       # [B, 2, 4]
+    up = gb.node("Unpack", "up", [pk], num=gb.a_int(2), T=f32,
+                 axis=gb.a_int(1))
+    cc = gb.node("ConcatV2", "cc",
+                 [f"{up}:0" if False else "up:0", "up:1",
+                  gb.const("ax", np.int32(1))],
+                 N=gb.a_int(2), T=f32)                   # [B, 8]
+    gb.mark_output("packed", pk)
+    gb.mark_output("cat", cc)
+    model = _load(tmp_path, gb.build())
+    x = np.arange(8, dtype=np.float32).reshape(2, 4)
+    y = -x
+    out = model.predict({"a": x, "b": y})
+    np.testing.assert_allclose(out["packed"],
+                               np.stack([x, y], axis=1))
+    np.testing.assert_allclose(out["cat"],
+                               np.concatenate([x, y], axis=1))
+
+
+def test_leaky_relu_and_pow(tmp_path):
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, [-1, 6], signature_name="x")
+    lr = gb.node("LeakyRelu", "lr", [x_ph], T=f32,
+                 alpha=gb.a_float(0.1))
+    p2 = gb.node("Pow", "p2", [x_ph, gb.const("e2", np.float32(2.0))],
+                 T=f32)
+    gb.mark_output("lr", lr)
+    gb.mark_output("p2", p2)
+    model = _load(tmp_path, gb.build())
+    x = np.linspace(-3, 3, 12, dtype=np.float32).reshape(2, 6)
+    out = model.predict({"x": x})
+    np.testing.assert_allclose(out["lr"], np.where(x > 0, x, 0.1 * x),
+                               rtol=1e-6)
+    np.testing.assert_allclose(out["p2"], x * x, rtol=1e-6)

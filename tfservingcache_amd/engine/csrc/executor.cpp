@@ -45,6 +45,7 @@ enum CallKind : int {
   K_QUANT_FP8,
   K_CAST,
   K_ARGMAX_LAST,
+  K_SCATTER,
 };
 
 struct Call {
@@ -165,6 +166,13 @@ static void launch_call(const Call& c, hipStream_t s) {
                            reinterpret_cast<float*>(c.ptrs[2]),
                            I[0], I[1], I[2]);
       break;
+    case K_SCATTER: {
+      // ints: [ndim, in_dims..., out_strides..., n_in]
+      int nd = int(I[0]);
+      launch_scatter(s, cp(0), p(1), nd, I.data() + 1, I.data() + 1 + nd,
+                     I[1 + 2 * nd]);
+      break;
+    }
     case K_CAST:
       launch_cast(s, reinterpret_cast<const void*>(c.ptrs[0]),
                   reinterpret_cast<void*>(c.ptrs[1]), I[0], int(I[1]));
@@ -360,6 +368,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.attr("K_QUANT_FP8") = int(tfsc::K_QUANT_FP8);
   mod.attr("K_CAST") = int(tfsc::K_CAST);
   mod.attr("K_ARGMAX_LAST") = int(tfsc::K_ARGMAX_LAST);
+  mod.attr("K_SCATTER") = int(tfsc::K_SCATTER);
 
   py::class_<tfsc::CallTemplate>(mod, "CallTemplate")
       .def(py::init([](py::list calls) {

@@ -55,6 +55,11 @@ void launch_pool(hipStream_t s, const ushort* x, ushort* y, bool is_max,
 void launch_transpose(hipStream_t s, const ushort* x, ushort* y, int ndim,
                       const int64_t* out_dims, const int64_t* in_strides,
                       int64_t n_out);
+// inverse of launch_transpose: LINEAR source -> STRIDED destination
+// (concat writes each input into an offset slice of the output)
+void launch_scatter(hipStream_t s, const ushort* x, ushort* y, int ndim,
+                    const int64_t* in_dims, const int64_t* out_strides,
+                    int64_t n_in);
 void launch_gather_rows(hipStream_t s, const ushort* table, const int* idx,
                         ushort* y, int64_t n_idx, int64_t row_elems);
 void launch_pad_nhwc(hipStream_t s, const ushort* x, ushort* y,

@@ -772,6 +772,41 @@ void launch_depthwise_conv(hipStream_t s, const ushort* x, const ushort* w,
 }
 
 // ---------------------------------------------------------------------------
+// scatter (linear src -> strided dst): the concat building block; the
+// exact inverse of k_transpose's gather
+// ---------------------------------------------------------------------------
+__global__ void k_scatter(const ushort* __restrict__ x,
+                          ushort* __restrict__ y, TransArgs ta,
+                          int64_t n) {
+  int64_t i0 = (int64_t)blockIdx.x * TPB + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * TPB;
+  for (int64_t i = i0; i < n; i += stride) {
+    int64_t rem = i, dst = 0;
+    #pragma unroll
+    for (int d = MAX_DIMS - 1; d >= 0; --d) {
+      if (d >= ta.ndim) continue;
+      int64_t c = rem % ta.out_dims[d];     // here: INPUT dims
+      rem /= ta.out_dims[d];
+      dst += c * ta.in_strides[d];          // here: OUTPUT strides
+    }
+    y[dst] = x[i];
+  }
+}
+
+void launch_scatter(hipStream_t s, const ushort* x, ushort* y, int ndim,
+                    const int64_t* in_dims, const int64_t* out_strides,
+                    int64_t n_in) {
+  TransArgs ta;
+  ta.ndim = ndim;
+  for (int d = 0; d < MAX_DIMS; ++d) {
+    ta.out_dims[d] = d < ndim ? in_dims[d] : 1;
+    ta.in_strides[d] = d < ndim ? out_strides[d] : 0;
+  }
+  hipLaunchKernelGGL(k_scatter, dim3(grid_for(n_in)), dim3(TPB), 0, s,
+                     x, y, ta, n_in);
+}
+
+// ---------------------------------------------------------------------------
 // dtype casts + rowwise argmax (classification heads / index ops)
 // ---------------------------------------------------------------------------
 __global__ void k_cast_i2f(const int* __restrict__ x,

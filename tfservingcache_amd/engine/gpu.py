@@ -799,6 +799,23 @@ class ExecContext:
                               [self._ptr(op.inputs[0]),
                                self._ptr(op.outputs[0])],
                               [rows, shape[-1]], []))
+            elif k == "concat":
+                out_shape = self.shapes[op.outputs[0]]
+                axis = p["axis"]
+                ostr = [1] * len(out_shape)
+                for d in range(len(out_shape) - 2, -1, -1):
+                    ostr[d] = ostr[d + 1] * out_shape[d + 1]
+                off_elems = 0
+                for inp in op.inputs:
+                    ish = self.shapes[inp]
+                    n_in = int(np.prod(ish)) if ish else 1
+                    calls.append((ext.K_SCATTER,
+                                  [self._ptr(inp),
+                                   self._ptr(op.outputs[0]) +
+                                   off_elems * 2],
+                                  [len(ish)] + list(ish) + ostr +
+                                  [n_in], []))
+                    off_elems += ish[axis] * ostr[axis]
             elif k == "pad":
                 shape = self.shapes[op.inputs[0]]
                 pads = p["pads"]

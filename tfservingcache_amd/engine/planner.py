@@ -428,9 +428,52 @@ class _Lowerer:
             self.ops.append(PlanOp("eltwise", [a, b_], [y],
                                    {"fn": _ELTWISE_BINARY[op]}))
             return
-        if op == "ConcatV2":
+        if op == "ConcatV2" or op == "Concat":
             self.lower_concat(nd, out)
             return
+        if op == "Pack":
+            self.lower_pack(nd, out)
+            return
+        if op == "Unpack":
+            self.lower_unpack(nd)
+            return
+        if op == "LeakyRelu":
+            x = self.tid(nd.input[0])
+            shape = self.tensors[x].shape
+            alpha = _attr_f(nd, "alpha", 0.2)
+            a_id = self.new_tensor((), "f32", "weight",
+                                   f"{nd.name}/alpha",
+                                   weight=np.float32(alpha))
+            sc = self.new_tensor(shape, "f32", "activation",
+                                 f"{nd.name}/scaled:0")
+            self.ops.append(PlanOp("eltwise", [x, a_id], [sc],
+                                   {"fn": "mul"}))
+            y = self.new_tensor(shape, "f32", "activation", out)
+            self.ops.append(PlanOp("eltwise", [x, sc], [y],
+                                   {"fn": "max"}))
+            return
+        if op == "Pow":
+            x = self.tid(nd.input[0])
+            e = self.const_value(nd.input[1])
+            ev = float(np.asarray(e).reshape(-1)[0]) if e is not None \
+                and np.asarray(e).size == 1 else None
+            shape = self.tensors[x].shape
+            if ev == 2.0:
+                y = self.new_tensor(shape, "f32", "activation", out)
+                self.ops.append(PlanOp("eltwise", [x], [y],
+                                       {"fn": "square"}))
+                return
+            if ev == 0.5:
+                y = self.new_tensor(shape, "f32", "activation", out)
+                self.ops.append(PlanOp("eltwise", [x], [y],
+                                       {"fn": "sqrt"}))
+                return
+            if ev == 1.0:
+                root = self.tensors[x]
+                self.new_tensor(root.shape, root.dtype, root.kind, out,
+                                alias_of=x)
+                return
+            raise PlanError(f"Pow exponent {ev} unsupported")
         if op == "Shape":
             # produce a const from the (possibly symbolic) shape
             shape = self.shape_of(nd.input[0])
@@ -1177,6 +1220,51 @@ class _Lowerer:
         self.ops.append(PlanOp("bn_act", [x, g_id, b_id], [y], {"act": act}))
         if out_name != out:
             self.by_name[out] = y
+
+    def lower_pack(self, nd: g.NodeDef, out: str) -> None:
+        """Pack (tf.stack): expand each input with a new unit axis
+        (reshape alias) and concat along it — reuses the validated
+        concat path."""
+        axis = _attr_i(nd, "axis", 0)
+        ids = [self.tid(r) for r in nd.input]
+        shapes = [self.tensors[i].shape for i in ids]
+        rank = len(shapes[0]) + 1
+        if axis < 0:
+            axis += rank
+        exp_ids = []
+        for i in ids:
+            sh = list(self.tensors[i].shape)
+            sh.insert(axis, 1)
+            exp_ids.append(self.new_tensor(
+                tuple(sh), "f32", "activation",
+                f"{nd.name}/exp{len(exp_ids)}", alias_of=i))
+        out_shape = list(shapes[0])
+        out_shape.insert(axis, len(ids))
+        y = self.new_tensor(tuple(out_shape), "f32", "activation", out)
+        self.ops.append(PlanOp("concat", exp_ids, [y], {"axis": axis}))
+
+    def lower_unpack(self, nd: g.NodeDef) -> None:
+        """Unpack (tf.unstack): one strided-copy per output slice."""
+        x = self.tid(nd.input[0])
+        xs = list(self.tensors[x].shape)
+        axis = _attr_i(nd, "axis", 0)
+        if axis < 0:
+            axis += len(xs)
+        if is_sym(xs[axis]):
+            raise PlanError("Unpack along the batch dim unsupported")
+        num = _attr_i(nd, "num", int(xs[axis]))
+        for j in range(num):
+            starts = [0] * len(xs)
+            steps = [1] * len(xs)
+            shrink = [False] * len(xs)
+            starts[axis] = j
+            shrink[axis] = True
+            out_dims = [d for q, d in enumerate(xs) if q != axis]
+            y = self.new_tensor(tuple(out_dims), "f32", "activation",
+                                f"{nd.name}:{j}")
+            self.ops.append(PlanOp("strided_copy", [x], [y],
+                                   {"starts": starts, "steps": steps,
+                                    "shrink": shrink}))
 
     def lower_concat(self, nd: g.NodeDef, out: str) -> None:
         axis_v = self.const_value(nd.input[-1])
