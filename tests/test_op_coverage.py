@@ -355,8 +355,7 @@ def test_pack_unpack_concat(tmp_path):
     up = gb.node("Unpack", "up", [pk], num=gb.a_int(2), T=f32,
                  axis=gb.a_int(1))
     cc = gb.node("ConcatV2", "cc",
-                 [f"{up}:0" if False else "up:0", "up:1",
-                  gb.const("ax", np.int32(1))],
+                 ["up:0", "up:1", gb.const("ax", np.int32(1))],
                  N=gb.a_int(2), T=f32)                   # [B, 8]
     gb.mark_output("packed", pk)
     gb.mark_output("cat", cc)
