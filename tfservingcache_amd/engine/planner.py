@@ -525,7 +525,8 @@ class _Lowerer:
                      if not (i in axes or (not axes and d == 1))]
         else:
             axis_v = self.const_value(nd.input[1])
-            axis = int(axis_v) if axis_v is not None else 0
+            axis = (int(np.asarray(axis_v).reshape(-1)[0])
+                    if axis_v is not None else 0)
             shape = list(xt.shape)
             if axis < 0:
                 axis += len(shape) + 1
@@ -1270,7 +1271,7 @@ class _Lowerer:
         axis_v = self.const_value(nd.input[-1])
         if axis_v is None:
             raise PlanError("dynamic ConcatV2 axis unsupported")
-        axis = int(axis_v)
+        axis = int(np.asarray(axis_v).reshape(-1)[0])
         ids = [self.tid(r) for r in nd.input[:-1]]
         shapes = [self.tensors[i].shape for i in ids]
         if axis < 0:
