@@ -13,3 +13,5 @@ if str(REPO_ROOT) not in sys.path:
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: tests that need a real MI355X GPU (run via gpurun)")
+    config.addinivalue_line(
+        "markers", "slow: long-running CPU tests (randomized end-to-end)")

@@ -1,0 +1,172 @@
+"""Hypothesis property tests over the pure-CPU subsystems.
+
+- Consistent-hash ring: minimal-disruption on member removal and the
+  distinct-replica contract of get_n (the reference leans on
+  stathat's consistent.GetN for both — cluster.go:116-130).
+- LRUCache: model-checked against an independent in-test mirror of the
+  byte-budget + MRU-order semantics (lrucache.go:43-101).
+- Planner/CPU executor: randomized MLP dimensions end-to-end vs a
+  plain numpy forward pass.
+"""
+import os
+import tempfile
+
+import numpy as np
+import pytest
+from hypothesis import given, settings, strategies as st
+
+from tfservingcache_amd.cachemanager.lrucache import LRUCache, Model
+from tfservingcache_amd.taskhandler.ring import ConsistentHashRing
+
+# ---------------------------------------------------------------------------
+# ring properties
+# ---------------------------------------------------------------------------
+
+_member_st = st.text(alphabet="abcdefgh0123", min_size=1, max_size=8).map(
+    lambda s: f"host-{s}:8093:8100:gpu0")
+_members_st = st.lists(_member_st, min_size=2, max_size=12, unique=True)
+_keys_st = st.lists(st.text(alphabet="mnopq/0123456789#", min_size=1,
+                            max_size=16),
+                    min_size=1, max_size=30)
+
+
+@given(members=_members_st, keys=_keys_st, data=st.data())
+@settings(max_examples=60, deadline=None)
+def test_ring_minimal_disruption_on_removal(members, keys, data):
+    """Removing one member must not remap keys it did not own: the
+    surviving vnodes keep their relative order, so every key whose
+    primary owner was a survivor keeps that owner."""
+    ring = ConsistentHashRing()
+    ring.set_members(members)
+    before = {k: ring.get(k) for k in keys}
+    removed = data.draw(st.sampled_from(members))
+    ring.set_members([mem for mem in members if mem != removed])
+    for k in keys:
+        if before[k] != removed:
+            assert ring.get(k) == before[k]
+        else:
+            assert ring.get(k) != removed
+
+
+@given(members=_members_st, keys=_keys_st,
+       n=st.integers(min_value=1, max_value=15))
+@settings(max_examples=60, deadline=None)
+def test_ring_get_n_distinct_and_prefix_stable(members, keys, n):
+    """get_n returns min(n, members) DISTINCT members, every one a real
+    member, and get_n(k, a) is a prefix of get_n(k, b) for a <= b (so
+    growing replicasPerModel only ADDS replica slots)."""
+    ring = ConsistentHashRing()
+    ring.set_members(members)
+    for k in keys:
+        owners = ring.get_n(k, n)
+        assert len(owners) == min(n, len(members))
+        assert len(set(owners)) == len(owners)
+        assert all(o in members for o in owners)
+        for a in range(1, len(owners) + 1):
+            assert ring.get_n(k, a) == owners[:a]
+
+
+@given(members=_members_st, keys=_keys_st)
+@settings(max_examples=40, deadline=None)
+def test_ring_assignment_is_deterministic(members, keys):
+    """Two independently-seeded rings over the same member set agree on
+    every key (routers need no coordination — taskhandler.go:84-93)."""
+    r1, r2 = ConsistentHashRing(), ConsistentHashRing()
+    r1.set_members(members)
+    r2.set_members(list(reversed(members)))
+    for k in keys:
+        assert r1.get_n(k, 3) == r2.get_n(k, 3)
+
+
+# ---------------------------------------------------------------------------
+# LRU cache model check
+# ---------------------------------------------------------------------------
+
+_op_st = st.one_of(
+    st.tuples(st.just("put"), st.integers(0, 9), st.integers(1, 50)),
+    st.tuples(st.just("get"), st.integers(0, 9), st.just(0)),
+    st.tuples(st.just("remove"), st.integers(0, 9), st.just(0)),
+)
+
+
+@given(ops=st.lists(_op_st, min_size=1, max_size=60),
+       cap=st.integers(min_value=10, max_value=120))
+@settings(max_examples=80, deadline=None)
+def test_lru_matches_reference_model(ops, cap):
+    """Replay a random op sequence against LRUCache and an independent
+    mirror of the spec: byte budget enforced by tail eviction, front =
+    MRU, replace-on-put, current_size always the sum of live entries."""
+    with tempfile.TemporaryDirectory() as td:
+        cache = LRUCache(td, cap)
+        mirror = []                      # [(name, size)] front = MRU
+
+        def msize():
+            return sum(s for _, s in mirror)
+
+        for kind, idx, size in ops:
+            name = f"m{idx}"
+            if kind == "put":
+                mirror[:] = [e for e in mirror if e[0] != name]
+                while mirror and msize() + size > cap:
+                    mirror.pop()         # evict LRU tail
+                mirror.insert(0, (name, size))
+                cache.put(Model(name=name, version=1,
+                                path=os.path.join(name, "1"),
+                                size_on_disk=size))
+            elif kind == "get":
+                got = cache.get(name, 1)
+                hit = any(e[0] == name for e in mirror)
+                assert (got is not None) == hit
+                if hit:
+                    e = next(e for e in mirror if e[0] == name)
+                    mirror.remove(e)
+                    mirror.insert(0, e)
+                    assert got.size_on_disk == e[1]
+            else:
+                removed = cache.remove(name, 1)
+                assert removed == any(e[0] == name for e in mirror)
+                mirror[:] = [e for e in mirror if e[0] != name]
+
+            assert cache.current_size == msize()
+            assert [(e.name, e.size_on_disk)
+                    for e in cache.list_models()] == mirror
+            # budget: only a single oversized entry may exceed the cap
+            assert cache.current_size <= cap or len(mirror) == 1
+
+
+# ---------------------------------------------------------------------------
+# planner/executor randomized shapes
+# ---------------------------------------------------------------------------
+
+@pytest.mark.slow
+@given(d_in=st.integers(1, 40), d_hidden=st.integers(1, 48),
+       d_out=st.integers(2, 24), batch=st.integers(1, 5),
+       seed=st.integers(0, 2 ** 16))
+@settings(max_examples=15, deadline=None)
+def test_mlp_random_dims_match_numpy(d_in, d_hidden, d_out, batch, seed):
+    """End-to-end plan+execute over arbitrary (non-tile-aligned) MLP
+    dims must match a plain numpy forward of the same weights."""
+    from tfservingcache_amd.engine.model import load_model_from_dir
+    from tfservingcache_amd.engine.savedmodel import write_saved_model
+    from tfservingcache_amd.models import build_mlp
+
+    with tempfile.TemporaryDirectory() as td:
+        d = os.path.join(td, "m", "1")
+        write_saved_model(build_mlp(d_in, d_hidden, d_out, seed=seed), d)
+        model = load_model_from_dir(d, "m", 1)
+
+        rng = np.random.default_rng(seed)
+        w1 = rng.standard_normal((d_in, d_hidden), dtype=np.float32) * 0.3
+        b1 = rng.standard_normal(d_hidden, dtype=np.float32) * 0.1
+        w2 = rng.standard_normal((d_hidden, d_out), dtype=np.float32) * 0.3
+        b2 = rng.standard_normal(d_out, dtype=np.float32) * 0.1
+
+        x = rng.standard_normal((batch, d_in), dtype=np.float32)
+        h = np.maximum(x @ w1 + b1, 0.0)
+        logits = h @ w2 + b2
+        e = np.exp(logits - logits.max(axis=-1, keepdims=True))
+        want = e / e.sum(axis=-1, keepdims=True)
+
+        out = model.predict({"x": x})
+        got = next(iter(out.values()))
+        np.testing.assert_allclose(got, want, rtol=2e-4, atol=2e-5)
