@@ -135,6 +135,126 @@ def test_lru_matches_reference_model(ops, cap):
 
 
 # ---------------------------------------------------------------------------
+# StridedSlice / Concat / Pack index math over random shapes
+# ---------------------------------------------------------------------------
+
+@st.composite
+def _slice_case(draw):
+    """A random [-1, H, W] shape plus per-dim slice specs that are
+    expressible as numpy basic indexing (positive strides; dim 0 is the
+    untouched batch dim, as the lowering requires)."""
+    h = draw(st.integers(2, 8))
+    w = draw(st.integers(2, 8))
+    specs = []
+    for size in (h, w):
+        kind = draw(st.sampled_from(["full", "range", "shrink"]))
+        if kind == "full":
+            specs.append(("full", 0, 0, draw(st.sampled_from([1, 2, 3]))))
+        elif kind == "shrink":
+            j = draw(st.integers(0, size - 1))
+            specs.append(("shrink", j, j + 1, 1))
+        else:
+            b = draw(st.integers(-size, size - 1))
+            canon = b if b >= 0 else b + size
+            e = draw(st.integers(canon + 1, size))
+            specs.append(("range", b, e, draw(st.sampled_from([1, 2, 3]))))
+    # at least one sliced dim, else the planner aliases instead
+    if all(k == "full" and s == 1 for k, _, _, s in specs):
+        specs[0] = ("shrink", 0, 1, 1)
+    return h, w, specs
+
+
+@given(case=_slice_case(), batch=st.integers(1, 4),
+       seed=st.integers(0, 999))
+@settings(max_examples=25, deadline=None)
+def test_strided_slice_random_specs(case, batch, seed):
+    from tfservingcache_amd.engine.model import load_model_from_dir
+    from tfservingcache_amd.engine.savedmodel import (GraphBuilder,
+                                                      write_saved_model)
+    h, w, specs = case
+    begin, end, strides = [0], [0], [1]
+    bm, em, shm = 1, 1, 0
+    np_index = [slice(None)]
+    for i, (kind, b, e, s) in enumerate(specs, start=1):
+        begin.append(b)
+        end.append(e)
+        strides.append(s)
+        if kind == "full":
+            bm |= 1 << i
+            em |= 1 << i
+            np_index.append(slice(None, None, s))
+        elif kind == "shrink":
+            shm |= 1 << i
+            np_index.append(b)
+        else:
+            np_index.append(slice(b, e, s))
+
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, [-1, h, w],
+                          signature_name="x")
+    ss = gb.node("StridedSlice", "ss",
+                 [x_ph,
+                  gb.const("b", np.array(begin, np.int32)),
+                  gb.const("e", np.array(end, np.int32)),
+                  gb.const("s", np.array(strides, np.int32))],
+                 T=f32, Index=gb.a_type(3),
+                 begin_mask=gb.a_int(bm), end_mask=gb.a_int(em),
+                 shrink_axis_mask=gb.a_int(shm),
+                 ellipsis_mask=gb.a_int(0), new_axis_mask=gb.a_int(0))
+    gb.mark_output("y", ss)
+
+    with tempfile.TemporaryDirectory() as td:
+        d = os.path.join(td, "m", "1")
+        write_saved_model(gb.build(), d)
+        model = load_model_from_dir(d, "m", 1)
+        x = np.random.default_rng(seed).standard_normal(
+            (batch, h, w)).astype(np.float32)
+        got = model.predict({"x": x})["y"]
+        want = x[tuple(np_index)]
+        assert got.shape == want.shape, (got.shape, want.shape)
+        np.testing.assert_allclose(got, want, rtol=1e-6)
+
+
+@given(h=st.integers(1, 6), w=st.integers(1, 6), batch=st.integers(1, 4),
+       axis=st.integers(1, 2), pack_axis=st.integers(1, 3),
+       seed=st.integers(0, 999))
+@settings(max_examples=25, deadline=None)
+def test_concat_pack_random_shapes(h, w, batch, axis, pack_axis, seed):
+    """Concat of [x, 2x] along a random non-batch axis and Pack along a
+    random axis vs numpy concatenate/stack."""
+    from tfservingcache_amd.engine.model import load_model_from_dir
+    from tfservingcache_amd.engine.savedmodel import (GraphBuilder,
+                                                      write_saved_model)
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, [-1, h, w],
+                          signature_name="x")
+    x2 = gb.node("Mul", "x2", [x_ph, gb.const("two", np.float32(2.0))],
+                 T=f32)
+    cc = gb.node("ConcatV2", "cc",
+                 [x_ph, x2, gb.const("ax", np.int32(axis))],
+                 N=gb.a_int(2), T=f32)
+    pk = gb.node("Pack", "pk", [x_ph, x2], N=gb.a_int(2), T=f32,
+                 axis=gb.a_int(pack_axis))
+    gb.mark_output("cat", cc)
+    gb.mark_output("packed", pk)
+
+    with tempfile.TemporaryDirectory() as td:
+        d = os.path.join(td, "m", "1")
+        write_saved_model(gb.build(), d)
+        model = load_model_from_dir(d, "m", 1)
+        x = np.random.default_rng(seed).standard_normal(
+            (batch, h, w)).astype(np.float32)
+        out = model.predict({"x": x})
+        np.testing.assert_allclose(
+            out["cat"], np.concatenate([x, 2 * x], axis=axis), rtol=1e-6)
+        np.testing.assert_allclose(
+            out["packed"], np.stack([x, 2 * x], axis=pack_axis),
+            rtol=1e-6)
+
+
+# ---------------------------------------------------------------------------
 # planner/executor randomized shapes
 # ---------------------------------------------------------------------------
 
