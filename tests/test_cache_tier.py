@@ -417,3 +417,84 @@ def test_rebuild_from_disk_after_restart(tmp_path):
     # without the flag the index starts empty (reference behavior)
     cache2 = LRUCache(str(cache_dir), 10 ** 9)
     assert cache2.list_models() == []
+
+
+@pytest.mark.slow
+def test_lru_thrash_soak(tmp_path):
+    """CPU miniature of the headline workload: more models than the
+    pool can hold, 8 concurrent threads, EVERY response checked against
+    that model's own numpy forward. Catches races in the single-flight
+    cold-load path under constant eviction (the reference's whole-
+    download mutex makes this trivially serial — cachemanager.go:114)."""
+    n_models, pool_cap = 10, 3
+    repo = tmp_path / "repo"
+    write_model_repo(
+        str(repo), [(f"mlp{i}", 1, "mlp") for i in range(n_models)],
+        builder_kwargs={f"mlp{i}": {"seed": i} for i in range(n_models)})
+    provider = DiskModelProvider(str(repo))
+    cache = LRUCache(str(tmp_path / "cache"), max_size_bytes=10 ** 9)
+    pool = ModelPool(make_cpu_loader(cache),
+                     max_concurrent_models=pool_cap)
+    cm = CacheManager(provider, cache, pool, model_fetch_timeout=60.0)
+    handler = LocalServingHandler(cm)
+
+    x = np.random.default_rng(99).standard_normal((2, 16)).astype(
+        np.float32)
+    want = {}
+    for i in range(n_models):
+        rng = np.random.default_rng(i)
+        w1 = rng.standard_normal((16, 32), dtype=np.float32) * 0.3
+        b1 = rng.standard_normal(32, dtype=np.float32) * 0.1
+        w2 = rng.standard_normal((32, 8), dtype=np.float32) * 0.3
+        b2 = rng.standard_normal(8, dtype=np.float32) * 0.1
+        h = np.maximum(x @ w1 + b1, 0.0)
+        logits = h @ w2 + b2
+        e = np.exp(logits - logits.max(axis=-1, keepdims=True))
+        want[f"mlp{i}"] = e / e.sum(axis=-1, keepdims=True)
+
+    order = [f"mlp{int(i)}"
+             for i in np.random.default_rng(0).integers(0, n_models, 400)]
+    it = iter(order)
+    lock = threading.Lock()
+    errors = []
+
+    def worker():
+        while True:
+            with lock:
+                name = next(it, None)
+            if name is None:
+                return
+            try:
+                req = m.PredictRequest(
+                    model_spec=m.ModelSpec(
+                        name=name, version=m.Int64Value(value=1)),
+                    inputs={"x": numpy_to_tensorproto(x)})
+                resp = handler.predict(req)
+                out = tensorproto_to_numpy(
+                    next(iter(resp.outputs.values())))
+                np.testing.assert_allclose(out, want[name],
+                                           rtol=2e-4, atol=2e-5)
+            except Exception as ex:       # noqa: BLE001
+                errors.append((name, repr(ex)))
+                return
+
+    threads = [threading.Thread(target=worker) for _ in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+    assert not errors, errors[:3]
+    # the pool may transiently exceed the cap by in-flight loads it
+    # refuses to cancel (modelpool.reload); one more request triggers a
+    # reconciling reload, after which it must converge to the cap
+    req = m.PredictRequest(
+        model_spec=m.ModelSpec(name="mlp0",
+                               version=m.Int64Value(value=1)),
+        inputs={"x": numpy_to_tensorproto(x)})
+    handler.predict(req)
+    deadline = time.time() + 10
+    while len(pool._entries) > pool_cap and \
+            time.time() < deadline:     # noqa: SLF001
+        time.sleep(0.05)
+        handler.predict(req)
+    assert len(pool._entries) <= pool_cap  # noqa: SLF001
