@@ -255,6 +255,59 @@ def test_concat_pack_random_shapes(h, w, batch, axis, pack_axis, seed):
 
 
 # ---------------------------------------------------------------------------
+# C++ dense-JSON parser (native REST fast path) vs Python json
+# ---------------------------------------------------------------------------
+
+def _probe():
+    pytest.importorskip("torch")
+    try:
+        from tfservingcache_amd.engine import _tfsc_engine as ext
+    except Exception:                    # noqa: BLE001
+        pytest.skip("engine extension not built")
+    return ext._rest_parse_probe
+
+
+@st.composite
+def _dense_array(draw):
+    shape = draw(st.lists(st.integers(1, 5), min_size=1, max_size=4))
+    n = int(np.prod(shape))
+    vals = draw(st.lists(
+        st.floats(allow_nan=False, allow_infinity=False, width=32),
+        min_size=n, max_size=n))
+    return np.array(vals, dtype=np.float32).reshape(shape)
+
+
+@given(x=_dense_array(), pretty=st.booleans())
+@settings(max_examples=60, deadline=None)
+def test_cpp_json_parser_random_floats(x, pretty):
+    """The hand-written C++ JSON number/nesting parser agrees with
+    Python json + numpy on arbitrary float32 payloads, compact or
+    whitespace-heavy."""
+    import json
+    probe = _probe()
+    body = json.dumps(x.tolist(), indent=1 if pretty else None).encode()
+    dims, vals = probe(body, False)
+    assert list(dims) == list(x.shape)
+    want = np.array(json.loads(body), dtype=np.float32).reshape(-1)
+    np.testing.assert_allclose(np.array(vals, dtype=np.float32), want,
+                               rtol=2e-6, atol=1e-37)
+
+
+@given(shape=st.lists(st.integers(1, 5), min_size=1, max_size=3),
+       seed=st.integers(0, 2 ** 31 - 1))
+@settings(max_examples=40, deadline=None)
+def test_cpp_json_parser_random_ints(shape, seed):
+    import json
+    probe = _probe()
+    rng = np.random.default_rng(seed)
+    x = rng.integers(-2 ** 31, 2 ** 31, size=shape, dtype=np.int64) \
+        .astype(np.int32)
+    dims, vals = probe(json.dumps(x.tolist()).encode(), True)
+    assert list(dims) == list(x.shape)
+    assert list(vals) == x.reshape(-1).tolist()
+
+
+# ---------------------------------------------------------------------------
 # planner/executor randomized shapes
 # ---------------------------------------------------------------------------
 
