@@ -308,6 +308,53 @@ def test_cpp_json_parser_random_ints(shape, seed):
 
 
 # ---------------------------------------------------------------------------
+# cross-codec: Python protobuf encoder -> C++ spec peek
+# ---------------------------------------------------------------------------
+
+@given(name=st.text(min_size=0, max_size=40),
+       version=st.one_of(st.none(),
+                         st.integers(0, 2 ** 62),
+                         st.sampled_from([0, 1, 127, 128, 16383, 16384,
+                                          2 ** 31 - 1, 2 ** 31, 2 ** 62])),
+       label=st.text(min_size=0, max_size=20),
+       rows=st.integers(1, 3))
+@settings(max_examples=60, deadline=None)
+def test_cpp_peek_spec_matches_python_encoder(name, version, label, rows):
+    """Two independent codec implementations must agree: requests built
+    by the Python wire encoder are peeked identically by the C++
+    fastpath parser (names/labels arbitrary unicode, versions across
+    varint-width boundaries, with a tensor payload present)."""
+    pytest.importorskip("torch")
+    try:
+        from tfservingcache_amd.engine import _tfsc_engine as ext
+    except Exception:                    # noqa: BLE001
+        pytest.skip("engine extension not built")
+    from tfservingcache_amd.wire import messages as m
+    from tfservingcache_amd.wire.tensor import numpy_to_tensorproto
+
+    spec = m.ModelSpec(
+        name=name,
+        version=(m.Int64Value(value=version)
+                 if version is not None else None),
+        version_label=label)
+    req = m.PredictRequest(
+        model_spec=spec,
+        inputs={"x": numpy_to_tensorproto(
+            np.zeros((rows, 4), dtype=np.float32))})
+    got_name, got_version, got_label = ext.peek_spec(req.encode())
+    assert got_name == name
+    assert got_label == label
+    if version is None or version == 0:
+        # version 0 encodes as an empty Int64Value submessage and means
+        # UNSET — the reference has the same quirk (its gRPC director
+        # formats GetVersion().GetValue(), so 0 == no version,
+        # tfservingproxy.go:248)
+        assert got_version is None
+    else:
+        assert got_version == version
+
+
+# ---------------------------------------------------------------------------
 # planner/executor randomized shapes
 # ---------------------------------------------------------------------------
 
