@@ -308,6 +308,58 @@ def test_cpp_json_parser_random_ints(shape, seed):
 
 
 # ---------------------------------------------------------------------------
+# Einsum trailing-K family over random equations/shapes
+# ---------------------------------------------------------------------------
+
+@st.composite
+def _einsum_case(draw):
+    """Random equation in the supported Dense-layer family: lhs =
+    prefix+K, rhs = K+suffix, out = prefix+suffix."""
+    n_pre = draw(st.integers(1, 3))
+    nk = draw(st.integers(1, 2))
+    n_suf = draw(st.integers(0, 2))
+    letters = "abcdefg"
+    pre = letters[:n_pre]
+    kk = letters[n_pre:n_pre + nk]
+    suf = letters[n_pre + nk:n_pre + nk + n_suf]
+    eq = f"{pre + kk},{kk + suf}->{pre + suf}"
+    pre_dims = [draw(st.integers(1, 4)) for _ in range(n_pre)]
+    k_dims = [draw(st.integers(1, 4)) for _ in range(nk)]
+    suf_dims = [draw(st.integers(1, 4)) for _ in range(n_suf)]
+    return eq, pre_dims, k_dims, suf_dims
+
+
+@given(case=_einsum_case(), seed=st.integers(0, 999))
+@settings(max_examples=30, deadline=None)
+def test_einsum_random_equations(case, seed):
+    from tfservingcache_amd.engine.model import load_model_from_dir
+    from tfservingcache_amd.engine.savedmodel import (GraphBuilder,
+                                                      write_saved_model)
+    eq, pre_dims, k_dims, suf_dims = case
+    rng = np.random.default_rng(seed)
+    w = (rng.standard_normal(tuple(k_dims + suf_dims)) * 0.3).astype(
+        np.float32)
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    ashape = [-1] + pre_dims[1:] + k_dims
+    x_ph = gb.placeholder("x", np.float32, ashape, signature_name="x")
+    e = gb.node("Einsum", "ein", [x_ph, gb.const("w", w)], T=f32,
+                equation=gb.a_str(eq), N=gb.a_ints([2]))
+    gb.mark_output("y", e)
+
+    with tempfile.TemporaryDirectory() as td:
+        d = os.path.join(td, "m", "1")
+        write_saved_model(gb.build(), d)
+        model = load_model_from_dir(d, "m", 1)
+        x = (rng.standard_normal(tuple(pre_dims + k_dims)) * 0.5) \
+            .astype(np.float32)
+        got = model.predict({"x": x})["y"]
+        want = np.einsum(eq, x, w)
+        assert got.shape == want.shape, (eq, got.shape, want.shape)
+        np.testing.assert_allclose(got, want, rtol=2e-4, atol=2e-5)
+
+
+# ---------------------------------------------------------------------------
 # cross-codec: Python protobuf encoder -> C++ spec peek
 # ---------------------------------------------------------------------------
 
