@@ -442,3 +442,33 @@ def test_mlp_random_dims_match_numpy(d_in, d_hidden, d_out, batch, seed):
         out = model.predict({"x": x})
         got = next(iter(out.values()))
         np.testing.assert_allclose(got, want, rtol=2e-4, atol=2e-5)
+
+
+# ---------------------------------------------------------------------------
+# config surface drift guard
+# ---------------------------------------------------------------------------
+
+def test_every_config_key_read_in_code_is_documented():
+    """Every dotted config key the package reads must appear in
+    config.yaml.example (possibly as a commented-out block) — guards
+    the README/config parity the reference promises its users
+    (README.md:27-68)."""
+    import pathlib
+    import re
+    root = pathlib.Path(__file__).resolve().parent.parent
+    keys = set()
+    for p in (root / "tfservingcache_amd").rglob("*.py"):
+        for mm in re.finditer(
+                r'get_(?:string|int|float|bool|dict|list)\(\s*f?"([^"]+)"',
+                p.read_text()):
+            if "{" not in mm.group(1):        # skip f-string templates
+                keys.add(mm.group(1))
+    doc = (root / "config.yaml.example").read_text().lower()
+    missing = []
+    for key in sorted(keys):
+        for seg in key.split("."):
+            if seg.lower() not in doc:
+                missing.append(key)
+                break
+    assert not missing, f"undocumented config keys: {missing}"
+    assert len(keys) >= 50      # the surface should not silently shrink
