@@ -472,3 +472,48 @@ def test_every_config_key_read_in_code_is_documented():
                 break
     assert not missing, f"undocumented config keys: {missing}"
     assert len(keys) >= 50      # the surface should not silently shrink
+
+
+# ---------------------------------------------------------------------------
+# DepthwiseConv2dNative over random geometry
+# ---------------------------------------------------------------------------
+
+@given(h=st.integers(3, 12), w=st.integers(3, 12),
+       c=st.sampled_from([1, 3, 4, 8]), k=st.sampled_from([1, 3, 5]),
+       stride=st.integers(1, 3),
+       padding=st.sampled_from(["SAME", "VALID"]),
+       batch=st.integers(1, 3), seed=st.integers(0, 999))
+@settings(max_examples=25, deadline=None)
+def test_depthwise_random_geometry(h, w, c, k, stride, padding, batch,
+                                   seed):
+    """DepthwiseConv2dNative lowering over random spatial sizes,
+    kernel sizes, strides and both paddings vs direct loops."""
+    from tests.test_op_coverage import _naive_depthwise
+    from tfservingcache_amd.engine.model import load_model_from_dir
+    from tfservingcache_amd.engine.savedmodel import (GraphBuilder,
+                                                      write_saved_model)
+    if padding == "VALID" and (h < k or w < k):
+        return                       # empty output; TF rejects too
+    rng = np.random.default_rng(seed)
+    wgt = (rng.standard_normal((k, k, c, 1)) * 0.3).astype(np.float32)
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("input", np.float32, [-1, h, w, c],
+                          signature_name="input")
+    d = gb.node("DepthwiseConv2dNative", "dw",
+                [x_ph, gb.const("w", wgt)], T=f32,
+                strides=gb.a_ints([1, stride, stride, 1]),
+                padding=gb.a_str(padding),
+                data_format=gb.a_str("NHWC"))
+    gb.mark_output("y", d)
+
+    with tempfile.TemporaryDirectory() as td:
+        vdir = os.path.join(td, "m", "1")
+        write_saved_model(gb.build(), vdir)
+        model = load_model_from_dir(vdir, "m", 1)
+        x = (rng.standard_normal((batch, h, w, c)) * 0.5).astype(
+            np.float32)
+        got = model.predict({"input": x})["y"]
+        want = _naive_depthwise(x, wgt, stride, padding)
+        assert got.shape == want.shape, (got.shape, want.shape)
+        np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)
