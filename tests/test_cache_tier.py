@@ -498,3 +498,27 @@ def test_lru_thrash_soak(tmp_path):
         time.sleep(0.05)
         handler.predict(req)
     assert len(pool._entries) <= pool_cap  # noqa: SLF001
+
+
+def test_shipped_testclient_runs(stack):
+    """The shipped scripts/testclient.py (parity with the reference's
+    cmd/testclient) works against a live cache gRPC server."""
+    import subprocess
+    import sys
+    from pathlib import Path
+    cm, handler = stack
+    server, _health = make_cache_grpc_server(handler)
+    port = server.add_insecure_port("127.0.0.1:0")
+    server.start()
+    try:
+        repo_root = Path(__file__).resolve().parent.parent
+        out = subprocess.run(
+            [sys.executable, str(repo_root / "scripts" / "testclient.py"),
+             f"127.0.0.1:{port}", "half_plus_two", "123"],
+            capture_output=True, text=True, timeout=60)
+        assert out.returncode == 0, out.stderr[-2000:]
+        assert "Predict output" in out.stdout
+        assert "2.5" in out.stdout and "4.5" in out.stdout
+        assert "Classify:" in out.stdout
+    finally:
+        server.stop(0)
