@@ -138,21 +138,37 @@ def clone_gpu_model(gpu_model, device: str):
     from ..engine.gpu import GpuModel
     import torch
 
+    from ..engine.gpu import _load_backend
+    _torch, ext = _load_backend()
+
     clone = GpuModel.__new__(GpuModel)
     clone.plan = gpu_model.plan
     clone.device = device
+    clone.dtype = getattr(gpu_model, "dtype", "bf16")
+    clone.model_name = getattr(gpu_model, "model_name", "")
+    clone.model_version = getattr(gpu_model, "model_version", 0)
     clone.max_batch = gpu_model.max_batch
     clone.use_graphs = gpu_model.use_graphs
     clone._contexts = {}
     clone.n_streams = getattr(gpu_model, "n_streams", 2)
+    clone._fast = ext.FastModel(clone.model_name or "model",
+                                int(clone.model_version),
+                                clone.n_streams)
     import threading
     clone._lock = threading.Lock()
     clone._released = False
+    # the clone re-derives transforms lazily from its peer-copied
+    # masters: blob-slot/arena state is deliberately NOT carried over
+    # (those views alias the SOURCE device's memory)
     with torch.cuda.device(device):
         clone._weights = {k: v.to(device, non_blocking=True)
                           for k, v in gpu_model._weights.items()}
         clone._gemm_weights = {k: v.to(device, non_blocking=True)
                                for k, v in gpu_model._gemm_weights.items()}
+        clone._gemm_weights_fp8 = {
+            k: tuple(t.to(device, non_blocking=True) for t in v)
+            for k, v in getattr(gpu_model,
+                                "_gemm_weights_fp8", {}).items()}
         clone._conv_weights = {k: v.to(device, non_blocking=True)
                                for k, v in gpu_model._conv_weights.items()}
         torch.cuda.synchronize(device)
