@@ -155,3 +155,18 @@ def test_single_flight_no_duplicate_fetch_under_waiters(tmp_path):
     assert not errs
     assert prov.loads == 1
     assert cm._in_flight == {}
+
+
+def test_warmup_corrupt_length_field(tmp_path):
+    """A warmup TFRecord with an absurd length field must be treated as
+    corrupt, not allocated."""
+    from tfservingcache_amd.engine.warmup import read_tfrecords
+    import struct
+    p = tmp_path / "warmup"
+    p.write_bytes(struct.pack("<Q", 1 << 40) + b"\0" * 12)
+    assert list(read_tfrecords(str(p))) == []
+    # a sane record still reads
+    rec = b"hello"
+    p.write_bytes(struct.pack("<Q", len(rec)) + b"\0" * 4 + rec +
+                  b"\0" * 4)
+    assert list(read_tfrecords(str(p))) == [rec]

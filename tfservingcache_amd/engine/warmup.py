@@ -62,6 +62,10 @@ def read_tfrecords(path: str) -> Iterator[bytes]:
             if len(header) < 8:
                 return
             (length,) = struct.unpack("<Q", header)
+            if length > 1 << 30:        # corrupt length field: stop
+                log.warning("warmup record length %d exceeds 1 GiB; "
+                            "treating file as corrupt", length)
+                return
             f.read(4)                   # length crc (unverified)
             data = f.read(length)
             if len(data) < length:
