@@ -15,6 +15,13 @@ import numpy as np
 import pytest
 from hypothesis import given, settings, strategies as st
 
+# TFSC_HYP_SCALE=N multiplies every max_examples (deep fuzz runs)
+_SCALE = float(os.environ.get("TFSC_HYP_SCALE", "1"))
+
+
+def _ex(n):
+    return max(1, int(n * _SCALE))
+
 from tfservingcache_amd.cachemanager.lrucache import LRUCache, Model
 from tfservingcache_amd.taskhandler.ring import ConsistentHashRing
 
@@ -31,7 +38,7 @@ _keys_st = st.lists(st.text(alphabet="mnopq/0123456789#", min_size=1,
 
 
 @given(members=_members_st, keys=_keys_st, data=st.data())
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=_ex(60), deadline=None)
 def test_ring_minimal_disruption_on_removal(members, keys, data):
     """Removing one member must not remap keys it did not own: the
     surviving vnodes keep their relative order, so every key whose
@@ -50,7 +57,7 @@ def test_ring_minimal_disruption_on_removal(members, keys, data):
 
 @given(members=_members_st, keys=_keys_st,
        n=st.integers(min_value=1, max_value=15))
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=_ex(60), deadline=None)
 def test_ring_get_n_distinct_and_prefix_stable(members, keys, n):
     """get_n returns min(n, members) DISTINCT members, every one a real
     member, and get_n(k, a) is a prefix of get_n(k, b) for a <= b (so
@@ -67,7 +74,7 @@ def test_ring_get_n_distinct_and_prefix_stable(members, keys, n):
 
 
 @given(members=_members_st, keys=_keys_st)
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=_ex(40), deadline=None)
 def test_ring_assignment_is_deterministic(members, keys):
     """Two independently-seeded rings over the same member set agree on
     every key (routers need no coordination — taskhandler.go:84-93)."""
@@ -91,7 +98,7 @@ _op_st = st.one_of(
 
 @given(ops=st.lists(_op_st, min_size=1, max_size=60),
        cap=st.integers(min_value=10, max_value=120))
-@settings(max_examples=80, deadline=None)
+@settings(max_examples=_ex(80), deadline=None)
 def test_lru_matches_reference_model(ops, cap):
     """Replay a random op sequence against LRUCache and an independent
     mirror of the spec: byte budget enforced by tail eviction, front =
@@ -166,7 +173,7 @@ def _slice_case(draw):
 
 @given(case=_slice_case(), batch=st.integers(1, 4),
        seed=st.integers(0, 999))
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=_ex(25), deadline=None)
 def test_strided_slice_random_specs(case, batch, seed):
     from tfservingcache_amd.engine.model import load_model_from_dir
     from tfservingcache_amd.engine.savedmodel import (GraphBuilder,
@@ -219,7 +226,7 @@ def test_strided_slice_random_specs(case, batch, seed):
 @given(h=st.integers(1, 6), w=st.integers(1, 6), batch=st.integers(1, 4),
        axis=st.integers(1, 2), pack_axis=st.integers(1, 3),
        seed=st.integers(0, 999))
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=_ex(25), deadline=None)
 def test_concat_pack_random_shapes(h, w, batch, axis, pack_axis, seed):
     """Concat of [x, 2x] along a random non-batch axis and Pack along a
     random axis vs numpy concatenate/stack."""
@@ -278,7 +285,7 @@ def _dense_array(draw):
 
 
 @given(x=_dense_array(), pretty=st.booleans())
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=_ex(60), deadline=None)
 def test_cpp_json_parser_random_floats(x, pretty):
     """The hand-written C++ JSON number/nesting parser agrees with
     Python json + numpy on arbitrary float32 payloads, compact or
@@ -295,7 +302,7 @@ def test_cpp_json_parser_random_floats(x, pretty):
 
 @given(shape=st.lists(st.integers(1, 5), min_size=1, max_size=3),
        seed=st.integers(0, 2 ** 31 - 1))
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=_ex(40), deadline=None)
 def test_cpp_json_parser_random_ints(shape, seed):
     import json
     probe = _probe()
@@ -330,7 +337,7 @@ def _einsum_case(draw):
 
 
 @given(case=_einsum_case(), seed=st.integers(0, 999))
-@settings(max_examples=30, deadline=None)
+@settings(max_examples=_ex(30), deadline=None)
 def test_einsum_random_equations(case, seed):
     from tfservingcache_amd.engine.model import load_model_from_dir
     from tfservingcache_amd.engine.savedmodel import (GraphBuilder,
@@ -370,7 +377,7 @@ def test_einsum_random_equations(case, seed):
                                           2 ** 31 - 1, 2 ** 31, 2 ** 62])),
        label=st.text(min_size=0, max_size=20),
        rows=st.integers(1, 3))
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=_ex(60), deadline=None)
 def test_cpp_peek_spec_matches_python_encoder(name, version, label, rows):
     """Two independent codec implementations must agree: requests built
     by the Python wire encoder are peeked identically by the C++
@@ -414,7 +421,7 @@ def test_cpp_peek_spec_matches_python_encoder(name, version, label, rows):
 @given(d_in=st.integers(1, 40), d_hidden=st.integers(1, 48),
        d_out=st.integers(2, 24), batch=st.integers(1, 5),
        seed=st.integers(0, 2 ** 16))
-@settings(max_examples=15, deadline=None)
+@settings(max_examples=_ex(15), deadline=None)
 def test_mlp_random_dims_match_numpy(d_in, d_hidden, d_out, batch, seed):
     """End-to-end plan+execute over arbitrary (non-tile-aligned) MLP
     dims must match a plain numpy forward of the same weights."""
@@ -483,7 +490,7 @@ def test_every_config_key_read_in_code_is_documented():
        stride=st.integers(1, 3),
        padding=st.sampled_from(["SAME", "VALID"]),
        batch=st.integers(1, 3), seed=st.integers(0, 999))
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=_ex(25), deadline=None)
 def test_depthwise_random_geometry(h, w, c, k, stride, padding, batch,
                                    seed):
     """DepthwiseConv2dNative lowering over random spatial sizes,
