@@ -6,6 +6,14 @@ import numpy as np
 import pytest
 from hypothesis import given, settings, strategies as st
 
+import os
+# TFSC_HYP_SCALE=N multiplies every max_examples (deep fuzz runs)
+_SCALE = float(os.environ.get("TFSC_HYP_SCALE", "1"))
+
+
+def _ex(n):
+    return max(1, int(n * _SCALE))
+
 from tfservingcache_amd.wire import messages as m
 from tfservingcache_amd.wire.pb import Message
 from tfservingcache_amd.wire.tensor import (numpy_to_tensorproto,
@@ -16,7 +24,7 @@ names = st.text(
     max_size=40)
 
 
-@settings(deadline=None, max_examples=150)
+@settings(deadline=None, max_examples=_ex(150))
 @given(name=names, version=st.integers(0, 2 ** 62),
        label=names, sig=names)
 def test_model_spec_round_trip(name, version, label, sig):
@@ -29,7 +37,7 @@ def test_model_spec_round_trip(name, version, label, sig):
     assert got.signature_name == sig
 
 
-@settings(deadline=None, max_examples=100)
+@settings(deadline=None, max_examples=_ex(100))
 @given(data=st.binary(max_size=400))
 def test_decoders_never_crash_on_garbage(data):
     """Arbitrary bytes: decode/peek either succeed or raise a clean
@@ -42,7 +50,7 @@ def test_decoders_never_crash_on_garbage(data):
             pass
 
 
-@settings(deadline=None, max_examples=60)
+@settings(deadline=None, max_examples=_ex(60))
 @given(shape=st.lists(st.integers(1, 6), min_size=0, max_size=4),
        dtype=st.sampled_from([np.float32, np.int32, np.int64,
                               np.float64]),
@@ -58,7 +66,7 @@ def test_tensor_round_trip(shape, dtype, splat):
     np.testing.assert_array_equal(out, arr)
 
 
-@settings(deadline=None, max_examples=80)
+@settings(deadline=None, max_examples=_ex(80))
 @given(inputs=st.dictionaries(
     names.filter(bool),
     st.lists(st.floats(-1e6, 1e6, width=32), min_size=1, max_size=8),
@@ -80,7 +88,7 @@ def test_predict_request_round_trip(inputs, name):
     assert peek.name == name
 
 
-@settings(deadline=None, max_examples=60)
+@settings(deadline=None, max_examples=_ex(60))
 @given(data=st.binary(max_size=200))
 def test_unknown_fields_preserved(data):
     """A message with unknown trailing fields re-encodes them verbatim
@@ -99,7 +107,7 @@ def test_unknown_fields_preserved(data):
     assert bytes(data) in bytes(re)
 
 
-@settings(deadline=None, max_examples=80)
+@settings(deadline=None, max_examples=_ex(80))
 @given(body=st.recursive(
     st.one_of(st.none(), st.booleans(), st.integers(-10, 10),
               st.floats(-100, 100, allow_nan=False), st.text(max_size=8)),
@@ -121,7 +129,7 @@ def test_rest_codec_never_crashes_on_garbage(body):
         pass
 
 
-@settings(deadline=None, max_examples=60)
+@settings(deadline=None, max_examples=_ex(60))
 @given(rows=st.integers(1, 5), cols=st.integers(1, 4),
        columnar=st.booleans())
 def test_rest_codec_round_trip(rows, cols, columnar):
