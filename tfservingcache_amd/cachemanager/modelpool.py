@@ -196,6 +196,9 @@ class ModelPool:
                 e.state = AVAILABLE
                 e.load_finished = time.monotonic()
                 self.load_durations.append(e.load_finished - e.load_started)
+                if len(self.load_durations) > 200_000:
+                    # long-lived servers: keep the recent half
+                    del self.load_durations[:100_000]
             avail = [x for x in self._entries.values()
                      if x.state == AVAILABLE and x.model is not None]
             mt.engine_pool_models.labels(self.device).set(len(avail))
