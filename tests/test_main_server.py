@@ -221,3 +221,22 @@ def test_logging_config(capsys):
     assert logging.getLogger().level == logging.WARNING
     # restore defaults for other tests
     setup_logging(Config({}))
+
+
+def test_example_config_parses_and_reads():
+    """config.yaml.example ships as the Docker image's default config —
+    it must load through Config and yield the documented values."""
+    from pathlib import Path
+    from tfservingcache_amd.config import Config
+    path = Path(__file__).resolve().parent.parent / "config.yaml.example"
+    cfg = Config.load(str(path))
+    assert cfg.get_int("proxyRestPort") == 8093
+    assert cfg.get_int("cacheGrpcPort") == 8095
+    assert cfg.get_string("modelProvider.type") == "diskProvider"
+    assert cfg.get_int("serving.maxConcurrentModels") == 16
+    assert cfg.get_int("serving.grpcMaxMsgSize") == 16 * 1024 * 1024
+    assert cfg.get_int("engine.maxBatch") == 64
+    assert cfg.get_int("proxy.replicasPerModel") == 2
+    assert cfg.get_string("healthProbe.modelName") == \
+        "__TFSERVINGCACHE_PROBE_CHECK__"
+    assert cfg.get_string("logging.level") == "info"
