@@ -485,18 +485,15 @@ def test_lru_thrash_soak(tmp_path):
         t.join(timeout=120)
     assert not errors, errors[:3]
     # the pool may transiently exceed the cap by in-flight loads it
-    # refuses to cancel (modelpool.reload); one more request triggers a
-    # reconciling reload, after which it must converge to the cap
-    req = m.PredictRequest(
-        model_spec=m.ModelSpec(name="mlp0",
-                               version=m.Int64Value(value=1)),
-        inputs={"x": numpy_to_tensorproto(x)})
-    handler.predict(req)
-    deadline = time.time() + 10
-    while len(pool._entries) > pool_cap and \
-            time.time() < deadline:     # noqa: SLF001
+    # refuses to cancel (modelpool.reload). A hit-path request never
+    # reloads, so drive reconciliation directly: once the in-flight
+    # loads land, a declarative reload must converge to the cap
+    deadline = time.time() + 20
+    while time.time() < deadline:
+        cm._reload_pool()               # noqa: SLF001
+        if len(pool._entries) <= pool_cap:  # noqa: SLF001
+            break
         time.sleep(0.05)
-        handler.predict(req)
     assert len(pool._entries) <= pool_cap  # noqa: SLF001
 
 
