@@ -385,3 +385,53 @@ def test_leaky_relu_and_pow(tmp_path):
     np.testing.assert_allclose(out["lr"], np.where(x > 0, x, 0.1 * x),
                                rtol=1e-6)
     np.testing.assert_allclose(out["p2"], x * x, rtol=1e-6)
+
+
+@pytest.mark.parametrize("mode,stride", [("avg", 1), ("avg", 2),
+                                         ("max", 2)])
+def test_pool_same_padding_tf_semantics(tmp_path, mode, stride):
+    """SAME-padded pooling: TF's AvgPool divides by the VALID cell
+    count only (excludes padding) — torch.nn.AvgPool2d with
+    count_include_pad=False emulates that; MaxPool is unaffected.
+    The GPU kernels already count valid cells (k_pool); this pins the
+    CPU reference to the same semantics."""
+    torch = pytest.importorskip("torch")
+    rng = np.random.default_rng(7)
+    H = W = 7
+    C = 4
+    k = 3
+    gb = GraphBuilder()
+    f32 = gb.a_type(1)
+    x_ph = gb.placeholder("x", np.float32, [-1, H, W, C],
+                          signature_name="x")
+    op = "AvgPool" if mode == "avg" else "MaxPool"
+    pl = gb.node(op, "pl", [x_ph], T=f32,
+                 ksize=gb.a_ints([1, k, k, 1]),
+                 strides=gb.a_ints([1, stride, stride, 1]),
+                 padding=gb.a_str("SAME"),
+                 data_format=gb.a_str("NHWC"))
+    gb.mark_output("y", pl)
+    model = _load(tmp_path, gb.build())
+
+    x = rng.standard_normal((2, H, W, C)).astype(np.float32)
+    got = model.predict({"x": x})["y"]
+
+    xt = torch.from_numpy(x).permute(0, 3, 1, 2)    # NCHW
+    # TF SAME padding for H=7,k=3: pad 1 on each side (stride 1) or
+    # asymmetric (stride 2: Ho=4, pad_h=2 -> 1+1)
+    Ho = -(-H // stride)
+    ph = max((Ho - 1) * stride + k - H, 0)
+    pt_, pb_ = ph // 2, ph - ph // 2
+    if mode == "avg":
+        # emulate asymmetric TF padding with symmetric torch pad when
+        # equal; here ph is even for these cases -> pt_ == pb_
+        assert pt_ == pb_
+        pool = torch.nn.AvgPool2d(k, stride=stride, padding=pt_,
+                                  count_include_pad=False)
+        want = pool(xt)
+    else:
+        pool = torch.nn.MaxPool2d(k, stride=stride, padding=pt_)
+        want = pool(xt)
+    want = want.permute(0, 2, 3, 1).numpy()
+    assert got.shape == want.shape, (got.shape, want.shape)
+    np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-6)

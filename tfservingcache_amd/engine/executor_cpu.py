@@ -283,5 +283,14 @@ class CpuExecutor:
         if p["mode"] == "max":
             out = stack.max(axis=0)
         else:
-            out = stack.mean(axis=0)
+            # TF AvgPool excludes padded cells from the divisor (the
+            # GPU kernels count valid cells the same way)
+            ones = np.pad(np.ones((1, x.shape[1], x.shape[2], 1),
+                                  dtype=np.float32),
+                          ((0, 0), (pt, pb), (pl, pr), (0, 0)))
+            cnt = np.zeros((1, Ho, Wo, 1), dtype=np.float32)
+            for i in range(kh):
+                for j in range(kw):
+                    cnt += ones[:, i:i + sh * Ho:sh, j:j + sw * Wo:sw, :]
+            out = stack.sum(axis=0) / np.maximum(cnt, 1.0)
         vals[op.outputs[0]] = out
