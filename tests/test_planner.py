@@ -137,3 +137,41 @@ def test_attention_fusion(tmp_path):
                                rtol=1e-5, atol=1e-6)
     np.testing.assert_allclose(a["pooled_output"], b["pooled_output"],
                                rtol=1e-5, atol=1e-6)
+
+
+def test_nchw_graphs_rejected_loudly(tmp_path):
+    """NCHW exports must raise PlanError, never silently compute NHWC
+    math (bias/BN/pool would broadcast along the wrong axis)."""
+    import numpy as np
+    from tfservingcache_amd.engine.planner import PlanError
+    from tfservingcache_amd.engine.savedmodel import GraphBuilder
+
+    def build(op):
+        gb = GraphBuilder()
+        f32 = gb.a_type(1)
+        x = gb.placeholder("x", np.float32, [-1, 4, 6, 6],
+                           signature_name="x")
+        if op == "BiasAdd":
+            nd = gb.node("BiasAdd", "y",
+                         [x, gb.const("b", np.zeros(4, np.float32))],
+                         T=f32, data_format=gb.a_str("NCHW"))
+        elif op == "MaxPool":
+            nd = gb.node("MaxPool", "y", [x], T=f32,
+                         ksize=gb.a_ints([1, 1, 2, 2]),
+                         strides=gb.a_ints([1, 1, 2, 2]),
+                         padding=gb.a_str("VALID"),
+                         data_format=gb.a_str("NCHW"))
+        else:
+            c = np.ones(4, np.float32)
+            nd = gb.node("FusedBatchNormV3", "y",
+                         [x, gb.const("s", c), gb.const("o", c * 0),
+                          gb.const("mn", c * 0), gb.const("vr", c)],
+                         T=f32, U=f32, epsilon=gb.a_float(1e-3),
+                         is_training=gb.a_bool(False),
+                         data_format=gb.a_str("NCHW"))
+        gb.mark_output("y", nd)
+        return gb.build()
+
+    for op in ("BiasAdd", "MaxPool", "FusedBatchNormV3"):
+        with pytest.raises(PlanError):
+            _load(tmp_path, build(op), name=f"nchw_{op.lower()}")

@@ -382,6 +382,8 @@ class _Lowerer:
             return
         if op == "BiasAdd":
             # unfused BiasAdd (producer wasn't conv/matmul)
+            if _attr_s(nd, "data_format", "NHWC") != "NHWC":
+                raise PlanError("only NHWC BiasAdd supported")
             x = self.tid(nd.input[0])
             b_id = self.tid(nd.input[1])
             shape = self.tensors[x].shape
@@ -549,11 +551,13 @@ class _Lowerer:
                 break
             if nxt.op == "BiasAdd" and bias is None and bn is None and \
                     residual is None and act == ACT_NONE and \
+                    _attr_s(nxt, "data_format", "NHWC") == "NHWC" and \
                     self.const_value(nxt.input[1]) is not None:
                 bias = np.asarray(self.weight_of(nxt.input[1]), dtype=np.float32)
             elif nxt.op in ("FusedBatchNorm", "FusedBatchNormV3") and \
                     bn is None and residual is None and act == ACT_NONE and \
                     not _attr_b(nxt, "is_training", False) and \
+                    _attr_s(nxt, "data_format", "NHWC") == "NHWC" and \
                     all(self.const_value(r) is not None for r in nxt.input[1:5]):
                 bn = tuple(np.asarray(self.weight_of(r), dtype=np.float32)
                            for r in nxt.input[1:5]) + \
@@ -1108,6 +1112,8 @@ class _Lowerer:
                                {"trans_a": ta, "trans_b": tb}))
 
     def lower_pool(self, nd: g.NodeDef, out: str) -> None:
+        if _attr_s(nd, "data_format", "NHWC") != "NHWC":
+            raise PlanError(f"only NHWC {nd.op} supported")
         x = self.tid(nd.input[0])
         ks = _attr_ints(nd, "ksize")
         st = _attr_ints(nd, "strides")
@@ -1197,6 +1203,8 @@ class _Lowerer:
     def lower_batchnorm_standalone(self, nd: g.NodeDef, out: str) -> None:
         """FusedBatchNorm whose producer isn't a conv — lower to the fused
         scale/shift (+ optional downstream relu) kernel."""
+        if _attr_s(nd, "data_format", "NHWC") != "NHWC":
+            raise PlanError("only NHWC FusedBatchNorm supported")
         x = self.tid(nd.input[0])
         scale, offset, mean, var = (np.asarray(self.weight_of(r),
                                                dtype=np.float32)
