@@ -175,3 +175,54 @@ def test_nchw_graphs_rejected_loudly(tmp_path):
     for op in ("BiasAdd", "MaxPool", "FusedBatchNormV3"):
         with pytest.raises(PlanError):
             _load(tmp_path, build(op), name=f"nchw_{op.lower()}")
+
+
+def test_unsupported_attrs_rejected_loudly(tmp_path):
+    """Attributes whose silent omission would change numerics must
+    raise PlanError: dilated convs, non-axis-0 Gather, PadV2 with a
+    non-zero constant."""
+    import numpy as np
+    from tfservingcache_amd.engine.planner import PlanError
+    from tfservingcache_amd.engine.savedmodel import GraphBuilder
+
+    def dilated_conv():
+        gb = GraphBuilder()
+        f32 = gb.a_type(1)
+        x = gb.placeholder("x", np.float32, [-1, 8, 8, 3],
+                           signature_name="x")
+        w = gb.const("w", np.zeros((3, 3, 3, 4), np.float32))
+        nd = gb.node("Conv2D", "y", [x, w], T=f32,
+                     strides=gb.a_ints([1, 1, 1, 1]),
+                     dilations=gb.a_ints([1, 2, 2, 1]),
+                     padding=gb.a_str("SAME"),
+                     data_format=gb.a_str("NHWC"))
+        gb.mark_output("y", nd)
+        return gb.build()
+
+    def gather_axis1():
+        gb = GraphBuilder()
+        f32 = gb.a_type(1)
+        t = gb.const("t", np.zeros((5, 7), np.float32))
+        idx = gb.placeholder("i", np.int32, [-1], signature_name="i")
+        nd = gb.node("GatherV2", "y",
+                     [t, idx, gb.const("ax", np.int32(1))],
+                     Tparams=f32, Tindices=gb.a_type(3),
+                     Taxis=gb.a_type(3))
+        gb.mark_output("y", nd)
+        return gb.build()
+
+    def padv2_nonzero():
+        gb = GraphBuilder()
+        f32 = gb.a_type(1)
+        x = gb.placeholder("x", np.float32, [-1, 4], signature_name="x")
+        nd = gb.node("PadV2", "y",
+                     [x, gb.const("p", np.array([[0, 0], [1, 1]],
+                                                np.int32)),
+                      gb.const("c", np.float32(-1.0))], T=f32)
+        gb.mark_output("y", nd)
+        return gb.build()
+
+    for i, build in enumerate((dilated_conv, gather_axis1,
+                               padv2_nonzero)):
+        with pytest.raises(PlanError):
+            _load(tmp_path, build(), name=f"bad_{i}")

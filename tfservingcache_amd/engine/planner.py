@@ -594,6 +594,9 @@ class _Lowerer:
         if _attr_s(nd, "data_format", "NHWC") != "NHWC":
             raise PlanError("only NHWC Conv2D supported")
         w = np.asarray(self.weight_of(w_ref), dtype=np.float32)  # [R,S,Cin,K]
+        dil = _attr_ints(nd, "dilations") or [1, 1, 1, 1]
+        if any(d != 1 for d in dil):
+            raise PlanError(f"dilated Conv2D unsupported ({dil})")
         strides = _attr_ints(nd, "strides") or [1, 1, 1, 1]
         padding = _attr_s(nd, "padding", "SAME")
         xs = self.tensors[x].shape                      # [B,H,W,C]
@@ -656,6 +659,9 @@ class _Lowerer:
         if M != 1:
             raise PlanError(
                 f"depth_multiplier {M} unsupported (only 1)")
+        dil = _attr_ints(nd, "dilations") or [1, 1, 1, 1]
+        if any(d != 1 for d in dil):
+            raise PlanError(f"dilated depthwise conv unsupported ({dil})")
         strides = _attr_ints(nd, "strides") or [1, 1, 1, 1]
         padding = _attr_s(nd, "padding", "SAME")
         xs = self.tensors[x].shape
@@ -1168,6 +1174,10 @@ class _Lowerer:
         pads_v = self.const_value(nd.input[1])
         if pads_v is None:
             raise PlanError("dynamic Pad unsupported")
+        if len(nd.input) > 2:           # PadV2 constant_values input
+            cv = self.const_value(nd.input[2])
+            if cv is None or float(np.asarray(cv).reshape(-1)[0]) != 0.0:
+                raise PlanError("PadV2 with non-zero constant unsupported")
         pads = pads_v.astype(int)
         xs = list(self.tensors[x].shape)
         shape = []
@@ -1192,6 +1202,12 @@ class _Lowerer:
         self.ops.append(PlanOp("transpose", [x], [y], {"perm": perm}))
 
     def lower_gather(self, nd: g.NodeDef, out: str) -> None:
+        if len(nd.input) > 2:           # GatherV2 axis input
+            axis_v = self.const_value(nd.input[2])
+            if axis_v is None or int(np.asarray(axis_v).reshape(-1)[0]) != 0:
+                raise PlanError("only axis-0 Gather supported")
+        if _attr_i(nd, "batch_dims", 0) != 0:
+            raise PlanError("batched Gather unsupported")
         table = self.tid(nd.input[0])
         idx = self.tid(nd.input[1])
         ts = self.tensors[table].shape
