@@ -519,3 +519,36 @@ def test_shipped_testclient_runs(stack):
         assert "Classify:" in out.stdout
     finally:
         server.stop(0)
+
+
+async def test_rest_label_routing(stack, aiohttp_client=None):
+    """TF Serving's /labels/<label> REST grammar (beyond the
+    reference's /versions-only regex): the label resolves through the
+    ReloadConfig version_labels map."""
+    cm, handler = stack
+    handler.handle_reload_config(m.ReloadConfigRequest(
+        config=m.ModelServerConfig(
+            model_config_list=m.ModelConfigList(config=[m.ModelConfig(
+                name="half_plus_two", base_path="/x",
+                model_version_policy=m.ServableVersionPolicy(
+                    specific=m.ServableVersionPolicySpecific(
+                        versions=[123])),
+                version_labels={"canary": 123})]))))
+    app = make_cache_rest_app(handler, metrics_render=mt.render)
+    server = TestServer(app)
+    client = TestClient(server)
+    await client.start_server()
+    try:
+        resp = await client.post(
+            "/v1/models/half_plus_two/labels/canary:predict",
+            json={"instances": [1.0, 2.0, 5.0]})
+        assert resp.status == 200
+        body = await resp.json()
+        assert body == {"predictions": [2.5, 3.0, 4.5]}
+        # unknown label -> error, not silently latest
+        resp = await client.post(
+            "/v1/models/half_plus_two/labels/nope:predict",
+            json={"instances": [1.0]})
+        assert resp.status in (400, 404)
+    finally:
+        await client.close()

@@ -35,7 +35,7 @@ log = logging.getLogger("tfsc.rest")
 
 URL_RE = re.compile(
     r"^/v1/models/(?P<modelName>[^/:]+)"
-    r"(/versions/(?P<version>[0-9]+))?"
+    r"(/versions/(?P<version>[0-9]+)|/labels/(?P<label>[^/:]+))?"
     r"(?P<rest>.*)$", re.IGNORECASE)
 
 _CODE_TO_HTTP = {
@@ -51,13 +51,17 @@ def _error_response(msg: str, code: int = 400) -> web.Response:
 
 
 def parse_model_url(path: str):
-    """Returns (model_name, version, verb) or None. verb in
-    {'predict','classify','regress','metadata',''(status)}."""
+    """Returns (model_name, version, label, verb) or None. verb in
+    {'predict','classify','regress','metadata',''(status)}. The
+    reference's grammar has /versions/N only (tfservingproxy.go:24);
+    /labels/<label> is TF Serving's own REST grammar, resolved through
+    the ReloadConfig version_labels map."""
     match = URL_RE.match(path)
     if not match:
         return None
     name = match.group("modelName")
     version = int(match.group("version") or 0)
+    label = match.group("label") or ""
     rest = match.group("rest") or ""
     verb = ""
     if rest.startswith(":"):
@@ -66,7 +70,7 @@ def parse_model_url(path: str):
         verb = "metadata"
     elif rest not in ("", "/"):
         return None
-    return name, version, verb
+    return name, version, label, verb
 
 
 # ---------------------------------------------------------------------------
@@ -84,24 +88,25 @@ def make_cache_rest_app(handler: LocalServingHandler,
             mt.proxy_requests_failed.labels("http").inc()
             return _error_response(
                 f"Malformed url: {request.path}", 404)
-        name, version, verb = parsed
+        name, version, label, verb = parsed
         loop = asyncio.get_running_loop()
         try:
             if verb == "predict":
                 body = await request.read()
                 return web.json_response(await loop.run_in_executor(
-                    None, _predict_sync, handler, name, version, body))
+                    None, _predict_sync, handler, name, version, body,
+                    label))
             if verb in ("classify", "regress"):
                 body = await request.read()
                 return web.json_response(await loop.run_in_executor(
                     None, _classify_regress_sync, handler, name, version,
-                    verb, body))
+                    verb, body, label))
             if verb == "metadata":
                 return web.json_response(await loop.run_in_executor(
-                    None, _metadata_sync, handler, name, version))
+                    None, _metadata_sync, handler, name, version, label))
             if verb == "":
                 return web.json_response(await loop.run_in_executor(
-                    None, _status_sync, handler, name, version))
+                    None, _status_sync, handler, name, version, label))
             return _error_response(f"unsupported method :{verb}", 400)
         except ServingError as e:
             mt.proxy_requests_failed.labels("http").inc()
@@ -127,7 +132,8 @@ def make_cache_rest_app(handler: LocalServingHandler,
     return app
 
 
-def _predict_sync(handler, name, version, body: bytes) -> dict:
+def _predict_sync(handler, name, version, body: bytes,
+                  label: str = "") -> dict:
     mt.proxy_requests_total.labels("http").inc()
     try:
         payload = json.loads(body or b"{}")
@@ -137,12 +143,13 @@ def _predict_sync(handler, name, version, body: bytes) -> dict:
         inputs, fmt, _sig = parse_predict_body(payload)
     except RestCodecError as e:
         raise ServingError(str(e))
-    outputs, _version = handler.predict_arrays(name, version, inputs)
+    outputs, _version = handler.predict_arrays(
+        name, version, inputs, version_label=label)
     return render_predict_response(outputs, fmt)
 
 
 def _classify_regress_sync(handler, name, version, verb,
-                           body: bytes) -> dict:
+                           body: bytes, label: str = "") -> dict:
     mt.proxy_requests_total.labels("http").inc()
     try:
         payload = json.loads(body or b"{}")
@@ -161,7 +168,8 @@ def _classify_regress_sync(handler, name, version, verb,
     feats = {k: np.asarray(v, dtype=np.float32) for k, v in cols.items()}
     if len(feats) == 1:
         feats = {"": next(iter(feats.values()))}
-    outputs, _version = handler.predict_arrays(name, version, feats)
+    outputs, _version = handler.predict_arrays(
+        name, version, feats, version_label=label)
     vals = next(iter(outputs.values()))
     if verb == "regress":
         return {"results": np.asarray(vals, dtype=np.float64).reshape(-1)
@@ -171,10 +179,12 @@ def _classify_regress_sync(handler, name, version, verb,
         [[str(i), float(s)] for i, s in enumerate(row)] for row in scores]}
 
 
-def _status_sync(handler, name, version) -> dict:
+def _status_sync(handler, name, version, label: str = "") -> dict:
     mt.proxy_requests_total.labels("http").inc()
     req = m.GetModelStatusRequest(model_spec=m.ModelSpec(
-        name=name, version=m.Int64Value(value=version) if version else None))
+        name=name,
+        version=m.Int64Value(value=version) if version else None,
+        version_label=label))
     resp = handler.get_model_status(req)
     return {"model_version_status": [
         {"version": str(s.version),
@@ -186,13 +196,15 @@ def _status_sync(handler, name, version) -> dict:
         for s in resp.model_version_status]}
 
 
-def _metadata_sync(handler, name, version) -> dict:
+def _metadata_sync(handler, name, version, label: str = "") -> dict:
     mt.proxy_requests_total.labels("http").inc()
     req = m.GetModelMetadataRequest(model_spec=m.ModelSpec(
-        name=name, version=m.Int64Value(value=version) if version else None),
+        name=name,
+        version=m.Int64Value(value=version) if version else None,
+        version_label=label),
         metadata_field=["signature_def"])
     resp = handler.get_model_metadata(req)
-    model, _v = handler.get_model(name, version)
+    model, _v = handler.get_model(name, version, version_label=label)
     sig = model.signature_def
     sig_json = {}
     if sig is not None:
@@ -236,7 +248,7 @@ def make_proxy_rest_app(pick_node: Callable[[str, int], str],
         if parsed is None:
             mt.proxy_requests_failed.labels("http").inc()
             return _error_response(f"Malformed url: {request.path}", 404)
-        name, version, _verb = parsed
+        name, version, _label, _verb = parsed
         try:
             target = pick_node(name, version)
         except Exception as e:      # noqa: BLE001
@@ -295,9 +307,10 @@ def make_rest_dispatcher(handler: LocalServingHandler,
                 mt.proxy_requests_failed.labels("http").inc()
                 return (404, "application/json", json.dumps(
                     {"error": f"Malformed url: {path}"}).encode())
-            name, version, verb = parsed
+            name, version, label, verb = parsed
             if verb == "predict":
-                payload = _predict_sync(handler, name, version, body)
+                payload = _predict_sync(handler, name, version, body,
+                                        label)
             elif verb in ("classify", "regress"):
                 payload = _classify_regress_sync(handler, name, version,
                                                  verb, body)
