@@ -116,3 +116,15 @@ def test_plane_service_world1(tmp_path):
         assert got == []
     finally:
         dist.destroy_process_group()
+
+
+def test_plane_metrics_on_scrape_endpoint():
+    """The replica-plane counters must be in the SERVED registry (a
+    counter on prometheus_client's default registry would silently
+    vanish from /monitoring)."""
+    from tfservingcache_amd.utils import metrics as mt
+    mt.plane_transfers.labels("send").inc(0)
+    mt.plane_bytes.labels("send").inc(0)
+    text = mt.render().decode()
+    assert "tfservingcache_replica_plane_transfers_total" in text
+    assert "tfservingcache_replica_plane_bytes_total" in text

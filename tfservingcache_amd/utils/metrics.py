@@ -57,10 +57,12 @@ engine_pool_bytes = Gauge(
     registry=REGISTRY)
 plane_transfers = Counter(
     "tfservingcache_replica_plane_transfers_total",
-    "Model byte-pushes over the RCCL/xGMI replica plane", ["role"])
+    "Model byte-pushes over the RCCL/xGMI replica plane", ["role"],
+    registry=REGISTRY)
 plane_bytes = Counter(
     "tfservingcache_replica_plane_bytes_total",
-    "Bytes moved over the RCCL/xGMI replica plane", ["role"])
+    "Bytes moved over the RCCL/xGMI replica plane", ["role"],
+    registry=REGISTRY)
 engine_pool_models = Gauge(
     "tfservingcache_engine_pool_models",
     "Number of models resident in the pool", ["device"], registry=REGISTRY)
