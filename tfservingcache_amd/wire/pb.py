@@ -368,12 +368,16 @@ class Message(metaclass=MessageMeta):
                         lst.append(v)
                 elif wire_type == WIRE_LEN and f.kind == "float":
                     n, pos = read_varint(data, pos)
+                    if pos + n > len(data):
+                        raise ValueError("truncated packed float field")
                     lst = getattr(msg, f.name)
                     for i in range(n // 4):
                         lst.append(_f32.unpack_from(data, pos + 4 * i)[0])
                     pos += n
                 elif wire_type == WIRE_LEN and f.kind == "double":
                     n, pos = read_varint(data, pos)
+                    if pos + n > len(data):
+                        raise ValueError("truncated packed double field")
                     lst = getattr(msg, f.name)
                     for i in range(n // 8):
                         lst.append(_f64.unpack_from(data, pos + 8 * i)[0])
