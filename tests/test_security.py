@@ -170,3 +170,39 @@ def test_warmup_corrupt_length_field(tmp_path):
     p.write_bytes(struct.pack("<Q", len(rec)) + b"\0" * 4 + rec +
                   b"\0" * 4)
     assert list(read_tfrecords(str(p))) == [rec]
+
+
+def test_malformed_tensorproto_dims_rejected_cleanly():
+    """dims/content mismatches and absurd dims must raise a clean
+    ValueError/ServingError from the conversion layer — never allocate
+    per attacker-declared dims or crash."""
+    from tfservingcache_amd.wire.tensor import tensorproto_to_numpy
+
+    # content 4 bytes, dims claim 100 elements
+    tp = m.TensorProto(
+        dtype=m.DT_FLOAT,
+        tensor_shape=m.TensorShapeProto(
+            dim=[m.TensorShapeDim(size=100)]),
+        tensor_content=b"\x00\x00\x80?")
+    with pytest.raises((ValueError, Exception)):
+        arr = tensorproto_to_numpy(tp)
+        assert arr.size == 100      # reaching here would be silent junk
+
+    # absurd dims: must not allocate ~8 EB
+    tp2 = m.TensorProto(
+        dtype=m.DT_FLOAT,
+        tensor_shape=m.TensorShapeProto(
+            dim=[m.TensorShapeDim(size=1 << 30),
+                 m.TensorShapeDim(size=1 << 30)]),
+        tensor_content=b"\x00" * 16)
+    with pytest.raises(Exception):
+        tensorproto_to_numpy(tp2)
+
+    # negative dim
+    tp3 = m.TensorProto(
+        dtype=m.DT_FLOAT,
+        tensor_shape=m.TensorShapeProto(
+            dim=[m.TensorShapeDim(size=-4)]),
+        tensor_content=b"\x00" * 16)
+    with pytest.raises(Exception):
+        tensorproto_to_numpy(tp3)

@@ -79,6 +79,10 @@ def tensorproto_to_numpy(tp: m.TensorProto) -> np.ndarray:
     if np_dtype is None:
         raise TensorCodecError(f"unsupported dtype {tp.dtype}")
     shape = tp.tensor_shape.sizes() if tp.tensor_shape is not None else None
+    if shape is not None and any(d < 0 for d in shape):
+        # negative dims are signature wildcards, not valid in a request
+        # tensor (numpy reshape would silently treat them as -1)
+        raise TensorCodecError(f"negative dim in tensor shape {shape}")
     if tp.tensor_content:
         arr = np.frombuffer(tp.tensor_content, dtype=np_dtype)
         if shape is not None:
