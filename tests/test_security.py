@@ -206,3 +206,25 @@ def test_malformed_tensorproto_dims_rejected_cleanly():
         tensor_content=b"\x00" * 16)
     with pytest.raises(Exception):
         tensorproto_to_numpy(tp3)
+
+
+def test_string_tensor_dims_bounded():
+    """DT_STRING dims beyond the actual value count must raise before
+    allocating (attacker-declared dims -> np.empty DoS)."""
+    from tfservingcache_amd.wire.tensor import tensorproto_to_numpy
+    tp = m.TensorProto(
+        dtype=m.DT_STRING,
+        tensor_shape=m.TensorShapeProto(
+            dim=[m.TensorShapeDim(size=1 << 30),
+                 m.TensorShapeDim(size=1 << 30)]),
+        string_val=[b"x"])
+    with pytest.raises(ValueError):
+        tensorproto_to_numpy(tp)
+    # sane string tensors still round-trip
+    tp2 = m.TensorProto(
+        dtype=m.DT_STRING,
+        tensor_shape=m.TensorShapeProto(
+            dim=[m.TensorShapeDim(size=2)]),
+        string_val=[b"a", b"bb"])
+    arr = tensorproto_to_numpy(tp2)
+    assert list(arr) == [b"a", b"bb"]

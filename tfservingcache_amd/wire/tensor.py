@@ -71,8 +71,17 @@ def tensorproto_to_numpy(tp: m.TensorProto) -> np.ndarray:
     """Decode a TensorProto into a numpy array (DT_BFLOAT16 -> uint16 words)."""
     if tp.dtype == m.DT_STRING:
         shape = tp.tensor_shape.sizes() if tp.tensor_shape else [len(tp.string_val)]
-        arr = np.empty(int(np.prod(shape)) if shape else 1, dtype=object)
-        for i, s in enumerate(tp.string_val):
+        if any(d < 0 for d in shape):
+            raise TensorCodecError(f"negative dim in tensor shape {shape}")
+        n = int(np.prod(shape)) if shape else 1
+        # dims are attacker-declared: never allocate past the actual
+        # payload (TF requires |string_val| == prod(dims), splat aside)
+        if n > max(len(tp.string_val), 1):
+            raise TensorCodecError(
+                f"string tensor dims {shape} exceed {len(tp.string_val)} "
+                "values")
+        arr = np.empty(n, dtype=object)
+        for i, s in enumerate(tp.string_val[:n]):
             arr[i] = s
         return arr.reshape(shape)
     np_dtype = DTYPE_TO_NP.get(tp.dtype)
