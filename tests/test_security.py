@@ -3,6 +3,8 @@ bytes, and single-flight lock lifetime (round-1 advisor findings)."""
 import threading
 import time
 
+import numpy as np
+
 import pytest
 
 from tfservingcache_amd.cachemanager import (CacheManager, LRUCache,
@@ -228,3 +230,25 @@ def test_string_tensor_dims_bounded():
         string_val=[b"a", b"bb"])
     arr = tensorproto_to_numpy(tp2)
     assert list(arr) == [b"a", b"bb"]
+
+
+def test_numeric_splat_expansion_bounded():
+    """The TF splat rule (repeat last value to fill dims) must not
+    expand to attacker-declared exabyte sizes."""
+    from tfservingcache_amd.wire.tensor import tensorproto_to_numpy
+    tp = m.TensorProto(
+        dtype=m.DT_FLOAT,
+        tensor_shape=m.TensorShapeProto(
+            dim=[m.TensorShapeDim(size=1 << 30),
+                 m.TensorShapeDim(size=1 << 30)]),
+        float_val=[1.0])
+    with pytest.raises(ValueError):
+        tensorproto_to_numpy(tp)
+    # the legitimate splat still works
+    tp2 = m.TensorProto(
+        dtype=m.DT_FLOAT,
+        tensor_shape=m.TensorShapeProto(
+            dim=[m.TensorShapeDim(size=4)]),
+        float_val=[7.0])
+    np.testing.assert_allclose(tensorproto_to_numpy(tp2),
+                               [7.0, 7.0, 7.0, 7.0])

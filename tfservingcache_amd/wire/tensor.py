@@ -109,7 +109,11 @@ def tensorproto_to_numpy(tp: m.TensorProto) -> np.ndarray:
         if arr.size == n:
             arr = arr.reshape(shape)
         elif arr.size and arr.size < n:
-            # TF splat rule: repeat the last value
+            # TF splat rule: repeat the last value. n is attacker-
+            # declared — bound the expansion (256M elements ~ 1 GiB f32)
+            if n > 1 << 28:
+                raise TensorCodecError(
+                    f"splat to {n} elements exceeds the 2^28 cap")
             arr = np.concatenate([arr, np.full(n - arr.size, arr[-1],
                                                dtype=arr.dtype)]).reshape(shape)
         elif arr.size == 0 and n == 0:
