@@ -226,3 +226,17 @@ def test_cpp_json_parser_probe():
     np.testing.assert_allclose(
         vals, np.array([1e-5, -2.5e3, 0.12345678901234568, -7.0],
                        dtype=np.float32), rtol=1e-6)
+
+
+def test_cpp_json_parser_depth_capped():
+    """A body of 100k '['s must raise cleanly, not overflow the native
+    stack (remote-crash DoS on the REST front-end)."""
+    import torch  # noqa: F401
+    from tfservingcache_amd.engine import _tfsc_engine as ext
+    deep = b"[" * 100000 + b"1" + b"]" * 100000
+    with pytest.raises(Exception):
+        ext._rest_parse_probe(deep, False)
+    # 8-D tensors (realistic rank ceiling) still parse
+    body = b"[" * 8 + b"1,2" + b"]" * 8
+    dims, vals = ext._rest_parse_probe(body, False)
+    assert list(dims) == [1] * 7 + [2]

@@ -248,6 +248,9 @@ static void parse_dense(JParser& j, int depth,
                         std::vector<int64_t>& dims,
                         std::vector<float>& fdata,
                         std::vector<int32_t>& idata, bool is_int) {
+  // recursion depth == tensor rank; cap it or a body of 100k '['s
+  // overflows the native stack (remote crash). Real tensors are <= 8-D.
+  if (depth > 32) throw RestFallback("nesting too deep");
   j.ws();
   if (j.p < j.end && *j.p == '[') {
     ++j.p;
