@@ -185,8 +185,9 @@ class LocalServingHandler:
         feats = self._examples_to_features(req.input)
         if len(feats) == 1:
             feats = {"": next(iter(feats.values()))}
-        outputs, version = self.predict_arrays(spec.name,
-                                               spec.version_value(), feats)
+        outputs, version = self.predict_arrays(
+            spec.name, spec.version_value(), feats,
+            version_label=spec.version_label)
         # TF classification signatures name their outputs 'scores' (and
         # 'classes'); fall back to the first output otherwise
         scores = None
@@ -213,8 +214,9 @@ class LocalServingHandler:
         feats = self._examples_to_features(req.input)
         if len(feats) == 1:
             feats = {"": next(iter(feats.values()))}
-        outputs, version = self.predict_arrays(spec.name,
-                                               spec.version_value(), feats)
+        outputs, version = self.predict_arrays(
+            spec.name, spec.version_value(), feats,
+            version_label=spec.version_label)
         vals = np.asarray(next(iter(outputs.values())),
                           dtype=np.float32).reshape(-1)
         return m.RegressionResponse(
@@ -227,7 +229,8 @@ class LocalServingHandler:
     def get_model_metadata(self, req: m.GetModelMetadataRequest
                            ) -> m.GetModelMetadataResponse:
         spec = req.model_spec or m.ModelSpec()
-        model, version = self.get_model(spec.name, spec.version_value())
+        model, version = self.get_model(spec.name, spec.version_value(),
+                                        spec.version_label)
         resp = m.GetModelMetadataResponse(model_spec=m.ModelSpec(
             name=spec.name, version=m.Int64Value(value=version)))
         sig = model.signature_def
@@ -315,7 +318,8 @@ class LocalServingHandler:
     # -- SessionRun --------------------------------------------------------
     def session_run(self, req: m.SessionRunRequest) -> m.SessionRunResponse:
         spec = req.model_spec or m.ModelSpec()
-        model, version = self.get_model(spec.name, spec.version_value())
+        model, version = self.get_model(spec.name, spec.version_value(),
+                                        spec.version_label)
         feeds = {nt.name: tensorproto_to_numpy(nt.tensor)
                  for nt in req.feed}
         try:
