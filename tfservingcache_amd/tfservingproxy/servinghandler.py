@@ -249,8 +249,15 @@ class LocalServingHandler:
         spec = req.model_spec or m.ModelSpec()
         if not spec.name:
             raise ServingError("missing model name")
-        entries = self.cm.pool.get_status(spec.name,
-                                          spec.version_value() or None)
+        version = spec.version_value()
+        if not version and spec.version_label:
+            labeled = self._resolve_label(spec.name, spec.version_label)
+            if labeled is None:
+                raise ServingError(
+                    f"unknown version label {spec.version_label!r} for "
+                    f"model {spec.name}", m.ERROR_NOT_FOUND)
+            version = labeled
+        entries = self.cm.pool.get_status(spec.name, version or None)
         resp = m.GetModelStatusResponse()
         if not entries:
             raise ServingError(
