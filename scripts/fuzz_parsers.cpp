@@ -186,7 +186,17 @@ int main(int argc, char** argv) {
       seed_predict_request(4, 64)};
   std::vector<std::string> json_seeds = {
       seed_json(2, 8), "[1,2,3]", "[[1],[2]]",
-      "[1e300, -0.5, 448, 0.00001]", "[[\"x\"]]", "[[[[1]]]]"};
+      "[1e300, -0.5, 448, 0.00001]", "[[\"x\"]]", "[[[[1]]]]",
+      // deep-nesting class (depth cap regression: a crafted body must
+      // throw RestFallback, never overflow the native stack)
+      std::string(1000, '[') + "1" + std::string(1000, ']'),
+      std::string(64, '[') + "1,2" + std::string(64, ']')};
+  // directed deep-nesting sweep beyond the mutation loop
+  for (int d : {16, 33, 100, 5000, 200000}) {
+    drive_json(std::string(size_t(d), '[') + "1" +
+               std::string(size_t(d), ']'));
+    drive_json(std::string(size_t(d), '['));
+  }
   for (long i = 0; i < iters; ++i) {
     drive_wire(mutate(wire_seeds[rng() % wire_seeds.size()]));
     drive_json(mutate(json_seeds[rng() % json_seeds.size()]));
