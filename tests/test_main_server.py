@@ -48,7 +48,7 @@ def make_cfg(tmp_path, with_discovery=False, name="n1"):
     return Config(values)
 
 
-def wait_http(port, path="/healthz", timeout=10):
+def wait_http(port, path="/healthz", timeout=30):
     deadline = time.time() + timeout
     while time.time() < deadline:
         try:

@@ -510,12 +510,25 @@ class Server:
             async def shutdown():
                 for r in self._runners:
                     await r.cleanup()
-            fut = asyncio.run_coroutine_threadsafe(shutdown(), self._loop)
+            coro = shutdown()
             try:
-                fut.result(timeout=5)
-            except Exception:       # noqa: BLE001
+                if self._loop.is_running() and not self._loop.is_closed():
+                    fut = asyncio.run_coroutine_threadsafe(coro, self._loop)
+                    coro = None
+                    try:
+                        fut.result(timeout=5)
+                    except Exception:       # noqa: BLE001
+                        pass
+            except RuntimeError:
+                pass                # loop died between check and schedule
+            finally:
+                if coro is not None:
+                    coro.close()    # never scheduled: silence the
+                                    # "never awaited" RuntimeWarning
+            try:
+                self._loop.call_soon_threadsafe(self._loop.stop)
+            except RuntimeError:
                 pass
-            self._loop.call_soon_threadsafe(self._loop.stop)
         self.cm.close()
 
 
