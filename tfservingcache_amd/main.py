@@ -363,6 +363,13 @@ class Server:
                 self._runners.append(runner)
         self._loop.run_until_complete(boot())
         self._loop.run_forever()
+        # stop() requested: drain callbacks and release the loop's
+        # resources (unclosed-loop ResourceWarning otherwise)
+        try:
+            self._loop.run_until_complete(asyncio.sleep(0))
+            self._loop.close()
+        except RuntimeError:
+            pass
 
     def _wire_replica_plane(self, replicas: int) -> None:
         """Cold-load fan-out: after this rank fetches a model from the
